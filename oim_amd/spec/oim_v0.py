@@ -1,0 +1,109 @@
+"""The oim.v0 API schema (authored from docs/spec.md).
+
+Wire-compatible with the reference's pkg/spec/oim/v0/oim.pb.go: same
+package, message names, field names/numbers and service methods
+(reference spec.md:25-196).  docs/spec.md in this repo is the literate
+source; tests/test_spec.py checks this module against it so the two
+cannot drift (the reference's Makefile:85-103 "spec.md -> oim.proto"
+extraction, done as a consistency test instead of codegen).
+"""
+
+from __future__ import annotations
+
+from ._build import Field, Message, Service, build_file
+
+PACKAGE = "oim.v0"
+
+MESSAGES = [
+    Message("SetValueRequest", [Field("value", 1, "Value")]),
+    Message("Value", [Field("path", 1, "string"), Field("value", 2, "string")]),
+    Message("SetValueReply", []),
+    Message("GetValuesRequest", [Field("path", 1, "string")]),
+    Message("GetValuesReply", [Field("values", 1, "Value", repeated=True)]),
+    Message(
+        "MapVolumeRequest",
+        [
+            Field("volume_id", 1, "string"),
+            Field("malloc", 2, "MallocParams", oneof="params"),
+            Field("ceph", 3, "CephParams", oneof="params"),
+        ],
+    ),
+    Message("MallocParams", []),
+    Message(
+        "CephParams",
+        [
+            Field("user_id", 1, "string"),
+            Field("secret", 2, "string"),
+            Field("monitors", 3, "string"),
+            Field("pool", 4, "string"),
+            Field("image", 5, "string"),
+        ],
+    ),
+    Message(
+        "MapVolumeReply",
+        [
+            Field("pci_address", 1, "PCIAddress"),
+            Field("scsi_disk", 2, "SCSIDisk"),
+        ],
+    ),
+    Message(
+        "PCIAddress",
+        [
+            Field("domain", 1, "uint32"),
+            Field("bus", 2, "uint32"),
+            Field("device", 3, "uint32"),
+            Field("function", 4, "uint32"),
+        ],
+    ),
+    Message("SCSIDisk", [Field("target", 1, "uint32"), Field("lun", 2, "uint32")]),
+    Message("UnmapVolumeRequest", [Field("volume_id", 1, "string")]),
+    Message("UnmapVolumeReply", []),
+    Message(
+        "ProvisionMallocBDevRequest",
+        [Field("bdev_name", 1, "string"), Field("size", 2, "int64")],
+    ),
+    Message("ProvisionMallocBDevReply", []),
+    Message("CheckMallocBDevRequest", [Field("bdev_name", 1, "string")]),
+    Message("CheckMallocBDevReply", []),
+]
+
+SERVICES = [
+    Service(
+        "Registry",
+        [
+            ("SetValue", "SetValueRequest", "SetValueReply"),
+            ("GetValues", "GetValuesRequest", "GetValuesReply"),
+        ],
+    ),
+    Service(
+        "Controller",
+        [
+            ("MapVolume", "MapVolumeRequest", "MapVolumeReply"),
+            ("UnmapVolume", "UnmapVolumeRequest", "UnmapVolumeReply"),
+            ("ProvisionMallocBDev", "ProvisionMallocBDevRequest", "ProvisionMallocBDevReply"),
+            ("CheckMallocBDev", "CheckMallocBDevRequest", "CheckMallocBDevReply"),
+        ],
+    ),
+]
+
+_classes = build_file(
+    name="oim_amd/oim.proto", package=PACKAGE, messages=MESSAGES, services=SERVICES
+)
+
+SetValueRequest = _classes["SetValueRequest"]
+Value = _classes["Value"]
+SetValueReply = _classes["SetValueReply"]
+GetValuesRequest = _classes["GetValuesRequest"]
+GetValuesReply = _classes["GetValuesReply"]
+MapVolumeRequest = _classes["MapVolumeRequest"]
+MallocParams = _classes["MallocParams"]
+CephParams = _classes["CephParams"]
+MapVolumeReply = _classes["MapVolumeReply"]
+PCIAddress = _classes["PCIAddress"]
+SCSIDisk = _classes["SCSIDisk"]
+UnmapVolumeRequest = _classes["UnmapVolumeRequest"]
+UnmapVolumeReply = _classes["UnmapVolumeReply"]
+ProvisionMallocBDevRequest = _classes["ProvisionMallocBDevRequest"]
+ProvisionMallocBDevReply = _classes["ProvisionMallocBDevReply"]
+CheckMallocBDevRequest = _classes["CheckMallocBDevRequest"]
+CheckMallocBDevReply = _classes["CheckMallocBDevReply"]
