@@ -1,0 +1,47 @@
+# Builds the hipstore native core: the oim_amd._hipstore Python module
+# (in-tree, travels to the GPU box) and the hipstored daemon.
+# hipcc cross-compiles gfx950 without a GPU present.
+
+ROCM ?= /opt/rocm
+HIPCC := $(ROCM)/bin/hipcc
+ARCH ?= gfx950
+PYTHON ?= python3
+
+PY_INC := $(shell $(PYTHON) -c "import sysconfig; print(sysconfig.get_paths()['include'])")
+PYBIND_INC := $(shell $(PYTHON) -c "import pybind11; print(pybind11.get_include())")
+EXT_SUFFIX := $(shell $(PYTHON) -c "import sysconfig; print(sysconfig.get_config_var('EXT_SUFFIX'))")
+
+CXXFLAGS := -O3 -std=c++17 -fPIC -Wall -Wextra -Wno-unused-parameter \
+            -Inative/include
+HIPFLAGS := --offload-arch=$(ARCH)
+
+B := native/build
+HDRS := $(wildcard native/include/hipstore/*.h)
+
+CORE_OBJS := $(B)/json.o $(B)/bdev.o $(B)/crc32c.o $(B)/rpc_server.o \
+             $(B)/methods.o $(B)/nbd.o $(B)/gpu.o
+
+.PHONY: all clean
+all: oim_amd/_hipstore$(EXT_SUFFIX) bin/hipstored
+
+$(B):
+	mkdir -p $(B)
+
+$(B)/%.o: native/src/%.cpp $(HDRS) | $(B)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(B)/gpu.o: native/src/gpu.hip $(HDRS) | $(B)
+	$(HIPCC) $(CXXFLAGS) $(HIPFLAGS) -c $< -o $@
+
+$(B)/pybind.o: native/src/pybind.cpp $(HDRS) | $(B)
+	$(HIPCC) $(CXXFLAGS) -fvisibility=hidden -I$(PY_INC) -I$(PYBIND_INC) -c $< -o $@
+
+oim_amd/_hipstore$(EXT_SUFFIX): $(CORE_OBJS) $(B)/pybind.o
+	$(HIPCC) $(HIPFLAGS) -shared $^ -o $@
+
+bin/hipstored: $(CORE_OBJS) $(B)/main.o
+	mkdir -p bin
+	$(HIPCC) $(HIPFLAGS) $^ -o $@
+
+clean:
+	rm -rf $(B) bin oim_amd/_hipstore*.so

@@ -1,0 +1,66 @@
+// GPU data-path engine API (implemented in gpu.hip).
+//
+// MI355X-native replacement for the reference's SPDK copy engine +
+// malloc bdev pair (reference lib/copy/copy_engine.c,
+// lib/bdev/malloc/bdev_malloc.c): block I/O executes as batched HIP
+// kernel launches on per-queue streams, 4 KiB tiles staged through LDS
+// (visible in rocprof as hipstore::k_copy_blocks), with pinned-host
+// descriptor rings read by the GPU directly.
+
+#pragma once
+
+#include <cstdint>
+#include <string>
+#include <vector>
+
+#include "hipstore/bdev.h"
+
+namespace hipstore {
+
+// True when at least one HIP device is present (cached).
+bool gpu_available();
+int gpu_device_count();
+// PCI BDF string ("0000:c1:00.0") of a device, for registry <id>/pci.
+std::string gpu_pci_address(int device);
+
+// Create an HBM-resident malloc bdev on `device`. Throws std::runtime_error
+// when no GPU or allocation fails. product_name stays "Malloc disk": the
+// HBM bdev IS the malloc bdev on this platform, and the reference
+// controller keys its keep-on-unmap logic on that string
+// (reference controller.go:205).
+BdevPtr create_hbm_bdev(const std::string& name, uint64_t block_size,
+                        uint64_t num_blocks, int device);
+
+// Pinned-host buffer helpers (fall back to plain malloc without a GPU).
+void* alloc_pinned(size_t bytes);
+void free_pinned(void* ptr);
+
+// Synchronous convenience wrappers (tests, NBD pump): create a
+// throwaway channel, submit, poll to completion. Returns IoStatus.
+int bdev_read_sync(Bdev* bdev, uint64_t offset, void* buf, uint64_t len);
+int bdev_write_sync(Bdev* bdev, uint64_t offset, const void* buf, uint64_t len);
+int bdev_fill_sync(Bdev* bdev, uint64_t offset, uint8_t value, uint64_t len);
+
+// fio-shaped benchmark harness (SPDK bdevperf analog): `workload` is
+// randread / randwrite / randrw; runs num_queues submitter threads each
+// holding queue_depth I/Os in flight for `seconds`, then reports
+// latency percentiles measured per-I/O.
+struct PerfResult {
+  double seconds = 0;
+  uint64_t io_count = 0;
+  double iops = 0;
+  double throughput_mbps = 0;
+  double lat_avg_us = 0;
+  double lat_p50_us = 0;
+  double lat_p90_us = 0;
+  double lat_p99_us = 0;
+  double lat_p999_us = 0;
+  double lat_max_us = 0;
+};
+
+PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
+                        uint32_t io_size, uint32_t queue_depth,
+                        int num_queues, double seconds,
+                        uint64_t max_ios = 0);
+
+}  // namespace hipstore

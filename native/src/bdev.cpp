@@ -1,0 +1,128 @@
+#include "hipstore/bdev.h"
+
+#include <cstring>
+#include <deque>
+#include <random>
+
+namespace hipstore {
+
+std::string Bdev::make_uuid() {
+  // Random v4-format UUID; no libuuid dependency.
+  static std::mutex mutex;
+  static std::mt19937_64 rng{std::random_device{}()};
+  std::lock_guard<std::mutex> lock(mutex);
+  uint64_t a = rng(), b = rng();
+  char buf[40];
+  snprintf(buf, sizeof(buf), "%08x-%04x-4%03x-%04x-%012llx",
+           static_cast<uint32_t>(a >> 32), static_cast<uint16_t>(a >> 16),
+           static_cast<uint16_t>(a & 0xFFF),
+           static_cast<uint16_t>(0x8000 | ((b >> 48) & 0x3FFF)),
+           static_cast<unsigned long long>(b & 0xFFFFFFFFFFFFULL));
+  return buf;
+}
+
+BdevManager& BdevManager::instance() {
+  static BdevManager manager;
+  return manager;
+}
+
+bool BdevManager::add(BdevPtr bdev) {
+  std::lock_guard<std::mutex> lock(mutex_);
+  return bdevs_.emplace(bdev->name(), std::move(bdev)).second;
+}
+
+bool BdevManager::remove(const std::string& name) {
+  std::lock_guard<std::mutex> lock(mutex_);
+  auto it = bdevs_.find(name);
+  if (it == bdevs_.end()) return false;
+  bdevs_.erase(it);
+  return true;
+}
+
+BdevPtr BdevManager::find(const std::string& name) const {
+  std::lock_guard<std::mutex> lock(mutex_);
+  auto it = bdevs_.find(name);
+  return it == bdevs_.end() ? nullptr : it->second;
+}
+
+std::vector<BdevPtr> BdevManager::list() const {
+  std::lock_guard<std::mutex> lock(mutex_);
+  std::vector<BdevPtr> out;
+  out.reserve(bdevs_.size());
+  for (const auto& [_, bdev] : bdevs_) out.push_back(bdev);
+  return out;
+}
+
+namespace {
+
+// Host-RAM bdev: the CI stand-in for the HBM bdev and the "SPDK Malloc
+// bdev on CPU" of BASELINE config 1. I/O executes at submit time (a
+// memcpy, like SPDK's default copy engine, reference
+// lib/copy/copy_engine.c mem_copy_submit) but completes at poll time to
+// preserve the async contract.
+class MallocChannel : public IoChannel {
+ public:
+  std::deque<std::pair<IoCompletion, int>> done;
+};
+
+class MallocBdev : public Bdev {
+ public:
+  MallocBdev(const std::string& name, uint64_t block_size, uint64_t num_blocks)
+      : Bdev(name, "Malloc disk", block_size, num_blocks),
+        data_(block_size * num_blocks, 0) {}
+
+  std::shared_ptr<IoChannel> get_channel() override {
+    return std::make_shared<MallocChannel>();
+  }
+
+  void submit(IoChannel* ch, IoRequest req) override {
+    auto* channel = static_cast<MallocChannel*>(ch);
+    int status = kIoOk;
+    if (!check_bounds(req)) {
+      status = kIoInvalid;
+    } else {
+      std::lock_guard<std::mutex> lock(mutex_);
+      switch (req.op) {
+        case IoOp::kRead:
+          memcpy(req.buffer, data_.data() + req.offset, req.length);
+          break;
+        case IoOp::kWrite:
+          memcpy(data_.data() + req.offset, req.buffer, req.length);
+          break;
+        case IoOp::kFill:
+          memset(data_.data() + req.offset, req.fill, req.length);
+          break;
+        case IoOp::kFlush:
+          break;
+      }
+    }
+    channel->done.emplace_back(std::move(req.on_complete), status);
+  }
+
+  int poll(IoChannel* ch) override {
+    auto* channel = static_cast<MallocChannel*>(ch);
+    // Only drain what was queued at entry: completion callbacks may
+    // resubmit, which appends to the deque again — draining to empty
+    // would never return to the caller.
+    size_t n = channel->done.size();
+    for (size_t i = 0; i < n; ++i) {
+      auto [cb, status] = std::move(channel->done.front());
+      channel->done.pop_front();
+      if (cb) cb(status);
+    }
+    return static_cast<int>(n);
+  }
+
+ private:
+  std::mutex mutex_;  // serializes overlapping I/O from many channels
+  std::vector<uint8_t> data_;
+};
+
+}  // namespace
+
+BdevPtr create_malloc_bdev(const std::string& name, uint64_t block_size,
+                           uint64_t num_blocks) {
+  return std::make_shared<MallocBdev>(name, block_size, num_blocks);
+}
+
+}  // namespace hipstore

@@ -1,0 +1,618 @@
+// MI355X (gfx950) data-path engine: HBM-resident malloc bdev + batched
+// LDS-staged block-copy kernels on per-queue HIP streams.
+//
+// Design (MI355X-first, not a port of SPDK's reactor model):
+//   - The bdev backing store is one hipMalloc region in HBM3E.
+//   - Each IoChannel owns a HIP stream, a pinned-host descriptor ring the
+//     GPU reads directly (zero-copy over PCIe), and an event pool.
+//   - submit() only queues; poll() coalesces every pending request into
+//     ONE kernel launch (a batch), records an event, and retires
+//     finished batches. This replaces SPDK's per-core poller loop
+//     (reference lib/bdev/malloc/bdev_malloc.c submit/complete) with a
+//     launch-batched GPU pipeline: at QD=32 a whole queue depth is one
+//     launch (~4 us) + one 128 KiB PCIe burst.
+//   - Kernels stage every 4 KiB block tile through LDS (16 B/lane
+//     vectorized, one wave per tile, 4 waves per workgroup): the staging
+//     hop costs little (LDS ~150 TB/s vs ~55 GB/s PCIe / 6.3 TB/s HBM)
+//     and gives the CRC32C/verify paths a resident tile to chew on.
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <atomic>
+#include <chrono>
+#include <cstring>
+#include <deque>
+#include <mutex>
+#include <random>
+#include <stdexcept>
+#include <thread>
+#include <vector>
+
+#include "hipstore/crc32c.h"
+#include "hipstore/engine.h"
+
+namespace hipstore {
+
+// CPU fallback defined in crc32c.cpp.
+void crc32c_cpu_fallback(Bdev* bdev, uint64_t offset, uint32_t block_size,
+                         uint32_t count, uint32_t* out);
+
+#define HIP_CHECK(expr)                                                   \
+  do {                                                                    \
+    hipError_t _e = (expr);                                               \
+    if (_e != hipSuccess) {                                               \
+      throw std::runtime_error(std::string("HIP error: ") +               \
+                               hipGetErrorString(_e) + " at " #expr);     \
+    }                                                                     \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// Kernels
+// ---------------------------------------------------------------------------
+
+namespace {
+
+constexpr uint32_t kTileBytes = 4096;   // one wave moves one tile
+constexpr uint32_t kWavesPerWg = 4;     // 256-thread workgroups
+constexpr uint32_t kRingSlots = 1 << 16;  // descriptor ring per channel
+constexpr uint32_t kMaxBatchTiles = kRingSlots / 4;
+
+struct BlockDesc {
+  const uint8_t* src;  // null => fill
+  uint8_t* dst;
+  uint32_t bytes;      // multiple of 16, <= kTileBytes
+  uint32_t fill;       // fill byte (replicated) when src == null
+};
+
+// One wave per descriptor; 16 B/lane vector moves staged through LDS.
+// Each lane re-reads exactly the bytes it wrote, so wave-internal
+// lgkmcnt ordering (compiler-inserted) is the only sync needed — no
+// __syncthreads() in the hot path.
+__global__ __launch_bounds__(kWavesPerWg * 64) void k_copy_blocks(
+    const BlockDesc* __restrict__ descs, uint32_t n) {
+  __shared__ __attribute__((aligned(16))) uint8_t lds_raw[kWavesPerWg * kTileBytes];
+  const uint32_t wave = threadIdx.x >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t idx = blockIdx.x * kWavesPerWg + wave;
+  if (idx >= n) return;
+  const BlockDesc d = descs[idx];
+  float4* lds = reinterpret_cast<float4*>(lds_raw + wave * kTileBytes);
+  const uint32_t n16 = d.bytes >> 4;  // 16-byte units
+  if (d.src != nullptr) {
+    const float4* __restrict__ src = reinterpret_cast<const float4*>(d.src);
+#pragma unroll 4
+    for (uint32_t i = lane; i < n16; i += 64) lds[i] = src[i];
+  } else {
+    const uint32_t b = d.fill & 0xFF;
+    const uint32_t word = b | (b << 8) | (b << 16) | (b << 24);
+    const float4 v = {__uint_as_float(word), __uint_as_float(word),
+                      __uint_as_float(word), __uint_as_float(word)};
+    for (uint32_t i = lane; i < n16; i += 64) lds[i] = v;
+  }
+  float4* __restrict__ dst = reinterpret_cast<float4*>(d.dst);
+#pragma unroll 4
+  for (uint32_t i = lane; i < n16; i += 64) dst[i] = lds[i];
+}
+
+// --- CRC32C (Castagnoli, reflected 0x82F63B78) -----------------------------
+//
+// One LANE per block: lanes of a wave digest 64 different blocks in
+// parallel, the 256-entry table lives in LDS (one copy per workgroup,
+// broadcast reads are conflict-free). Parallelism comes from batch
+// width (count >= a few thousand blocks fills the chip), which matches
+// the NVMe-oF/TCP digest use: one CRC per in-flight 4 KiB PDU.
+__global__ __launch_bounds__(256) void k_crc32c_blocks(
+    const uint8_t* __restrict__ base, uint32_t block_size, uint32_t count,
+    uint32_t* __restrict__ out) {
+  __shared__ uint32_t table[256];
+  // Build the table once per workgroup (cheap: 8 iterations/entry).
+  if (threadIdx.x < 256) {
+    uint32_t crc = threadIdx.x;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      crc = (crc & 1u) ? (crc >> 1) ^ 0x82F63B78u : crc >> 1;
+    }
+    table[threadIdx.x] = crc;
+  }
+  __syncthreads();
+  const uint32_t idx = blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= count) return;
+  const uint8_t* p = base + static_cast<uint64_t>(idx) * block_size;
+  uint32_t crc = 0xFFFFFFFFu;
+  // 4 bytes per iteration through a word load (block_size % 4 == 0 is
+  // guaranteed by the 512-byte block granularity).
+  const uint32_t* pw = reinterpret_cast<const uint32_t*>(p);
+  for (uint32_t i = 0; i < block_size / 4; ++i) {
+    uint32_t w = pw[i];
+#pragma unroll
+    for (int b = 0; b < 4; ++b) {
+      crc = (crc >> 8) ^ table[(crc ^ (w & 0xFFu)) & 0xFFu];
+      w >>= 8;
+    }
+  }
+  out[idx] = ~crc;
+}
+
+int g_device_count = -1;
+
+int device_count_cached() {
+  static std::once_flag once;
+  std::call_once(once, [] {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) n = 0;
+    g_device_count = n;
+  });
+  return g_device_count;
+}
+
+}  // namespace
+
+bool gpu_available() { return device_count_cached() > 0; }
+int gpu_device_count() { return device_count_cached(); }
+
+std::string gpu_pci_address(int device) {
+  char bus_id[32] = {0};
+  if (hipDeviceGetPCIBusId(bus_id, sizeof(bus_id), device) != hipSuccess) {
+    return "";
+  }
+  // hip returns "0000:c1:00.0" style already; normalize to lower-case.
+  std::string s(bus_id);
+  std::transform(s.begin(), s.end(), s.begin(), ::tolower);
+  return s;
+}
+
+void* alloc_pinned(size_t bytes) {
+  if (!gpu_available()) return malloc(bytes);
+  void* p = nullptr;
+  HIP_CHECK(hipHostMalloc(&p, bytes, hipHostMallocMapped));
+  return p;
+}
+
+void free_pinned(void* ptr) {
+  if (ptr == nullptr) return;
+  if (!gpu_available()) {
+    free(ptr);
+    return;
+  }
+  (void)hipHostFree(ptr);
+}
+
+// ---------------------------------------------------------------------------
+// HBM bdev
+// ---------------------------------------------------------------------------
+
+namespace {
+
+// Host pointer -> pointer the GPU may dereference.
+template <typename T>
+T* device_view(T* host_ptr) {
+  void* dev = nullptr;
+  HIP_CHECK(hipHostGetDevicePointer(&dev, const_cast<void*>(
+      reinterpret_cast<const void*>(host_ptr)), 0));
+  return reinterpret_cast<T*>(dev);
+}
+
+struct PendingIo {
+  IoRequest req;
+  int status;
+};
+
+struct InflightBatch {
+  hipEvent_t event = nullptr;
+  uint32_t slots = 0;  // descriptor-ring slots to free on completion
+  std::vector<std::pair<IoCompletion, int>> completions;
+};
+
+class HbmChannel : public IoChannel {
+ public:
+  HbmChannel(int device, uint8_t* base, uint64_t size) : base_(base), size_(size) {
+    HIP_CHECK(hipSetDevice(device));
+    HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&ring_),
+                            kRingSlots * sizeof(BlockDesc), hipHostMallocMapped));
+    ring_dev_ = device_view(ring_);
+    device_ = device;
+  }
+
+  ~HbmChannel() override {
+    // Drain so no kernel touches the ring after it is freed.
+    (void)hipStreamSynchronize(stream_);
+    for (auto& batch : inflight_) (void)hipEventDestroy(batch.event);
+    for (auto event : event_pool_) (void)hipEventDestroy(event);
+    (void)hipHostFree(ring_);
+    (void)hipStreamDestroy(stream_);
+  }
+
+  void enqueue(IoRequest req, int status) {
+    pending_.push_back(PendingIo{std::move(req), status});
+  }
+
+  // Launch every pending request as one batch (if ring space allows).
+  // Batches occupy CONTIGUOUS descriptor-ring slots [head, head+n) — the
+  // kernel takes a flat pointer — so a batch never wraps: it is capped
+  // at the space to the end of the ring, and the head snaps to 0 when
+  // the ring is empty.
+  void kick() {
+    if (pending_.empty()) return;
+    if (used_slots_ == 0) ring_head_ = 0;  // empty ring: reclaim tail space
+    const uint32_t free_slots = kRingSlots - used_slots_;
+    const uint32_t space_to_end = kRingSlots - ring_head_;
+    const uint32_t cap =
+        std::min({free_slots, space_to_end, kMaxBatchTiles});
+    uint32_t n = 0;
+    size_t take = 0;
+    std::vector<std::pair<IoCompletion, int>> completions;
+    for (const PendingIo& io : pending_) {
+      uint32_t tiles = 0;
+      if (io.status == kIoOk && io.req.op != IoOp::kFlush) {
+        tiles = static_cast<uint32_t>(
+            (io.req.length + kTileBytes - 1) / kTileBytes);
+      }
+      if (n + tiles > cap) break;
+      n += tiles;
+      ++take;
+    }
+    if (take == 0) return;  // no contiguous space yet; retire() will free it
+    uint32_t slot = ring_head_;
+    for (size_t k = 0; k < take; ++k) {
+      PendingIo& io = pending_[k];
+      if (io.status == kIoOk && io.req.op != IoOp::kFlush) {
+        expand(io.req, &slot);
+      }
+      completions.emplace_back(std::move(io.req.on_complete), io.status);
+    }
+    pending_.erase(pending_.begin(), pending_.begin() + take);
+    if (n > 0) {
+      const uint32_t first = ring_head_;
+      const uint32_t grid = (n + kWavesPerWg - 1) / kWavesPerWg;
+      hipLaunchKernelGGL(k_copy_blocks, dim3(grid), dim3(kWavesPerWg * 64), 0,
+                         stream_, ring_dev_ + first, n);
+    }
+    hipEvent_t event = get_event();
+    HIP_CHECK(hipEventRecord(event, stream_));
+    ring_head_ = (ring_head_ + n) % kRingSlots;
+    used_slots_ += n;
+    inflight_.push_back(InflightBatch{event, n, std::move(completions)});
+  }
+
+  int retire(bool wait) {
+    int completed = 0;
+    while (!inflight_.empty()) {
+      InflightBatch& batch = inflight_.front();
+      hipError_t st = wait ? hipEventSynchronize(batch.event)
+                           : hipEventQuery(batch.event);
+      if (st == hipErrorNotReady) break;
+      if (st != hipSuccess) throw std::runtime_error("hip event failure");
+      for (auto& [cb, status] : batch.completions) {
+        if (cb) cb(status);
+        ++completed;
+      }
+      used_slots_ -= batch.slots;
+      event_pool_.push_back(batch.event);
+      inflight_.pop_front();
+    }
+    return completed;
+  }
+
+  bool idle() const { return pending_.empty() && inflight_.empty(); }
+
+ private:
+  void expand(const IoRequest& req, uint32_t* slot) {
+    // Split a request into <=4 KiB tiles, one ring descriptor each.
+    // kick() capped the batch so [ring_head_, ring_head_+n) never wraps.
+    uint64_t done = 0;
+    while (done < req.length) {
+      const uint32_t bytes = static_cast<uint32_t>(
+          std::min<uint64_t>(kTileBytes, req.length - done));
+      BlockDesc& d = ring_[*slot];
+      if (req.op == IoOp::kRead) {
+        d.src = base_ + req.offset + done;
+        d.dst = buf_device(static_cast<uint8_t*>(req.buffer) + done);
+      } else if (req.op == IoOp::kWrite) {
+        d.src = buf_device(static_cast<uint8_t*>(req.buffer) + done);
+        d.dst = base_ + req.offset + done;
+      } else {  // kFill
+        d.src = nullptr;
+        d.dst = base_ + req.offset + done;
+      }
+      d.bytes = bytes;
+      d.fill = req.fill;
+      ++*slot;
+      done += bytes;
+    }
+  }
+
+  // The caller's buffer may be pinned (device-visible) or pageable.
+  // Pinned path: translate; pageable would need a bounce buffer — the
+  // daemon and bench always use alloc_pinned, so reject pageable here.
+  uint8_t* buf_device(uint8_t* host) {
+    void* dev = nullptr;
+    hipError_t err = hipHostGetDevicePointer(&dev, host, 0);
+    if (err != hipSuccess) {
+      throw std::runtime_error(
+          "hipstore: I/O buffer is not pinned host memory (use alloc_pinned)");
+    }
+    return static_cast<uint8_t*>(dev);
+  }
+
+  hipEvent_t get_event() {
+    if (!event_pool_.empty()) {
+      hipEvent_t e = event_pool_.back();
+      event_pool_.pop_back();
+      return e;
+    }
+    hipEvent_t e = nullptr;
+    HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+    return e;
+  }
+
+  uint8_t* base_;
+  uint64_t size_;
+  int device_ = 0;
+  hipStream_t stream_ = nullptr;
+  BlockDesc* ring_ = nullptr;      // pinned host
+  BlockDesc* ring_dev_ = nullptr;  // GPU view of the ring
+  uint32_t ring_head_ = 0;
+  uint32_t used_slots_ = 0;
+  std::vector<PendingIo> pending_;
+  std::deque<InflightBatch> inflight_;
+  std::vector<hipEvent_t> event_pool_;
+};
+
+class HbmBdev : public Bdev {
+ public:
+  HbmBdev(const std::string& name, uint64_t block_size, uint64_t num_blocks,
+          int device)
+      : Bdev(name, "Malloc disk", block_size, num_blocks), device_(device) {
+    HIP_CHECK(hipSetDevice(device));
+    HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&base_), size_bytes()));
+    HIP_CHECK(hipMemset(base_, 0, size_bytes()));
+  }
+
+  ~HbmBdev() override {
+    (void)hipSetDevice(device_);
+    (void)hipFree(base_);
+  }
+
+  int device() const { return device_; }
+  void* device_base() override { return base_; }
+  int gpu_device() const override { return device_; }
+
+  std::shared_ptr<IoChannel> get_channel() override {
+    HIP_CHECK(hipSetDevice(device_));
+    return std::make_shared<HbmChannel>(device_, base_, size_bytes());
+  }
+
+  void submit(IoChannel* ch, IoRequest req) override {
+    auto* channel = static_cast<HbmChannel*>(ch);
+    int status = kIoOk;
+    if (req.op != IoOp::kFlush) {
+      // A request larger than half the descriptor ring could never form
+      // a batch; reject it rather than deadlock the queue.
+      if (!check_bounds(req) ||
+          req.length > static_cast<uint64_t>(kMaxBatchTiles) * kTileBytes) {
+        status = kIoInvalid;
+      }
+    }
+    channel->enqueue(std::move(req), status);
+  }
+
+  int poll(IoChannel* ch) override {
+    auto* channel = static_cast<HbmChannel*>(ch);
+    (void)hipSetDevice(device_);  // launches must come from the bdev's device
+    int completed = channel->retire(/*wait=*/false);
+    channel->kick();
+    completed += channel->retire(/*wait=*/false);
+    return completed;
+  }
+
+ private:
+  uint8_t* base_ = nullptr;
+  int device_;
+};
+
+}  // namespace
+
+BdevPtr create_hbm_bdev(const std::string& name, uint64_t block_size,
+                        uint64_t num_blocks, int device) {
+  if (!gpu_available()) {
+    throw std::runtime_error("hipstore: no HIP device for HBM bdev");
+  }
+  if (device < 0 || device >= gpu_device_count()) {
+    throw std::runtime_error("hipstore: bad device index");
+  }
+  return std::make_shared<HbmBdev>(name, block_size, num_blocks, device);
+}
+
+void crc32c_hbm_blocks(Bdev* bdev, uint64_t offset, uint32_t block_size,
+                       uint32_t count, uint32_t* out) {
+  if (block_size % 4 != 0) {
+    throw std::runtime_error("crc32c: block_size must be a multiple of 4");
+  }
+  if (offset % bdev->block_size() != 0 ||
+      offset + static_cast<uint64_t>(block_size) * count > bdev->size_bytes()) {
+    throw std::runtime_error("crc32c: range out of bounds");
+  }
+  void* base = bdev->device_base();
+  if (base == nullptr) {
+    crc32c_cpu_fallback(bdev, offset, block_size, count, out);
+    return;
+  }
+  HIP_CHECK(hipSetDevice(bdev->gpu_device()));
+  uint32_t* out_dev = static_cast<uint32_t*>(alloc_pinned(count * 4));
+  const uint32_t grid = (count + 255) / 256;
+  hipLaunchKernelGGL(k_crc32c_blocks, dim3(grid), dim3(256), 0, nullptr,
+                     static_cast<const uint8_t*>(base) + offset, block_size,
+                     count, device_view(out_dev));
+  HIP_CHECK(hipDeviceSynchronize());
+  memcpy(out, out_dev, count * 4);
+  free_pinned(out_dev);
+}
+
+// ---------------------------------------------------------------------------
+// Synchronous helpers
+// ---------------------------------------------------------------------------
+
+namespace {
+
+int run_sync(Bdev* bdev, IoRequest req) {
+  auto channel = bdev->get_channel();
+  int result = kIoFailed;
+  bool done = false;
+  req.on_complete = [&](int status) {
+    result = status;
+    done = true;
+  };
+  bdev->submit(channel.get(), std::move(req));
+  while (!done) bdev->poll(channel.get());
+  return result;
+}
+
+}  // namespace
+
+int bdev_read_sync(Bdev* bdev, uint64_t offset, void* buf, uint64_t len) {
+  IoRequest req;
+  req.op = IoOp::kRead;
+  req.offset = offset;
+  req.length = len;
+  req.buffer = buf;
+  return run_sync(bdev, std::move(req));
+}
+
+int bdev_write_sync(Bdev* bdev, uint64_t offset, const void* buf, uint64_t len) {
+  IoRequest req;
+  req.op = IoOp::kWrite;
+  req.offset = offset;
+  req.length = len;
+  req.buffer = const_cast<void*>(buf);
+  return run_sync(bdev, std::move(req));
+}
+
+int bdev_fill_sync(Bdev* bdev, uint64_t offset, uint8_t value, uint64_t len) {
+  IoRequest req;
+  req.op = IoOp::kFill;
+  req.offset = offset;
+  req.length = len;
+  req.fill = value;
+  return run_sync(bdev, std::move(req));
+}
+
+// ---------------------------------------------------------------------------
+// bdevperf harness
+// ---------------------------------------------------------------------------
+
+PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
+                        uint32_t io_size, uint32_t queue_depth,
+                        int num_queues, double seconds, uint64_t max_ios) {
+  using clock = std::chrono::steady_clock;
+  const bool do_read = workload != "randwrite";
+  const bool do_write = workload == "randwrite" || workload == "randrw";
+  const uint64_t units = bdev->size_bytes() / io_size;
+  if (units == 0) throw std::runtime_error("bdev smaller than io_size");
+
+  struct QueueStats {
+    uint64_t ios = 0;
+    std::vector<uint32_t> lat_us;
+  };
+  std::vector<QueueStats> stats(num_queues);
+  std::vector<std::thread> threads;
+  std::atomic<bool> failed{false};
+  const auto t0 = clock::now();
+  const auto deadline = t0 + std::chrono::duration<double>(seconds);
+  const uint64_t per_queue_cap =
+      max_ios ? (max_ios + num_queues - 1) / num_queues : 0;
+
+  for (int q = 0; q < num_queues; ++q) {
+    threads.emplace_back([&, q] {
+      try {
+        auto channel = bdev->get_channel();
+        uint8_t* buf = static_cast<uint8_t*>(
+            alloc_pinned(static_cast<size_t>(io_size) * queue_depth));
+        std::mt19937_64 rng(0x9E3779B97F4A7C15ULL ^ (q * 0x8DA6B343));
+        QueueStats& st = stats[q];
+        st.lat_us.reserve(1 << 20);
+        std::vector<clock::time_point> submit_ts(queue_depth);
+        uint32_t inflight = 0;
+        bool stopping = false;
+        uint64_t submitted = 0;
+
+        std::function<void(uint32_t)> submit_slot = [&](uint32_t slot) {
+          IoRequest req;
+          bool write = do_write && (!do_read || (rng() & 1));
+          req.op = write ? IoOp::kWrite : IoOp::kRead;
+          req.offset = (rng() % units) * io_size;
+          req.length = io_size;
+          req.buffer = buf + static_cast<size_t>(slot) * io_size;
+          submit_ts[slot] = clock::now();
+          req.on_complete = [&, slot](int status) {
+            if (status != kIoOk) failed.store(true);
+            const auto now = clock::now();
+            st.lat_us.push_back(static_cast<uint32_t>(
+                std::chrono::duration_cast<std::chrono::microseconds>(
+                    now - submit_ts[slot]).count()));
+            ++st.ios;
+            --inflight;
+            if (!stopping) {
+              submit_slot(slot);
+              ++inflight;
+              ++submitted;
+            }
+          };
+          bdev->submit(channel.get(), std::move(req));
+        };
+
+        for (uint32_t slot = 0; slot < queue_depth; ++slot) {
+          submit_slot(slot);
+          ++inflight;
+          ++submitted;
+        }
+        while (true) {
+          bdev->poll(channel.get());
+          if (!stopping &&
+              (clock::now() >= deadline ||
+               (per_queue_cap && submitted >= per_queue_cap))) {
+            stopping = true;
+          }
+          if (stopping && inflight == 0) break;
+        }
+        free_pinned(buf);
+      } catch (const std::exception&) {
+        failed.store(true);
+      }
+    });
+  }
+  for (auto& thread : threads) thread.join();
+  const double elapsed =
+      std::chrono::duration<double>(clock::now() - t0).count();
+  if (failed.load()) throw std::runtime_error("bdevperf: I/O failures");
+
+  PerfResult result;
+  result.seconds = elapsed;
+  std::vector<uint32_t> all;
+  for (auto& st : stats) {
+    result.io_count += st.ios;
+    all.insert(all.end(), st.lat_us.begin(), st.lat_us.end());
+  }
+  result.iops = result.io_count / elapsed;
+  result.throughput_mbps =
+      result.io_count * static_cast<double>(io_size) / elapsed / 1e6;
+  if (!all.empty()) {
+    std::sort(all.begin(), all.end());
+    double sum = 0;
+    for (uint32_t v : all) sum += v;
+    auto pct = [&](double p) {
+      size_t i = static_cast<size_t>(p * (all.size() - 1));
+      return static_cast<double>(all[i]);
+    };
+    result.lat_avg_us = sum / all.size();
+    result.lat_p50_us = pct(0.50);
+    result.lat_p90_us = pct(0.90);
+    result.lat_p99_us = pct(0.99);
+    result.lat_p999_us = pct(0.999);
+    result.lat_max_us = all.back();
+  }
+  return result;
+}
+
+}  // namespace hipstore

@@ -1,0 +1,87 @@
+// hipstored — the MI355X data-path daemon.
+//
+// Replaces the reference's SPDK vhost app (reference app/vhost/vhost.c):
+// serves the SPDK-compatible JSON-RPC socket, backs malloc bdevs with
+// MI355X HBM3E (hipMalloc) and moves blocks with LDS-staged HIP kernels
+// on per-queue streams. --cpu forces host-RAM bdevs (CI / config 1).
+
+#include <getopt.h>
+#include <signal.h>
+#include <unistd.h>
+
+#include <condition_variable>
+#include <cstdio>
+#include <cstring>
+#include <mutex>
+
+#include "hipstore/engine.h"
+#include "hipstore/nbd.h"
+#include "hipstore/rpc.h"
+
+namespace {
+
+std::mutex g_mutex;
+std::condition_variable g_cv;
+bool g_stop = false;
+
+void handle_signal(int) {
+  {
+    std::lock_guard<std::mutex> lock(g_mutex);
+    g_stop = true;
+  }
+  g_cv.notify_all();
+}
+
+void usage(const char* argv0) {
+  fprintf(stderr,
+          "usage: %s [-S socket] [-d device] [-C] \n"
+          "  -S PATH   JSON-RPC unix socket (default /var/tmp/hipstored.sock)\n"
+          "  -d N      HIP device index backing malloc bdevs (default 0)\n"
+          "  -C        CPU mode: host-RAM bdevs even when a GPU is present\n",
+          argv0);
+}
+
+}  // namespace
+
+int main(int argc, char** argv) {
+  std::string socket_path = "/var/tmp/hipstored.sock";
+  int device = 0;
+  bool cpu_only = false;
+
+  int opt;
+  while ((opt = getopt(argc, argv, "S:d:Ch")) != -1) {
+    switch (opt) {
+      case 'S': socket_path = optarg; break;
+      case 'd': device = atoi(optarg); break;
+      case 'C': cpu_only = true; break;
+      case 'h': usage(argv[0]); return 0;
+      default: usage(argv[0]); return 2;
+    }
+  }
+
+  const bool use_hbm = !cpu_only && hipstore::gpu_available();
+  fprintf(stderr, "hipstored: socket=%s mode=%s device=%d gpus=%d\n",
+          socket_path.c_str(), use_hbm ? "hbm" : "cpu", device,
+          hipstore::gpu_device_count());
+
+  hipstore::RpcServer server(socket_path);
+  hipstore::register_storage_methods(&server, use_hbm, device);
+  try {
+    server.start();
+  } catch (const std::exception& e) {
+    fprintf(stderr, "hipstored: %s\n", e.what());
+    return 1;
+  }
+  fprintf(stderr, "hipstored: ready\n");
+
+  signal(SIGINT, handle_signal);
+  signal(SIGTERM, handle_signal);
+  {
+    std::unique_lock<std::mutex> lock(g_mutex);
+    g_cv.wait(lock, [] { return g_stop; });
+  }
+  fprintf(stderr, "hipstored: shutting down\n");
+  hipstore::nbd_stop_all();
+  server.stop();
+  return 0;
+}

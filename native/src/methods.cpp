@@ -1,0 +1,354 @@
+// The SPDK-wire-compatible RPC method set.
+//
+// Exactly the methods the reference's Go client invokes
+// (reference pkg/spdk/spdk.go), same parameter and result shapes, plus
+// two native extras (get_rpc_methods, perf_run). "Not found" errors use
+// kInvalidParams like SPDK does; all reference callers tolerate that
+// (reference local.go:53-57, controller.go:76,204,239).
+
+#include <cinttypes>
+#include <mutex>
+
+#include "hipstore/bdev.h"
+#include "hipstore/engine.h"
+#include "hipstore/json.h"
+#include "hipstore/nbd.h"
+#include "hipstore/rpc.h"
+
+namespace hipstore {
+
+namespace {
+
+constexpr int kMaxScsiTargets = 8;  // reference vhost_scsi.c:952 default
+
+struct ScsiLun {
+  int id;
+  std::string bdev_name;
+};
+
+struct ScsiTarget {
+  bool used = false;
+  std::string target_name;
+  std::vector<ScsiLun> luns;
+};
+
+struct VhostController {
+  std::string cpumask;
+  ScsiTarget targets[kMaxScsiTargets];
+};
+
+struct VhostState {
+  std::mutex mutex;
+  std::map<std::string, VhostController> controllers;
+};
+
+VhostState& vhost_state() {
+  static VhostState s;
+  return s;
+}
+
+int g_malloc_seq = 0;
+std::mutex g_name_mutex;
+
+Json bdev_to_json(const BdevPtr& bdev) {
+  JsonObject io_types;
+  io_types["read"] = Json(true);
+  io_types["write"] = Json(true);
+  io_types["unmap"] = Json(true);
+  io_types["write_zeroes"] = Json(true);
+  io_types["flush"] = Json(true);
+  io_types["reset"] = Json(false);
+  io_types["nvme_admin"] = Json(false);
+  io_types["nvme_io"] = Json(false);
+  JsonObject o;
+  o["name"] = Json(bdev->name());
+  o["product_name"] = Json(bdev->product_name());
+  o["uuid"] = Json(bdev->uuid());
+  o["block_size"] = Json(static_cast<int64_t>(bdev->block_size()));
+  o["num_blocks"] = Json(static_cast<int64_t>(bdev->num_blocks()));
+  o["claimed"] = Json(bdev->claimed());
+  o["supported_io_types"] = Json(std::move(io_types));
+  if (bdev->gpu_device() >= 0) {
+    JsonObject hbm;
+    hbm["device"] = Json(static_cast<int64_t>(bdev->gpu_device()));
+    hbm["pci_address"] = Json(gpu_pci_address(bdev->gpu_device()));
+    JsonObject ds;
+    ds["hbm"] = Json(std::move(hbm));
+    o["driver_specific"] = Json(std::move(ds));
+  }
+  return Json(std::move(o));
+}
+
+[[noreturn]] void not_found(const std::string& what) {
+  throw RpcError{kInvalidParams, what + " does not exist"};
+}
+
+}  // namespace
+
+void register_storage_methods(RpcServer* server, bool use_hbm, int device) {
+  auto& manager = BdevManager::instance();
+
+  server->register_method("get_bdevs", [&manager](const Json& p) {
+    const std::string name = p.get_string("name");
+    JsonArray out;
+    if (!name.empty()) {
+      BdevPtr bdev = manager.find(name);
+      if (!bdev) not_found("bdev " + name);
+      out.push_back(bdev_to_json(bdev));
+    } else {
+      for (const auto& bdev : manager.list()) out.push_back(bdev_to_json(bdev));
+    }
+    return Json(std::move(out));
+  });
+
+  server->register_method("delete_bdev", [&manager](const Json& p) {
+    const std::string name = p.get_string("name");
+    BdevPtr bdev = manager.find(name);
+    if (!bdev) not_found("bdev " + name);
+    if (bdev->claimed()) {
+      throw RpcError{kInvalidParams, "bdev " + name + " is claimed"};
+    }
+    manager.remove(name);
+    return Json(JsonObject{});
+  });
+
+  server->register_method(
+      "construct_malloc_bdev", [&manager, use_hbm, device](const Json& p) {
+        const int64_t num_blocks = p.get_int("num_blocks");
+        const int64_t block_size = p.get_int("block_size");
+        if (num_blocks <= 0 || block_size <= 0 || block_size % 512 != 0) {
+          throw RpcError{kInvalidParams, "invalid num_blocks/block_size"};
+        }
+        std::string name = p.get_string("name");
+        {
+          std::lock_guard<std::mutex> lock(g_name_mutex);
+          if (name.empty()) name = "Malloc" + std::to_string(g_malloc_seq++);
+        }
+        if (manager.find(name)) {
+          throw RpcError{kInvalidParams, "bdev " + name + " already exists"};
+        }
+        BdevPtr bdev;
+        if (use_hbm && gpu_available()) {
+          bdev = create_hbm_bdev(name, block_size, num_blocks, device);
+        } else {
+          bdev = create_malloc_bdev(name, block_size, num_blocks);
+        }
+        manager.add(bdev);
+        return Json(name);
+      });
+
+  server->register_method(
+      "construct_rbd_bdev", [&manager, use_hbm, device](const Json& p) {
+        // Ceph RBD path. Without a reachable cluster (this environment
+        // has no network) the image is emulated by a local backing
+        // store of `config.emu_size_mb` (default 1 GiB) — the
+        // ceph-csi parameter plumbing (user_id/pool/image/monitors/
+        // secret) is exercised end-to-end either way.
+        const std::string pool = p.get_string("pool_name");
+        const std::string image = p.get_string("rbd_name");
+        const int64_t block_size = p.get_int("block_size", 512);
+        if (pool.empty() || image.empty()) {
+          throw RpcError{kInvalidParams, "pool_name and rbd_name required"};
+        }
+        if (block_size <= 0 || block_size % 512 != 0) {
+          throw RpcError{kInvalidParams, "invalid block_size"};
+        }
+        std::string name = p.get_string("name");
+        if (name.empty()) name = pool + "/" + image;
+        if (manager.find(name)) {
+          throw RpcError{kInvalidParams, "bdev " + name + " already exists"};
+        }
+        int64_t emu_mb = 1024;
+        if (const Json* config = p.get("config")) {
+          if (config->is_object()) {
+            emu_mb = config->get_int("emu_size_mb", emu_mb);
+          }
+        }
+        const uint64_t num_blocks = emu_mb * 1024 * 1024 / block_size;
+        BdevPtr inner;
+        if (use_hbm && gpu_available()) {
+          inner = create_hbm_bdev(name, block_size, num_blocks, device);
+        } else {
+          inner = create_malloc_bdev(name, block_size, num_blocks);
+        }
+        // Re-badge as an RBD disk: UnmapVolume deletes these
+        // (product_name != "Malloc disk", reference controller.go:205).
+        inner->set_product("Ceph Rbd Disk");
+        manager.add(inner);
+        return Json(name);
+      });
+
+  server->register_method("start_nbd_disk", [&manager](const Json& p) {
+    const std::string bdev_name = p.get_string("bdev_name");
+    const std::string device = p.get_string("nbd_device");
+    if (!manager.find(bdev_name)) not_found("bdev " + bdev_name);
+    nbd_start(bdev_name, device);
+    return Json(true);
+  });
+
+  server->register_method("get_nbd_disks", [](const Json&) {
+    JsonArray out;
+    for (const auto& [bdev_name, device] : nbd_list()) {
+      JsonObject o;
+      o["bdev_name"] = Json(bdev_name);
+      o["nbd_device"] = Json(device);
+      out.push_back(Json(std::move(o)));
+    }
+    return Json(std::move(out));
+  });
+
+  server->register_method("stop_nbd_disk", [](const Json& p) {
+    nbd_stop(p.get_string("nbd_device"));
+    return Json(true);
+  });
+
+  server->register_method("construct_vhost_scsi_controller", [](const Json& p) {
+    const std::string name = p.get_string("ctrlr");
+    if (name.empty()) throw RpcError{kInvalidParams, "ctrlr required"};
+    auto& state = vhost_state();
+    std::lock_guard<std::mutex> lock(state.mutex);
+    if (state.controllers.count(name)) {
+      throw RpcError{kInvalidParams, "controller " + name + " already exists"};
+    }
+    state.controllers[name].cpumask = p.get_string("cpumask", "0x1");
+    return Json(JsonObject{});
+  });
+
+  server->register_method("add_vhost_scsi_lun", [&manager](const Json& p) {
+    const std::string ctrlr = p.get_string("ctrlr");
+    const int64_t target_num = p.get_int("scsi_target_num", -1);
+    const std::string bdev_name = p.get_string("bdev_name");
+    auto& state = vhost_state();
+    std::lock_guard<std::mutex> lock(state.mutex);
+    auto it = state.controllers.find(ctrlr);
+    if (it == state.controllers.end()) not_found("controller " + ctrlr);
+    if (target_num < 0 || target_num >= kMaxScsiTargets) {
+      throw RpcError{kInvalidParams, "scsi_target_num out of range"};
+    }
+    BdevPtr bdev = manager.find(bdev_name);
+    if (!bdev) not_found("bdev " + bdev_name);
+    ScsiTarget& target = it->second.targets[target_num];
+    if (target.used) {
+      throw RpcError{kInvalidParams, "target already occupied"};
+    }
+    if (!bdev->claim()) {
+      throw RpcError{kInvalidParams, "bdev " + bdev_name + " is claimed"};
+    }
+    target.used = true;
+    target.target_name = "Target " + std::to_string(target_num);
+    target.luns = {ScsiLun{0, bdev_name}};
+    return Json(JsonObject{});
+  });
+
+  server->register_method("remove_vhost_scsi_target", [&manager](const Json& p) {
+    const std::string ctrlr = p.get_string("ctrlr");
+    const int64_t target_num = p.get_int("scsi_target_num", -1);
+    auto& state = vhost_state();
+    std::lock_guard<std::mutex> lock(state.mutex);
+    auto it = state.controllers.find(ctrlr);
+    if (it == state.controllers.end()) not_found("controller " + ctrlr);
+    if (target_num < 0 || target_num >= kMaxScsiTargets ||
+        !it->second.targets[target_num].used) {
+      throw RpcError{kInvalidParams, "no such target"};
+    }
+    ScsiTarget& target = it->second.targets[target_num];
+    for (const ScsiLun& lun : target.luns) {
+      if (BdevPtr bdev = manager.find(lun.bdev_name)) bdev->release();
+    }
+    target = ScsiTarget{};
+    return Json(JsonObject{});
+  });
+
+  server->register_method("remove_vhost_controller", [&manager](const Json& p) {
+    const std::string ctrlr = p.get_string("ctrlr");
+    auto& state = vhost_state();
+    std::lock_guard<std::mutex> lock(state.mutex);
+    auto it = state.controllers.find(ctrlr);
+    if (it == state.controllers.end()) not_found("controller " + ctrlr);
+    for (ScsiTarget& target : it->second.targets) {
+      for (const ScsiLun& lun : target.luns) {
+        if (BdevPtr bdev = manager.find(lun.bdev_name)) bdev->release();
+      }
+    }
+    state.controllers.erase(it);
+    return Json(JsonObject{});
+  });
+
+  server->register_method("get_vhost_controllers", [](const Json&) {
+    auto& state = vhost_state();
+    std::lock_guard<std::mutex> lock(state.mutex);
+    JsonArray out;
+    for (const auto& [name, ctrl] : state.controllers) {
+      JsonArray scsi;
+      for (int t = 0; t < kMaxScsiTargets; ++t) {
+        const ScsiTarget& target = ctrl.targets[t];
+        if (!target.used) continue;
+        JsonArray luns;
+        for (const ScsiLun& lun : target.luns) {
+          JsonObject lo;
+          lo["id"] = Json(static_cast<int64_t>(lun.id));
+          lo["bdev_name"] = Json(lun.bdev_name);
+          luns.push_back(Json(std::move(lo)));
+        }
+        JsonObject to;
+        to["target_name"] = Json(target.target_name);
+        to["id"] = Json(static_cast<int64_t>(t));
+        to["scsi_dev_num"] = Json(static_cast<int64_t>(t));
+        to["luns"] = Json(std::move(luns));
+        scsi.push_back(Json(std::move(to)));
+      }
+      JsonObject backend;
+      backend["scsi"] = Json(std::move(scsi));
+      JsonObject o;
+      o["ctrlr"] = Json(name);
+      o["cpumask"] = Json(ctrl.cpumask);
+      o["backend_specific"] = Json(std::move(backend));
+      out.push_back(Json(std::move(o)));
+    }
+    return Json(std::move(out));
+  });
+
+  // --- native extras -------------------------------------------------------
+
+  server->register_method("get_rpc_methods", [](const Json&) {
+    JsonArray out;
+    for (const char* name :
+         {"get_bdevs", "delete_bdev", "construct_malloc_bdev",
+          "construct_rbd_bdev", "start_nbd_disk", "get_nbd_disks",
+          "stop_nbd_disk", "construct_vhost_scsi_controller",
+          "add_vhost_scsi_lun", "remove_vhost_scsi_target",
+          "remove_vhost_controller", "get_vhost_controllers",
+          "get_rpc_methods", "perf_run"}) {
+      out.push_back(Json(name));
+    }
+    return Json(std::move(out));
+  });
+
+  server->register_method("perf_run", [&manager](const Json& p) {
+    // In-daemon bdevperf: the fio-shaped measurement loop runs next to
+    // the data path, so the RPC socket is not in the hot path.
+    BdevPtr bdev = manager.find(p.get_string("bdev_name"));
+    if (!bdev) not_found("bdev " + p.get_string("bdev_name"));
+    PerfResult r = run_bdevperf(
+        bdev.get(), p.get_string("workload", "randread"),
+        static_cast<uint32_t>(p.get_int("io_size", 4096)),
+        static_cast<uint32_t>(p.get_int("queue_depth", 32)),
+        static_cast<int>(p.get_int("num_queues", 1)),
+        p.get("seconds") ? p.get("seconds")->as_double() : 2.0,
+        static_cast<uint64_t>(p.get_int("max_ios", 0)));
+    JsonObject o;
+    o["seconds"] = Json(r.seconds);
+    o["io_count"] = Json(static_cast<int64_t>(r.io_count));
+    o["iops"] = Json(r.iops);
+    o["throughput_mbps"] = Json(r.throughput_mbps);
+    o["lat_avg_us"] = Json(r.lat_avg_us);
+    o["lat_p50_us"] = Json(r.lat_p50_us);
+    o["lat_p90_us"] = Json(r.lat_p90_us);
+    o["lat_p99_us"] = Json(r.lat_p99_us);
+    o["lat_p999_us"] = Json(r.lat_p999_us);
+    o["lat_max_us"] = Json(r.lat_max_us);
+    return Json(std::move(o));
+  });
+}
+
+}  // namespace hipstore

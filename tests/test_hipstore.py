@@ -1,0 +1,126 @@
+"""hipstored daemon + client integration (CPU mode).
+
+Counterpart of the reference's pkg/spdk/spdk_test.go: every RPC wrapper
+against a real daemon, including SPDK error-code behavior.
+"""
+
+import pytest
+
+from oim_amd import hipstore
+from oim_amd.hipstore import RpcError
+
+from fixtures import hipstored  # noqa: F401  (fixture)
+
+
+@pytest.fixture
+def client(hipstored):  # noqa: F811
+    c = hipstore.Client(hipstored.socket_path)
+    yield c
+    c.close()
+
+
+class TestBdevs:
+    def test_construct_get_delete(self, client):
+        name = hipstore.construct_malloc_bdev(client, num_blocks=2048, block_size=512,
+                                              name="test-bdev")
+        assert name == "test-bdev"
+        bdevs = hipstore.get_bdevs(client, name="test-bdev")
+        assert len(bdevs) == 1
+        assert bdevs[0].product_name == "Malloc disk"
+        assert bdevs[0].size_bytes == 1024 * 1024
+        assert not bdevs[0].claimed
+        hipstore.delete_bdev(client, "test-bdev")
+        with pytest.raises(RpcError) as excinfo:
+            hipstore.get_bdevs(client, name="test-bdev")
+        assert excinfo.value.is_not_found()
+
+    def test_auto_name(self, client):
+        name = hipstore.construct_malloc_bdev(client, num_blocks=1024, block_size=512)
+        assert name.startswith("Malloc")
+        hipstore.delete_bdev(client, name)
+
+    def test_duplicate_name_rejected(self, client):
+        hipstore.construct_malloc_bdev(client, 1024, 512, name="dup")
+        with pytest.raises(RpcError):
+            hipstore.construct_malloc_bdev(client, 1024, 512, name="dup")
+        hipstore.delete_bdev(client, "dup")
+
+    def test_invalid_block_size(self, client):
+        with pytest.raises(RpcError) as excinfo:
+            hipstore.construct_malloc_bdev(client, 1024, 100)
+        assert excinfo.value.code == hipstore.ERROR_INVALID_PARAMS
+
+    def test_delete_missing(self, client):
+        with pytest.raises(RpcError) as excinfo:
+            hipstore.delete_bdev(client, "no-such")
+        assert excinfo.value.is_not_found()
+
+    def test_rbd_emulation(self, client):
+        name = hipstore.construct_rbd_bdev(
+            client, pool_name="rbd", rbd_name="img0", block_size=512,
+            user_id="admin", config={"mon_host": "127.0.0.1:6789", "key": "x"},
+        )
+        bdevs = hipstore.get_bdevs(client, name=name)
+        assert bdevs[0].product_name == "Ceph Rbd Disk"
+        hipstore.delete_bdev(client, name)
+
+
+class TestVhost:
+    def test_controller_lifecycle(self, client):
+        hipstore.construct_vhost_scsi_controller(client, "vhost.0")
+        hipstore.construct_malloc_bdev(client, 1024, 512, name="vol-a")
+        hipstore.add_vhost_scsi_lun(client, "vhost.0", 2, "vol-a")
+        # bdev is now claimed
+        assert hipstore.get_bdevs(client, "vol-a")[0].claimed
+        with pytest.raises(RpcError):
+            hipstore.delete_bdev(client, "vol-a")
+        controllers = hipstore.get_vhost_controllers(client)
+        assert len(controllers) == 1
+        ctrl = controllers[0]
+        assert ctrl.controller == "vhost.0"
+        assert len(ctrl.scsi_targets) == 1
+        target = ctrl.scsi_targets[0]
+        assert target.scsi_dev_num == 2
+        assert target.luns[0].bdev_name == "vol-a"
+        assert target.luns[0].lun == 0
+        # occupied target rejected
+        hipstore.construct_malloc_bdev(client, 1024, 512, name="vol-b")
+        with pytest.raises(RpcError):
+            hipstore.add_vhost_scsi_lun(client, "vhost.0", 2, "vol-b")
+        hipstore.remove_vhost_scsi_target(client, "vhost.0", 2)
+        assert not hipstore.get_bdevs(client, "vol-a")[0].claimed
+        hipstore.remove_vhost_controller(client, "vhost.0")
+        assert hipstore.get_vhost_controllers(client) == []
+
+    def test_missing_bdev(self, client):
+        hipstore.construct_vhost_scsi_controller(client, "vhost.1")
+        with pytest.raises(RpcError) as excinfo:
+            hipstore.add_vhost_scsi_lun(client, "vhost.1", 0, "ghost")
+        assert excinfo.value.is_not_found()
+        hipstore.remove_vhost_controller(client, "vhost.1")
+
+
+class TestPerf:
+    def test_perf_run_cpu(self, client):
+        hipstore.construct_malloc_bdev(client, 8192, 512, name="perf0")
+        result = hipstore.perf_run(client, "perf0", io_size=4096,
+                                   queue_depth=8, num_queues=1, seconds=0.2)
+        assert result["io_count"] > 0
+        assert result["iops"] > 1000
+        assert result["lat_p99_us"] >= result["lat_p50_us"]
+        hipstore.delete_bdev(client, "perf0")
+
+
+class TestProtocol:
+    def test_unknown_method(self, client):
+        with pytest.raises(RpcError) as excinfo:
+            client.invoke("bogus_method")
+        assert excinfo.value.code == hipstore.client.ERROR_METHOD_NOT_FOUND
+
+    def test_pipelined_requests(self, hipstored):  # noqa: F811
+        # Two clients on one daemon; interleaved calls.
+        with hipstore.Client(hipstored.socket_path) as a, \
+             hipstore.Client(hipstored.socket_path) as b:
+            hipstore.construct_malloc_bdev(a, 1024, 512, name="pipe")
+            assert hipstore.get_bdevs(b, "pipe")[0].name == "pipe"
+            hipstore.delete_bdev(b, "pipe")
