@@ -1,0 +1,153 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: 4 KiB random-read IOPS + p99 latency on the
+HBM-resident Malloc bdev (BASELINE.json metric), fio-shaped synthetic
+I/O at QD=32 per queue.
+
+One rank per GPU (torch.distributed over RCCL when launched via
+torch.distributed.run); each rank drives its own GPU's bdev — the
+"one accelerator card per GPU" model (weak scaling). A *step* is a
+fixed batch of STEP_IOS random 4 KiB reads per GPU; rank 0 prints one
+JSON line with the whole-job aggregate.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+"""
+
+import argparse
+import json
+import os
+import sys
+
+REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO_ROOT)
+
+STEP_IOS = 131072  # 4 KiB reads per GPU per step (512 MiB moved)
+
+
+def main() -> int:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=20)
+    parser.add_argument("--warmup", type=int, default=5)
+    parser.add_argument("--io-size", type=int, default=4096)
+    parser.add_argument("--queue-depth", type=int, default=32)
+    parser.add_argument("--num-queues", type=int, default=8)
+    parser.add_argument("--bdev-gb", type=float, default=8.0)
+    parser.add_argument("--workload", default="randread")
+    args = parser.parse_args()
+
+    import torch
+    import torch.distributed as dist
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world_size > 1
+    use_gpu = torch.cuda.is_available()
+
+    if distributed:
+        backend = "nccl" if use_gpu else "gloo"
+        dist.init_process_group(backend=backend)
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+
+    from oim_amd import _hipstore as hs
+
+    block = args.io_size
+    num_blocks = int(args.bdev_gb * (1 << 30)) // block
+    if use_gpu:
+        if not hs.gpu_available():
+            raise RuntimeError(
+                "torch sees a GPU but oim_amd._hipstore does not — "
+                "native extension not loaded; rebuild with `make`")
+        bdev = hs.create_hbm_bdev(f"bench-{rank}", block, num_blocks,
+                                  device=local_rank)
+        backend_name = "hbm"
+    else:
+        # CPU fallback so the bench is testable in the no-GPU container.
+        num_blocks = min(num_blocks, (1 << 30) // block)
+        bdev = hs.create_malloc_bdev(f"bench-{rank}", block, num_blocks)
+        backend_name = "cpu"
+
+    # Touch every block once so HBM pages are resident (fill pattern).
+    bdev.fill(0, 0x5A, bdev.size_bytes)
+
+    def run_step(n_ios: int) -> dict:
+        return hs.run_bdevperf(bdev, args.workload, args.io_size,
+                               args.queue_depth, args.num_queues,
+                               seconds=3600.0, max_ios=n_ios)
+
+    def barrier():
+        if distributed:
+            dist.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        run_step(STEP_IOS)
+
+    barrier()
+    import time
+    t0 = time.perf_counter()
+    p99s, p50s = [], []
+    ios_done = 0
+    for _ in range(args.steps):
+        r = run_step(STEP_IOS)
+        ios_done += r["io_count"]
+        p99s.append(r["lat_p99_us"])
+        p50s.append(r["lat_p50_us"])
+    barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks of elapsed; aggregate IOs over ranks.
+    if distributed:
+        stats = torch.tensor([elapsed, float(ios_done), max(p99s)],
+                             dtype=torch.float64)
+        if use_gpu:
+            stats = stats.cuda()
+        gathered = [torch.zeros_like(stats) for _ in range(world_size)]
+        dist.all_gather(gathered, stats)
+        elapsed = max(g[0].item() for g in gathered)
+        total_ios = sum(g[1].item() for g in gathered)
+        p99 = max(g[2].item() for g in gathered)
+    else:
+        total_ios = float(ios_done)
+        p99 = max(p99s)
+
+    value = total_ios / elapsed
+    if rank == 0:
+        out = {
+            "metric": "4KiB_randread_IOPS",
+            "value": round(value, 1),
+            "unit": "IOPS",
+            "n_gpus": world_size if distributed else args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed * 1000 / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "uint8",
+            "data": "synthetic",
+            "config": {
+                "model": "hbm-malloc-bdev",
+                "global_batch": args.queue_depth * args.num_queues,
+                "seq_len": args.io_size,
+                "parallelism": f"1-card-per-gpu-x{world_size if distributed else 1}",
+                "workload": args.workload,
+                "io_size": args.io_size,
+                "queue_depth": args.queue_depth,
+                "num_queues": args.num_queues,
+                "bdev_gb": args.bdev_gb,
+                "step_ios_per_gpu": STEP_IOS,
+                "backend": backend_name,
+                "p99_us": round(p99, 1),
+            },
+        }
+        print(json.dumps(out))
+    if distributed:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,121 @@
+"""MI355X GPU tests: HBM bdev numerics + kernels vs CPU references.
+
+All marked `gpu`; run via `pytest -m gpu` on a GPU box. Every data-path
+result is compared against plain host-side byte operations (the CPU
+reference), and CRC32C against the bit-exact software table.
+"""
+
+import os
+import random
+
+import pytest
+
+from oim_amd import _hipstore as hs
+from oim_amd import hipstore
+
+from fixtures import launch_hipstored
+
+pytestmark = pytest.mark.gpu
+
+needs_gpu = pytest.mark.skipif(not hs.gpu_available(), reason="no HIP device")
+
+
+@needs_gpu
+class TestHbmBdev:
+    @pytest.fixture(scope="class")
+    def bdev(self):
+        # 64 MiB HBM bdev
+        return hs.create_hbm_bdev("gpu-test", 512, 131072, device=0)
+
+    def test_roundtrip_single_block(self, bdev):
+        payload = bytes(random.getrandbits(8) for _ in range(512))
+        bdev.write(0, payload)
+        assert bdev.read(0, 512) == payload
+
+    def test_roundtrip_large_offsets(self, bdev):
+        rng = random.Random(7)
+        blob = bytes(rng.getrandbits(8) for _ in range(64 * 1024))
+        offset = 32 * 1024 * 1024 + 4096
+        bdev.write(offset, blob)
+        assert bdev.read(offset, len(blob)) == blob
+        # Neighbors untouched (bdev zero-initialized)
+        assert bdev.read(offset - 512, 512) == b"\x00" * 512
+        assert bdev.read(offset + len(blob), 512) == b"\x00" * 512
+
+    def test_fill(self, bdev):
+        bdev.fill(1024 * 1024, 0xCD, 8192)
+        assert bdev.read(1024 * 1024, 8192) == b"\xcd" * 8192
+
+    def test_unaligned_tail(self, bdev):
+        # 512-byte granularity I/O smaller than one 4 KiB tile
+        payload = bytes(random.getrandbits(8) for _ in range(1536))
+        bdev.write(2048, payload)
+        assert bdev.read(2048, 1536) == payload
+
+    def test_out_of_bounds(self, bdev):
+        with pytest.raises(RuntimeError):
+            bdev.read(bdev.size_bytes, 512)
+        with pytest.raises(RuntimeError):
+            bdev.write(bdev.size_bytes - 512, b"x" * 1024)
+
+    def test_persistence_across_channels(self, bdev):
+        # Data survives independent I/O sessions (the Malloc bdev
+        # contract: survives Map/Unmap, reference spec.md:116-119).
+        bdev.write(4096, b"\xee" * 512)
+        assert bdev.read(4096, 512) == b"\xee" * 512
+
+
+@needs_gpu
+class TestCrc32c:
+    def test_gpu_matches_software(self):
+        bdev = hs.create_hbm_bdev("crc-test", 4096, 256, device=0)
+        rng = random.Random(13)
+        blocks = [bytes(rng.getrandbits(8) for _ in range(4096)) for _ in range(16)]
+        for i, blk in enumerate(blocks):
+            bdev.write(i * 4096, blk)
+        gpu = hs.crc32c_gpu_blocks(bdev, 0, 4096, 16)
+        sw = [hs.crc32c(blk) for blk in blocks]
+        assert list(gpu) == sw
+
+    def test_known_answer(self):
+        assert hs.crc32c(b"123456789") == 0xE3069283
+
+
+@needs_gpu
+class TestPerf:
+    def test_randread_sanity(self):
+        bdev = hs.create_hbm_bdev("perf-gpu", 4096, 262144, device=0)  # 1 GiB
+        bdev.fill(0, 0x5A, bdev.size_bytes)
+        r = hs.run_bdevperf(bdev, "randread", 4096, 32, 4, 1.0)
+        assert r["io_count"] > 0
+        # An MI355X must beat 100k IOPS trivially; guards against a
+        # silently serialized path.
+        assert r["iops"] > 100_000, r
+        assert r["lat_p99_us"] < 100_000
+
+    def test_randwrite_correct_and_fast(self):
+        bdev = hs.create_hbm_bdev("perfw-gpu", 4096, 65536, device=0)
+        r = hs.run_bdevperf(bdev, "randwrite", 4096, 32, 2, 0.5)
+        assert r["iops"] > 50_000, r
+
+
+@needs_gpu
+class TestDaemonHbm:
+    def test_daemon_gpu_mode(self, tmp_path):
+        fixture = launch_hipstored(tmp_path, cpu=False)
+        try:
+            with hipstore.Client(fixture.socket_path) as client:
+                name = hipstore.construct_malloc_bdev(
+                    client, num_blocks=262144, block_size=4096, name="hbm0")
+                bdevs = hipstore.get_bdevs(client, name)
+                assert bdevs[0].product_name == "Malloc disk"
+                assert "hbm" in bdevs[0].driver_specific
+                pci = bdevs[0].driver_specific["hbm"]["pci_address"]
+                assert pci.count(":") == 2
+                result = hipstore.perf_run(client, "hbm0", io_size=4096,
+                                           queue_depth=32, num_queues=4,
+                                           seconds=1.0)
+                assert result["iops"] > 100_000, result
+                hipstore.delete_bdev(client, name)
+        finally:
+            fixture.stop()
