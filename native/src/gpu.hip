@@ -456,17 +456,35 @@ void crc32c_hbm_blocks(Bdev* bdev, uint64_t offset, uint32_t block_size,
 
 namespace {
 
+// Large synchronous requests are chunked below the HBM queue's
+// per-request cap (the descriptor ring holds 64 MiB of 4 KiB tiles).
+constexpr uint64_t kSyncChunk = 32ull << 20;
+
 int run_sync(Bdev* bdev, IoRequest req) {
   auto channel = bdev->get_channel();
-  int result = kIoFailed;
-  bool done = false;
-  req.on_complete = [&](int status) {
-    result = status;
-    done = true;
-  };
-  bdev->submit(channel.get(), std::move(req));
-  while (!done) bdev->poll(channel.get());
-  return result;
+  uint64_t done_bytes = 0;
+  const uint64_t total = req.length;
+  while (done_bytes < total || total == 0) {
+    IoRequest chunk;
+    chunk.op = req.op;
+    chunk.fill = req.fill;
+    chunk.offset = req.offset + done_bytes;
+    chunk.length = std::min<uint64_t>(kSyncChunk, total - done_bytes);
+    chunk.buffer = req.buffer == nullptr
+                       ? nullptr
+                       : static_cast<uint8_t*>(req.buffer) + done_bytes;
+    int result = kIoFailed;
+    bool done = false;
+    chunk.on_complete = [&](int status) {
+      result = status;
+      done = true;
+    };
+    bdev->submit(channel.get(), std::move(chunk));
+    while (!done) bdev->poll(channel.get());
+    if (result != kIoOk || total == 0) return result;
+    done_bytes += std::min<uint64_t>(kSyncChunk, total - done_bytes);
+  }
+  return kIoOk;
 }
 
 }  // namespace
