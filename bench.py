@@ -71,10 +71,13 @@ def main() -> int:
     # Touch every block once so HBM pages are resident (fill pattern).
     bdev.fill(0, 0x5A, bdev.size_bytes)
 
+    # Persistent queues (threads + HIP streams + pinned rings) across
+    # steps: a step measures steady-state I/O, not queue setup.
+    session = hs.PerfSession(bdev, args.workload, args.io_size,
+                             args.queue_depth, args.num_queues)
+
     def run_step(n_ios: int) -> dict:
-        return hs.run_bdevperf(bdev, args.workload, args.io_size,
-                               args.queue_depth, args.num_queues,
-                               seconds=3600.0, max_ios=n_ios)
+        return session.step(n_ios)
 
     def barrier():
         if distributed:

@@ -63,4 +63,23 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
                         int num_queues, double seconds,
                         uint64_t max_ios = 0);
 
+// Persistent-queue variant for stepped benchmarking (bench.py): queue
+// threads, their I/O channels (HIP streams + pinned rings) and buffers
+// live across step() calls, so a step measures steady-state IOPS
+// without per-call setup.
+class PerfSession {
+ public:
+  PerfSession(BdevPtr bdev, std::string workload, uint32_t io_size,
+              uint32_t queue_depth, int num_queues);
+  ~PerfSession();
+
+  // Run until `total_ios` I/Os complete (spread over the queues);
+  // returns the step's aggregate stats.
+  PerfResult step(uint64_t total_ios);
+
+ private:
+  struct Impl;
+  std::unique_ptr<Impl> impl_;
+};
+
 }  // namespace hipstore

@@ -21,8 +21,10 @@
 #include <algorithm>
 #include <atomic>
 #include <chrono>
+#include <condition_variable>
 #include <cstring>
 #include <deque>
+#include <functional>
 #include <mutex>
 #include <random>
 #include <stdexcept>
@@ -615,6 +617,186 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
   result.iops = result.io_count / elapsed;
   result.throughput_mbps =
       result.io_count * static_cast<double>(io_size) / elapsed / 1e6;
+  if (!all.empty()) {
+    std::sort(all.begin(), all.end());
+    double sum = 0;
+    for (uint32_t v : all) sum += v;
+    auto pct = [&](double p) {
+      size_t i = static_cast<size_t>(p * (all.size() - 1));
+      return static_cast<double>(all[i]);
+    };
+    result.lat_avg_us = sum / all.size();
+    result.lat_p50_us = pct(0.50);
+    result.lat_p90_us = pct(0.90);
+    result.lat_p99_us = pct(0.99);
+    result.lat_p999_us = pct(0.999);
+    result.lat_max_us = all.back();
+  }
+  return result;
+}
+
+// ---------------------------------------------------------------------------
+// PerfSession: persistent queues for stepped benchmarking
+// ---------------------------------------------------------------------------
+
+struct PerfSession::Impl {
+  BdevPtr bdev;
+  std::string workload;
+  uint32_t io_size;
+  uint32_t queue_depth;
+  int num_queues;
+
+  std::mutex mutex;
+  std::condition_variable cv;
+  uint64_t epoch = 0;           // incremented per step
+  uint64_t per_queue_ios = 0;   // target for the current epoch
+  bool stop = false;
+
+  struct QueueStats {
+    uint64_t done_epoch = 0;
+    uint64_t ios = 0;
+    std::vector<uint32_t> lat_us;
+    bool failed = false;
+  };
+  std::vector<QueueStats> stats;
+  std::vector<std::thread> threads;
+
+  void worker(int q) {
+    using clock = std::chrono::steady_clock;
+    try {
+      auto channel = bdev->get_channel();
+      uint8_t* buf = static_cast<uint8_t*>(
+          alloc_pinned(static_cast<size_t>(io_size) * queue_depth));
+      std::mt19937_64 rng(0x9E3779B97F4A7C15ULL ^ (q * 0x8DA6B343));
+      const uint64_t units = bdev->size_bytes() / io_size;
+      const bool do_read = workload != "randwrite";
+      const bool do_write = workload == "randwrite" || workload == "randrw";
+      std::vector<clock::time_point> submit_ts(queue_depth);
+      QueueStats& st = stats[q];
+
+      uint64_t my_epoch = 0;
+      while (true) {
+        {
+          std::unique_lock<std::mutex> lock(mutex);
+          cv.wait(lock, [&] { return stop || epoch > my_epoch; });
+          if (stop) break;
+          my_epoch = epoch;
+        }
+        const uint64_t target = per_queue_ios;
+        uint64_t completed = 0, submitted = 0;
+        uint32_t inflight = 0;
+        st.lat_us.clear();
+
+        std::function<void(uint32_t)> submit_slot = [&](uint32_t slot) {
+          IoRequest req;
+          bool write = do_write && (!do_read || (rng() & 1));
+          req.op = write ? IoOp::kWrite : IoOp::kRead;
+          req.offset = (rng() % units) * io_size;
+          req.length = io_size;
+          req.buffer = buf + static_cast<size_t>(slot) * io_size;
+          submit_ts[slot] = clock::now();
+          req.on_complete = [&, slot](int status) {
+            if (status != kIoOk) st.failed = true;
+            st.lat_us.push_back(static_cast<uint32_t>(
+                std::chrono::duration_cast<std::chrono::microseconds>(
+                    clock::now() - submit_ts[slot]).count()));
+            ++completed;
+            --inflight;
+            if (submitted < target) {
+              submit_slot(slot);
+              ++inflight;
+              ++submitted;
+            }
+          };
+          bdev->submit(channel.get(), std::move(req));
+        };
+
+        const uint32_t initial =
+            static_cast<uint32_t>(std::min<uint64_t>(queue_depth, target));
+        for (uint32_t slot = 0; slot < initial; ++slot) {
+          submit_slot(slot);
+          ++inflight;
+          ++submitted;
+        }
+        while (inflight > 0) bdev->poll(channel.get());
+        st.ios = completed;
+        {
+          std::lock_guard<std::mutex> lock(mutex);
+          st.done_epoch = my_epoch;
+        }
+        cv.notify_all();
+      }
+      free_pinned(buf);
+    } catch (const std::exception&) {
+      std::lock_guard<std::mutex> lock(mutex);
+      stats[q].failed = true;
+      stats[q].done_epoch = ~0ull;  // never blocks step()
+      cv.notify_all();
+    }
+  }
+};
+
+PerfSession::PerfSession(BdevPtr bdev, std::string workload, uint32_t io_size,
+                         uint32_t queue_depth, int num_queues)
+    : impl_(new Impl) {
+  impl_->bdev = std::move(bdev);
+  impl_->workload = std::move(workload);
+  impl_->io_size = io_size;
+  impl_->queue_depth = queue_depth;
+  impl_->num_queues = num_queues;
+  impl_->stats.resize(num_queues);
+  if (impl_->bdev->size_bytes() < io_size) {
+    throw std::runtime_error("bdev smaller than io_size");
+  }
+  for (int q = 0; q < num_queues; ++q) {
+    impl_->threads.emplace_back([this, q] { impl_->worker(q); });
+  }
+}
+
+PerfSession::~PerfSession() {
+  {
+    std::lock_guard<std::mutex> lock(impl_->mutex);
+    impl_->stop = true;
+  }
+  impl_->cv.notify_all();
+  for (auto& t : impl_->threads) t.join();
+}
+
+PerfResult PerfSession::step(uint64_t total_ios) {
+  using clock = std::chrono::steady_clock;
+  auto& im = *impl_;
+  const auto t0 = clock::now();
+  uint64_t this_epoch;
+  {
+    std::lock_guard<std::mutex> lock(im.mutex);
+    im.per_queue_ios = (total_ios + im.num_queues - 1) / im.num_queues;
+    this_epoch = ++im.epoch;
+  }
+  im.cv.notify_all();
+  {
+    std::unique_lock<std::mutex> lock(im.mutex);
+    im.cv.wait(lock, [&] {
+      for (auto& st : im.stats) {
+        if (st.done_epoch < this_epoch) return false;
+      }
+      return true;
+    });
+  }
+  const double elapsed = std::chrono::duration<double>(clock::now() - t0).count();
+
+  PerfResult result;
+  result.seconds = elapsed;
+  std::vector<uint32_t> all;
+  bool failed = false;
+  for (auto& st : im.stats) {
+    result.io_count += st.ios;
+    failed |= st.failed;
+    all.insert(all.end(), st.lat_us.begin(), st.lat_us.end());
+  }
+  if (failed) throw std::runtime_error("perf session: I/O failures");
+  result.iops = result.io_count / elapsed;
+  result.throughput_mbps =
+      result.io_count * static_cast<double>(im.io_size) / elapsed / 1e6;
   if (!all.empty()) {
     std::sort(all.begin(), all.end());
     double sum = 0;

@@ -94,6 +94,20 @@ PYBIND11_MODULE(_hipstore, m) {
   m.def("create_hbm_bdev", &create_hbm_bdev, py::arg("name"),
         py::arg("block_size"), py::arg("num_blocks"), py::arg("device") = 0);
 
+  py::class_<PerfSession>(m, "PerfSession")
+      .def(py::init<BdevPtr, std::string, uint32_t, uint32_t, int>(),
+           py::arg("bdev"), py::arg("workload") = "randread",
+           py::arg("io_size") = 4096, py::arg("queue_depth") = 32,
+           py::arg("num_queues") = 8)
+      .def("step", [](PerfSession& s, uint64_t total_ios) {
+        PerfResult r;
+        {
+          py::gil_scoped_release release;
+          r = s.step(total_ios);
+        }
+        return perf_to_dict(r);
+      }, py::arg("total_ios"));
+
   m.def(
       "run_bdevperf",
       [](BdevPtr bdev, const std::string& workload, uint32_t io_size,
