@@ -386,3 +386,8 @@ class ControllerServer:
     def stop(self) -> None:
         self.controller.stop()
         self.server.stop()
+
+    def run(self) -> None:
+        """start() + block until SIGINT/SIGTERM."""
+        self.server.run(
+            lambda s: spec.add_controller_to_server(self.controller, s))

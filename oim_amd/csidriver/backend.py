@@ -1,0 +1,29 @@
+"""The OIMBackend seam (reference oim-driver.go:71-78): what differs
+between local mode (drive hipstored directly, device via kernel NBD)
+and remote mode (drive a controller through the registry proxy, device
+via PCI/SCSI discovery)."""
+
+from __future__ import annotations
+
+from typing import Dict, Optional, Tuple
+
+
+class OIMBackend:
+    def create_volume(self, name: str, size: int) -> Tuple[str, Dict[str, str]]:
+        """Returns (volume_id, volume_context)."""
+        raise NotImplementedError
+
+    def delete_volume(self, volume_id: str) -> None:
+        raise NotImplementedError
+
+    def check_volume_exists(self, volume_id: str) -> bool:
+        raise NotImplementedError
+
+    def create_device(self, volume_id: str,
+                      volume_context: Dict[str, str]) -> str:
+        """Makes the volume appear as a host block device; returns its
+        /dev path. Idempotent."""
+        raise NotImplementedError
+
+    def delete_device(self, volume_id: str) -> None:
+        raise NotImplementedError

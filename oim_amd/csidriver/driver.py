@@ -1,0 +1,305 @@
+"""The OIM CSI driver: Identity/Controller/Node services over one
+endpoint (reference pkg/oim-csi-driver/oim-driver.go, controllerserver.go,
+nodeserver.go).
+
+Supported modes (mutually exclusive, oim-driver.go:216-221):
+  - local:  --hipstored-socket: drive hipstored directly, NBD device
+  - remote: --oim-registry-address + --controller-id: drive a remote
+            controller through the registry proxy
+
+CSI version: v1 only. The reference also carried a CSI 0.3 twin for
+old ceph-csi compatibility; this rebuild keeps the emulation hook
+(--emulate=ceph-csi) on the v1 surface instead (SURVEY.md section 7.3
+recommends exactly this simplification).
+"""
+
+from __future__ import annotations
+
+import os
+import threading
+from typing import Dict, Optional
+
+import grpc
+
+from .. import __version__
+from ..common.server import NonBlockingGRPCServer
+from ..common.tracing import LogServerInterceptor
+from ..log import from_context
+from ..spec import csi_v1 as csi
+from ..spec.rpc_csi import (
+    CSIControllerServicer,
+    CSIIdentityServicer,
+    CSINodeServicer,
+    add_csi_controller_to_server,
+    add_csi_identity_to_server,
+    add_csi_node_to_server,
+)
+from .backend import OIMBackend
+from .cephemu import EMULATIONS
+from .mount import Mounter
+
+
+class _KeyedMutex:
+    def __init__(self):
+        self._guard = threading.Lock()
+        self._locks: Dict[str, threading.Lock] = {}
+
+    def get(self, key: str) -> threading.Lock:
+        with self._guard:
+            return self._locks.setdefault(key, threading.Lock())
+
+
+class IdentityServer(CSIIdentityServicer):
+    def __init__(self, driver_name: str):
+        self.driver_name = driver_name
+
+    def GetPluginInfo(self, request, context):
+        return csi.GetPluginInfoResponse(
+            name=self.driver_name, vendor_version=__version__)
+
+    def GetPluginCapabilities(self, request, context):
+        response = csi.GetPluginCapabilitiesResponse()
+        cap = response.capabilities.add()
+        cap.service.type = csi.PLUGIN_CAPABILITY_CONTROLLER_SERVICE
+        return response
+
+    def Probe(self, request, context):
+        response = csi.ProbeResponse()
+        response.ready.value = True
+        return response
+
+
+def _validate_capabilities(caps, context):
+    """Reject block volumes and multi-writer modes
+    (controllerserver.go:29-47)."""
+    for cap in caps:
+        if cap.WhichOneof("access_type") == "block":
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "block volumes are not supported")
+        if cap.access_mode.mode in (
+                csi.ACCESS_MODE_MULTI_NODE_SINGLE_WRITER,
+                csi.ACCESS_MODE_MULTI_NODE_MULTI_WRITER):
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "multi-node writers are not supported")
+
+
+class ControllerServer(CSIControllerServicer):
+    def __init__(self, backend: OIMBackend):
+        self.backend = backend
+        self._name_mutex = _KeyedMutex()
+
+    def CreateVolume(self, request, context):
+        if not request.name:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT, "missing name")
+        if not request.volume_capabilities:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume capabilities")
+        _validate_capabilities(request.volume_capabilities, context)
+        size = request.capacity_range.required_bytes or (1 << 20)
+        with self._name_mutex.get(request.name):
+            try:
+                volume_id, volume_context = self.backend.create_volume(
+                    request.name, size)
+            except ValueError as exc:
+                context.abort(grpc.StatusCode.INVALID_ARGUMENT, str(exc))
+            except grpc.RpcError as exc:
+                context.abort(exc.code(), exc.details())
+        response = csi.CreateVolumeResponse()
+        response.volume.volume_id = volume_id
+        response.volume.capacity_bytes = size
+        for key, value in volume_context.items():
+            response.volume.volume_context[key] = value
+        return response
+
+    def DeleteVolume(self, request, context):
+        if not request.volume_id:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT, "missing volume id")
+        with self._name_mutex.get(request.volume_id):
+            try:
+                self.backend.delete_volume(request.volume_id)
+            except grpc.RpcError as exc:
+                context.abort(exc.code(), exc.details())
+        return csi.DeleteVolumeResponse()
+
+    def ValidateVolumeCapabilities(self, request, context):
+        if not request.volume_id:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT, "missing volume id")
+        if not request.volume_capabilities:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume capabilities")
+        if not self.backend.check_volume_exists(request.volume_id):
+            context.abort(grpc.StatusCode.NOT_FOUND,
+                          f"volume {request.volume_id} not found")
+        response = csi.ValidateVolumeCapabilitiesResponse()
+        ok = True
+        for cap in request.volume_capabilities:
+            if cap.WhichOneof("access_type") == "block":
+                ok = False
+            if cap.access_mode.mode not in (
+                    csi.ACCESS_MODE_SINGLE_NODE_WRITER,
+                    csi.ACCESS_MODE_SINGLE_NODE_READER_ONLY,
+                    csi.ACCESS_MODE_MULTI_NODE_READER_ONLY):
+                ok = False
+        if ok:
+            for cap in request.volume_capabilities:
+                response.confirmed.volume_capabilities.add().CopyFrom(cap)
+        else:
+            response.message = "unsupported volume capabilities"
+        return response
+
+    def ControllerGetCapabilities(self, request, context):
+        response = csi.ControllerGetCapabilitiesResponse()
+        cap = response.capabilities.add()
+        cap.rpc.type = csi.CTRL_CAP_CREATE_DELETE_VOLUME
+        return response
+
+
+class NodeServer(CSINodeServicer):
+    def __init__(self, node_id: str, backend: OIMBackend,
+                 mounter: Optional[Mounter] = None):
+        self.node_id = node_id
+        self.backend = backend
+        self.mounter = mounter or Mounter()
+        self._volume_mutex = _KeyedMutex()
+
+    def NodeStageVolume(self, request, context):
+        if not request.volume_id:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT, "missing volume id")
+        if not request.staging_target_path:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing staging target path")
+        if request.volume_capability.WhichOneof("access_type") == "block":
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "block volumes are not supported")
+        with self._volume_mutex.get(request.volume_id):
+            # Already staged? (nodeserver.go:169-183)
+            if self.mounter.is_mount_point(request.staging_target_path):
+                return csi.NodeStageVolumeResponse()
+            stage_info = {
+                "volume_context": dict(request.volume_context),
+                "secrets": dict(request.secrets),
+                "staging_path": request.staging_target_path,
+            }
+            try:
+                device = self.backend.create_device(request.volume_id,
+                                                    stage_info)
+            except TimeoutError as exc:
+                context.abort(grpc.StatusCode.DEADLINE_EXCEEDED, str(exc))
+            except grpc.RpcError as exc:
+                context.abort(exc.code(), exc.details())
+            except (RuntimeError, ValueError) as exc:
+                context.abort(grpc.StatusCode.INTERNAL, str(exc))
+            fs_type = request.volume_capability.mount.fs_type or "ext4"
+            os.makedirs(request.staging_target_path, exist_ok=True)
+            try:
+                self.mounter.format_and_mount(
+                    device, request.staging_target_path, fs_type,
+                    request.volume_capability.mount.mount_flags)
+            except Exception as exc:  # noqa: BLE001 - surface as INTERNAL
+                context.abort(grpc.StatusCode.INTERNAL,
+                              f"format/mount failed: {exc}")
+        return csi.NodeStageVolumeResponse()
+
+    def NodeUnstageVolume(self, request, context):
+        if not request.volume_id or not request.staging_target_path:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume id or staging path")
+        with self._volume_mutex.get(request.volume_id):
+            if self.mounter.is_mount_point(request.staging_target_path):
+                try:
+                    self.mounter.unmount(request.staging_target_path)
+                except Exception as exc:  # noqa: BLE001
+                    context.abort(grpc.StatusCode.INTERNAL, str(exc))
+            try:
+                self.backend.delete_device(request.volume_id)
+            except grpc.RpcError as exc:
+                context.abort(exc.code(), exc.details())
+        return csi.NodeUnstageVolumeResponse()
+
+    def NodePublishVolume(self, request, context):
+        if (not request.volume_id or not request.staging_target_path
+                or not request.target_path):
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume id or paths")
+        with self._volume_mutex.get(request.volume_id):
+            if self.mounter.is_mount_point(request.target_path):
+                return csi.NodePublishVolumeResponse()
+            os.makedirs(request.target_path, exist_ok=True)
+            try:
+                self.mounter.bind_mount(request.staging_target_path,
+                                        request.target_path,
+                                        readonly=request.readonly)
+            except Exception as exc:  # noqa: BLE001
+                context.abort(grpc.StatusCode.INTERNAL, str(exc))
+        return csi.NodePublishVolumeResponse()
+
+    def NodeUnpublishVolume(self, request, context):
+        if not request.volume_id or not request.target_path:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume id or target path")
+        with self._volume_mutex.get(request.volume_id):
+            if self.mounter.is_mount_point(request.target_path):
+                try:
+                    self.mounter.unmount(request.target_path)
+                except Exception as exc:  # noqa: BLE001
+                    context.abort(grpc.StatusCode.INTERNAL, str(exc))
+        return csi.NodeUnpublishVolumeResponse()
+
+    def NodeGetCapabilities(self, request, context):
+        response = csi.NodeGetCapabilitiesResponse()
+        cap = response.capabilities.add()
+        cap.rpc.type = csi.NODE_CAP_STAGE_UNSTAGE_VOLUME
+        return response
+
+    def NodeGetInfo(self, request, context):
+        return csi.NodeGetInfoResponse(node_id=self.node_id)
+
+
+class OIMDriver:
+    """Assembles the three services on one endpoint
+    (reference Start, oim-driver.go:275-292)."""
+
+    def __init__(self, driver_name: str, node_id: str, endpoint: str,
+                 backend: OIMBackend, mounter: Optional[Mounter] = None):
+        self.identity = IdentityServer(driver_name)
+        self.controller = ControllerServer(backend)
+        self.node = NodeServer(node_id, backend, mounter)
+        self.server = NonBlockingGRPCServer(
+            endpoint=endpoint, interceptors=[LogServerInterceptor()])
+
+    def start(self) -> None:
+        def register(server):
+            add_csi_identity_to_server(self.identity, server)
+            add_csi_controller_to_server(self.controller, server)
+            add_csi_node_to_server(self.node, server)
+
+        self.server.start(register)
+        from_context().info("CSI driver started", endpoint=self.server.addr())
+
+    def addr(self) -> str:
+        return self.server.addr()
+
+    def stop(self) -> None:
+        self.server.stop()
+
+    def run(self) -> None:
+        def register(server):
+            add_csi_identity_to_server(self.identity, server)
+            add_csi_controller_to_server(self.controller, server)
+            add_csi_node_to_server(self.node, server)
+
+        self.server.run(register)
+
+
+def make_params_mapper(emulate: str):
+    """Resolve --emulate to a params mapper (oim-driver.go:244-269)."""
+    from .remote import malloc_params
+
+    if not emulate:
+        return malloc_params
+    try:
+        return EMULATIONS[emulate]
+    except KeyError:
+        raise ValueError(
+            f"unknown emulation {emulate!r}; supported: "
+            f"{', '.join(EMULATIONS)}") from None

@@ -1,0 +1,104 @@
+"""Local backend: hipstored on this node, device via kernel NBD
+(reference pkg/oim-csi-driver/local.go).
+
+Volume sizing rules kept from the reference (local.go:50-84): round up
+to 512 bytes, minimum 1 MiB, maximum 1 TiB."""
+
+from __future__ import annotations
+
+import os
+from typing import Dict, Tuple
+
+from .. import hipstore
+from ..log import from_context
+from .backend import OIMBackend
+
+MIN_VOLUME_SIZE = 1 << 20
+MAX_VOLUME_SIZE = 1 << 40
+
+
+def round_to_blocks(size: int) -> int:
+    size = max(size, MIN_VOLUME_SIZE)
+    return (size + 511) // 512 * 512
+
+
+class LocalBackend(OIMBackend):
+    def __init__(self, hipstored_socket: str, nbd_prefix: str = "/dev/nbd"):
+        self.socket = hipstored_socket
+        self.nbd_prefix = nbd_prefix
+
+    def _client(self) -> hipstore.Client:
+        return hipstore.Client(self.socket)
+
+    def create_volume(self, name: str, size: int) -> Tuple[str, Dict[str, str]]:
+        if size > MAX_VOLUME_SIZE:
+            raise ValueError(f"volume too large: {size}")
+        size = round_to_blocks(size)
+        with self._client() as client:
+            try:
+                bdevs = hipstore.get_bdevs(client, name)
+            except hipstore.RpcError as err:
+                if not err.is_not_found():
+                    raise
+                bdevs = []
+            if bdevs:
+                if bdevs[0].size_bytes != size:
+                    raise ValueError(
+                        f"volume {name} exists with different size")
+                return name, {}
+            hipstore.construct_malloc_bdev(
+                client, num_blocks=size // 512, block_size=512, name=name)
+        return name, {}
+
+    def delete_volume(self, volume_id: str) -> None:
+        with self._client() as client:
+            try:
+                hipstore.delete_bdev(client, volume_id)
+            except hipstore.RpcError as err:
+                if not err.is_not_found():
+                    raise
+
+    def check_volume_exists(self, volume_id: str) -> bool:
+        with self._client() as client:
+            try:
+                return bool(hipstore.get_bdevs(client, volume_id))
+            except hipstore.RpcError as err:
+                if err.is_not_found():
+                    return False
+                raise
+
+    def _find_exported(self, client, volume_id: str) -> str:
+        for disk in hipstore.get_nbd_disks(client):
+            if disk.bdev_name == volume_id:
+                return disk.nbd_device
+        return ""
+
+    def create_device(self, volume_id: str, volume_context) -> str:
+        with self._client() as client:
+            # Reuse an existing export (local.go:128,208-219).
+            device = self._find_exported(client, volume_id)
+            if device:
+                return device
+            in_use = {d.nbd_device for d in hipstore.get_nbd_disks(client)}
+            # Probe /dev/nbd0.. for a free device (local.go:139-176; the
+            # reference notes the size==0 probe is racy — daemon-side
+            # bookkeeping covers our own exports, the probe covers
+            # foreign users of the device).
+            for i in range(16):
+                candidate = f"{self.nbd_prefix}{i}"
+                if candidate in in_use or not os.path.exists(candidate):
+                    continue
+                try:
+                    hipstore.start_nbd_disk(client, volume_id, candidate)
+                except hipstore.RpcError as err:
+                    from_context().warn("NBD attach failed, trying next",
+                                        device=candidate, error=str(err))
+                    continue
+                return candidate
+            raise RuntimeError("no free NBD device found")
+
+    def delete_device(self, volume_id: str) -> None:
+        with self._client() as client:
+            device = self._find_exported(client, volume_id)
+            if device:
+                hipstore.stop_nbd_disk(client, device)

@@ -10,6 +10,7 @@ through asyncio/threads started with copy_context, plus explicit
 """
 
 from .level import Level, parse_level
+from . import logger
 from .logger import (
     Logger,
     SimpleLogger,
@@ -20,6 +21,7 @@ from .logger import (
     with_logger,
     with_fields,
     add_flags,
+    init_from_args,
 )
 from .testlog import TestLogger
 
@@ -35,5 +37,6 @@ __all__ = [
     "with_logger",
     "with_fields",
     "add_flags",
+    "init_from_args",
     "TestLogger",
 ]

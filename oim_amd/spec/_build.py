@@ -86,14 +86,22 @@ def build_file(
     package: str,
     messages: Sequence[Message],
     services: Sequence[Service] = (),
+    dependencies: Sequence[str] = (),
     pool: Optional[descriptor_pool.DescriptorPool] = None,
 ) -> Dict[str, type]:
-    """Register the schema and return {message_name: message_class}."""
+    """Register the schema and return {message_name: message_class}.
+
+    `dependencies` lists imported .proto files already present in the
+    pool (e.g. "google/protobuf/wrappers.proto"); fields may then use
+    fully-qualified external type names like ".google.protobuf.BoolValue".
+    """
     pool = pool or descriptor_pool.Default()
     file_proto = descriptor_pb2.FileDescriptorProto()
     file_proto.name = name
     file_proto.package = package
     file_proto.syntax = "proto3"
+    for dep in dependencies:
+        file_proto.dependency.append(dep)
 
     for message in messages:
         msg_proto = file_proto.message_type.add()

@@ -1,0 +1,55 @@
+"""CLI entry-point tests: oimctl against a live registry; flag parsing."""
+
+import subprocess
+import sys
+import threading
+
+import pytest
+
+from oim_amd.cmd import oimctl
+from oim_amd.registry import MemRegistryDB, Registry, RegistryServer
+
+
+@pytest.fixture
+def live_registry(tmp_path):
+    registry = Registry(db=MemRegistryDB())
+    server = RegistryServer(f"unix://{tmp_path}/reg.sock", registry)
+    server.start()
+    yield registry, server
+    server.stop()
+
+
+class TestOimctl:
+    def test_set_get_delete(self, live_registry, capsys):
+        registry, server = live_registry
+        endpoint = server.addr()
+        assert oimctl.main(["--registry", endpoint, "set",
+                            "host-0/pci", "0000:00:15.0"]) == 0
+        assert oimctl.main(["--registry", endpoint, "get"]) == 0
+        out = capsys.readouterr().out
+        assert "host-0/pci: 0000:00:15.0" in out
+        assert oimctl.main(["--registry", endpoint, "delete", "host-0/pci"]) == 0
+        assert oimctl.main(["--registry", endpoint, "get"]) == 0
+        assert "host-0" not in capsys.readouterr().out
+
+
+class TestMainModules:
+    def test_help_screens(self):
+        """Every binary parses --help (flag wiring sanity)."""
+        for module in ("oim_amd.cmd.oim_registry", "oim_amd.cmd.oim_controller",
+                       "oim_amd.cmd.oim_csi_driver", "oim_amd.cmd.oimctl"):
+            proc = subprocess.run(
+                [sys.executable, "-m", module, "--help"],
+                capture_output=True, text=True, timeout=60)
+            assert proc.returncode == 0, proc.stderr
+            assert "usage" in proc.stdout.lower()
+
+    def test_csi_driver_mode_validation(self):
+        from oim_amd.cmd import oim_csi_driver
+
+        with pytest.raises(SystemExit):
+            oim_csi_driver.main(["--nodeid", "n1"])  # no mode selected
+        with pytest.raises(SystemExit):
+            oim_csi_driver.main([
+                "--nodeid", "n1", "--hipstored-socket", "/x",
+                "--oim-registry-address", "tcp://y:1"])  # both modes
