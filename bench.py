@@ -1,13 +1,17 @@
 #!/usr/bin/env python3
 """Flagship benchmark: 4 KiB random-read IOPS + p99 latency on the
-HBM-resident Malloc bdev (BASELINE.json metric), fio-shaped synthetic
-I/O at QD=32 per queue.
+HBM-resident Malloc bdev, provisioned through the CSI control plane
+(BASELINE.json metric: "4KiB randread IOPS + p99 lat, Malloc bdev via
+CSI"), fio-shaped synthetic I/O at QD=32 per queue.
 
 One rank per GPU (torch.distributed over RCCL when launched via
-torch.distributed.run); each rank drives its own GPU's bdev — the
-"one accelerator card per GPU" model (weak scaling). A *step* is a
-fixed batch of STEP_IOS random 4 KiB reads per GPU; rank 0 prints one
-JSON line with the whole-job aggregate.
+torch.distributed.run); each rank runs the full per-card stack —
+hipstored daemon on its GPU + registry + controller + CSI driver — and
+provisions its bdev via CSI CreateVolume -> registry proxy ->
+controller -> hipstored, then attaches it with MapVolume (the
+reference's signature flow). A *step* is a fixed batch of STEP_IOS
+random 4 KiB reads per GPU measured in-daemon on persistent queues;
+rank 0 prints one JSON line with the whole-job aggregate (weak scaling).
 
 Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
 """
@@ -15,12 +19,99 @@ Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
 import argparse
 import json
 import os
+import subprocess
 import sys
+import tempfile
+import time
 
 REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO_ROOT)
 
 STEP_IOS = 131072  # 4 KiB reads per GPU per step (512 MiB moved)
+
+
+def start_stack(rank, local_rank, use_gpu, args, tmp):
+    """Per-rank control plane + data daemon; returns (client, cleanup)."""
+    import grpc
+
+    from oim_amd import hipstore, spec
+    from oim_amd.common.server import grpc_target
+    from oim_amd.controller import Controller, ControllerServer
+    from oim_amd.csidriver import OIMDriver, RemoteBackend
+    from oim_amd.registry import MemRegistryDB, Registry, RegistryServer
+    from oim_amd.spec import csi_v1 as csi
+    from oim_amd.spec.rpc_csi import CSIControllerStub
+
+    daemon_sock = os.path.join(tmp, "hipstored.sock")
+    cmd = [os.path.join(REPO_ROOT, "bin", "hipstored"), "-S", daemon_sock,
+           "-d", str(local_rank)]
+    if not use_gpu:
+        cmd.append("-C")
+    daemon = subprocess.Popen(cmd, stderr=subprocess.DEVNULL)
+    deadline = time.time() + 60
+    while not os.path.exists(daemon_sock):
+        if daemon.poll() is not None:
+            raise RuntimeError("hipstored exited early")
+        if time.time() > deadline:
+            raise RuntimeError("hipstored did not start")
+        time.sleep(0.05)
+
+    registry = Registry(db=MemRegistryDB())
+    reg_server = RegistryServer(f"unix://{tmp}/registry.sock", registry)
+    reg_server.start()
+    controller_id = f"gpu-{rank}"
+    controller = Controller(
+        controller_id=controller_id,
+        hipstored_socket=daemon_sock,
+        controller_address=f"unix://{tmp}/controller.sock",
+        registry_address=reg_server.addr(),
+        registry_delay=3600.0,
+    )
+    ctrl_server = ControllerServer(f"unix://{tmp}/controller.sock", controller)
+    ctrl_server.start()
+    controller.register()
+    backend = RemoteBackend(
+        registry_address=reg_server.addr(), controller_id=controller_id)
+    driver = OIMDriver(
+        driver_name="oim-malloc", node_id=f"bench-node-{rank}",
+        endpoint=f"unix://{tmp}/csi.sock", backend=backend)
+    driver.start()
+
+    # CSI CreateVolume -> proxy -> controller -> hipstored (HBM bdev).
+    volume = f"bench-{rank}"
+    size = int(args.bdev_gb * (1 << 30))
+    with grpc.insecure_channel(grpc_target(f"unix://{tmp}/csi.sock")) as ch:
+        stub = CSIControllerStub(ch)
+        request = csi.CreateVolumeRequest(name=volume)
+        request.capacity_range.required_bytes = size
+        cap = request.volume_capabilities.add()
+        cap.mount.fs_type = "ext4"
+        cap.access_mode.mode = csi.ACCESS_MODE_SINGLE_NODE_WRITER
+        stub.CreateVolume(request, timeout=120)
+    # MapVolume through the proxy: attach the bdev to a SCSI target of
+    # this card (completes the reference's NodeStage control path; the
+    # host-device hotplug itself needs a kernel NBD/virtio path and is
+    # exercised in tests, not in the timed bench).
+    with grpc.insecure_channel(grpc_target(reg_server.addr())) as ch:
+        stub = spec.ControllerStub(ch)
+        stub.MapVolume(
+            spec.MapVolumeRequest(volume_id=volume, malloc=spec.MallocParams()),
+            metadata=((spec.CONTROLLER_ID_KEY, controller_id),), timeout=60)
+
+    client = hipstore.Client(daemon_sock, timeout=3600)
+
+    def cleanup():
+        client.close()
+        driver.stop()
+        ctrl_server.stop()
+        reg_server.stop()
+        daemon.terminate()
+        try:
+            daemon.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            daemon.kill()
+
+    return client, volume, cleanup
 
 
 def main() -> int:
@@ -45,63 +136,57 @@ def main() -> int:
     use_gpu = torch.cuda.is_available()
 
     if distributed:
-        backend = "nccl" if use_gpu else "gloo"
-        dist.init_process_group(backend=backend)
+        dist.init_process_group(backend="nccl" if use_gpu else "gloo")
     if use_gpu:
         torch.cuda.set_device(local_rank)
-
-    from oim_amd import _hipstore as hs
-
-    block = args.io_size
-    num_blocks = int(args.bdev_gb * (1 << 30)) // block
-    if use_gpu:
+        from oim_amd import _hipstore as hs
         if not hs.gpu_available():
             raise RuntimeError(
                 "torch sees a GPU but oim_amd._hipstore does not — "
                 "native extension not loaded; rebuild with `make`")
-        bdev = hs.create_hbm_bdev(f"bench-{rank}", block, num_blocks,
-                                  device=local_rank)
-        backend_name = "hbm"
-    else:
-        # CPU fallback so the bench is testable in the no-GPU container.
-        num_blocks = min(num_blocks, (1 << 30) // block)
-        bdev = hs.create_malloc_bdev(f"bench-{rank}", block, num_blocks)
-        backend_name = "cpu"
 
-    # Touch every block once so HBM pages are resident (fill pattern).
-    bdev.fill(0, 0x5A, bdev.size_bytes)
+    if not use_gpu:
+        args.bdev_gb = min(args.bdev_gb, 1.0)
 
-    # Persistent queues (threads + HIP streams + pinned rings) across
-    # steps: a step measures steady-state I/O, not queue setup.
-    session = hs.PerfSession(bdev, args.workload, args.io_size,
-                             args.queue_depth, args.num_queues)
+    with tempfile.TemporaryDirectory(prefix="oim-bench-") as tmp:
+        client, volume, cleanup = start_stack(rank, local_rank, use_gpu,
+                                              args, tmp)
+        try:
+            session = client.invoke("perf_session_start", {
+                "bdev_name": volume,
+                "workload": args.workload,
+                "io_size": args.io_size,
+                "queue_depth": args.queue_depth,
+                "num_queues": args.num_queues,
+            })["session_id"]
 
-    def run_step(n_ios: int) -> dict:
-        return session.step(n_ios)
+            def run_step(n_ios):
+                return client.invoke("perf_session_step", {
+                    "session_id": session, "total_ios": n_ios})
 
-    def barrier():
-        if distributed:
-            dist.barrier()
-        if use_gpu:
-            torch.cuda.synchronize()
+            def barrier():
+                if distributed:
+                    dist.barrier()
+                if use_gpu:
+                    torch.cuda.synchronize()
 
-    for _ in range(args.warmup):
-        run_step(STEP_IOS)
+            for _ in range(args.warmup):
+                run_step(STEP_IOS)
 
-    barrier()
-    import time
-    t0 = time.perf_counter()
-    p99s, p50s = [], []
-    ios_done = 0
-    for _ in range(args.steps):
-        r = run_step(STEP_IOS)
-        ios_done += r["io_count"]
-        p99s.append(r["lat_p99_us"])
-        p50s.append(r["lat_p50_us"])
-    barrier()
-    elapsed = time.perf_counter() - t0
+            barrier()
+            t0 = time.perf_counter()
+            p99s = []
+            ios_done = 0
+            for _ in range(args.steps):
+                r = run_step(STEP_IOS)
+                ios_done += r["io_count"]
+                p99s.append(r["lat_p99_us"])
+            barrier()
+            elapsed = time.perf_counter() - t0
+            client.invoke("perf_session_stop", {"session_id": session})
+        finally:
+            cleanup()
 
-    # MAX over ranks of elapsed; aggregate IOs over ranks.
     if distributed:
         stats = torch.tensor([elapsed, float(ios_done), max(p99s)],
                              dtype=torch.float64)
@@ -132,7 +217,7 @@ def main() -> int:
             "dtype": "uint8",
             "data": "synthetic",
             "config": {
-                "model": "hbm-malloc-bdev",
+                "model": "hbm-malloc-bdev-via-csi",
                 "global_batch": args.queue_depth * args.num_queues,
                 "seq_len": args.io_size,
                 "parallelism": f"1-card-per-gpu-x{world_size if distributed else 1}",
@@ -142,7 +227,8 @@ def main() -> int:
                 "num_queues": args.num_queues,
                 "bdev_gb": args.bdev_gb,
                 "step_ios_per_gpu": STEP_IOS,
-                "backend": backend_name,
+                "backend": "hbm" if use_gpu else "cpu",
+                "provisioning": "csi-createvolume+proxy-mapvolume",
                 "p99_us": round(p99, 1),
             },
         }

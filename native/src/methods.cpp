@@ -318,10 +318,68 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device) {
           "stop_nbd_disk", "construct_vhost_scsi_controller",
           "add_vhost_scsi_lun", "remove_vhost_scsi_target",
           "remove_vhost_controller", "get_vhost_controllers",
-          "get_rpc_methods", "perf_run"}) {
+          "get_rpc_methods", "perf_run", "perf_session_start",
+          "perf_session_step", "perf_session_stop"}) {
       out.push_back(Json(name));
     }
     return Json(std::move(out));
+  });
+
+  // Stepped benchmarking with persistent queues (bench.py contract):
+  // perf_session_start -> id; perf_session_step runs a fixed I/O count
+  // on the session's live queues; perf_session_stop tears down.
+  struct Sessions {
+    std::mutex mutex;
+    std::map<int64_t, std::shared_ptr<PerfSession>> by_id;
+    int64_t next_id = 1;
+  };
+  auto sessions = std::make_shared<Sessions>();
+
+  server->register_method(
+      "perf_session_start", [&manager, sessions](const Json& p) {
+        BdevPtr bdev = manager.find(p.get_string("bdev_name"));
+        if (!bdev) not_found("bdev " + p.get_string("bdev_name"));
+        auto session = std::make_shared<PerfSession>(
+            bdev, p.get_string("workload", "randread"),
+            static_cast<uint32_t>(p.get_int("io_size", 4096)),
+            static_cast<uint32_t>(p.get_int("queue_depth", 32)),
+            static_cast<int>(p.get_int("num_queues", 8)));
+        std::lock_guard<std::mutex> lock(sessions->mutex);
+        int64_t id = sessions->next_id++;
+        sessions->by_id[id] = std::move(session);
+        JsonObject o;
+        o["session_id"] = Json(id);
+        return Json(std::move(o));
+      });
+
+  server->register_method("perf_session_step", [sessions](const Json& p) {
+    std::shared_ptr<PerfSession> session;  // keeps the session alive
+    {                                      // across a concurrent stop
+      std::lock_guard<std::mutex> lock(sessions->mutex);
+      auto it = sessions->by_id.find(p.get_int("session_id"));
+      if (it == sessions->by_id.end()) not_found("perf session");
+      session = it->second;
+    }
+    PerfResult r = session->step(
+        static_cast<uint64_t>(p.get_int("total_ios", 1 << 17)));
+    JsonObject o;
+    o["seconds"] = Json(r.seconds);
+    o["io_count"] = Json(static_cast<int64_t>(r.io_count));
+    o["iops"] = Json(r.iops);
+    o["throughput_mbps"] = Json(r.throughput_mbps);
+    o["lat_avg_us"] = Json(r.lat_avg_us);
+    o["lat_p50_us"] = Json(r.lat_p50_us);
+    o["lat_p90_us"] = Json(r.lat_p90_us);
+    o["lat_p99_us"] = Json(r.lat_p99_us);
+    o["lat_p999_us"] = Json(r.lat_p999_us);
+    o["lat_max_us"] = Json(r.lat_max_us);
+    return Json(std::move(o));
+  });
+
+  server->register_method("perf_session_stop", [sessions](const Json& p) {
+    std::lock_guard<std::mutex> lock(sessions->mutex);
+    sessions->by_id.erase(p.get_int("session_id"));
+    return Json(JsonObject{});
   });
 
   server->register_method("perf_run", [&manager](const Json& p) {

@@ -1,0 +1,57 @@
+"""bench.py contract tests (CPU): single-process and 2-rank gloo runs
+must emit the JSON line the driver parses."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+BENCH = os.path.join(REPO_ROOT, "bench.py")
+
+REQUIRED_KEYS = {
+    "metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+    "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config",
+}
+
+
+def last_json_line(output: str) -> dict:
+    for line in reversed(output.strip().splitlines()):
+        if line.startswith("{"):
+            return json.loads(line)
+    raise AssertionError(f"no JSON line in output:\n{output}")
+
+
+@pytest.mark.skipif(not os.path.exists(os.path.join(REPO_ROOT, "bin", "hipstored")),
+                    reason="hipstored not built")
+class TestBenchContract:
+    def test_single_process(self):
+        proc = subprocess.run(
+            [sys.executable, BENCH, "--steps", "2", "--warmup", "1"],
+            capture_output=True, text=True, timeout=600, cwd=REPO_ROOT)
+        assert proc.returncode == 0, proc.stderr[-2000:]
+        result = last_json_line(proc.stdout)
+        assert REQUIRED_KEYS.issubset(result.keys())
+        assert result["metric"] == "4KiB_randread_IOPS"
+        assert result["value"] > 0
+        assert result["n_gpus"] == 1
+        assert result["steps"] == 2
+        assert result["config"]["p99_us"] > 0
+        assert result["scaling"] == "weak"
+
+    def test_two_rank_gloo(self):
+        env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT="29517")
+        proc = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", "29517", BENCH,
+             "--gpus", "2", "--steps", "2", "--warmup", "0"],
+            capture_output=True, text=True, timeout=600, cwd=REPO_ROOT,
+            env=env)
+        assert proc.returncode == 0, proc.stderr[-2000:]
+        result = last_json_line(proc.stdout)
+        assert result["n_gpus"] == 2
+        # aggregate over both ranks: 2 ranks x steps x STEP_IOS I/Os
+        assert result["value"] > 0
