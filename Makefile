@@ -19,7 +19,7 @@ B := native/build
 HDRS := $(wildcard native/include/hipstore/*.h)
 
 CORE_OBJS := $(B)/json.o $(B)/bdev.o $(B)/crc32c.o $(B)/rpc_server.o \
-             $(B)/methods.o $(B)/nbd.o $(B)/gpu.o
+             $(B)/methods.o $(B)/nbd.o $(B)/composite.o $(B)/gpu.o
 
 .PHONY: all clean
 all: oim_amd/_hipstore$(EXT_SUFFIX) bin/hipstored

@@ -10,6 +10,7 @@
 #include <mutex>
 
 #include "hipstore/bdev.h"
+#include "hipstore/composite.h"
 #include "hipstore/engine.h"
 #include "hipstore/json.h"
 #include "hipstore/nbd.h"
@@ -324,6 +325,88 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device) {
     }
     return Json(std::move(out));
   });
+
+  // --- composite bdevs (BASELINE config 5) -------------------------------
+  // Striped / replicated malloc bdevs spanning the node's GPUs. The
+  // `devices` param lists HIP device indices (one HBM child each); with
+  // --cpu (or no GPU) `replicas` host-RAM children emulate the layout
+  // for CI. These methods are native extensions — SPDK's raid bdev has
+  // a different RPC shape and no GPU notion.
+  auto make_children = [&manager, use_hbm](
+                           const Json& p, const std::string& name,
+                           int64_t num_blocks, int64_t block_size,
+                           size_t count_hint) {
+    std::vector<BdevPtr> children;
+    std::vector<int> devices;
+    if (const Json* d = p.get("devices")) {
+      for (const Json& v : d->as_array()) {
+        devices.push_back(static_cast<int>(v.as_int()));
+      }
+    }
+    size_t count = devices.empty() ? count_hint : devices.size();
+    if (count == 0) throw RpcError{kInvalidParams, "no devices/replicas"};
+    for (size_t i = 0; i < count; ++i) {
+      const std::string child_name = name + "." + std::to_string(i);
+      if (use_hbm && gpu_available() && !devices.empty()) {
+        children.push_back(create_hbm_bdev(child_name, block_size,
+                                           num_blocks, devices[i]));
+      } else {
+        children.push_back(
+            create_malloc_bdev(child_name, block_size, num_blocks));
+      }
+    }
+    return children;
+  };
+
+  server->register_method(
+      "construct_striped_malloc_bdev",
+      [make_children](const Json& p) {
+        const std::string name = p.get_string("name");
+        const int64_t num_blocks = p.get_int("num_blocks");   // per child
+        const int64_t block_size = p.get_int("block_size", 512);
+        const int64_t stripe_kb = p.get_int("stripe_size_kb", 128);
+        const int64_t replicas = p.get_int("count", 2);
+        if (name.empty() || num_blocks <= 0 || block_size <= 0) {
+          throw RpcError{kInvalidParams, "name/num_blocks/block_size required"};
+        }
+        if (manager.find(name)) {
+          throw RpcError{kInvalidParams, "bdev " + name + " already exists"};
+        }
+        try {
+          auto children = make_children(p, name, num_blocks, block_size,
+                                        replicas);
+          BdevPtr bdev = create_striped_bdev(name, std::move(children),
+                                             stripe_kb * 1024);
+          manager.add(bdev);
+        } catch (const std::exception& e) {
+          throw RpcError{kInvalidParams, e.what()};
+        }
+        return Json(name);
+      });
+
+  server->register_method(
+      "construct_replicated_malloc_bdev",
+      [make_children](const Json& p) {
+        const std::string name = p.get_string("name");
+        const int64_t num_blocks = p.get_int("num_blocks");
+        const int64_t block_size = p.get_int("block_size", 512);
+        const int64_t replicas = p.get_int("count", 2);
+        if (name.empty() || num_blocks <= 0 || block_size <= 0) {
+          throw RpcError{kInvalidParams, "name/num_blocks/block_size required"};
+        }
+        if (manager.find(name)) {
+          throw RpcError{kInvalidParams, "bdev " + name + " already exists"};
+        }
+        try {
+          auto children = make_children(p, name, num_blocks, block_size,
+                                        replicas);
+          BdevPtr bdev = create_replicated_bdev(name, std::move(children));
+          manager.add(bdev);
+        } catch (const std::exception& e) {
+          throw RpcError{kInvalidParams, e.what()};
+        }
+        return Json(name);
+      });
 
   // Stepped benchmarking with persistent queues (bench.py contract):
   // perf_session_start -> id; perf_session_step runs a fixed I/O count

@@ -13,6 +13,7 @@
 #include <stdexcept>
 
 #include "hipstore/bdev.h"
+#include "hipstore/composite.h"
 #include "hipstore/crc32c.h"
 #include "hipstore/engine.h"
 
@@ -93,6 +94,10 @@ PYBIND11_MODULE(_hipstore, m) {
         py::arg("block_size"), py::arg("num_blocks"));
   m.def("create_hbm_bdev", &create_hbm_bdev, py::arg("name"),
         py::arg("block_size"), py::arg("num_blocks"), py::arg("device") = 0);
+  m.def("create_striped_bdev", &create_striped_bdev, py::arg("name"),
+        py::arg("children"), py::arg("stripe_size") = 131072);
+  m.def("create_replicated_bdev", &create_replicated_bdev, py::arg("name"),
+        py::arg("children"));
 
   py::class_<PerfSession>(m, "PerfSession")
       .def(py::init<BdevPtr, std::string, uint32_t, uint32_t, int>(),
