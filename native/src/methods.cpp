@@ -332,7 +332,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device) {
   // --cpu (or no GPU) `replicas` host-RAM children emulate the layout
   // for CI. These methods are native extensions — SPDK's raid bdev has
   // a different RPC shape and no GPU notion.
-  auto make_children = [&manager, use_hbm](
+  auto make_children = [use_hbm](
                            const Json& p, const std::string& name,
                            int64_t num_blocks, int64_t block_size,
                            size_t count_hint) {
@@ -360,7 +360,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device) {
 
   server->register_method(
       "construct_striped_malloc_bdev",
-      [make_children](const Json& p) {
+      [&manager, make_children](const Json& p) {
         const std::string name = p.get_string("name");
         const int64_t num_blocks = p.get_int("num_blocks");   // per child
         const int64_t block_size = p.get_int("block_size", 512);
@@ -386,7 +386,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device) {
 
   server->register_method(
       "construct_replicated_malloc_bdev",
-      [make_children](const Json& p) {
+      [&manager, make_children](const Json& p) {
         const std::string name = p.get_string("name");
         const int64_t num_blocks = p.get_int("num_blocks");
         const int64_t block_size = p.get_int("block_size", 512);
