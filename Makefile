@@ -19,7 +19,9 @@ B := native/build
 HDRS := $(wildcard native/include/hipstore/*.h)
 
 CORE_OBJS := $(B)/json.o $(B)/bdev.o $(B)/crc32c.o $(B)/rpc_server.o \
-             $(B)/methods.o $(B)/nbd.o $(B)/composite.o $(B)/gpu.o
+             $(B)/methods.o $(B)/nbd.o $(B)/composite.o \
+             $(B)/nvmf_common.o $(B)/nvmf_target.o $(B)/nvmf_initiator.o \
+             $(B)/gpu.o
 
 .PHONY: all clean
 all: oim_amd/_hipstore$(EXT_SUFFIX) bin/hipstored

@@ -20,6 +20,12 @@ namespace hipstore {
 // so crc32c_sw(0, ...) matches the usual CRC-32C value.
 uint32_t crc32c_sw(uint32_t init, const void* data, size_t len);
 
+// CRC of a concatenation: crc32c(A||B) from crc32c(A), crc32c(B) and
+// len(B) (zlib crc32_combine construction on the Castagnoli poly).
+// Lets the GPU produce per-4KiB-block CRCs in parallel while the host
+// folds them into one NVMe/TCP data digest.
+uint32_t crc32c_combine(uint32_t crc1, uint32_t crc2, size_t len2);
+
 // Per-block CRC32C of `count` consecutive `block_size`-byte blocks
 // starting at byte `offset` of the bdev. HBM bdevs compute on-GPU;
 // other bdevs fall back to reading + software CRC (CI path).

@@ -41,6 +41,49 @@ uint32_t crc32c_sw(uint32_t init, const void* data, size_t len) {
   return ~crc;
 }
 
+namespace {
+
+// GF(2) 32x32 matrix ops over the CRC register (zlib crc32_combine).
+uint32_t gf2_times(const uint32_t* mat, uint32_t vec) {
+  uint32_t sum = 0;
+  int i = 0;
+  while (vec) {
+    if (vec & 1) sum ^= mat[i];
+    vec >>= 1;
+    ++i;
+  }
+  return sum;
+}
+
+void gf2_square(uint32_t* square, const uint32_t* mat) {
+  for (int i = 0; i < 32; ++i) square[i] = gf2_times(mat, mat[i]);
+}
+
+}  // namespace
+
+uint32_t crc32c_combine(uint32_t crc1, uint32_t crc2, size_t len2) {
+  if (len2 == 0) return crc1;
+  uint32_t even[32];  // operator for 2^(k) zero bytes
+  uint32_t odd[32];
+  // operator for one zero BIT: shift-right with poly feedback
+  odd[0] = kPolyReflected;
+  for (int i = 1; i < 32; ++i) odd[i] = 1u << (i - 1);
+  gf2_square(even, odd);  // operator for 2 zero bits
+  gf2_square(odd, even);  // operator for 4 zero bits
+  // Square-and-multiply over len2 zero BYTES: the first loop square
+  // yields the 8-bit (1 byte) operator (zlib crc32_combine ladder).
+  do {
+    gf2_square(even, odd);
+    if (len2 & 1) crc1 = gf2_times(even, crc1);
+    len2 >>= 1;
+    if (len2 == 0) break;
+    gf2_square(odd, even);
+    if (len2 & 1) crc1 = gf2_times(odd, crc1);
+    len2 >>= 1;
+  } while (len2);
+  return crc1 ^ crc2;
+}
+
 void crc32c_cpu_fallback(Bdev* bdev, uint64_t offset, uint32_t block_size,
                          uint32_t count, uint32_t* out) {
   std::vector<uint8_t> buf(block_size);

@@ -14,6 +14,7 @@
 #include "hipstore/engine.h"
 #include "hipstore/json.h"
 #include "hipstore/nbd.h"
+#include "hipstore/nvmf.h"
 #include "hipstore/rpc.h"
 
 namespace hipstore {
@@ -325,6 +326,75 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device) {
     }
     return Json(std::move(out));
   });
+
+  // --- NVMe-oF/TCP (BASELINE config 3) -----------------------------------
+  // nvmf_create_target exports existing bdevs as namespaces (nsid =
+  // position in `bdevs` + 1); construct_nvme_tcp_bdev connects an
+  // initiator bdev to any NVMe/TCP target (ours or a foreign one).
+  struct Targets {
+    std::mutex mutex;
+    std::map<std::string, std::shared_ptr<NvmfTcpTarget>> by_nqn;
+  };
+  auto targets = std::make_shared<Targets>();
+
+  server->register_method(
+      "nvmf_create_target", [&manager, targets](const Json& p) {
+        const std::string subnqn =
+            p.get_string("subnqn", "nqn.2026-01.com.amd:oim-amd");
+        std::lock_guard<std::mutex> lock(targets->mutex);
+        if (targets->by_nqn.count(subnqn)) {
+          throw RpcError{kInvalidParams, "target " + subnqn + " exists"};
+        }
+        auto target = start_nvmf_tcp_target(
+            p.get_string("listen_addr"),
+            static_cast<uint16_t>(p.get_int("port", 0)), subnqn,
+            p.get_bool("digests", true));
+        if (const Json* bdevs = p.get("bdevs")) {
+          for (const Json& name : bdevs->as_array()) {
+            BdevPtr bdev = manager.find(name.as_string());
+            if (!bdev) {
+              target->stop();
+              not_found("bdev " + name.as_string());
+            }
+            target->add_namespace(bdev);
+          }
+        }
+        JsonObject o;
+        o["port"] = Json(static_cast<int64_t>(target->port()));
+        o["subnqn"] = Json(subnqn);
+        targets->by_nqn[subnqn] = std::move(target);
+        return Json(std::move(o));
+      });
+
+  server->register_method("nvmf_delete_target", [targets](const Json& p) {
+    std::lock_guard<std::mutex> lock(targets->mutex);
+    auto it = targets->by_nqn.find(p.get_string("subnqn"));
+    if (it == targets->by_nqn.end()) not_found("nvmf target");
+    it->second->stop();
+    targets->by_nqn.erase(it);
+    return Json(JsonObject{});
+  });
+
+  server->register_method(
+      "construct_nvme_tcp_bdev", [&manager](const Json& p) {
+        const std::string name = p.get_string("name");
+        if (name.empty()) throw RpcError{kInvalidParams, "name required"};
+        if (manager.find(name)) {
+          throw RpcError{kInvalidParams, "bdev " + name + " already exists"};
+        }
+        try {
+          BdevPtr bdev = create_nvmf_tcp_bdev(
+              name, p.get_string("traddr", "127.0.0.1"),
+              static_cast<uint16_t>(p.get_int("trsvcid")),
+              p.get_string("subnqn", "nqn.2026-01.com.amd:oim-amd"),
+              static_cast<uint32_t>(p.get_int("nsid", 1)),
+              p.get_bool("digests", true));
+          manager.add(bdev);
+        } catch (const std::exception& e) {
+          throw RpcError{kInternalError, e.what()};
+        }
+        return Json(name);
+      });
 
   // --- composite bdevs (BASELINE config 5) -------------------------------
   // Striped / replicated malloc bdevs spanning the node's GPUs. The

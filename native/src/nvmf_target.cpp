@@ -1,0 +1,458 @@
+// NVMe-oF/TCP target: exports hipstored bdevs as namespaces.
+//
+// Counterpart of the SPDK nvmf target the reference's initiator would
+// talk to (reference lib/nvme/nvme_tcp.c peer); one thread per
+// connection (the control plane's scale; the polled initiator is the
+// measured data path). HBM-namespace C2HData digests come from the GPU
+// CRC32C kernel + host combine.
+
+#include <arpa/inet.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <cstring>
+#include <map>
+#include <mutex>
+#include <thread>
+#include <vector>
+
+#include "hipstore/engine.h"
+#include "hipstore/nvmf.h"
+#include "nvmf_common.h"
+
+namespace hipstore {
+
+namespace {
+
+using namespace nvmf;
+
+// CRC32C of an aligned HBM range via the GPU kernel, else software.
+uint32_t range_crc32c(Bdev* bdev, uint64_t offset, const void* host_copy,
+                      uint32_t len) {
+  if (bdev->device_base() != nullptr && len >= 16384 && len % 4096 == 0 &&
+      offset % 4096 == 0) {
+    const uint32_t blocks = len / 4096;
+    std::vector<uint32_t> crcs(blocks);
+    crc32c_hbm_blocks(bdev, offset, 4096, blocks, crcs.data());
+    uint32_t crc = crcs[0];
+    for (uint32_t i = 1; i < blocks; ++i) {
+      crc = crc32c_combine(crc, crcs[i], 4096);
+    }
+    return crc;
+  }
+  return crc32c_sw(0, host_copy, len);
+}
+
+class TargetImpl : public NvmfTcpTarget,
+                   public std::enable_shared_from_this<TargetImpl> {
+ public:
+  TargetImpl(const std::string& addr, uint16_t port, std::string subnqn,
+             bool digests)
+      : subnqn_(std::move(subnqn)), digests_(digests) {
+    listen_fd_ = socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
+    if (listen_fd_ < 0) throw std::runtime_error("nvmf target: socket failed");
+    int one = 1;
+    setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    sockaddr_in sa{};
+    sa.sin_family = AF_INET;
+    sa.sin_port = htons(port);
+    sa.sin_addr.s_addr =
+        addr.empty() ? htonl(INADDR_LOOPBACK) : inet_addr(addr.c_str());
+    if (bind(listen_fd_, reinterpret_cast<sockaddr*>(&sa), sizeof(sa)) < 0 ||
+        listen(listen_fd_, 16) < 0) {
+      close(listen_fd_);
+      throw std::runtime_error("nvmf target: bind/listen failed");
+    }
+    socklen_t slen = sizeof(sa);
+    getsockname(listen_fd_, reinterpret_cast<sockaddr*>(&sa), &slen);
+    port_ = ntohs(sa.sin_port);
+  }
+
+  ~TargetImpl() override { stop(); }
+
+  void start() {
+    running_.store(true);
+    accept_thread_ = std::thread([this] { accept_loop(); });
+  }
+
+  uint16_t port() const override { return port_; }
+
+  void add_namespace(BdevPtr bdev) override {
+    std::lock_guard<std::mutex> lock(mutex_);
+    namespaces_.push_back(std::move(bdev));
+  }
+
+  void stop() override {
+    if (!running_.exchange(false)) return;
+    shutdown(listen_fd_, SHUT_RDWR);
+    close(listen_fd_);
+    if (accept_thread_.joinable()) accept_thread_.join();
+    std::vector<std::thread> conns;
+    {
+      std::lock_guard<std::mutex> lock(mutex_);
+      conns.swap(connections_);
+    }
+    for (auto& t : conns) {
+      if (t.joinable()) t.join();
+    }
+  }
+
+ private:
+  BdevPtr ns(uint32_t nsid) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    if (nsid == 0 || nsid > namespaces_.size()) return nullptr;
+    return namespaces_[nsid - 1];
+  }
+
+  void accept_loop() {
+    while (running_.load()) {
+      int fd = accept(listen_fd_, nullptr, nullptr);
+      if (fd < 0) {
+        if (!running_.load()) break;
+        continue;
+      }
+      int one = 1;
+      setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+      std::lock_guard<std::mutex> lock(mutex_);
+      connections_.emplace_back([this, fd] {
+        serve(fd);
+        close(fd);
+      });
+    }
+  }
+
+  struct PendingWrite {
+    Sqe sqe;
+    std::vector<uint8_t> data;
+    uint32_t received = 0;
+  };
+
+  void serve(int fd) {
+    // --- ICReq / ICResp handshake ---------------------------------------
+    IcReq icreq{};
+    if (!read_exact(fd, &icreq, sizeof(icreq)) || icreq.ch.type != kIcReq) {
+      return;
+    }
+    const bool hdgst = digests_ && (icreq.dgst & 0x1);
+    const bool ddgst = digests_ && (icreq.dgst & 0x2);
+    IcResp icresp{};
+    icresp.ch = {kIcResp, 0, sizeof(IcResp), 0, sizeof(IcResp)};
+    icresp.pfv = 0;
+    icresp.cpda = 0;
+    icresp.dgst = (hdgst ? 0x1 : 0) | (ddgst ? 0x2 : 0);
+    icresp.maxh2cdata = kMaxDataPerPdu;
+    if (!write_exact(fd, &icresp, sizeof(icresp))) return;
+
+    uint32_t cc = 0;
+    uint16_t next_ttag = 1;
+    std::map<uint16_t, PendingWrite> pending_writes;  // by ttag
+    std::vector<uint8_t> header(256);
+
+    auto send_cqe = [&](uint16_t cid, uint16_t sc, uint64_t result = 0) {
+      struct {
+        CommonHeader ch;
+        Cqe cqe;
+      } __attribute__((packed)) resp{};
+      resp.ch = {kCapsuleResp, static_cast<uint8_t>(hdgst ? kFlagHdgst : 0),
+                 24, 0, static_cast<uint32_t>(24 + (hdgst ? 4 : 0))};
+      resp.cqe.set_result64(result);
+      resp.cqe.set_cid(cid);
+      resp.cqe.set_status(sc);
+      return send_pdu(fd, &resp, 24, hdgst, nullptr, 0, 0, false);
+    };
+
+    auto send_c2h_data = [&](uint16_t cid, const uint8_t* data, uint32_t len,
+                             uint32_t digest_all) {
+      // Split into <= kMaxDataPerPdu chunks; per-PDU DDGST.
+      uint32_t offset = 0;
+      while (offset < len) {
+        const uint32_t chunk = std::min(len - offset, kMaxDataPerPdu);
+        DataHeader dh{};
+        const uint8_t pdo = 24 + (hdgst ? 4 : 0);
+        const bool last = offset + chunk >= len;
+        dh.ch = {kC2HData,
+                 static_cast<uint8_t>((hdgst ? kFlagHdgst : 0) |
+                                      (ddgst ? kFlagDdgst : 0) |
+                                      (last ? kFlagLast : 0)),
+                 24, pdo,
+                 static_cast<uint32_t>(pdo + chunk + (ddgst ? 4 : 0))};
+        dh.cccid = cid;
+        dh.datao = offset;
+        dh.datal = chunk;
+        uint32_t dd = 0;
+        if (ddgst) {
+          dd = (chunk == len && digest_all != 0)
+                   ? digest_all  // whole-range digest already on hand
+                   : crc32c_sw(0, data + offset, chunk);
+        }
+        if (!send_pdu(fd, &dh, 24, hdgst, data + offset, chunk, dd, ddgst)) {
+          return false;
+        }
+        offset += chunk;
+      }
+      return true;
+    };
+
+    auto channel_cache = std::map<Bdev*, std::shared_ptr<IoChannel>>();
+    auto bdev_io = [&](Bdev* bdev, IoOp op, uint64_t off, void* buf,
+                       uint64_t len, uint8_t fill) {
+      auto it = channel_cache.find(bdev);
+      if (it == channel_cache.end()) {
+        it = channel_cache.emplace(bdev, bdev->get_channel()).first;
+      }
+      int result = kIoFailed;
+      bool done = false;
+      IoRequest req;
+      req.op = op;
+      req.offset = off;
+      req.length = len;
+      req.buffer = buf;
+      req.fill = fill;
+      req.on_complete = [&](int status) {
+        result = status;
+        done = true;
+      };
+      bdev->submit(it->second.get(), std::move(req));
+      while (!done) bdev->poll(it->second.get());
+      return result;
+    };
+
+    auto handle_command = [&](const Sqe& sqe, std::vector<uint8_t>& capsule_data) {
+      if (sqe.opc() == kOpcFabrics) {
+        switch (sqe.fctype()) {
+          case kFctypeConnect: {
+            // cntlid in dw0.
+            return send_cqe(sqe.cid(), kScSuccess, 1);
+          }
+          case kFctypePropertySet: {
+            uint32_t ofst = sqe.cdw(11);
+            if (ofst == kPropCc) cc = sqe.cdw(12);
+            return send_cqe(sqe.cid(), kScSuccess);
+          }
+          case kFctypePropertyGet: {
+            uint32_t ofst = sqe.cdw(11);
+            uint64_t value = 0;
+            if (ofst == kPropCap) {
+              value = 0x3FFull | (1ull << 37);  // MQES=1023, NVM cmd set
+            } else if (ofst == kPropCsts) {
+              value = (cc & 1) ? 1 : 0;  // RDY follows CC.EN
+            } else if (ofst == kPropCc) {
+              value = cc;
+            }
+            return send_cqe(sqe.cid(), kScSuccess, value);
+          }
+          default:
+            return send_cqe(sqe.cid(), kScInvalidField);
+        }
+      }
+      switch (sqe.opc()) {
+        case kOpcIdentify: {
+          const uint8_t cns = sqe.cdw(10) & 0xFF;
+          std::vector<uint8_t> data(4096, 0);
+          if (cns == 0x00) {  // namespace
+            BdevPtr bdev = ns(sqe.nsid());
+            if (!bdev) return send_cqe(sqe.cid(), kScInvalidField);
+            uint64_t nsze = bdev->num_blocks();
+            memcpy(data.data() + 0, &nsze, 8);
+            memcpy(data.data() + 8, &nsze, 8);
+            memcpy(data.data() + 16, &nsze, 8);
+            uint8_t lbads = 0;
+            for (uint64_t bs = bdev->block_size(); bs > 1; bs >>= 1) ++lbads;
+            data[128 + 2] = lbads;  // LBAF0.LBADS
+          } else if (cns == 0x01) {  // controller: zeros suffice
+          } else {
+            return send_cqe(sqe.cid(), kScInvalidField);
+          }
+          if (!send_c2h_data(sqe.cid(), data.data(), data.size(), 0)) {
+            return false;
+          }
+          return send_cqe(sqe.cid(), kScSuccess);
+        }
+        case kOpcKeepAlive:
+          return send_cqe(sqe.cid(), kScSuccess);
+        case kOpcRead: {
+          BdevPtr bdev = ns(sqe.nsid());
+          if (!bdev) return send_cqe(sqe.cid(), kScInvalidField);
+          const uint64_t slba =
+              sqe.cdw(10) | (static_cast<uint64_t>(sqe.cdw(11)) << 32);
+          const uint32_t nlb = (sqe.cdw(12) & 0xFFFF) + 1;
+          const uint64_t off = slba * bdev->block_size();
+          const uint64_t len = static_cast<uint64_t>(nlb) * bdev->block_size();
+          if (off + len > bdev->size_bytes()) {
+            return send_cqe(sqe.cid(), kScLbaOutOfRange);
+          }
+          uint8_t* bounce = static_cast<uint8_t*>(alloc_pinned(len));
+          int status = bdev_io(bdev.get(), IoOp::kRead, off, bounce, len, 0);
+          if (status != kIoOk) {
+            free_pinned(bounce);
+            return send_cqe(sqe.cid(), kScInternalError);
+          }
+          // GPU digest for HBM namespaces (per-4KiB kernel + combine).
+          uint32_t digest = 0;
+          if (ddgst && len <= kMaxDataPerPdu) {
+            digest = range_crc32c(bdev.get(), off, bounce, len);
+          }
+          bool ok = send_c2h_data(sqe.cid(), bounce, len, digest);
+          free_pinned(bounce);
+          if (!ok) return false;
+          return send_cqe(sqe.cid(), kScSuccess);
+        }
+        case kOpcWrite: {
+          BdevPtr bdev = ns(sqe.nsid());
+          if (!bdev) return send_cqe(sqe.cid(), kScInvalidField);
+          const uint64_t slba =
+              sqe.cdw(10) | (static_cast<uint64_t>(sqe.cdw(11)) << 32);
+          const uint32_t nlb = (sqe.cdw(12) & 0xFFFF) + 1;
+          const uint64_t len = static_cast<uint64_t>(nlb) * bdev->block_size();
+          if (slba * bdev->block_size() + len > bdev->size_bytes()) {
+            return send_cqe(sqe.cid(), kScLbaOutOfRange);
+          }
+          if (!capsule_data.empty()) {
+            // In-capsule data (small writes).
+            if (capsule_data.size() != len) {
+              return send_cqe(sqe.cid(), kScInvalidField);
+            }
+            uint8_t* bounce = static_cast<uint8_t*>(alloc_pinned(len));
+            memcpy(bounce, capsule_data.data(), len);
+            int status = bdev_io(bdev.get(), IoOp::kWrite,
+                                 slba * bdev->block_size(), bounce, len, 0);
+            free_pinned(bounce);
+            return send_cqe(sqe.cid(),
+                            status == kIoOk ? kScSuccess : kScInternalError);
+          }
+          // Solicit the data with one R2T covering the whole transfer.
+          PendingWrite pw;
+          pw.sqe = sqe;
+          pw.data.resize(len);
+          uint16_t ttag = next_ttag++;
+          pending_writes[ttag] = std::move(pw);
+          DataHeader r2t{};
+          r2t.ch = {kR2T, static_cast<uint8_t>(hdgst ? kFlagHdgst : 0), 24, 0,
+                    static_cast<uint32_t>(24 + (hdgst ? 4 : 0))};
+          r2t.cccid = sqe.cid();
+          r2t.ttag = ttag;
+          r2t.datao = 0;
+          r2t.datal = len;
+          return send_pdu(fd, &r2t, 24, hdgst, nullptr, 0, 0, false);
+        }
+        case kOpcFlush:
+          return send_cqe(sqe.cid(), kScSuccess);
+        case kOpcWriteZeroes: {
+          BdevPtr bdev = ns(sqe.nsid());
+          if (!bdev) return send_cqe(sqe.cid(), kScInvalidField);
+          const uint64_t slba =
+              sqe.cdw(10) | (static_cast<uint64_t>(sqe.cdw(11)) << 32);
+          const uint32_t nlb = (sqe.cdw(12) & 0xFFFF) + 1;
+          int status = bdev_io(bdev.get(), IoOp::kFill,
+                               slba * bdev->block_size(), nullptr,
+                               static_cast<uint64_t>(nlb) * bdev->block_size(),
+                               0);
+          return send_cqe(sqe.cid(),
+                          status == kIoOk ? kScSuccess : kScInternalError);
+        }
+        default:
+          return send_cqe(sqe.cid(), kScInvalidOpcode);
+      }
+    };
+
+    // --- PDU loop --------------------------------------------------------
+    while (running_.load()) {
+      CommonHeader ch;
+      if (!read_exact(fd, &ch, sizeof(ch))) break;
+      if (ch.hlen < sizeof(ch) || ch.plen < ch.hlen) break;
+      header.resize(ch.hlen);
+      memcpy(header.data(), &ch, sizeof(ch));
+      if (!read_exact(fd, header.data() + sizeof(ch), ch.hlen - sizeof(ch))) {
+        break;
+      }
+      if (hdgst) {
+        uint32_t hd;
+        if (!read_exact(fd, &hd, 4)) break;
+        if (hd != crc32c_sw(0, header.data(), ch.hlen)) break;
+      }
+      // Data length: everything between the data offset and the digest.
+      const uint32_t pdo = ch.pdo ? ch.pdo : ch.hlen + (hdgst ? 4 : 0);
+      uint32_t data_len = 0;
+      bool has_ddgst = false;
+      if (ch.plen > pdo) {
+        has_ddgst = ddgst;
+        data_len = ch.plen - pdo - (has_ddgst ? 4 : 0);
+        // skip padding between header(+hdgst) and pdo
+        uint32_t pad = pdo - ch.hlen - (hdgst ? 4 : 0);
+        if (pad > 16) break;
+        uint8_t padbuf[16];
+        if (pad && !read_exact(fd, padbuf, pad)) break;
+      }
+      std::vector<uint8_t> data(data_len);
+      if (data_len > 0 && !read_exact(fd, data.data(), data_len)) break;
+      if (has_ddgst && data_len > 0) {
+        uint32_t dd;
+        if (!read_exact(fd, &dd, 4)) break;
+        if (dd != crc32c_sw(0, data.data(), data_len)) {
+          // data corruption: fail hard (a real target sends TermReq)
+          break;
+        }
+      }
+
+      if (ch.type == kCapsuleCmd) {
+        Sqe sqe;
+        memcpy(sqe.bytes, header.data() + 8, 64);
+        if (!handle_command(sqe, data)) break;
+      } else if (ch.type == kH2CData) {
+        DataHeader dh;
+        memcpy(&dh, header.data(), sizeof(dh));
+        auto it = pending_writes.find(dh.ttag);
+        if (it == pending_writes.end()) break;
+        PendingWrite& pw = it->second;
+        if (dh.datao + data_len > pw.data.size()) break;
+        memcpy(pw.data.data() + dh.datao, data.data(), data_len);
+        pw.received += data_len;
+        if (pw.received >= pw.data.size()) {
+          BdevPtr bdev = ns(pw.sqe.nsid());
+          const uint64_t slba = pw.sqe.cdw(10) |
+                                (static_cast<uint64_t>(pw.sqe.cdw(11)) << 32);
+          uint8_t* bounce =
+              static_cast<uint8_t*>(alloc_pinned(pw.data.size()));
+          memcpy(bounce, pw.data.data(), pw.data.size());
+          int status =
+              bdev_io(bdev.get(), IoOp::kWrite, slba * bdev->block_size(),
+                      bounce, pw.data.size(), 0);
+          free_pinned(bounce);
+          uint16_t cid = pw.sqe.cid();
+          pending_writes.erase(it);
+          if (!send_cqe(cid, status == kIoOk ? kScSuccess : kScInternalError)) {
+            break;
+          }
+        }
+      } else {
+        break;  // unexpected PDU
+      }
+    }
+  }
+
+  std::string subnqn_;
+  bool digests_;
+  int listen_fd_ = -1;
+  uint16_t port_ = 0;
+  std::atomic<bool> running_{false};
+  std::thread accept_thread_;
+  std::mutex mutex_;
+  std::vector<std::thread> connections_;
+  std::vector<BdevPtr> namespaces_;
+};
+
+}  // namespace
+
+std::shared_ptr<NvmfTcpTarget> start_nvmf_tcp_target(
+    const std::string& listen_addr, uint16_t port, const std::string& subnqn,
+    bool enable_digests) {
+  auto target = std::make_shared<TargetImpl>(listen_addr, port, subnqn,
+                                             enable_digests);
+  target->start();
+  return target;
+}
+
+}  // namespace hipstore

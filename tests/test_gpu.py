@@ -132,6 +132,33 @@ class TestCompositeGpu:
 
 
 @needs_gpu
+class TestNvmfGpu:
+    def test_hbm_namespace_gpu_digest_roundtrip(self):
+        """NVMe/TCP loopback with an HBM-resident namespace: C2HData
+        digests for aligned >=16 KiB reads are produced by the GPU
+        CRC32C kernel + host combine; the initiator verifies them in
+        software, so a mismatch anywhere fails the read."""
+        backing = hs.create_hbm_bdev("nvmf-hbm", 4096, 16384, device=0)
+        target = hs.start_nvmf_tcp_target(
+            "", 0, "nqn.2026-01.com.amd:gpu-ns", True)
+        target.add_namespace(backing)
+        try:
+            bdev = hs.create_nvmf_tcp_bdev(
+                "nvmf-gpu", "127.0.0.1", target.port,
+                "nqn.2026-01.com.amd:gpu-ns")
+            rng = random.Random(31)
+            data = bytes(rng.getrandbits(8) for _ in range(128 * 1024))
+            bdev.write(0, data)
+            # 128 KiB aligned read -> GPU-digested C2HData
+            assert bdev.read(0, len(data)) == data
+            assert backing.read(0, len(data)) == data
+            r = hs.run_bdevperf(bdev, "randread", 65536, 8, 2, 0.5)
+            assert r["iops"] > 100, r
+        finally:
+            target.stop()
+
+
+@needs_gpu
 class TestDaemonHbm:
     def test_daemon_gpu_mode(self, tmp_path):
         fixture = launch_hipstored(tmp_path, cpu=False)
