@@ -121,7 +121,9 @@ class SimpleLogger(Logger):
         formatter: Optional[Formatter] = None,
     ):
         self.level = level
-        self.output = output if output is not None else sys.stderr
+        # None = "current sys.stderr", resolved per log call so stream
+        # redirection (tests, daemons re-opening stderr) is honored.
+        self.output = output
         self.formatter = formatter or Formatter()
         self._mutex = threading.Lock()
 
@@ -129,8 +131,12 @@ class SimpleLogger(Logger):
         if level < self.level:
             return
         line = self.formatter.format(level, msg, fields)
+        stream = self.output if self.output is not None else sys.stderr
         with self._mutex:
-            print(line, file=self.output, flush=True)
+            try:
+                print(line, file=stream, flush=True)
+            except ValueError:
+                pass  # stream closed under us (interpreter shutdown)
 
 
 class NullLogger(Logger):
