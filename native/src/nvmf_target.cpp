@@ -151,6 +151,22 @@ class TargetImpl : public NvmfTcpTarget,
     std::map<uint16_t, PendingWrite> pending_writes;  // by ttag
     std::vector<uint8_t> header(256);
 
+    // Reusable pinned bounce arena: hipHostMalloc costs ~1 ms, so a
+    // per-I/O allocation collapses the connection to ~1K IOPS. Grown
+    // geometrically, freed with the connection.
+    uint8_t* bounce_arena = nullptr;
+    size_t bounce_capacity = 0;
+    auto bounce = [&](size_t len) -> uint8_t* {
+      if (len > bounce_capacity) {
+        size_t want = std::max<size_t>(len, 256 * 1024);
+        want = std::max(want, bounce_capacity * 2);
+        free_pinned(bounce_arena);
+        bounce_arena = static_cast<uint8_t*>(alloc_pinned(want));
+        bounce_capacity = want;
+      }
+      return bounce_arena;
+    };
+
     auto send_cqe = [&](uint16_t cid, uint16_t sc, uint64_t result = 0) {
       struct {
         CommonHeader ch;
@@ -284,20 +300,17 @@ class TargetImpl : public NvmfTcpTarget,
           if (off + len > bdev->size_bytes()) {
             return send_cqe(sqe.cid(), kScLbaOutOfRange);
           }
-          uint8_t* bounce = static_cast<uint8_t*>(alloc_pinned(len));
-          int status = bdev_io(bdev.get(), IoOp::kRead, off, bounce, len, 0);
+          uint8_t* buf = bounce(len);
+          int status = bdev_io(bdev.get(), IoOp::kRead, off, buf, len, 0);
           if (status != kIoOk) {
-            free_pinned(bounce);
             return send_cqe(sqe.cid(), kScInternalError);
           }
           // GPU digest for HBM namespaces (per-4KiB kernel + combine).
           uint32_t digest = 0;
           if (ddgst && len <= kMaxDataPerPdu) {
-            digest = range_crc32c(bdev.get(), off, bounce, len);
+            digest = range_crc32c(bdev.get(), off, buf, len);
           }
-          bool ok = send_c2h_data(sqe.cid(), bounce, len, digest);
-          free_pinned(bounce);
-          if (!ok) return false;
+          if (!send_c2h_data(sqe.cid(), buf, len, digest)) return false;
           return send_cqe(sqe.cid(), kScSuccess);
         }
         case kOpcWrite: {
@@ -315,11 +328,10 @@ class TargetImpl : public NvmfTcpTarget,
             if (capsule_data.size() != len) {
               return send_cqe(sqe.cid(), kScInvalidField);
             }
-            uint8_t* bounce = static_cast<uint8_t*>(alloc_pinned(len));
-            memcpy(bounce, capsule_data.data(), len);
+            uint8_t* buf = bounce(len);
+            memcpy(buf, capsule_data.data(), len);
             int status = bdev_io(bdev.get(), IoOp::kWrite,
-                                 slba * bdev->block_size(), bounce, len, 0);
-            free_pinned(bounce);
+                                 slba * bdev->block_size(), buf, len, 0);
             return send_cqe(sqe.cid(),
                             status == kIoOk ? kScSuccess : kScInternalError);
           }
@@ -414,13 +426,11 @@ class TargetImpl : public NvmfTcpTarget,
           BdevPtr bdev = ns(pw.sqe.nsid());
           const uint64_t slba = pw.sqe.cdw(10) |
                                 (static_cast<uint64_t>(pw.sqe.cdw(11)) << 32);
-          uint8_t* bounce =
-              static_cast<uint8_t*>(alloc_pinned(pw.data.size()));
-          memcpy(bounce, pw.data.data(), pw.data.size());
+          uint8_t* buf = bounce(pw.data.size());
+          memcpy(buf, pw.data.data(), pw.data.size());
           int status =
               bdev_io(bdev.get(), IoOp::kWrite, slba * bdev->block_size(),
-                      bounce, pw.data.size(), 0);
-          free_pinned(bounce);
+                      buf, pw.data.size(), 0);
           uint16_t cid = pw.sqe.cid();
           pending_writes.erase(it);
           if (!send_cqe(cid, status == kIoOk ? kScSuccess : kScInternalError)) {
@@ -431,6 +441,7 @@ class TargetImpl : public NvmfTcpTarget,
         break;  // unexpected PDU
       }
     }
+    free_pinned(bounce_arena);
   }
 
   std::string subnqn_;
