@@ -45,6 +45,8 @@ def start_stack(rank, local_rank, use_gpu, args, tmp):
     daemon_sock = os.path.join(tmp, "hipstored.sock")
     cmd = [os.path.join(REPO_ROOT, "bin", "hipstored"), "-S", daemon_sock,
            "-d", str(local_rank)]
+    if args.engine == "persistent":
+        cmd.append("-P")
     if not use_gpu:
         cmd.append("-C")
     daemon = subprocess.Popen(cmd, stderr=subprocess.DEVNULL)
@@ -124,6 +126,10 @@ def main() -> int:
     parser.add_argument("--num-queues", type=int, default=8)
     parser.add_argument("--bdev-gb", type=float, default=8.0)
     parser.add_argument("--workload", default="randread")
+    parser.add_argument("--engine", default="persistent",
+                        choices=["batched", "persistent"],
+                        help="HBM I/O engine: batched kernel launches or "
+                             "the on-GPU polling service kernel")
     args = parser.parse_args()
 
     import torch
@@ -228,6 +234,7 @@ def main() -> int:
                 "bdev_gb": args.bdev_gb,
                 "step_ios_per_gpu": STEP_IOS,
                 "backend": "hbm" if use_gpu else "cpu",
+                "engine": args.engine,
                 "provisioning": "csi-createvolume+proxy-mapvolume",
                 "p99_us": round(p99, 1),
             },

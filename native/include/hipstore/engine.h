@@ -28,8 +28,12 @@ std::string gpu_pci_address(int device);
 // HBM bdev IS the malloc bdev on this platform, and the reference
 // controller keys its keep-on-unmap logic on that string
 // (reference controller.go:205).
+// `persistent` selects the on-GPU polling service kernel (lowest
+// latency; waves poll the submission ring and self-exit when idle)
+// instead of batched per-poll launches.
 BdevPtr create_hbm_bdev(const std::string& name, uint64_t block_size,
-                        uint64_t num_blocks, int device);
+                        uint64_t num_blocks, int device,
+                        bool persistent = false);
 
 // Pinned-host buffer helpers (fall back to plain malloc without a GPU).
 void* alloc_pinned(size_t bytes);

@@ -65,6 +65,7 @@ class RpcServer {
 // Registers the SPDK-compatible method set
 // against the global BdevManager. `use_hbm` selects HBM-resident malloc
 // bdevs on `device`; false (or no GPU) falls back to host RAM.
-void register_storage_methods(RpcServer* server, bool use_hbm, int device);
+void register_storage_methods(RpcServer* server, bool use_hbm, int device,
+                              bool persistent = false);
 
 }  // namespace hipstore

@@ -37,7 +37,8 @@ void usage(const char* argv0) {
           "usage: %s [-S socket] [-d device] [-C] \n"
           "  -S PATH   JSON-RPC unix socket (default /var/tmp/hipstored.sock)\n"
           "  -d N      HIP device index backing malloc bdevs (default 0)\n"
-          "  -C        CPU mode: host-RAM bdevs even when a GPU is present\n",
+          "  -C        CPU mode: host-RAM bdevs even when a GPU is present\n"
+          "  -P        persistent engine: on-GPU polling service kernels\n",
           argv0);
 }
 
@@ -47,13 +48,15 @@ int main(int argc, char** argv) {
   std::string socket_path = "/var/tmp/hipstored.sock";
   int device = 0;
   bool cpu_only = false;
+  bool persistent = false;
 
   int opt;
-  while ((opt = getopt(argc, argv, "S:d:Ch")) != -1) {
+  while ((opt = getopt(argc, argv, "S:d:CPh")) != -1) {
     switch (opt) {
       case 'S': socket_path = optarg; break;
       case 'd': device = atoi(optarg); break;
       case 'C': cpu_only = true; break;
+      case 'P': persistent = true; break;
       case 'h': usage(argv[0]); return 0;
       default: usage(argv[0]); return 2;
     }
@@ -65,7 +68,7 @@ int main(int argc, char** argv) {
           hipstore::gpu_device_count());
 
   hipstore::RpcServer server(socket_path);
-  hipstore::register_storage_methods(&server, use_hbm, device);
+  hipstore::register_storage_methods(&server, use_hbm, device, persistent);
   try {
     server.start();
   } catch (const std::exception& e) {

@@ -87,7 +87,8 @@ Json bdev_to_json(const BdevPtr& bdev) {
 
 }  // namespace
 
-void register_storage_methods(RpcServer* server, bool use_hbm, int device) {
+void register_storage_methods(RpcServer* server, bool use_hbm, int device,
+                              bool persistent) {
   auto& manager = BdevManager::instance();
 
   server->register_method("get_bdevs", [&manager](const Json& p) {
@@ -115,7 +116,8 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device) {
   });
 
   server->register_method(
-      "construct_malloc_bdev", [&manager, use_hbm, device](const Json& p) {
+      "construct_malloc_bdev",
+      [&manager, use_hbm, device, persistent](const Json& p) {
         const int64_t num_blocks = p.get_int("num_blocks");
         const int64_t block_size = p.get_int("block_size");
         if (num_blocks <= 0 || block_size <= 0 || block_size % 512 != 0) {
@@ -131,7 +133,8 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device) {
         }
         BdevPtr bdev;
         if (use_hbm && gpu_available()) {
-          bdev = create_hbm_bdev(name, block_size, num_blocks, device);
+          bdev = create_hbm_bdev(name, block_size, num_blocks, device,
+                                 p.get_bool("persistent", persistent));
         } else {
           bdev = create_malloc_bdev(name, block_size, num_blocks);
         }

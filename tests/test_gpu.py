@@ -66,6 +66,47 @@ class TestHbmBdev:
 
 
 @needs_gpu
+class TestPersistentEngine:
+    """The on-GPU polling service kernel: same numerics contract as the
+    batched engine, verified against host byte references."""
+
+    @pytest.fixture(scope="class")
+    def bdev(self):
+        return hs.create_hbm_bdev("gpu-pers", 512, 131072, device=0,
+                                  persistent=True)
+
+    def test_roundtrip(self, bdev):
+        rng = random.Random(41)
+        blob = bytes(rng.getrandbits(8) for _ in range(64 * 1024))
+        bdev.write(8192, blob)
+        assert bdev.read(8192, len(blob)) == blob
+
+    def test_fill_and_small_blocks(self, bdev):
+        bdev.fill(0, 0x3C, 4096)
+        assert bdev.read(0, 512) == b"\x3c" * 512
+        payload = bytes(random.getrandbits(8) for _ in range(512))
+        bdev.write(512, payload)
+        assert bdev.read(512, 512) == payload
+
+    def test_idle_exit_and_relaunch(self, bdev):
+        """The service kernel self-exits when idle (~1s) and must
+        transparently relaunch on the next submission."""
+        import time
+        bdev.write(0, b"\x11" * 512)
+        time.sleep(2.5)  # beyond the idle timeout
+        bdev.write(512, b"\x22" * 512)
+        assert bdev.read(0, 512) == b"\x11" * 512
+        assert bdev.read(512, 512) == b"\x22" * 512
+
+    def test_perf_sanity(self):
+        bdev = hs.create_hbm_bdev("gpu-pers-perf", 4096, 262144, device=0,
+                                  persistent=True)
+        bdev.fill(0, 0x5A, bdev.size_bytes)
+        r = hs.run_bdevperf(bdev, "randread", 4096, 32, 4, 1.0)
+        assert r["iops"] > 100_000, r
+
+
+@needs_gpu
 class TestCrc32c:
     def test_gpu_matches_software(self):
         bdev = hs.create_hbm_bdev("crc-test", 4096, 256, device=0)
