@@ -160,17 +160,53 @@ struct PersistentCtl {
   volatile unsigned long long* sq_tail;  // host-written monotonic count
   volatile unsigned long long* cq;  // per-slot completion: seq = idx+1
   volatile uint32_t* stop;     // host-set stop flag
-  // Device memory:
-  unsigned long long* claim_counter;  // monotonic claim index (64-bit:
-                                      // never wraps within a deployment)
+  // Device memory. Only the LEADER wave touches host memory while
+  // idle: with many queues, dozens of waves spin-reading a pinned
+  // word would storm PCIe with small reads (measured: 4 channels x 16
+  // spinning waves collapsed throughput ~30x). Workers poll the
+  // device-resident mirror instead (relaxed agent loads = L2-served
+  // sc1, no L1 invalidates — the monotonic counter IS the flag).
+  unsigned long long* claim_counter;  // monotonic claim index
+  unsigned long long* known_tail;     // leader-published tail mirror
+  uint32_t* exit_flag;                // leader-published stop/idle exit
   uint32_t ring_mask;
-  uint32_t idle_spins;         // spins before self-exit
+  uint32_t idle_spins;         // leader spins before self-exit
 };
 
 __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
   __shared__ __attribute__((aligned(16))) uint8_t lds_raw[kTileBytes];
   const uint32_t lane = threadIdx.x & 63;
   float4* lds = reinterpret_cast<float4*>(lds_raw);
+
+  if (blockIdx.x == 0) {
+    // Leader wave: the only PCIe poller. Mirrors sq_tail into device
+    // memory and publishes stop/idle exits.
+    if (lane != 0) return;
+    unsigned long long known = __hip_atomic_load(
+        ctl.known_tail, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    uint32_t spins = 0;
+    while (true) {
+      unsigned long long tail = __hip_atomic_load(
+          const_cast<const unsigned long long*>(ctl.sq_tail),
+          __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM);
+      if (tail > known) {
+        known = tail;
+        __hip_atomic_store(ctl.known_tail, known, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        spins = 0;
+      }
+      if (__hip_atomic_load(const_cast<const uint32_t*>(ctl.stop),
+                            __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_SYSTEM) != 0 ||
+          ++spins > ctl.idle_spins) {
+        __hip_atomic_store(ctl.exit_flag, 1u, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        return;
+      }
+      __builtin_amdgcn_s_sleep(8);
+    }
+  }
+
   while (true) {
     // One lane claims the next descriptor index; the wave follows
     // (64-bit broadcast as two 32-bit halves).
@@ -186,20 +222,17 @@ __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
       claim = (static_cast<unsigned long long>(static_cast<uint32_t>(hi))
                << 32) | static_cast<uint32_t>(lo);
     }
-    // Wait for the host to publish it (or stop / go idle).
-    uint32_t spins = 0;
+    // Wait for the leader to publish it (or exit). Device-memory
+    // polling only: no PCIe traffic from workers while idle.
     while (true) {
       unsigned long long tail = __hip_atomic_load(
-          const_cast<const unsigned long long*>(ctl.sq_tail),
-          __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM);
+          ctl.known_tail, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
       if (tail > claim) break;
-      if (__hip_atomic_load(const_cast<const uint32_t*>(ctl.stop),
-                            __ATOMIC_RELAXED,
-                            __HIP_MEMORY_SCOPE_SYSTEM) != 0) {
+      if (__hip_atomic_load(ctl.exit_flag, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT) != 0) {
         return;
       }
-      if (++spins > ctl.idle_spins) return;  // idle self-exit
-      __builtin_amdgcn_s_sleep(32);
+      __builtin_amdgcn_s_sleep(16);
     }
     const BlockDesc d = ctl.sq[claim & ctl.ring_mask];
     const uint32_t n16 = d.bytes >> 4;
@@ -477,8 +510,9 @@ class HbmPersistentChannel : public IoChannel {
     *sq_tail_ = 0;
     *stop_ = 0;
     memset(const_cast<unsigned long long*>(cq_), 0, kRing * 8);
-    HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&claim_ctr_), 8));
-    HIP_CHECK(hipMemset(claim_ctr_, 0, 8));
+    // Device words: [0]=claim counter, [1]=known_tail, [2]=exit_flag.
+    HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&claim_ctr_), 24));
+    HIP_CHECK(hipMemset(claim_ctr_, 0, 24));
     desc_io_.resize(kRing, nullptr);
     launch();
   }
@@ -542,11 +576,19 @@ class HbmPersistentChannel : public IoChannel {
       ++completed;
     }
     drain_pending();
-    // Liveness: if work is outstanding but the service kernel has
-    // idle-exited, relaunch from the completed prefix (descriptor
-    // replay is idempotent).
-    if (completed_ < tail_ && hipStreamQuery(stream_) == hipSuccess) {
-      launch();
+    if (completed > 0) last_progress_ = std::chrono::steady_clock::now();
+    // Liveness: if work is outstanding but nothing completed for a
+    // while, the service kernel may have idle-exited in the submit
+    // race window — relaunch from the completed prefix (descriptor
+    // replay is idempotent). The 1 ms stall gate keeps hipStreamQuery
+    // (a locked runtime call) out of the hot path: unthrottled, 8
+    // polling threads serialized on it and collapsed throughput.
+    if (completed_ < tail_) {
+      const auto now = std::chrono::steady_clock::now();
+      if (now - last_progress_ > std::chrono::milliseconds(1) &&
+          hipStreamQuery(stream_) == hipSuccess) {
+        launch();
+      }
     }
     return completed;
   }
@@ -560,8 +602,9 @@ class HbmPersistentChannel : public IoChannel {
  private:
   void launch() {
     (void)hipSetDevice(device_);
-    const unsigned long long prefix = completed_;
-    HIP_CHECK(hipMemcpyAsync(claim_ctr_, &prefix, 8, hipMemcpyHostToDevice,
+    // [claim, known_tail, exit_flag] reset to the completed prefix.
+    const unsigned long long init[3] = {completed_, completed_, 0};
+    HIP_CHECK(hipMemcpyAsync(claim_ctr_, init, 24, hipMemcpyHostToDevice,
                              stream_));
     PersistentCtl ctl;
     ctl.sq = device_view(sq_);
@@ -569,10 +612,13 @@ class HbmPersistentChannel : public IoChannel {
     ctl.cq = device_view(const_cast<unsigned long long*>(cq_));
     ctl.stop = device_view(const_cast<uint32_t*>(stop_));
     ctl.claim_counter = claim_ctr_;
+    ctl.known_tail = claim_ctr_ + 1;
+    ctl.exit_flag = reinterpret_cast<uint32_t*>(claim_ctr_ + 2);
     ctl.ring_mask = kRing - 1;
     ctl.idle_spins = kIdleSpins;
-    hipLaunchKernelGGL(k_persistent_copy, dim3(kWorkers), dim3(64), 0,
+    hipLaunchKernelGGL(k_persistent_copy, dim3(kWorkers + 1), dim3(64), 0,
                        stream_, ctl);
+    last_progress_ = std::chrono::steady_clock::now();
   }
 
   void drain_pending() {
@@ -630,6 +676,7 @@ class HbmPersistentChannel : public IoChannel {
   volatile unsigned long long* cq_ = nullptr;
   volatile uint32_t* stop_ = nullptr;
   unsigned long long* claim_ctr_ = nullptr;
+  std::chrono::steady_clock::time_point last_progress_;
   uint64_t tail_ = 0;
   uint64_t completed_ = 0;
   std::vector<IoState*> desc_io_;
