@@ -234,7 +234,18 @@ __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
       }
       __builtin_amdgcn_s_sleep(16);
     }
-    const BlockDesc d = ctl.sq[claim & ctl.ring_mask];
+    // Volatile field loads: the descriptor address is computable BEFORE
+    // the wait loop, so a plain load could be hoisted above it by the
+    // compiler and read the slot before the host wrote it (observed as
+    // null-dst faults / phantom zero-byte completions). Volatile pins
+    // the reads after the tail match; host memory is uncached on the
+    // GPU side, so the values are then current by construction.
+    const volatile BlockDesc* vd = &ctl.sq[claim & ctl.ring_mask];
+    BlockDesc d;
+    d.src = vd->src;
+    d.dst = vd->dst;
+    d.bytes = vd->bytes;
+    d.fill = vd->fill;
     const uint32_t n16 = d.bytes >> 4;
     if (d.src != nullptr) {
       const float4* __restrict__ src = reinterpret_cast<const float4*>(d.src);
