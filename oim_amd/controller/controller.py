@@ -160,7 +160,12 @@ class Controller(spec.ControllerServicer):
                                     scsi_disk=spec.SCSIDisk(
                                         target=target.scsi_dev_num, lun=0),
                                 )
-                # Attach to the first free target 0..7 (controller.go:131-148).
+                # Attach to the first free target 0..7 (controller.go:
+                # 131-148). Concurrent MapVolume calls for DIFFERENT
+                # volumes hold different keyed mutexes and can race for
+                # the same free target; "occupied" just means another
+                # volume won it — try the next one.
+                last_err = None
                 for target_num in range(MAX_SCSI_TARGETS):
                     if target_num in targets_used:
                         continue
@@ -168,11 +173,19 @@ class Controller(spec.ControllerServicer):
                         hipstore.add_vhost_scsi_lun(
                             client, self.vhost_controller, target_num, volume_id)
                     except hipstore.RpcError as err:
+                        if "occupied" in err.message:
+                            last_err = err
+                            continue
                         context.abort(grpc.StatusCode.INTERNAL, str(err))
                     return spec.MapVolumeReply(
                         pci_address=self._reply_pci(),
                         scsi_disk=spec.SCSIDisk(target=target_num, lun=0),
                     )
+                if last_err is not None:
+                    context.abort(
+                        grpc.StatusCode.RESOURCE_EXHAUSTED,
+                        f"no free SCSI target on {self.vhost_controller}: "
+                        f"{last_err}")
                 context.abort(
                     grpc.StatusCode.RESOURCE_EXHAUSTED,
                     f"no free SCSI target on {self.vhost_controller}",
