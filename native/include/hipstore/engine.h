@@ -35,6 +35,10 @@ BdevPtr create_hbm_bdev(const std::string& name, uint64_t block_size,
                         uint64_t num_blocks, int device,
                         bool persistent = false);
 
+// Persistent-engine debug counters: 0=launches, 1=relaunches,
+// 2=stall queries.
+uint64_t persistent_stat(int which);
+
 // Pinned-host buffer helpers (fall back to plain malloc without a GPU).
 void* alloc_pinned(size_t bytes);
 void free_pinned(void* ptr);

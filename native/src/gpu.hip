@@ -271,6 +271,13 @@ __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
   }
 }
 
+// Debug counters for the persistent engine (read via pybind
+// persistent_stats(); helps separate relaunch thrash from scheduling
+// issues on multi-queue collapses).
+std::atomic<uint64_t> g_pers_launches{0};
+std::atomic<uint64_t> g_pers_relaunches{0};
+std::atomic<uint64_t> g_pers_stall_queries{0};
+
 int g_device_count = -1;
 
 int device_count_cached() {
@@ -284,6 +291,15 @@ int device_count_cached() {
 }
 
 }  // namespace
+
+uint64_t persistent_stat(int which) {
+  switch (which) {
+    case 0: return g_pers_launches.load();
+    case 1: return g_pers_relaunches.load();
+    case 2: return g_pers_stall_queries.load();
+    default: return 0;
+  }
+}
 
 bool gpu_available() { return device_count_cached() > 0; }
 int gpu_device_count() { return device_count_cached(); }
@@ -596,9 +612,13 @@ class HbmPersistentChannel : public IoChannel {
     // polling threads serialized on it and collapsed throughput.
     if (completed_ < tail_) {
       const auto now = std::chrono::steady_clock::now();
-      if (now - last_progress_ > std::chrono::milliseconds(1) &&
-          hipStreamQuery(stream_) == hipSuccess) {
-        launch();
+      if (now - last_progress_ > std::chrono::milliseconds(1)) {
+        g_pers_stall_queries.fetch_add(1, std::memory_order_relaxed);
+        if (hipStreamQuery(stream_) == hipSuccess) {
+          g_pers_relaunches.fetch_add(1, std::memory_order_relaxed);
+          launch();
+        }
+        last_progress_ = now;  // gate the query itself to 1/ms
       }
     }
     return completed;
@@ -612,6 +632,7 @@ class HbmPersistentChannel : public IoChannel {
 
  private:
   void launch() {
+    g_pers_launches.fetch_add(1, std::memory_order_relaxed);
     (void)hipSetDevice(device_);
     // [claim, known_tail, exit_flag] reset to the completed prefix.
     const unsigned long long init[3] = {completed_, completed_, 0};

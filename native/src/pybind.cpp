@@ -147,6 +147,14 @@ PYBIND11_MODULE(_hipstore, m) {
       py::arg("num_queues") = 1, py::arg("seconds") = 2.0,
       py::arg("max_ios") = 0);
 
+  m.def("persistent_stats", [] {
+    py::dict d;
+    d["launches"] = hipstore::persistent_stat(0);
+    d["relaunches"] = hipstore::persistent_stat(1);
+    d["stall_queries"] = hipstore::persistent_stat(2);
+    return d;
+  });
+
   m.def("crc32c_combine", &crc32c_combine, py::arg("crc1"), py::arg("crc2"),
         py::arg("len2"));
 
