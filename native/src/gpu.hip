@@ -764,10 +764,15 @@ class HbmBdev : public Bdev {
   }
 
   int poll(IoChannel* ch) override {
-    (void)hipSetDevice(device_);  // launches must come from the bdev's device
     if (persistent_) {
+      // No HIP API calls on this path: poll() runs in a tight loop on
+      // every submitter thread, and even hipSetDevice takes the
+      // runtime's global lock — 4+ spinning threads convoyed on it and
+      // starved each other (measured: one of four queues fell to ~10
+      // completions/s). launch() sets the device when actually needed.
       return static_cast<HbmPersistentChannel*>(ch)->poll();
     }
+    (void)hipSetDevice(device_);  // launches must come from the bdev's device
     auto* channel = static_cast<HbmChannel*>(ch);
     int completed = channel->retire(/*wait=*/false);
     channel->kick();
