@@ -24,6 +24,10 @@ import sys
 import tempfile
 import time
 
+# Persistent service kernels need one real HW queue each (ROCm default
+# is 4; beyond that MES time-slices them). Must precede HIP init.
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "24")
+
 REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO_ROOT)
 
@@ -123,7 +127,7 @@ def main() -> int:
     parser.add_argument("--warmup", type=int, default=5)
     parser.add_argument("--io-size", type=int, default=4096)
     parser.add_argument("--queue-depth", type=int, default=32)
-    parser.add_argument("--num-queues", type=int, default=8)
+    parser.add_argument("--num-queues", type=int, default=12)
     parser.add_argument("--bdev-gb", type=float, default=8.0)
     parser.add_argument("--workload", default="randread")
     parser.add_argument("--engine", default="persistent",

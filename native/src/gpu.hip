@@ -17,6 +17,7 @@
 //     and gives the CRC32C/verify paths a resident tile to chew on.
 
 #include <hip/hip_runtime.h>
+#include <stdlib.h>
 
 #include <algorithm>
 #include <atomic>
@@ -279,6 +280,18 @@ std::atomic<uint64_t> g_pers_relaunches{0};
 std::atomic<uint64_t> g_pers_stall_queries{0};
 
 int g_device_count = -1;
+
+// ROCm defaults to 4 hardware queues per process; more concurrent
+// streams get gang-scheduled (preempted) onto them, which time-slices
+// persistent service kernels — measured as a 30-50x multi-queue
+// collapse with second-long p999s. Raise the cap before the HIP
+// runtime initializes (no effect if the embedding process already
+// initialized HIP; hipstored sets it in main() too). Users can
+// override by exporting their own value.
+struct HwQueueEnvInit {
+  HwQueueEnvInit() { setenv("GPU_MAX_HW_QUEUES", "24", /*overwrite=*/0); }
+};
+HwQueueEnvInit g_hw_queue_env_init;
 
 int device_count_cached() {
   static std::once_flag once;

@@ -45,6 +45,11 @@ void usage(const char* argv0) {
 }  // namespace
 
 int main(int argc, char** argv) {
+  // Before any HIP call: the persistent engine needs one real hardware
+  // queue per service kernel (ROCm default 4 -> gang-scheduling
+  // preempts persistent kernels; see native/src/gpu.hip).
+  setenv("GPU_MAX_HW_QUEUES", "24", /*overwrite=*/0);
+
   std::string socket_path = "/var/tmp/hipstored.sock";
   int device = 0;
   bool cpu_only = false;
