@@ -67,6 +67,7 @@ struct BlockDesc {
   uint32_t bytes;      // multiple of 16, <= kTileBytes
   uint32_t fill;       // fill byte (replicated) when src == null
 };
+static_assert(sizeof(BlockDesc) == 24, "persistent engine reads 3 x u64");
 
 // One wave per descriptor; 16 B/lane vector moves staged through LDS.
 // Each lane re-reads exactly the bytes it wrote, so wave-internal
@@ -235,18 +236,25 @@ __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
       }
       __builtin_amdgcn_s_sleep(16);
     }
-    // Volatile field loads: the descriptor address is computable BEFORE
-    // the wait loop, so a plain load could be hoisted above it by the
+    // Volatile loads: the descriptor address is computable BEFORE the
+    // wait loop, so a plain load could be hoisted above it by the
     // compiler and read the slot before the host wrote it (observed as
     // null-dst faults / phantom zero-byte completions). Volatile pins
     // the reads after the tail match; host memory is uncached on the
-    // GPU side, so the values are then current by construction.
-    const volatile BlockDesc* vd = &ctl.sq[claim & ctl.ring_mask];
+    // GPU side, so the values are then current by construction. Three
+    // independent u64 loads pipeline over PCIe (vs 4 dependent field
+    // reads at ~1 us each).
+    const volatile unsigned long long* vd =
+        reinterpret_cast<const volatile unsigned long long*>(
+            &ctl.sq[claim & ctl.ring_mask]);
+    const unsigned long long w0 = vd[0];
+    const unsigned long long w1 = vd[1];
+    const unsigned long long w2 = vd[2];
     BlockDesc d;
-    d.src = vd->src;
-    d.dst = vd->dst;
-    d.bytes = vd->bytes;
-    d.fill = vd->fill;
+    d.src = reinterpret_cast<const uint8_t*>(w0);
+    d.dst = reinterpret_cast<uint8_t*>(w1);
+    d.bytes = static_cast<uint32_t>(w2 & 0xFFFFFFFFu);
+    d.fill = static_cast<uint32_t>(w2 >> 32);
     const uint32_t n16 = d.bytes >> 4;
     if (d.src != nullptr) {
       const float4* __restrict__ src = reinterpret_cast<const float4*>(d.src);
@@ -530,8 +538,18 @@ class HbmChannel : public IoChannel {
 class HbmPersistentChannel : public IoChannel {
  public:
   static constexpr uint32_t kRing = 32768;      // descriptors (>= 2x max request tiles)
-  static constexpr uint32_t kWorkers = 16;      // waves (1-wave workgroups)
   static constexpr uint32_t kIdleSpins = 500000;  // ~1 s of s_sleep polling
+
+  // Worker waves per queue (1-wave workgroups). 16 default; QD-deep
+  // queues benefit from more (HIPSTORE_PERSISTENT_WORKERS overrides).
+  static uint32_t workers() {
+    static uint32_t w = [] {
+      const char* env = getenv("HIPSTORE_PERSISTENT_WORKERS");
+      int v = env ? atoi(env) : 16;
+      return static_cast<uint32_t>(std::min(std::max(v, 1), 64));
+    }();
+    return w;
+  }
 
   HbmPersistentChannel(int device, uint8_t* base) : base_(base), device_(device) {
     HIP_CHECK(hipSetDevice(device));
@@ -661,7 +679,7 @@ class HbmPersistentChannel : public IoChannel {
     ctl.exit_flag = reinterpret_cast<uint32_t*>(claim_ctr_ + 2);
     ctl.ring_mask = kRing - 1;
     ctl.idle_spins = kIdleSpins;
-    hipLaunchKernelGGL(k_persistent_copy, dim3(kWorkers + 1), dim3(64), 0,
+    hipLaunchKernelGGL(k_persistent_copy, dim3(workers() + 1), dim3(64), 0,
                        stream_, ctl);
     last_progress_ = std::chrono::steady_clock::now();
   }
@@ -983,7 +1001,9 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
           ++submitted;
         }
         while (true) {
-          bdev->poll(channel.get());
+          if (bdev->poll(channel.get()) == 0) {
+            __builtin_ia32_pause();  // spinning submitter hygiene
+          }
           if (!stopping &&
               (clock::now() >= deadline ||
                (per_queue_cap && submitted >= per_queue_cap))) {
@@ -1113,7 +1133,9 @@ struct PerfSession::Impl {
           ++inflight;
           ++submitted;
         }
-        while (inflight > 0) bdev->poll(channel.get());
+        while (inflight > 0) {
+          if (bdev->poll(channel.get()) == 0) __builtin_ia32_pause();
+        }
         st.ios = completed;
         {
           std::lock_guard<std::mutex> lock(mutex);

@@ -38,7 +38,8 @@ void usage(const char* argv0) {
           "  -S PATH   JSON-RPC unix socket (default /var/tmp/hipstored.sock)\n"
           "  -d N      HIP device index backing malloc bdevs (default 0)\n"
           "  -C        CPU mode: host-RAM bdevs even when a GPU is present\n"
-          "  -P        persistent engine: on-GPU polling service kernels\n",
+          "  -P        persistent engine: on-GPU polling service kernels (default)\n"
+          "  -B        batched engine: per-poll kernel launches\n",
           argv0);
 }
 
@@ -53,15 +54,16 @@ int main(int argc, char** argv) {
   std::string socket_path = "/var/tmp/hipstored.sock";
   int device = 0;
   bool cpu_only = false;
-  bool persistent = false;
+  bool persistent = true;  // the measured-better engine is the default
 
   int opt;
-  while ((opt = getopt(argc, argv, "S:d:CPh")) != -1) {
+  while ((opt = getopt(argc, argv, "S:d:CPBh")) != -1) {
     switch (opt) {
       case 'S': socket_path = optarg; break;
       case 'd': device = atoi(optarg); break;
       case 'C': cpu_only = true; break;
       case 'P': persistent = true; break;
+      case 'B': persistent = false; break;
       case 'h': usage(argv[0]); return 0;
       default: usage(argv[0]); return 2;
     }
