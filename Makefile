@@ -14,6 +14,7 @@ EXT_SUFFIX := $(shell $(PYTHON) -c "import sysconfig; print(sysconfig.get_config
 CXXFLAGS := -O3 -std=c++17 -fPIC -Wall -Wextra -Wno-unused-parameter \
             -Inative/include
 HIPFLAGS := --offload-arch=$(ARCH)
+LDLIBS := -L$(ROCM)/lib -lrccl
 
 B := native/build
 HDRS := $(wildcard native/include/hipstore/*.h)
@@ -39,11 +40,11 @@ $(B)/pybind.o: native/src/pybind.cpp $(HDRS) | $(B)
 	$(HIPCC) $(CXXFLAGS) -fvisibility=hidden -I$(PY_INC) -I$(PYBIND_INC) -c $< -o $@
 
 oim_amd/_hipstore$(EXT_SUFFIX): $(CORE_OBJS) $(B)/pybind.o
-	$(HIPCC) $(HIPFLAGS) -shared $^ -o $@
+	$(HIPCC) $(HIPFLAGS) -shared $^ $(LDLIBS) -o $@
 
 bin/hipstored: $(CORE_OBJS) $(B)/main.o
 	mkdir -p bin
-	$(HIPCC) $(HIPFLAGS) $^ -o $@
+	$(HIPCC) $(HIPFLAGS) $^ $(LDLIBS) -o $@
 
 clean:
 	rm -rf $(B) bin oim_amd/_hipstore*.so

@@ -1,9 +1,12 @@
 #include "hipstore/composite.h"
 
 #include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+#include <stdlib.h>
 
 #include <deque>
 #include <memory>
+#include <mutex>
 #include <set>
 #include <stdexcept>
 
@@ -179,6 +182,45 @@ class StripedBdev : public Bdev {
 // Replicated bdev
 // ---------------------------------------------------------------------------
 
+// RCCL communicator clique over the replicas' devices + one stream per
+// device, shared by the bdev's channels (serialized by a mutex: NCCL
+// comms are not concurrency-safe). Broadcast fan-out beats a serial
+// chain of peer copies from replica count >= 3: RCCL rings pipeline
+// over several xGMI links instead of re-sending from the primary.
+struct RcclClique {
+  std::vector<ncclComm_t> comms;
+  std::vector<hipStream_t> streams;
+  std::vector<int> devices;
+  std::mutex mutex;
+  bool ok = false;
+
+  explicit RcclClique(const std::vector<int>& devs) : devices(devs) {
+    comms.resize(devs.size(), nullptr);
+    if (ncclCommInitAll(comms.data(), static_cast<int>(devs.size()),
+                        devs.data()) != ncclSuccess) {
+      return;
+    }
+    streams.resize(devs.size(), nullptr);
+    for (size_t i = 0; i < devs.size(); ++i) {
+      if (hipSetDevice(devs[i]) != hipSuccess ||
+          hipStreamCreateWithFlags(&streams[i], hipStreamNonBlocking) !=
+              hipSuccess) {
+        return;
+      }
+    }
+    ok = true;
+  }
+
+  ~RcclClique() {
+    for (auto s : streams) {
+      if (s != nullptr) (void)hipStreamDestroy(s);
+    }
+    for (auto c : comms) {
+      if (c != nullptr) (void)ncclCommDestroy(c);
+    }
+  }
+};
+
 class ReplicatedBdev : public Bdev {
  public:
   ReplicatedBdev(const std::string& name, std::vector<BdevPtr> children)
@@ -210,7 +252,21 @@ class ReplicatedBdev : public Bdev {
         }
       }
     }
+    // RCCL broadcast fan-out for >= 3 HBM replicas (HIPSTORE_RCCL=0
+    // disables; peer copies remain the fallback and the 2-replica
+    // path, where a single xGMI link is already optimal).
+    const char* env = getenv("HIPSTORE_RCCL");
+    if (peer_copy_ && children_.size() >= 3 && (!env || atoi(env) != 0)) {
+      std::vector<int> devices;
+      for (const auto& child : children_) {
+        devices.push_back(child->gpu_device());
+      }
+      rccl_ = std::make_unique<RcclClique>(devices);
+      if (!rccl_->ok) rccl_.reset();
+    }
   }
+
+  bool uses_rccl() const { return rccl_ != nullptr; }
 
   bool peer_copy() const { return peer_copy_; }
 
@@ -259,7 +315,9 @@ class ReplicatedBdev : public Bdev {
     auto state = std::make_shared<SplitState>();
     state->on_complete = std::move(req.on_complete);
     if (peer_copy_ && req.op == IoOp::kWrite) {
-      // Host -> primary HBM once, then primary -> replicas over xGMI.
+      // Host -> primary HBM once, then primary -> replicas over xGMI:
+      // RCCL ring broadcast for >= 3 replicas, direct peer copies
+      // otherwise.
       state->remaining = static_cast<int>(n);  // primary + n-1 replicas
       IoRequest primary = req;
       const uint64_t offset = req.offset;
@@ -273,6 +331,9 @@ class ReplicatedBdev : public Bdev {
           return;
         }
         state->child_done(kIoOk);
+        if (rccl_ != nullptr && broadcast_rccl(channel, state, offset, length)) {
+          return;
+        }
         (void)hipSetDevice(children_[0]->gpu_device());
         uint8_t* src =
             static_cast<uint8_t*>(children_[0]->device_base()) + offset;
@@ -317,8 +378,41 @@ class ReplicatedBdev : public Bdev {
   const std::vector<BdevPtr>& children() const { return children_; }
 
  private:
+  // One ncclBroadcast across the clique (root = primary), completion
+  // tracked per replica stream. Returns false to fall back to peer
+  // copies (state untouched except on success).
+  bool broadcast_rccl(CompositeChannel* channel,
+                      const std::shared_ptr<SplitState>& state,
+                      uint64_t offset, uint64_t length) {
+    std::lock_guard<std::mutex> lock(rccl_->mutex);
+    if (ncclGroupStart() != ncclSuccess) return false;
+    bool ok = true;
+    for (size_t i = 0; i < children_.size(); ++i) {
+      void* buf = static_cast<uint8_t*>(children_[i]->device_base()) + offset;
+      if (ncclBroadcast(buf, buf, length, ncclChar, /*root=*/0,
+                        rccl_->comms[i], rccl_->streams[i]) != ncclSuccess) {
+        ok = false;
+      }
+    }
+    if (ncclGroupEnd() != ncclSuccess || !ok) return false;
+    // Completion: one event per REPLICA stream (the primary already
+    // holds the data; its stream entry just orders the collective).
+    for (size_t i = 1; i < children_.size(); ++i) {
+      (void)hipSetDevice(children_[i]->gpu_device());
+      hipEvent_t event = channel->get_event();
+      if (hipEventRecord(event, rccl_->streams[i]) != hipSuccess) {
+        channel->event_pool.push_back(event);
+        state->child_done(kIoFailed);
+        continue;
+      }
+      channel->pending_reps.push_back({event, state});
+    }
+    return true;
+  }
+
   std::vector<BdevPtr> children_;
   bool peer_copy_ = false;
+  std::unique_ptr<RcclClique> rccl_;
 };
 
 }  // namespace

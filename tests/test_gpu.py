@@ -171,6 +171,20 @@ class TestCompositeGpu:
         for child in children:
             assert child.read(8192, len(data)) == data
 
+    @pytest.mark.skipif(hs.gpu_device_count() < 3,
+                        reason="needs >=3 GPUs for RCCL broadcast fan-out")
+    def test_replicated_rccl_broadcast(self):
+        """>=3 replicas use ncclBroadcast over xGMI for the fan-out."""
+        n = min(hs.gpu_device_count(), 4)
+        children = [hs.create_hbm_bdev(f"rc-{i}", 512, 65536, device=i)
+                    for i in range(n)]
+        bdev = hs.create_replicated_bdev("rc", children)
+        rng = random.Random(17)
+        data = bytes(rng.getrandbits(8) for _ in range(64 * 4096))
+        bdev.write(4096, data)
+        for child in children:
+            assert child.read(4096, len(data)) == data
+
 
 @needs_gpu
 class TestNvmfGpu:
