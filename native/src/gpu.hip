@@ -846,23 +846,36 @@ void crc32c_hbm_blocks(Bdev* bdev, uint64_t offset, uint32_t block_size,
     return;
   }
   HIP_CHECK(hipSetDevice(bdev->gpu_device()));
-  uint32_t* out_dev = static_cast<uint32_t*>(alloc_pinned(count * 4));
-  const uint32_t grid = (count + 255) / 256;
-  // Own non-blocking stream + stream sync: a device-wide sync would
-  // block on live persistent service kernels.
-  hipStream_t stream = nullptr;
-  HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
-  hipLaunchKernelGGL(k_crc32c_blocks, dim3(grid), dim3(256), 0, stream,
-                     static_cast<const uint8_t*>(base) + offset, block_size,
-                     count, device_view(out_dev));
-  hipError_t sync_err = hipStreamSynchronize(stream);
-  (void)hipStreamDestroy(stream);
-  if (sync_err != hipSuccess) {
-    free_pinned(out_dev);
-    throw std::runtime_error("crc32c kernel failed");
+  // Thread-local stream + result buffer: hipHostMalloc and stream
+  // creation cost ~1 ms each, which collapsed per-I/O digests (NVMe/TCP
+  // targets call this once per large read). Own non-blocking stream +
+  // stream sync, never a device-wide sync (it would block on live
+  // persistent service kernels).
+  struct DigestCtx {
+    hipStream_t stream = nullptr;
+    uint32_t* out = nullptr;
+    uint32_t capacity = 0;
+
+    ~DigestCtx() {
+      if (stream != nullptr) (void)hipStreamDestroy(stream);
+      free_pinned(out);
+    }
+  };
+  thread_local DigestCtx ctx;
+  if (ctx.stream == nullptr) {
+    HIP_CHECK(hipStreamCreateWithFlags(&ctx.stream, hipStreamNonBlocking));
   }
-  memcpy(out, out_dev, count * 4);
-  free_pinned(out_dev);
+  if (count > ctx.capacity) {
+    free_pinned(ctx.out);
+    ctx.capacity = std::max(count, std::max(1024u, ctx.capacity * 2));
+    ctx.out = static_cast<uint32_t*>(alloc_pinned(ctx.capacity * 4));
+  }
+  const uint32_t grid = (count + 255) / 256;
+  hipLaunchKernelGGL(k_crc32c_blocks, dim3(grid), dim3(256), 0, ctx.stream,
+                     static_cast<const uint8_t*>(base) + offset, block_size,
+                     count, device_view(ctx.out));
+  HIP_CHECK(hipStreamSynchronize(ctx.stream));
+  memcpy(out, ctx.out, count * 4);
 }
 
 // ---------------------------------------------------------------------------
