@@ -7,13 +7,17 @@
 // CRC32C kernel + host combine.
 
 #include <arpa/inet.h>
+#include <fcntl.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/socket.h>
 #include <unistd.h>
 
 #include <atomic>
+#include <cerrno>
+#include <chrono>
 #include <cstring>
+#include <functional>
 #include <map>
 #include <mutex>
 #include <thread>
@@ -146,25 +150,46 @@ class TargetImpl : public NvmfTcpTarget,
     icresp.maxh2cdata = kMaxDataPerPdu;
     if (!write_exact(fd, &icresp, sizeof(icresp))) return;
 
+    // The connection is served as a polled pipeline: the socket goes
+    // non-blocking, commands submit ASYNCHRONOUSLY to the bdev channel
+    // and completions (fired from poll on this thread) send the
+    // C2HData/CQE — so a connection sustains its full queue depth
+    // instead of one command at a time. Outbound sends stay blocking
+    // (write_exact spins on EAGAIN).
     uint32_t cc = 0;
     uint16_t next_ttag = 1;
     std::map<uint16_t, PendingWrite> pending_writes;  // by ttag
-    std::vector<uint8_t> header(256);
+    bool dead = false;
+    int inflight = 0;
 
-    // Reusable pinned bounce arena: hipHostMalloc costs ~1 ms, so a
-    // per-I/O allocation collapses the connection to ~1K IOPS. Grown
-    // geometrically, freed with the connection.
-    uint8_t* bounce_arena = nullptr;
-    size_t bounce_capacity = 0;
-    auto bounce = [&](size_t len) -> uint8_t* {
-      if (len > bounce_capacity) {
-        size_t want = std::max<size_t>(len, 256 * 1024);
-        want = std::max(want, bounce_capacity * 2);
-        free_pinned(bounce_arena);
-        bounce_arena = static_cast<uint8_t*>(alloc_pinned(want));
-        bounce_capacity = want;
+    // Reusable pinned buffers: hipHostMalloc costs ~1 ms, so per-I/O
+    // allocation is off the table; concurrent in-flight commands each
+    // borrow from this pool.
+    struct PoolBuf {
+      uint8_t* ptr;
+      size_t cap;
+    };
+    std::vector<PoolBuf> pool;
+    auto pool_get = [&](size_t len) -> PoolBuf {
+      for (auto it = pool.begin(); it != pool.end(); ++it) {
+        if (it->cap >= len) {
+          PoolBuf b = *it;
+          pool.erase(it);
+          return b;
+        }
       }
-      return bounce_arena;
+      size_t cap = std::max<size_t>(len, 128 * 1024);
+      return PoolBuf{static_cast<uint8_t*>(alloc_pinned(cap)), cap};
+    };
+    auto pool_put = [&](PoolBuf b) { pool.push_back(b); };
+
+    std::map<Bdev*, std::shared_ptr<IoChannel>> channels;
+    auto channel_of = [&](Bdev* bdev) -> IoChannel* {
+      auto it = channels.find(bdev);
+      if (it == channels.end()) {
+        it = channels.emplace(bdev, bdev->get_channel()).first;
+      }
+      return it->second.get();
     };
 
     auto send_cqe = [&](uint16_t cid, uint16_t sc, uint64_t result = 0) {
@@ -177,12 +202,12 @@ class TargetImpl : public NvmfTcpTarget,
       resp.cqe.set_result64(result);
       resp.cqe.set_cid(cid);
       resp.cqe.set_status(sc);
-      return send_pdu(fd, &resp, 24, hdgst, nullptr, 0, 0, false);
+      if (!send_pdu(fd, &resp, 24, hdgst, nullptr, 0, 0, false)) dead = true;
+      return !dead;
     };
 
     auto send_c2h_data = [&](uint16_t cid, const uint8_t* data, uint32_t len,
                              uint32_t digest_all) {
-      // Split into <= kMaxDataPerPdu chunks; per-PDU DDGST.
       uint32_t offset = 0;
       while (offset < len) {
         const uint32_t chunk = std::min(len - offset, kMaxDataPerPdu);
@@ -205,6 +230,7 @@ class TargetImpl : public NvmfTcpTarget,
                    : crc32c_sw(0, data + offset, chunk);
         }
         if (!send_pdu(fd, &dh, 24, hdgst, data + offset, chunk, dd, ddgst)) {
+          dead = true;
           return false;
         }
         offset += chunk;
@@ -212,40 +238,30 @@ class TargetImpl : public NvmfTcpTarget,
       return true;
     };
 
-    auto channel_cache = std::map<Bdev*, std::shared_ptr<IoChannel>>();
-    auto bdev_io = [&](Bdev* bdev, IoOp op, uint64_t off, void* buf,
-                       uint64_t len, uint8_t fill) {
-      auto it = channel_cache.find(bdev);
-      if (it == channel_cache.end()) {
-        it = channel_cache.emplace(bdev, bdev->get_channel()).first;
-      }
-      int result = kIoFailed;
-      bool done = false;
+    // Async bdev ops; `done` runs inline from the poll loop below.
+    auto submit_io = [&](BdevPtr bdev, IoOp op, uint64_t off, void* buf,
+                         uint64_t len, std::function<void(int)> done) {
       IoRequest req;
       req.op = op;
       req.offset = off;
       req.length = len;
       req.buffer = buf;
-      req.fill = fill;
-      req.on_complete = [&](int status) {
-        result = status;
-        done = true;
+      req.fill = 0;
+      ++inflight;
+      req.on_complete = [&inflight, done = std::move(done)](int status) {
+        --inflight;
+        done(status);
       };
-      bdev->submit(it->second.get(), std::move(req));
-      while (!done) bdev->poll(it->second.get());
-      return result;
+      bdev->submit(channel_of(bdev.get()), std::move(req));
     };
 
     auto handle_command = [&](const Sqe& sqe, std::vector<uint8_t>& capsule_data) {
       if (sqe.opc() == kOpcFabrics) {
         switch (sqe.fctype()) {
-          case kFctypeConnect: {
-            // cntlid in dw0.
-            return send_cqe(sqe.cid(), kScSuccess, 1);
-          }
+          case kFctypeConnect:
+            return send_cqe(sqe.cid(), kScSuccess, 1);  // cntlid in dw0
           case kFctypePropertySet: {
-            uint32_t ofst = sqe.cdw(11);
-            if (ofst == kPropCc) cc = sqe.cdw(12);
+            if (sqe.cdw(11) == kPropCc) cc = sqe.cdw(12);
             return send_cqe(sqe.cid(), kScSuccess);
           }
           case kFctypePropertyGet: {
@@ -300,18 +316,26 @@ class TargetImpl : public NvmfTcpTarget,
           if (off + len > bdev->size_bytes()) {
             return send_cqe(sqe.cid(), kScLbaOutOfRange);
           }
-          uint8_t* buf = bounce(len);
-          int status = bdev_io(bdev.get(), IoOp::kRead, off, buf, len, 0);
-          if (status != kIoOk) {
-            return send_cqe(sqe.cid(), kScInternalError);
-          }
-          // GPU digest for HBM namespaces (per-4KiB kernel + combine).
-          uint32_t digest = 0;
-          if (ddgst && len <= kMaxDataPerPdu) {
-            digest = range_crc32c(bdev.get(), off, buf, len);
-          }
-          if (!send_c2h_data(sqe.cid(), buf, len, digest)) return false;
-          return send_cqe(sqe.cid(), kScSuccess);
+          PoolBuf pb = pool_get(len);
+          const uint16_t cid = sqe.cid();
+          submit_io(bdev, IoOp::kRead, off, pb.ptr, len,
+                    [&, bdev, pb, off, len, cid](int status) {
+                      if (status != kIoOk) {
+                        send_cqe(cid, kScInternalError);
+                      } else {
+                        // GPU digest for HBM namespaces (per-4KiB
+                        // kernel + host combine).
+                        uint32_t digest = 0;
+                        if (ddgst && len <= kMaxDataPerPdu) {
+                          digest = range_crc32c(bdev.get(), off, pb.ptr, len);
+                        }
+                        if (send_c2h_data(cid, pb.ptr, len, digest)) {
+                          send_cqe(cid, kScSuccess);
+                        }
+                      }
+                      pool_put(pb);
+                    });
+          return true;
         }
         case kOpcWrite: {
           BdevPtr bdev = ns(sqe.nsid());
@@ -319,8 +343,9 @@ class TargetImpl : public NvmfTcpTarget,
           const uint64_t slba =
               sqe.cdw(10) | (static_cast<uint64_t>(sqe.cdw(11)) << 32);
           const uint32_t nlb = (sqe.cdw(12) & 0xFFFF) + 1;
+          const uint64_t off = slba * bdev->block_size();
           const uint64_t len = static_cast<uint64_t>(nlb) * bdev->block_size();
-          if (slba * bdev->block_size() + len > bdev->size_bytes()) {
+          if (off + len > bdev->size_bytes()) {
             return send_cqe(sqe.cid(), kScLbaOutOfRange);
           }
           if (!capsule_data.empty()) {
@@ -328,12 +353,16 @@ class TargetImpl : public NvmfTcpTarget,
             if (capsule_data.size() != len) {
               return send_cqe(sqe.cid(), kScInvalidField);
             }
-            uint8_t* buf = bounce(len);
-            memcpy(buf, capsule_data.data(), len);
-            int status = bdev_io(bdev.get(), IoOp::kWrite,
-                                 slba * bdev->block_size(), buf, len, 0);
-            return send_cqe(sqe.cid(),
-                            status == kIoOk ? kScSuccess : kScInternalError);
+            PoolBuf pb = pool_get(len);
+            memcpy(pb.ptr, capsule_data.data(), len);
+            const uint16_t cid = sqe.cid();
+            submit_io(bdev, IoOp::kWrite, off, pb.ptr, len,
+                      [&, pb, cid](int status) {
+                        send_cqe(cid, status == kIoOk ? kScSuccess
+                                                      : kScInternalError);
+                        pool_put(pb);
+                      });
+            return true;
           }
           // Solicit the data with one R2T covering the whole transfer.
           PendingWrite pw;
@@ -348,100 +377,151 @@ class TargetImpl : public NvmfTcpTarget,
           r2t.ttag = ttag;
           r2t.datao = 0;
           r2t.datal = len;
-          return send_pdu(fd, &r2t, 24, hdgst, nullptr, 0, 0, false);
+          if (!send_pdu(fd, &r2t, 24, hdgst, nullptr, 0, 0, false)) {
+            dead = true;
+            return false;
+          }
+          return true;
         }
-        case kOpcFlush:
-          return send_cqe(sqe.cid(), kScSuccess);
+        case kOpcFlush: {
+          BdevPtr bdev = ns(sqe.nsid());
+          if (!bdev) return send_cqe(sqe.cid(), kScSuccess);
+          const uint16_t cid = sqe.cid();
+          submit_io(bdev, IoOp::kFlush, 0, nullptr, 0, [&, cid](int status) {
+            send_cqe(cid, status == kIoOk ? kScSuccess : kScInternalError);
+          });
+          return true;
+        }
         case kOpcWriteZeroes: {
           BdevPtr bdev = ns(sqe.nsid());
           if (!bdev) return send_cqe(sqe.cid(), kScInvalidField);
           const uint64_t slba =
               sqe.cdw(10) | (static_cast<uint64_t>(sqe.cdw(11)) << 32);
           const uint32_t nlb = (sqe.cdw(12) & 0xFFFF) + 1;
-          int status = bdev_io(bdev.get(), IoOp::kFill,
-                               slba * bdev->block_size(), nullptr,
-                               static_cast<uint64_t>(nlb) * bdev->block_size(),
-                               0);
-          return send_cqe(sqe.cid(),
-                          status == kIoOk ? kScSuccess : kScInternalError);
+          const uint16_t cid = sqe.cid();
+          submit_io(bdev, IoOp::kFill, slba * bdev->block_size(), nullptr,
+                    static_cast<uint64_t>(nlb) * bdev->block_size(),
+                    [&, cid](int status) {
+                      send_cqe(cid,
+                               status == kIoOk ? kScSuccess : kScInternalError);
+                    });
+          return true;
         }
         default:
           return send_cqe(sqe.cid(), kScInvalidOpcode);
       }
     };
 
-    // --- PDU loop --------------------------------------------------------
-    while (running_.load()) {
-      CommonHeader ch;
-      if (!read_exact(fd, &ch, sizeof(ch))) break;
-      if (ch.hlen < sizeof(ch) || ch.plen < ch.hlen) break;
-      header.resize(ch.hlen);
-      memcpy(header.data(), &ch, sizeof(ch));
-      if (!read_exact(fd, header.data() + sizeof(ch), ch.hlen - sizeof(ch))) {
-        break;
-      }
+    auto handle_pdu = [&](const uint8_t* pdu, const CommonHeader& chh) {
       if (hdgst) {
         uint32_t hd;
-        if (!read_exact(fd, &hd, 4)) break;
-        if (hd != crc32c_sw(0, header.data(), ch.hlen)) break;
-      }
-      // Data length: everything between the data offset and the digest.
-      const uint32_t pdo = ch.pdo ? ch.pdo : ch.hlen + (hdgst ? 4 : 0);
-      uint32_t data_len = 0;
-      bool has_ddgst = false;
-      if (ch.plen > pdo) {
-        has_ddgst = ddgst;
-        data_len = ch.plen - pdo - (has_ddgst ? 4 : 0);
-        // skip padding between header(+hdgst) and pdo
-        uint32_t pad = pdo - ch.hlen - (hdgst ? 4 : 0);
-        if (pad > 16) break;
-        uint8_t padbuf[16];
-        if (pad && !read_exact(fd, padbuf, pad)) break;
-      }
-      std::vector<uint8_t> data(data_len);
-      if (data_len > 0 && !read_exact(fd, data.data(), data_len)) break;
-      if (has_ddgst && data_len > 0) {
-        uint32_t dd;
-        if (!read_exact(fd, &dd, 4)) break;
-        if (dd != crc32c_sw(0, data.data(), data_len)) {
-          // data corruption: fail hard (a real target sends TermReq)
-          break;
+        memcpy(&hd, pdu + chh.hlen, 4);
+        if (hd != crc32c_sw(0, pdu, chh.hlen)) {
+          dead = true;
+          return;
         }
       }
-
-      if (ch.type == kCapsuleCmd) {
+      const uint32_t pdo = chh.pdo ? chh.pdo : chh.hlen + (hdgst ? 4 : 0);
+      uint32_t data_len = 0;
+      if (chh.plen > pdo) data_len = chh.plen - pdo - (ddgst ? 4 : 0);
+      if (ddgst && data_len > 0) {
+        uint32_t dd;
+        memcpy(&dd, pdu + pdo + data_len, 4);
+        if (dd != crc32c_sw(0, pdu + pdo, data_len)) {
+          dead = true;  // a real target sends C2HTermReq
+          return;
+        }
+      }
+      if (chh.type == kCapsuleCmd) {
         Sqe sqe;
-        memcpy(sqe.bytes, header.data() + 8, 64);
-        if (!handle_command(sqe, data)) break;
-      } else if (ch.type == kH2CData) {
+        memcpy(sqe.bytes, pdu + 8, 64);
+        std::vector<uint8_t> capsule_data(pdu + pdo, pdu + pdo + data_len);
+        handle_command(sqe, capsule_data);
+      } else if (chh.type == kH2CData) {
         DataHeader dh;
-        memcpy(&dh, header.data(), sizeof(dh));
+        memcpy(&dh, pdu, sizeof(dh));
         auto it = pending_writes.find(dh.ttag);
-        if (it == pending_writes.end()) break;
+        if (it == pending_writes.end()) {
+          dead = true;
+          return;
+        }
         PendingWrite& pw = it->second;
-        if (dh.datao + data_len > pw.data.size()) break;
-        memcpy(pw.data.data() + dh.datao, data.data(), data_len);
+        if (dh.datao + data_len > pw.data.size()) {
+          dead = true;
+          return;
+        }
+        memcpy(pw.data.data() + dh.datao, pdu + pdo, data_len);
         pw.received += data_len;
         if (pw.received >= pw.data.size()) {
           BdevPtr bdev = ns(pw.sqe.nsid());
           const uint64_t slba = pw.sqe.cdw(10) |
                                 (static_cast<uint64_t>(pw.sqe.cdw(11)) << 32);
-          uint8_t* buf = bounce(pw.data.size());
-          memcpy(buf, pw.data.data(), pw.data.size());
-          int status =
-              bdev_io(bdev.get(), IoOp::kWrite, slba * bdev->block_size(),
-                      buf, pw.data.size(), 0);
-          uint16_t cid = pw.sqe.cid();
+          PoolBuf pb = pool_get(pw.data.size());
+          memcpy(pb.ptr, pw.data.data(), pw.data.size());
+          const uint16_t cid = pw.sqe.cid();
+          const uint64_t len = pw.data.size();
           pending_writes.erase(it);
-          if (!send_cqe(cid, status == kIoOk ? kScSuccess : kScInternalError)) {
-            break;
-          }
+          submit_io(bdev, IoOp::kWrite, slba * bdev->block_size(), pb.ptr,
+                    len, [&, pb, cid](int status) {
+                      send_cqe(cid, status == kIoOk ? kScSuccess
+                                                    : kScInternalError);
+                      pool_put(pb);
+                    });
         }
       } else {
-        break;  // unexpected PDU
+        dead = true;  // unexpected PDU
       }
+    };
+
+    // --- polled PDU/completion loop --------------------------------------
+    {
+      int flags = fcntl(fd, F_GETFL, 0);
+      fcntl(fd, F_SETFL, flags | O_NONBLOCK);
     }
-    free_pinned(bounce_arena);
+    std::string rx;
+    while (running_.load() && !dead) {
+      char tmp[65536];
+      while (true) {
+        ssize_t n = recv(fd, tmp, sizeof(tmp), 0);
+        if (n > 0) {
+          rx.append(tmp, n);
+          if (rx.size() > (64u << 20)) dead = true;
+          continue;
+        }
+        if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) break;
+        if (n < 0 && errno == EINTR) continue;
+        dead = true;  // n == 0: peer closed
+        break;
+      }
+      size_t start = 0;
+      while (!dead && rx.size() - start >= sizeof(CommonHeader) &&
+             inflight < 256) {
+        CommonHeader chh;
+        memcpy(&chh, rx.data() + start, sizeof(chh));
+        if (chh.hlen < sizeof(chh) || chh.plen < chh.hlen ||
+            chh.plen > (64u << 20)) {
+          dead = true;
+          break;
+        }
+        if (rx.size() - start < chh.plen) break;
+        handle_pdu(reinterpret_cast<const uint8_t*>(rx.data()) + start, chh);
+        start += chh.plen;
+      }
+      rx.erase(0, start);
+      int completed = 0;
+      for (auto& [bdev, channel] : channels) {
+        completed += bdev->poll(channel.get());
+      }
+      if (completed == 0) __builtin_ia32_pause();
+    }
+    // Drain in-flight completions so pool buffers are not freed under
+    // a pending I/O.
+    const auto deadline =
+        std::chrono::steady_clock::now() + std::chrono::seconds(10);
+    while (inflight > 0 && std::chrono::steady_clock::now() < deadline) {
+      for (auto& [bdev, channel] : channels) bdev->poll(channel.get());
+    }
+    for (auto& pb : pool) free_pinned(pb.ptr);
   }
 
   std::string subnqn_;
