@@ -1,0 +1,172 @@
+"""Failure-recovery properties (reference SURVEY.md section 5):
+per-operation dialing means component restarts never strand clients;
+CSI edge cases around idempotency."""
+
+import grpc
+import pytest
+
+from oim_amd import spec
+from oim_amd.common.server import grpc_target
+from oim_amd.controller import Controller, ControllerServer
+from oim_amd.csidriver import LocalBackend, Mounter, FakeExec, OIMDriver
+from oim_amd.registry import MemRegistryDB, Registry, RegistryServer
+from oim_amd.spec import csi_v1 as csi
+from oim_amd.spec.rpc_csi import CSIControllerStub
+
+from fixtures import hipstored  # noqa: F401
+
+
+class TestControllerRestart:
+    def test_proxy_survives_controller_restart(self, hipstored, tmp_path):  # noqa: F811
+        """The registry dials the controller per request
+        (reference registry.go:196-210, README.md:49): kill and restart
+        the controller and the same client keeps working."""
+        registry = Registry(db=MemRegistryDB())
+        reg_server = RegistryServer(f"unix://{tmp_path}/reg.sock", registry)
+        reg_server.start()
+        endpoint = f"unix://{tmp_path}/ctrl.sock"
+
+        def start_controller():
+            controller = Controller(controller_id="c0",
+                                    hipstored_socket=hipstored.socket_path)
+            server = ControllerServer(endpoint, controller)
+            server.start()
+            return server
+
+        ctrl = start_controller()
+        registry.db.store(["c0", "address"], endpoint)
+        metadata = ((spec.CONTROLLER_ID_KEY, "c0"),)
+        try:
+            with grpc.insecure_channel(grpc_target(reg_server.addr())) as ch:
+                stub = spec.ControllerStub(ch)
+                stub.ProvisionMallocBDev(
+                    spec.ProvisionMallocBDevRequest(bdev_name="rv", size=1 << 20),
+                    metadata=metadata, timeout=30)
+                # restart the controller process-equivalent
+                ctrl.stop()
+                with pytest.raises(grpc.RpcError):
+                    stub.CheckMallocBDev(
+                        spec.CheckMallocBDevRequest(bdev_name="rv"),
+                        metadata=metadata, timeout=5)
+                ctrl = start_controller()
+                # same channel, no re-dial needed by the client: the
+                # proxy's per-request dial picks up the new controller
+                stub.CheckMallocBDev(
+                    spec.CheckMallocBDevRequest(bdev_name="rv"),
+                    metadata=metadata, timeout=30)
+                stub.ProvisionMallocBDev(
+                    spec.ProvisionMallocBDevRequest(bdev_name="rv", size=0),
+                    metadata=metadata, timeout=30)
+        finally:
+            ctrl.stop()
+            reg_server.stop()
+
+    def test_daemon_data_survives_controller_restart(self, hipstored, tmp_path):  # noqa: F811
+        """Bdev state lives in hipstored, not the controller: a volume
+        provisioned before a controller restart is still mapped after."""
+        controller = Controller(controller_id="c1",
+                                hipstored_socket=hipstored.socket_path)
+        server = ControllerServer(f"unix://{tmp_path}/c1.sock", controller)
+        server.start()
+
+        class Ctx:
+            def abort(self, code, details):
+                raise AssertionError(f"{code}: {details}")
+
+        controller.ProvisionMallocBDev(
+            spec.ProvisionMallocBDevRequest(bdev_name="persist", size=1 << 20),
+            Ctx())
+        reply1 = controller.MapVolume(
+            spec.MapVolumeRequest(volume_id="persist",
+                                  malloc=spec.MallocParams()), Ctx())
+        server.stop()
+        # "restarted" controller sees the existing mapping (idempotent
+        # MapVolume against daemon state)
+        controller2 = Controller(controller_id="c1",
+                                 hipstored_socket=hipstored.socket_path)
+        server2 = ControllerServer(f"unix://{tmp_path}/c1b.sock", controller2)
+        server2.start()
+        try:
+            reply2 = controller2.MapVolume(
+                spec.MapVolumeRequest(volume_id="persist",
+                                      malloc=spec.MallocParams()), Ctx())
+            assert reply1.scsi_disk.target == reply2.scsi_disk.target
+            controller2.UnmapVolume(
+                spec.UnmapVolumeRequest(volume_id="persist"), Ctx())
+            controller2.ProvisionMallocBDev(
+                spec.ProvisionMallocBDevRequest(bdev_name="persist", size=0),
+                Ctx())
+        finally:
+            server2.stop()
+
+
+class TestCSISanity:
+    """csi-test-sanity-style edge cases (reference oim-driver_test.go
+    used the upstream suite; these are the load-bearing subset)."""
+
+    @pytest.fixture
+    def driver(self, hipstored, tmp_path):  # noqa: F811
+        backend = LocalBackend(hipstored.socket_path)
+        driver = OIMDriver(driver_name="oim-local", node_id="n0",
+                           endpoint=f"unix://{tmp_path}/csi.sock",
+                           backend=backend, mounter=Mounter(FakeExec()))
+        driver.start()
+        yield driver
+        driver.stop()
+
+    def _stub(self, driver):
+        channel = grpc.insecure_channel(grpc_target(driver.addr()))
+        return CSIControllerStub(channel), channel
+
+    def _create(self, stub, name, size=1 << 20):
+        request = csi.CreateVolumeRequest(name=name)
+        request.capacity_range.required_bytes = size
+        cap = request.volume_capabilities.add()
+        cap.mount.fs_type = "ext4"
+        cap.access_mode.mode = csi.ACCESS_MODE_SINGLE_NODE_WRITER
+        return stub.CreateVolume(request, timeout=30)
+
+    def test_create_idempotent_same_size(self, driver):
+        stub, channel = self._stub(driver)
+        with channel:
+            a = self._create(stub, "sanity-1")
+            b = self._create(stub, "sanity-1")
+            assert a.volume.volume_id == b.volume.volume_id
+            stub.DeleteVolume(csi.DeleteVolumeRequest(
+                volume_id=a.volume.volume_id), timeout=30)
+
+    def test_create_conflicting_size_rejected(self, driver):
+        stub, channel = self._stub(driver)
+        with channel:
+            self._create(stub, "sanity-2", 1 << 20)
+            with pytest.raises(grpc.RpcError) as excinfo:
+                self._create(stub, "sanity-2", 2 << 20)
+            assert excinfo.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+            stub.DeleteVolume(csi.DeleteVolumeRequest(volume_id="sanity-2"),
+                              timeout=30)
+
+    def test_delete_nonexistent_ok(self, driver):
+        stub, channel = self._stub(driver)
+        with channel:
+            stub.DeleteVolume(csi.DeleteVolumeRequest(volume_id="ghost"),
+                              timeout=30)
+
+    def test_create_missing_capabilities(self, driver):
+        stub, channel = self._stub(driver)
+        with channel:
+            request = csi.CreateVolumeRequest(name="nocaps")
+            request.capacity_range.required_bytes = 1 << 20
+            with pytest.raises(grpc.RpcError) as excinfo:
+                stub.CreateVolume(request, timeout=30)
+            assert excinfo.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+
+    def test_validate_missing_volume(self, driver):
+        stub, channel = self._stub(driver)
+        with channel:
+            request = csi.ValidateVolumeCapabilitiesRequest(volume_id="ghost")
+            cap = request.volume_capabilities.add()
+            cap.mount.fs_type = "ext4"
+            cap.access_mode.mode = csi.ACCESS_MODE_SINGLE_NODE_WRITER
+            with pytest.raises(grpc.RpcError) as excinfo:
+                stub.ValidateVolumeCapabilities(request, timeout=30)
+            assert excinfo.value.code() == grpc.StatusCode.NOT_FOUND
