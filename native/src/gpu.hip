@@ -470,6 +470,7 @@ class HbmChannel : public IoChannel {
   }
 
   bool idle() const { return pending_.empty() && inflight_.empty(); }
+  bool has_pending() const { return !pending_.empty(); }
 
  private:
   void expand(const IoRequest& req, uint32_t* slot) {
@@ -803,11 +804,15 @@ class HbmBdev : public Bdev {
       // completions/s). launch() sets the device when actually needed.
       return static_cast<HbmPersistentChannel*>(ch)->poll();
     }
-    (void)hipSetDevice(device_);  // launches must come from the bdev's device
     auto* channel = static_cast<HbmChannel*>(ch);
     int completed = channel->retire(/*wait=*/false);
-    channel->kick();
-    completed += channel->retire(/*wait=*/false);
+    if (channel->has_pending()) {
+      // Launches must come from the bdev's device; only pay the
+      // runtime call when there is something to launch.
+      (void)hipSetDevice(device_);
+      channel->kick();
+      completed += channel->retire(/*wait=*/false);
+    }
     return completed;
   }
 

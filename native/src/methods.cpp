@@ -405,7 +405,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   // --cpu (or no GPU) `replicas` host-RAM children emulate the layout
   // for CI. These methods are native extensions — SPDK's raid bdev has
   // a different RPC shape and no GPU notion.
-  auto make_children = [use_hbm](
+  auto make_children = [use_hbm, persistent](
                            const Json& p, const std::string& name,
                            int64_t num_blocks, int64_t block_size,
                            size_t count_hint) {
@@ -421,8 +421,9 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
     for (size_t i = 0; i < count; ++i) {
       const std::string child_name = name + "." + std::to_string(i);
       if (use_hbm && gpu_available() && !devices.empty()) {
-        children.push_back(create_hbm_bdev(child_name, block_size,
-                                           num_blocks, devices[i]));
+        children.push_back(create_hbm_bdev(
+            child_name, block_size, num_blocks, devices[i],
+            p.get_bool("persistent", persistent)));
       } else {
         children.push_back(
             create_malloc_bdev(child_name, block_size, num_blocks));
