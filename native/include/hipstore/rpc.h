@@ -7,6 +7,7 @@
 
 #include <atomic>
 #include <functional>
+#include <memory>
 #include <map>
 #include <mutex>
 #include <string>
@@ -57,7 +58,9 @@ class RpcServer {
   int listen_fd_ = -1;
   std::atomic<bool> running_{false};
   std::thread accept_thread_;
-  std::vector<std::thread> connections_;
+  // (thread, finished flag) so the accept loop can reap exited ones.
+  std::vector<std::pair<std::thread, std::shared_ptr<std::atomic<bool>>>>
+      connections_;
   std::map<std::string, RpcMethod> methods_;
   mutable std::mutex mutex_;
 };

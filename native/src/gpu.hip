@@ -999,9 +999,11 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
           req.on_complete = [&, slot](int status) {
             if (status != kIoOk) failed.store(true);
             const auto now = clock::now();
-            st.lat_us.push_back(static_cast<uint32_t>(
-                std::chrono::duration_cast<std::chrono::microseconds>(
-                    now - submit_ts[slot]).count()));
+            if (st.lat_us.size() < (16u << 20)) {  // bound long runs
+              st.lat_us.push_back(static_cast<uint32_t>(
+                  std::chrono::duration_cast<std::chrono::microseconds>(
+                      now - submit_ts[slot]).count()));
+            }
             ++st.ios;
             --inflight;
             if (!stopping) {

@@ -94,12 +94,12 @@ class TargetImpl : public NvmfTcpTarget,
     shutdown(listen_fd_, SHUT_RDWR);
     close(listen_fd_);
     if (accept_thread_.joinable()) accept_thread_.join();
-    std::vector<std::thread> conns;
+    decltype(connections_) conns;
     {
       std::lock_guard<std::mutex> lock(mutex_);
       conns.swap(connections_);
     }
-    for (auto& t : conns) {
+    for (auto& [t, done] : conns) {
       if (t.joinable()) t.join();
     }
   }
@@ -121,10 +121,22 @@ class TargetImpl : public NvmfTcpTarget,
       int one = 1;
       setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
       std::lock_guard<std::mutex> lock(mutex_);
-      connections_.emplace_back([this, fd] {
-        serve(fd);
-        close(fd);
-      });
+      for (auto it = connections_.begin(); it != connections_.end();) {
+        if (it->second->load()) {
+          it->first.join();
+          it = connections_.erase(it);
+        } else {
+          ++it;
+        }
+      }
+      auto done = std::make_shared<std::atomic<bool>>(false);
+      connections_.emplace_back(
+          std::thread([this, fd, done] {
+            serve(fd);
+            close(fd);
+            done->store(true);
+          }),
+          done);
     }
   }
 
@@ -531,7 +543,8 @@ class TargetImpl : public NvmfTcpTarget,
   std::atomic<bool> running_{false};
   std::thread accept_thread_;
   std::mutex mutex_;
-  std::vector<std::thread> connections_;
+  std::vector<std::pair<std::thread, std::shared_ptr<std::atomic<bool>>>>
+      connections_;
   std::vector<BdevPtr> namespaces_;
 };
 

@@ -47,7 +47,7 @@ void RpcServer::stop() {
   ::shutdown(listen_fd_, SHUT_RDWR);
   ::close(listen_fd_);
   if (accept_thread_.joinable()) accept_thread_.join();
-  for (auto& t : connections_) {
+  for (auto& [t, done] : connections_) {
     if (t.joinable()) t.join();
   }
   connections_.clear();
@@ -62,7 +62,23 @@ void RpcServer::accept_loop() {
       continue;
     }
     std::lock_guard<std::mutex> lock(mutex_);
-    connections_.emplace_back([this, fd] { serve_connection(fd); });
+    // Reap finished connection threads so a long-lived daemon does not
+    // accumulate joinable handles (one per past client).
+    for (auto it = connections_.begin(); it != connections_.end();) {
+      if (it->second->load()) {
+        it->first.join();
+        it = connections_.erase(it);
+      } else {
+        ++it;
+      }
+    }
+    auto done = std::make_shared<std::atomic<bool>>(false);
+    connections_.emplace_back(
+        std::thread([this, fd, done] {
+          serve_connection(fd);
+          done->store(true);
+        }),
+        done);
   }
 }
 

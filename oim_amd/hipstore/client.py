@@ -81,13 +81,20 @@ class Client:
 
     def _read_response(self) -> dict:
         while True:
-            text = self._buffer.decode("utf-8", errors="strict")
+            try:
+                text = self._buffer.decode("utf-8", errors="strict")
+            except UnicodeDecodeError:
+                # recv() split a multi-byte sequence: need more bytes
+                text = ""
             stripped = text.lstrip()
             offset = len(text) - len(stripped)
             if stripped:
                 try:
                     value, end = self._decoder.raw_decode(stripped)
-                    self._buffer = self._buffer[offset + end:]
+                    # Slice in BYTES: offset/end are character counts and
+                    # differ from byte counts for non-ASCII payloads.
+                    consumed = len(text[:offset + end].encode("utf-8"))
+                    self._buffer = self._buffer[consumed:]
                     return value
                 except json.JSONDecodeError:
                     pass  # incomplete; read more
