@@ -245,10 +245,33 @@ class NodeServer(CSINodeServicer):
                     context.abort(grpc.StatusCode.INTERNAL, str(exc))
         return csi.NodeUnpublishVolumeResponse()
 
+    def NodeGetVolumeStats(self, request, context):
+        path = request.volume_path or request.staging_target_path
+        if not request.volume_id or not path:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume id or path")
+        if not os.path.exists(path):
+            context.abort(grpc.StatusCode.NOT_FOUND, f"no such path: {path}")
+        st = os.statvfs(path)
+        response = csi.NodeGetVolumeStatsResponse()
+        usage = response.usage.add()
+        usage.unit = csi.USAGE_UNIT_BYTES
+        usage.total = st.f_frsize * st.f_blocks
+        usage.available = st.f_frsize * st.f_bavail
+        usage.used = st.f_frsize * (st.f_blocks - st.f_bfree)
+        inodes = response.usage.add()
+        inodes.unit = csi.USAGE_UNIT_INODES
+        inodes.total = st.f_files
+        inodes.available = st.f_favail
+        inodes.used = st.f_files - st.f_ffree
+        return response
+
     def NodeGetCapabilities(self, request, context):
         response = csi.NodeGetCapabilitiesResponse()
         cap = response.capabilities.add()
         cap.rpc.type = csi.NODE_CAP_STAGE_UNSTAGE_VOLUME
+        cap = response.capabilities.add()
+        cap.rpc.type = csi.NODE_CAP_GET_VOLUME_STATS
         return response
 
     def NodeGetInfo(self, request, context):

@@ -44,6 +44,10 @@ NODE_CAP_UNKNOWN = 0
 NODE_CAP_STAGE_UNSTAGE_VOLUME = 1
 NODE_CAP_GET_VOLUME_STATS = 2
 
+# VolumeUsage.Unit
+USAGE_UNIT_BYTES = 1
+USAGE_UNIT_INODES = 2
+
 MESSAGES = [
     # Identity
     Message("GetPluginInfoRequest", []),
@@ -182,6 +186,27 @@ MESSAGES = [
         "NodeServiceCapability",
         [Field("rpc", 1, "NodeServiceCapability.RPC", oneof="type")],
     ),
+    Message(
+        "NodeGetVolumeStatsRequest",
+        [
+            Field("volume_id", 1, "string"),
+            Field("volume_path", 2, "string"),
+            Field("staging_target_path", 3, "string"),
+        ],
+    ),
+    Message(
+        "NodeGetVolumeStatsResponse",
+        [Field("usage", 1, "VolumeUsage", repeated=True)],
+    ),
+    Message(
+        "VolumeUsage",
+        [
+            Field("available", 1, "int64"),
+            Field("total", 2, "int64"),
+            Field("used", 3, "int64"),
+            Field("unit", 4, "int32"),  # 1=BYTES, 2=INODES
+        ],
+    ),
     Message("NodeGetInfoRequest", []),
     Message(
         "NodeGetInfoResponse",
@@ -268,6 +293,8 @@ SERVICES = [
              "NodePublishVolumeResponse"),
             ("NodeUnpublishVolume", "NodeUnpublishVolumeRequest",
              "NodeUnpublishVolumeResponse"),
+            ("NodeGetVolumeStats", "NodeGetVolumeStatsRequest",
+             "NodeGetVolumeStatsResponse"),
             ("NodeGetCapabilities", "NodeGetCapabilitiesRequest",
              "NodeGetCapabilitiesResponse"),
             ("NodeGetInfo", "NodeGetInfoRequest", "NodeGetInfoResponse"),
@@ -289,7 +316,6 @@ UNIMPLEMENTED_CONTROLLER_METHODS = (
     "ControllerExpandVolume",
 )
 UNIMPLEMENTED_NODE_METHODS = (
-    "NodeGetVolumeStats",
     "NodeExpandVolume",
 )
 

@@ -57,6 +57,9 @@ class CSIControllerServicer:
 
 
 class CSINodeServicer:
+    def NodeGetVolumeStats(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "NodeGetVolumeStats")
+
     def NodeStageVolume(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "NodeStageVolume")
 
@@ -114,6 +117,8 @@ def add_csi_node_to_server(servicer, server):
                                     csi.NodePublishVolumeRequest),
         "NodeUnpublishVolume": _unary(servicer.NodeUnpublishVolume,
                                       csi.NodeUnpublishVolumeRequest),
+        "NodeGetVolumeStats": _unary(servicer.NodeGetVolumeStats,
+                                     csi.NodeGetVolumeStatsRequest),
         "NodeGetCapabilities": _unary(servicer.NodeGetCapabilities,
                                       csi.NodeGetCapabilitiesRequest),
         "NodeGetInfo": _unary(servicer.NodeGetInfo, csi.NodeGetInfoRequest),
@@ -170,6 +175,8 @@ class CSINodeStub:
                                         csi.NodePublishVolumeResponse)
         self.NodeUnpublishVolume = method("NodeUnpublishVolume",
                                           csi.NodeUnpublishVolumeResponse)
+        self.NodeGetVolumeStats = method("NodeGetVolumeStats",
+                                         csi.NodeGetVolumeStatsResponse)
         self.NodeGetCapabilities = method("NodeGetCapabilities",
                                           csi.NodeGetCapabilitiesResponse)
         self.NodeGetInfo = method("NodeGetInfo", csi.NodeGetInfoResponse)

@@ -170,3 +170,34 @@ class TestCSISanity:
             with pytest.raises(grpc.RpcError) as excinfo:
                 stub.ValidateVolumeCapabilities(request, timeout=30)
             assert excinfo.value.code() == grpc.StatusCode.NOT_FOUND
+
+
+class TestNodeGetVolumeStats:
+    def test_stats_of_mounted_path(self, hipstored, tmp_path):  # noqa: F811
+        from oim_amd.spec.rpc_csi import CSINodeStub
+
+        backend = LocalBackend(hipstored.socket_path)
+        driver = OIMDriver(driver_name="oim-local", node_id="n0",
+                           endpoint=f"unix://{tmp_path}/csi2.sock",
+                           backend=backend, mounter=Mounter(FakeExec()))
+        driver.start()
+        try:
+            channel = grpc.insecure_channel(grpc_target(driver.addr()))
+            with channel:
+                stub = CSINodeStub(channel)
+                response = stub.NodeGetVolumeStats(
+                    csi.NodeGetVolumeStatsRequest(
+                        volume_id="v", volume_path=str(tmp_path)),
+                    timeout=30)
+                by_unit = {u.unit: u for u in response.usage}
+                assert csi.USAGE_UNIT_BYTES in by_unit
+                assert by_unit[csi.USAGE_UNIT_BYTES].total > 0
+                assert csi.USAGE_UNIT_INODES in by_unit
+                with pytest.raises(grpc.RpcError) as excinfo:
+                    stub.NodeGetVolumeStats(
+                        csi.NodeGetVolumeStatsRequest(
+                            volume_id="v", volume_path="/no/such/path"),
+                        timeout=30)
+                assert excinfo.value.code() == grpc.StatusCode.NOT_FOUND
+        finally:
+            driver.stop()
