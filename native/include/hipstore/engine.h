@@ -43,6 +43,14 @@ uint64_t persistent_stat(int which);
 void* alloc_pinned(size_t bytes);
 void free_pinned(void* ptr);
 
+// Device-side range copy between HBM bdevs (volume clone / stripe
+// rebuild): same-device copies run the LDS-staged block kernel at HBM
+// rates; cross-device copies go over xGMI peer access. Returns
+// IoStatus; kIoInvalid when either bdev is not HBM-resident or ranges
+// are out of bounds/misaligned.
+int hbm_copy_sync(Bdev* src, uint64_t src_offset, Bdev* dst,
+                  uint64_t dst_offset, uint64_t length);
+
 // Synchronous convenience wrappers (tests, NBD pump): create a
 // throwaway channel, submit, poll to completion. Returns IoStatus.
 int bdev_read_sync(Bdev* bdev, uint64_t offset, void* buf, uint64_t len);

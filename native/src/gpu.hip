@@ -883,6 +883,66 @@ void crc32c_hbm_blocks(Bdev* bdev, uint64_t offset, uint32_t block_size,
   memcpy(out, ctx.out, count * 4);
 }
 
+int hbm_copy_sync(Bdev* src, uint64_t src_offset, Bdev* dst,
+                  uint64_t dst_offset, uint64_t length) {
+  uint8_t* src_base = static_cast<uint8_t*>(src->device_base());
+  uint8_t* dst_base = static_cast<uint8_t*>(dst->device_base());
+  if (src_base == nullptr || dst_base == nullptr) return kIoInvalid;
+  if (length == 0 || length % 16 != 0 ||
+      src_offset + length > src->size_bytes() ||
+      dst_offset + length > dst->size_bytes()) {
+    return kIoInvalid;
+  }
+  const int src_dev = src->gpu_device();
+  const int dst_dev = dst->gpu_device();
+  try {
+    HIP_CHECK(hipSetDevice(dst_dev));
+    hipStream_t stream = nullptr;
+    HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+    hipError_t err;
+    if (src_dev != dst_dev) {
+      // Cross-GPU: one xGMI peer copy (SDMA saturates the link).
+      (void)hipSetDevice(src_dev);
+      (void)hipDeviceEnablePeerAccess(dst_dev, 0);
+      (void)hipSetDevice(dst_dev);
+      (void)hipDeviceEnablePeerAccess(src_dev, 0);
+      err = hipMemcpyPeerAsync(dst_base + dst_offset, dst_dev,
+                               src_base + src_offset, src_dev, length,
+                               stream);
+    } else {
+      // Same device: tile through the LDS-staged copy kernel in
+      // batches of descriptors built in pinned memory.
+      const uint32_t tiles_per_launch = 1u << 16;
+      BlockDesc* descs = static_cast<BlockDesc*>(
+          alloc_pinned(tiles_per_launch * sizeof(BlockDesc)));
+      BlockDesc* descs_dev = device_view(descs);
+      uint64_t done = 0;
+      err = hipSuccess;
+      while (done < length && err == hipSuccess) {
+        uint32_t n = 0;
+        while (n < tiles_per_launch && done < length) {
+          const uint32_t bytes = static_cast<uint32_t>(
+              std::min<uint64_t>(kTileBytes, length - done));
+          descs[n] = {src_base + src_offset + done,
+                      dst_base + dst_offset + done, bytes, 0};
+          ++n;
+          done += bytes;
+        }
+        const uint32_t grid = (n + kWavesPerWg - 1) / kWavesPerWg;
+        hipLaunchKernelGGL(k_copy_blocks, dim3(grid), dim3(kWavesPerWg * 64),
+                           0, stream, descs_dev, n);
+        err = hipStreamSynchronize(stream);
+      }
+      free_pinned(descs);
+    }
+    if (err == hipSuccess) err = hipStreamSynchronize(stream);
+    (void)hipStreamDestroy(stream);
+    return err == hipSuccess ? kIoOk : kIoFailed;
+  } catch (const std::exception&) {
+    return kIoFailed;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Synchronous helpers
 // ---------------------------------------------------------------------------

@@ -482,6 +482,22 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         return Json(name);
       });
 
+  server->register_method("bdev_copy", [&manager](const Json& p) {
+    // Device-side clone/rebuild: HBM->HBM at memory/xGMI rates.
+    BdevPtr src = manager.find(p.get_string("src"));
+    BdevPtr dst = manager.find(p.get_string("dst"));
+    if (!src) not_found("bdev " + p.get_string("src"));
+    if (!dst) not_found("bdev " + p.get_string("dst"));
+    const uint64_t length = p.get_int("length",
+                                      static_cast<int64_t>(src->size_bytes()));
+    int status = hbm_copy_sync(src.get(), p.get_int("src_offset", 0),
+                               dst.get(), p.get_int("dst_offset", 0), length);
+    if (status != kIoOk) {
+      throw RpcError{kInvalidParams, "bdev_copy failed (HBM bdevs only)"};
+    }
+    return Json(JsonObject{});
+  });
+
   // Stepped benchmarking with persistent queues (bench.py contract):
   // perf_session_start -> id; perf_session_step runs a fixed I/O count
   // on the session's live queues; perf_session_stop tears down.

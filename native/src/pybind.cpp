@@ -147,6 +147,23 @@ PYBIND11_MODULE(_hipstore, m) {
       py::arg("num_queues") = 1, py::arg("seconds") = 2.0,
       py::arg("max_ios") = 0);
 
+  m.def("hbm_copy", [](BdevPtr src, uint64_t src_offset, BdevPtr dst,
+                       uint64_t dst_offset, uint64_t length) {
+    int status;
+    {
+      py::gil_scoped_release release;
+      status = hbm_copy_sync(src.get(), src_offset, dst.get(), dst_offset,
+                             length);
+    }
+    if (status != kIoOk) {
+      throw std::runtime_error("hbm_copy failed: status " +
+                               std::to_string(status));
+    }
+  }, py::arg("src"), py::arg("src_offset"), py::arg("dst"),
+     py::arg("dst_offset"), py::arg("length"),
+     "Device-side range copy between HBM bdevs (LDS-staged kernel "
+     "same-device, xGMI peer copy cross-device)");
+
   m.def("persistent_stats", [] {
     py::dict d;
     d["launches"] = hipstore::persistent_stat(0);

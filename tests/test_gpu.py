@@ -187,6 +187,38 @@ class TestCompositeGpu:
 
 
 @needs_gpu
+class TestHbmCopy:
+    def test_same_device_copy(self):
+        a = hs.create_hbm_bdev("cp-a", 4096, 16384, device=0)  # 64 MiB
+        b = hs.create_hbm_bdev("cp-b", 4096, 16384, device=0)
+        rng = random.Random(51)
+        blob = bytes(rng.getrandbits(8) for _ in range(1 << 20))
+        a.write(4096, blob)
+        hs.hbm_copy(a, 4096, b, 8192, len(blob))
+        assert b.read(8192, len(blob)) == blob
+        # sub-tile granularity (16-byte aligned lengths)
+        hs.hbm_copy(a, 4096, b, 0, 512)
+        assert b.read(0, 512) == blob[:512]
+
+    def test_rejects_cpu_bdev(self):
+        a = hs.create_hbm_bdev("cp-c", 4096, 1024, device=0)
+        c = hs.create_malloc_bdev("cp-d", 4096, 1024)
+        with pytest.raises(RuntimeError):
+            hs.hbm_copy(a, 0, c, 0, 4096)
+
+    @pytest.mark.skipif(hs.gpu_device_count() < 2,
+                        reason="needs >=2 GPUs for xGMI copy")
+    def test_cross_device_copy(self):
+        a = hs.create_hbm_bdev("cp-x0", 4096, 16384, device=0)
+        b = hs.create_hbm_bdev("cp-x1", 4096, 16384, device=1)
+        rng = random.Random(53)
+        blob = bytes(rng.getrandbits(8) for _ in range(1 << 20))
+        a.write(0, blob)
+        hs.hbm_copy(a, 0, b, 0, len(blob))
+        assert b.read(0, len(blob)) == blob
+
+
+@needs_gpu
 class TestNvmfGpu:
     def test_hbm_namespace_gpu_digest_roundtrip(self):
         """NVMe/TCP loopback with an HBM-resident namespace: C2HData
