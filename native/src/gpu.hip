@@ -26,8 +26,10 @@
 #include <cstring>
 #include <deque>
 #include <functional>
+#include <map>
 #include <mutex>
 #include <random>
+#include <set>
 #include <stdexcept>
 #include <thread>
 #include <vector>
@@ -302,6 +304,182 @@ __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
 std::atomic<uint64_t> g_pers_launches{0};
 std::atomic<uint64_t> g_pers_relaunches{0};
 std::atomic<uint64_t> g_pers_stall_queries{0};
+
+// --- shared service kernel (one per device) --------------------------------
+//
+// The per-queue persistent kernel needs one hardware queue per channel
+// (GPU_MAX_HW_QUEUES caps that per process). The shared service is the
+// scalable form: ONE resident kernel per device multiplexes up to 64
+// submission rings. The leader workgroup polls every active ring's
+// host tail IN PARALLEL (lane l owns slot l, so a full sweep is one
+// PCIe round trip, not 64); worker waves sweep the slots and claim
+// descriptors with a bounded CAS (claim-only-when-available — a blind
+// fetch_add would strand a worker on one ring while others have work).
+// Same liveness contract as the per-queue kernel: idle self-exit +
+// idempotent-replay relaunch.
+
+struct SharedSlot {
+  // Written by the host (hipMemcpy) while inactive; read by waves via
+  // relaxed agent loads. 8-byte fields so every access is one atom.
+  const BlockDesc* sq;                    // pinned host ring (device view)
+  volatile unsigned long long* sq_tail;   // pinned host tail (device view)
+  volatile unsigned long long* cq;        // pinned host CQ (device view)
+  unsigned long long claim;               // device-side claim counter
+  unsigned long long known_tail;          // leader-mirrored tail
+  unsigned long long ring_mask_active;    // low 32: mask, high 32: active
+};
+static_assert(sizeof(SharedSlot) == 48, "8-byte atom layout");
+
+constexpr uint32_t kSharedSlots = 64;
+
+struct SharedCtl {
+  SharedSlot* slots;            // device memory, kSharedSlots entries
+  volatile uint32_t* stop;      // pinned host
+  uint32_t* exit_flag;          // device
+  uint32_t idle_spins;
+};
+
+// Slot fields as 8-byte words for uniform agent-scope access (sc1
+// loads bypass the CU's L1, which may hold stale lines from a previous
+// occupant of a reused slot): [0]=sq [1]=sq_tail [2]=cq [3]=claim
+// [4]=known_tail [5]=mask|active<<32.
+__device__ __forceinline__ unsigned long long slot_ld(
+    SharedSlot* slot, int word) {
+  return __hip_atomic_load(
+      reinterpret_cast<unsigned long long*>(slot) + word, __ATOMIC_RELAXED,
+      __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__device__ __forceinline__ void slot_st(SharedSlot* slot, int word,
+                                        unsigned long long value) {
+  __hip_atomic_store(reinterpret_cast<unsigned long long*>(slot) + word,
+                     value, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__device__ __forceinline__ unsigned long long wave_bcast_u64(
+    unsigned long long v) {
+  int lo = __shfl(static_cast<int>(v & 0xFFFFFFFFu), 0);
+  int hi = __shfl(static_cast<int>(v >> 32), 0);
+  return (static_cast<unsigned long long>(static_cast<uint32_t>(hi)) << 32) |
+         static_cast<uint32_t>(lo);
+}
+
+__global__ __launch_bounds__(64) void k_shared_service(SharedCtl ctl) {
+  __shared__ __attribute__((aligned(16))) uint8_t lds_raw[kTileBytes];
+  const uint32_t lane = threadIdx.x & 63;
+  float4* lds = reinterpret_cast<float4*>(lds_raw);
+
+  if (blockIdx.x == 0) {
+    // Leader wave: lane l mirrors slot l's host tail — a full sweep of
+    // all 64 slots is ONE parallel PCIe round trip.
+    SharedSlot* slot = &ctl.slots[lane];
+    unsigned long long known = 0;
+    uint32_t spins = 0;
+    while (true) {
+      bool advanced = false;
+      if (slot_ld(slot, 5) >> 32) {
+        volatile unsigned long long* tail_ptr =
+            reinterpret_cast<volatile unsigned long long*>(slot_ld(slot, 1));
+        const unsigned long long tail = __hip_atomic_load(
+            const_cast<const unsigned long long*>(tail_ptr),
+            __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM);
+        if (tail > known) {
+          known = tail;
+          slot_st(slot, 4, known);
+          advanced = true;
+        }
+      } else {
+        known = 0;  // slot detached: forget its generation
+      }
+      const bool stop_now =
+          lane == 0 &&
+          __hip_atomic_load(const_cast<const uint32_t*>(ctl.stop),
+                            __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_SYSTEM) != 0;
+      if (__any(advanced)) spins = 0;
+      if (__any(stop_now) || ++spins > ctl.idle_spins) {
+        if (lane == 0) {
+          __hip_atomic_store(ctl.exit_flag, 1u, __ATOMIC_RELEASE,
+                             __HIP_MEMORY_SCOPE_AGENT);
+        }
+        return;
+      }
+      __builtin_amdgcn_s_sleep(8);
+    }
+  }
+
+  // Worker wave: sweep the slots, claim and serve at most one
+  // descriptor per slot per sweep (fairness across rings). A bounded
+  // CAS claims work only when available — a blind fetch_add would park
+  // this worker on one ring while others have work.
+  const uint32_t start = blockIdx.x;  // stagger sweep origins
+  while (true) {
+    bool progress = false;
+    for (uint32_t i = 0; i < kSharedSlots; ++i) {
+      SharedSlot* slot = &ctl.slots[(start + i) % kSharedSlots];
+      const unsigned long long ma = slot_ld(slot, 5);
+      if (!(ma >> 32)) continue;
+      const uint32_t ring_mask = static_cast<uint32_t>(ma);
+      const unsigned long long kt = slot_ld(slot, 4);
+      unsigned long long c = slot_ld(slot, 3);
+      unsigned long long claim = ~0ull;
+      while (c < kt) {
+        unsigned long long witnessed;
+        if (lane == 0) {
+          witnessed = atomicCAS(&slot->claim, c, c + 1);
+        }
+        witnessed = wave_bcast_u64(witnessed);
+        if (witnessed == c) {
+          claim = c;
+          break;
+        }
+        c = witnessed;
+      }
+      if (claim == ~0ull) continue;
+      // Serve it (volatile loads: pinned after the claim, never hoisted).
+      const BlockDesc* sq = reinterpret_cast<const BlockDesc*>(slot_ld(slot, 0));
+      const volatile unsigned long long* vd =
+          reinterpret_cast<const volatile unsigned long long*>(
+              &sq[claim & ring_mask]);
+      const unsigned long long w0 = vd[0];
+      const unsigned long long w1 = vd[1];
+      const unsigned long long w2 = vd[2];
+      const uint8_t* src = reinterpret_cast<const uint8_t*>(w0);
+      uint8_t* dst = reinterpret_cast<uint8_t*>(w1);
+      const uint32_t bytes = static_cast<uint32_t>(w2 & 0xFFFFFFFFu);
+      const uint32_t n16 = bytes >> 4;
+      if (src != nullptr) {
+        const float4* __restrict__ s4 = reinterpret_cast<const float4*>(src);
+#pragma unroll 4
+        for (uint32_t k = lane; k < n16; k += 64) lds[k] = s4[k];
+      } else {
+        const uint32_t b = static_cast<uint32_t>(w2 >> 32) & 0xFF;
+        const uint32_t word = b | (b << 8) | (b << 16) | (b << 24);
+        const float4 v = {__uint_as_float(word), __uint_as_float(word),
+                          __uint_as_float(word), __uint_as_float(word)};
+        for (uint32_t k = lane; k < n16; k += 64) lds[k] = v;
+      }
+      float4* __restrict__ d4 = reinterpret_cast<float4*>(dst);
+#pragma unroll 4
+      for (uint32_t k = lane; k < n16; k += 64) d4[k] = lds[k];
+      __threadfence_system();
+      if (lane == 0) {
+        volatile unsigned long long* cq =
+            reinterpret_cast<volatile unsigned long long*>(slot_ld(slot, 2));
+        __hip_atomic_store(&cq[claim & ring_mask], claim + 1,
+                           __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
+      }
+      progress = true;
+    }
+    if (!progress) {
+      if (__hip_atomic_load(ctl.exit_flag, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT) != 0) {
+        return;
+      }
+      __builtin_amdgcn_s_sleep(16);
+    }
+  }
+}
 
 int g_device_count = -1;
 
@@ -765,6 +943,307 @@ class HbmPersistentChannel : public IoChannel {
   std::vector<std::pair<IoCompletion, int>> immediate_;
 };
 
+// Host side of the shared service kernel: one per device, multiplexing
+// every shared-mode channel through the 64-slot table.
+class SharedService {
+ public:
+  static constexpr uint32_t kIdleSpins = 500000;
+
+  static std::shared_ptr<SharedService>& instance(int device) {
+    static std::mutex mutex;
+    static std::map<int, std::shared_ptr<SharedService>> services;
+    std::lock_guard<std::mutex> lock(mutex);
+    auto& service = services[device];
+    if (!service) service = std::make_shared<SharedService>(device);
+    return service;
+  }
+
+  explicit SharedService(int device) : device_(device) {
+    HIP_CHECK(hipSetDevice(device));
+    HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+    HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&slots_dev_),
+                        kSharedSlots * sizeof(SharedSlot) + 8));
+    HIP_CHECK(hipMemset(slots_dev_, 0, kSharedSlots * sizeof(SharedSlot) + 8));
+    exit_flag_dev_ = reinterpret_cast<uint32_t*>(slots_dev_ + kSharedSlots);
+    void* p_stop = nullptr;
+    HIP_CHECK(hipHostMalloc(&p_stop, 4, hipHostMallocMapped));
+    stop_ = static_cast<volatile uint32_t*>(p_stop);
+    *stop_ = 0;
+    memset(mirror_, 0, sizeof(mirror_));
+    memset(used_, 0, sizeof(used_));
+    memset(prefix_, 0, sizeof(prefix_));
+  }
+
+  // The service lives for the process (kernels self-exit when idle, so
+  // an abandoned service costs nothing); no teardown path needed.
+
+  int attach(const BlockDesc* sq_dev, unsigned long long* tail_dev,
+             unsigned long long* cq_dev, uint32_t ring_mask) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    for (uint32_t i = 0; i < kSharedSlots; ++i) {
+      if (used_[i]) continue;
+      used_[i] = true;
+      prefix_[i] = 0;
+      SharedSlot& m = mirror_[i];
+      m.sq = sq_dev;
+      m.sq_tail = tail_dev;
+      m.cq = cq_dev;
+      m.claim = 0;
+      m.known_tail = 0;
+      m.ring_mask_active = ring_mask;  // active bit clear
+      (void)hipSetDevice(device_);
+      // Config words first, the active bit last (stream-ordered copies
+      // so a running kernel never sees active before the pointers).
+      HIP_CHECK(hipMemcpyAsync(&slots_dev_[i], &m, 40, hipMemcpyHostToDevice,
+                               stream_));
+      m.ring_mask_active = ring_mask | (1ull << 32);
+      HIP_CHECK(hipMemcpyAsync(
+          reinterpret_cast<unsigned long long*>(&slots_dev_[i]) + 5,
+          &m.ring_mask_active, 8, hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+      ensure_running_locked(/*force_check=*/true);
+      return static_cast<int>(i);
+    }
+    throw std::runtime_error("shared service: no free ring slots (64 max)");
+  }
+
+  void detach(int slot) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    // The owning channel drained first, so no worker can win a claim
+    // on this slot; clearing the active bit stops sweeps touching it.
+    (void)hipSetDevice(device_);
+    mirror_[slot].ring_mask_active &= 0xFFFFFFFFull;
+    (void)hipMemcpy(
+        reinterpret_cast<unsigned long long*>(&slots_dev_[slot]) + 5,
+        &mirror_[slot].ring_mask_active, 8, hipMemcpyHostToDevice);
+    used_[slot] = false;
+  }
+
+  void update_prefix(int slot, uint64_t completed) {
+    // Racy-read tolerable: only consumed under mutex_ in relaunch.
+    prefix_[slot] = completed;
+  }
+
+  // Called by channels when work is outstanding but completions stall:
+  // relaunch the service kernel if it idle-exited (claims reset to each
+  // channel's completed prefix; descriptor replay is idempotent).
+  void ensure_running() {
+    std::lock_guard<std::mutex> lock(mutex_);
+    ensure_running_locked(false);
+  }
+
+ private:
+  void ensure_running_locked(bool force_check) {
+    const auto now = std::chrono::steady_clock::now();
+    if (!force_check && now - last_check_ < std::chrono::milliseconds(1)) {
+      return;
+    }
+    last_check_ = now;
+    if (hipStreamQuery(stream_) != hipSuccess) return;  // still running
+    (void)hipSetDevice(device_);
+    for (uint32_t i = 0; i < kSharedSlots; ++i) {
+      if (!used_[i]) continue;
+      mirror_[i].claim = prefix_[i];
+      mirror_[i].known_tail = prefix_[i];
+    }
+    HIP_CHECK(hipMemcpyAsync(slots_dev_, mirror_,
+                             kSharedSlots * sizeof(SharedSlot),
+                             hipMemcpyHostToDevice, stream_));
+    HIP_CHECK(hipMemsetAsync(exit_flag_dev_, 0, 4, stream_));
+    SharedCtl ctl;
+    ctl.slots = slots_dev_;
+    ctl.stop = device_view(const_cast<uint32_t*>(stop_));
+    ctl.exit_flag = exit_flag_dev_;
+    ctl.idle_spins = kIdleSpins;
+    const char* env = getenv("HIPSTORE_SHARED_WORKERS");
+    int workers = env ? atoi(env) : 48;
+    workers = std::min(std::max(workers, 1), 255);
+    hipLaunchKernelGGL(k_shared_service, dim3(workers + 1), dim3(64), 0,
+                       stream_, ctl);
+    g_pers_launches.fetch_add(1, std::memory_order_relaxed);
+  }
+
+  int device_;
+  hipStream_t stream_ = nullptr;
+  SharedSlot* slots_dev_ = nullptr;
+  uint32_t* exit_flag_dev_ = nullptr;
+  volatile uint32_t* stop_ = nullptr;
+  std::mutex mutex_;
+  SharedSlot mirror_[kSharedSlots];
+  bool used_[kSharedSlots];
+  uint64_t prefix_[kSharedSlots];
+  std::chrono::steady_clock::time_point last_check_{};
+};
+
+// Channel whose ring is serviced by the per-device shared kernel.
+// Ring/completion bookkeeping mirrors HbmPersistentChannel; the
+// difference is who runs the GPU side.
+class HbmSharedChannel : public IoChannel {
+ public:
+  static constexpr uint32_t kRing = 32768;
+
+  HbmSharedChannel(int device, uint8_t* base) : base_(base) {
+    service_ = SharedService::instance(device);
+    HIP_CHECK(hipSetDevice(device));
+    HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&sq_),
+                            kRing * sizeof(BlockDesc), hipHostMallocMapped));
+    void* p_tail = nullptr;
+    void* p_cq = nullptr;
+    HIP_CHECK(hipHostMalloc(&p_tail, 8, hipHostMallocMapped));
+    HIP_CHECK(hipHostMalloc(&p_cq, kRing * 8, hipHostMallocMapped));
+    sq_tail_ = static_cast<volatile unsigned long long*>(p_tail);
+    cq_ = static_cast<volatile unsigned long long*>(p_cq);
+    *sq_tail_ = 0;
+    memset(const_cast<unsigned long long*>(cq_), 0, kRing * 8);
+    desc_io_.resize(kRing, nullptr);
+    slot_ = service_->attach(
+        device_view(sq_), device_view(const_cast<unsigned long long*>(sq_tail_)),
+        device_view(const_cast<unsigned long long*>(cq_)), kRing - 1);
+  }
+
+  ~HbmSharedChannel() override {
+    // Drain so no worker holds a claim below tail, then detach.
+    const auto deadline =
+        std::chrono::steady_clock::now() + std::chrono::seconds(10);
+    while (completed_ < tail_ &&
+           std::chrono::steady_clock::now() < deadline) {
+      poll();
+    }
+    service_->detach(slot_);
+    (void)hipHostFree(sq_);
+    (void)hipHostFree(const_cast<unsigned long long*>(sq_tail_));
+    (void)hipHostFree(const_cast<unsigned long long*>(cq_));
+    std::set<IoState*> leftovers(desc_io_.begin(), desc_io_.end());
+    for (IoState* state : leftovers) {
+      if (state != nullptr) delete state;  // undrained multi-tile IOs
+    }
+  }
+
+  struct IoState {
+    IoCompletion cb;
+    uint32_t remaining;
+    int status;
+  };
+
+  void enqueue(IoRequest req, int status) {
+    if (status != kIoOk) {
+      immediate_.emplace_back(std::move(req.on_complete), status);
+      return;
+    }
+    if (req.op == IoOp::kFlush) {
+      flushes_.push_back({tail_, std::move(req.on_complete)});
+      return;
+    }
+    pending_.push_back(std::move(req));
+    drain_pending();
+  }
+
+  int poll() {
+    int completed = 0;
+    for (auto& [cb, status] : immediate_) {
+      if (cb) cb(status);
+      ++completed;
+    }
+    immediate_.clear();
+    while (completed_ < tail_) {
+      unsigned long long seq = __atomic_load_n(
+          const_cast<const unsigned long long*>(&cq_[completed_ % kRing]),
+          __ATOMIC_ACQUIRE);
+      if (seq != completed_ + 1) break;
+      IoState* state = desc_io_[completed_ % kRing];
+      desc_io_[completed_ % kRing] = nullptr;
+      ++completed_;
+      if (state != nullptr && --state->remaining == 0) {
+        if (state->cb) state->cb(state->status);
+        delete state;
+        ++completed;
+      }
+    }
+    if (completed > 0) {
+      service_->update_prefix(slot_, completed_);
+      last_progress_ = std::chrono::steady_clock::now();
+    }
+    while (!flushes_.empty() && flushes_.front().first <= completed_) {
+      if (flushes_.front().second) flushes_.front().second(kIoOk);
+      flushes_.pop_front();
+      ++completed;
+    }
+    drain_pending();
+    if (completed_ < tail_ &&
+        std::chrono::steady_clock::now() - last_progress_ >
+            std::chrono::milliseconds(1)) {
+      service_->ensure_running();
+      last_progress_ = std::chrono::steady_clock::now();
+    }
+    return completed;
+  }
+
+  bool has_capacity(uint32_t tiles) const {
+    return tail_ - completed_ + tiles <= kRing;
+  }
+
+ private:
+  void drain_pending() {
+    while (!pending_.empty()) {
+      IoRequest& req = pending_.front();
+      const uint32_t tiles = static_cast<uint32_t>(
+          (req.length + kTileBytes - 1) / kTileBytes);
+      if (!has_capacity(tiles)) return;
+      auto* state = new IoState{std::move(req.on_complete), tiles, kIoOk};
+      uint64_t done = 0;
+      uint64_t t = tail_;
+      while (done < req.length) {
+        const uint32_t bytes = static_cast<uint32_t>(
+            std::min<uint64_t>(kTileBytes, req.length - done));
+        BlockDesc& d = sq_[t % kRing];
+        if (req.op == IoOp::kRead) {
+          d.src = base_ + req.offset + done;
+          d.dst = buf_device(static_cast<uint8_t*>(req.buffer) + done);
+        } else if (req.op == IoOp::kWrite) {
+          d.src = buf_device(static_cast<uint8_t*>(req.buffer) + done);
+          d.dst = base_ + req.offset + done;
+        } else {
+          d.src = nullptr;
+          d.dst = base_ + req.offset + done;
+        }
+        d.bytes = bytes;
+        d.fill = req.fill;
+        desc_io_[t % kRing] = state;
+        ++t;
+        done += bytes;
+      }
+      tail_ = t;
+      __atomic_store_n(const_cast<unsigned long long*>(sq_tail_), tail_,
+                       __ATOMIC_RELEASE);
+      pending_.pop_front();
+    }
+  }
+
+  uint8_t* buf_device(uint8_t* host) {
+    void* dev = nullptr;
+    hipError_t err = hipHostGetDevicePointer(&dev, host, 0);
+    if (err != hipSuccess) {
+      throw std::runtime_error(
+          "hipstore: I/O buffer is not pinned host memory (use alloc_pinned)");
+    }
+    return static_cast<uint8_t*>(dev);
+  }
+
+  uint8_t* base_;
+  std::shared_ptr<SharedService> service_;
+  int slot_ = -1;
+  BlockDesc* sq_ = nullptr;
+  volatile unsigned long long* sq_tail_ = nullptr;
+  volatile unsigned long long* cq_ = nullptr;
+  uint64_t tail_ = 0;
+  uint64_t completed_ = 0;
+  std::vector<IoState*> desc_io_;
+  std::deque<IoRequest> pending_;
+  std::deque<std::pair<uint64_t, IoCompletion>> flushes_;
+  std::vector<std::pair<IoCompletion, int>> immediate_;
+  std::chrono::steady_clock::time_point last_progress_{};
+};
+
 class HbmBdev : public Bdev {
  public:
   HbmBdev(const std::string& name, uint64_t block_size, uint64_t num_blocks,
@@ -772,6 +1251,12 @@ class HbmBdev : public Bdev {
       : Bdev(name, "Malloc disk", block_size, num_blocks),
         device_(device),
         persistent_(persistent) {
+    if (persistent_) {
+      // HIPSTORE_SHARED=1 selects the shared per-device service kernel
+      // (one HW queue total) instead of a kernel per channel.
+      const char* env = getenv("HIPSTORE_SHARED");
+      shared_ = env != nullptr && atoi(env) != 0;
+    }
     HIP_CHECK(hipSetDevice(device));
     HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&base_), size_bytes()));
     HIP_CHECK(hipMemset(base_, 0, size_bytes()));
@@ -789,6 +1274,7 @@ class HbmBdev : public Bdev {
   std::shared_ptr<IoChannel> get_channel() override {
     HIP_CHECK(hipSetDevice(device_));
     if (persistent_) {
+      if (shared_) return std::make_shared<HbmSharedChannel>(device_, base_);
       return std::make_shared<HbmPersistentChannel>(device_, base_);
     }
     return std::make_shared<HbmChannel>(device_, base_, size_bytes());
@@ -804,7 +1290,9 @@ class HbmBdev : public Bdev {
         status = kIoInvalid;
       }
     }
-    if (persistent_) {
+    if (persistent_ && shared_) {
+      static_cast<HbmSharedChannel*>(ch)->enqueue(std::move(req), status);
+    } else if (persistent_) {
       static_cast<HbmPersistentChannel*>(ch)->enqueue(std::move(req), status);
     } else {
       static_cast<HbmChannel*>(ch)->enqueue(std::move(req), status);
@@ -817,7 +1305,9 @@ class HbmBdev : public Bdev {
       // every submitter thread, and even hipSetDevice takes the
       // runtime's global lock — 4+ spinning threads convoyed on it and
       // starved each other (measured: one of four queues fell to ~10
-      // completions/s). launch() sets the device when actually needed.
+      // completions/s). launch()/ensure_running() set the device when
+      // actually needed.
+      if (shared_) return static_cast<HbmSharedChannel*>(ch)->poll();
       return static_cast<HbmPersistentChannel*>(ch)->poll();
     }
     auto* channel = static_cast<HbmChannel*>(ch);
@@ -836,6 +1326,7 @@ class HbmBdev : public Bdev {
   uint8_t* base_ = nullptr;
   int device_;
   bool persistent_;
+  bool shared_ = false;
 };
 
 }  // namespace
