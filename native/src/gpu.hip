@@ -961,6 +961,11 @@ class SharedService {
   explicit SharedService(int device) : device_(device) {
     HIP_CHECK(hipSetDevice(device));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+    // Slot config copies go on their OWN stream: an async copy queued
+    // on the kernel's stream would wait behind the resident service
+    // kernel and the slot would only activate after its idle-exit
+    // (measured: first completions at 1.4-2.9 s).
+    HIP_CHECK(hipStreamCreateWithFlags(&config_stream_, hipStreamNonBlocking));
     HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&slots_dev_),
                         kSharedSlots * sizeof(SharedSlot) + 8));
     HIP_CHECK(hipMemset(slots_dev_, 0, kSharedSlots * sizeof(SharedSlot) + 8));
@@ -995,12 +1000,12 @@ class SharedService {
       // Config words first, the active bit last (stream-ordered copies
       // so a running kernel never sees active before the pointers).
       HIP_CHECK(hipMemcpyAsync(&slots_dev_[i], &m, 40, hipMemcpyHostToDevice,
-                               stream_));
+                               config_stream_));
       m.ring_mask_active = ring_mask | (1ull << 32);
       HIP_CHECK(hipMemcpyAsync(
           reinterpret_cast<unsigned long long*>(&slots_dev_[i]) + 5,
-          &m.ring_mask_active, 8, hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipStreamSynchronize(stream_));
+          &m.ring_mask_active, 8, hipMemcpyHostToDevice, config_stream_));
+      HIP_CHECK(hipStreamSynchronize(config_stream_));
       ensure_running_locked(/*force_check=*/true);
       return static_cast<int>(i);
     }
@@ -1013,9 +1018,11 @@ class SharedService {
     // on this slot; clearing the active bit stops sweeps touching it.
     (void)hipSetDevice(device_);
     mirror_[slot].ring_mask_active &= 0xFFFFFFFFull;
-    (void)hipMemcpy(
+    (void)hipMemcpyAsync(
         reinterpret_cast<unsigned long long*>(&slots_dev_[slot]) + 5,
-        &mirror_[slot].ring_mask_active, 8, hipMemcpyHostToDevice);
+        &mirror_[slot].ring_mask_active, 8, hipMemcpyHostToDevice,
+        config_stream_);
+    (void)hipStreamSynchronize(config_stream_);
     used_[slot] = false;
   }
 
@@ -1065,6 +1072,7 @@ class SharedService {
 
   int device_;
   hipStream_t stream_ = nullptr;
+  hipStream_t config_stream_ = nullptr;
   SharedSlot* slots_dev_ = nullptr;
   uint32_t* exit_flag_dev_ = nullptr;
   volatile uint32_t* stop_ = nullptr;
