@@ -48,3 +48,13 @@ bin/hipstored: $(CORE_OBJS) $(B)/main.o
 
 clean:
 	rm -rf $(B) bin oim_amd/_hipstore*.so
+
+.PHONY: test gpu-test bench
+test: all
+	$(PYTHON) -m pytest tests -q -m "not gpu"
+
+gpu-test: all
+	$(PYTHON) -m pytest tests -q -m gpu
+
+bench: all
+	$(PYTHON) bench.py --steps 20 --warmup 5
