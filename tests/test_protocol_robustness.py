@@ -1,0 +1,93 @@
+"""Negative-path protocol tests: the daemon and NVMe/TCP target must
+survive malformed clients (parse errors answered or connection dropped,
+process stays healthy)."""
+
+import socket
+
+import pytest
+
+from oim_amd import _hipstore as hs
+from oim_amd import hipstore
+
+from fixtures import hipstored  # noqa: F401
+
+
+class TestJsonRpcRobustness:
+    def test_malformed_json_then_healthy(self, hipstored):  # noqa: F811
+        with socket.socket(socket.AF_UNIX, socket.SOCK_STREAM) as raw:
+            raw.connect(hipstored.socket_path)
+            raw.sendall(b'{"jsonrpc": "2.0", "method": [BROKEN')
+            raw.settimeout(10)
+            reply = raw.recv(65536)
+            assert b"-32700" in reply or reply == b""  # parse error or drop
+        # daemon still serves new clients
+        with hipstore.Client(hipstored.socket_path) as client:
+            assert "get_bdevs" in client.invoke("get_rpc_methods")
+
+    def test_non_object_request(self, hipstored):  # noqa: F811
+        with socket.socket(socket.AF_UNIX, socket.SOCK_STREAM) as raw:
+            raw.connect(hipstored.socket_path)
+            raw.sendall(b"[1,2,3]")
+            raw.settimeout(10)
+            reply = raw.recv(65536)
+            assert b"-32600" in reply  # invalid request
+        with hipstore.Client(hipstored.socket_path) as client:
+            assert client.invoke("get_bdevs") == []
+
+    def test_missing_params_defaults(self, hipstored):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            with pytest.raises(hipstore.RpcError):
+                client.invoke("construct_malloc_bdev")  # no params at all
+
+    def test_huge_method_name(self, hipstored):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            with pytest.raises(hipstore.RpcError) as excinfo:
+                client.invoke("x" * 100000)
+            assert excinfo.value.code == hipstore.client.ERROR_METHOD_NOT_FOUND
+
+
+class TestNvmfRobustness:
+    def test_garbage_after_handshake_does_not_kill_target(self):
+        backing = hs.create_malloc_bdev("robust-ns", 512, 2048)
+        target = hs.start_nvmf_tcp_target("", 0, "nqn.robust", True)
+        target.add_namespace(backing)
+        try:
+            with socket.create_connection(("127.0.0.1", target.port),
+                                          timeout=10) as raw:
+                raw.sendall(b"\x00" * 128)  # ICReq-sized zeros: bad type
+            with socket.create_connection(("127.0.0.1", target.port),
+                                          timeout=10) as raw:
+                raw.sendall(b"GET / HTTP/1.0\r\n\r\n")  # not NVMe/TCP at all
+            # target still serves a real initiator
+            bdev = hs.create_nvmf_tcp_bdev("robust-init", "127.0.0.1",
+                                           target.port, "nqn.robust")
+            bdev.write(0, b"\xaa" * 512)
+            assert bdev.read(0, 512) == b"\xaa" * 512
+        finally:
+            target.stop()
+
+    def test_oversized_plen_drops_connection(self):
+        backing = hs.create_malloc_bdev("robust-ns2", 512, 2048)
+        target = hs.start_nvmf_tcp_target("", 0, "nqn.robust2", False)
+        target.add_namespace(backing)
+        try:
+            import struct
+
+            with socket.create_connection(("127.0.0.1", target.port),
+                                          timeout=10) as raw:
+                # valid ICReq first
+                icreq = struct.pack("<BBBBI", 0x00, 0, 128, 0, 128)
+                icreq += struct.pack("<HBBI", 0, 0, 0, 15) + b"\x00" * 112
+                raw.sendall(icreq)
+                icresp = raw.recv(128)
+                assert len(icresp) == 128
+                # then a PDU claiming a 1 GiB payload: must be rejected
+                raw.sendall(struct.pack("<BBBBI", 0x04, 0, 72, 0, 1 << 30))
+                raw.settimeout(10)
+                assert raw.recv(16) == b""  # connection dropped
+            bdev = hs.create_nvmf_tcp_bdev("robust-init2", "127.0.0.1",
+                                           target.port, "nqn.robust2", 1,
+                                           False)
+            assert bdev.num_blocks == 2048
+        finally:
+            target.stop()
