@@ -28,6 +28,18 @@ import time
 # is 4; beyond that MES time-slices them). Must precede HIP init.
 os.environ.setdefault("GPU_MAX_HW_QUEUES", "24")
 
+
+def _set_affinity_base(num_queues: int) -> None:
+    """Pin each rank's daemon queue threads to a disjoint core range
+    (SPDK-reactor style; spinning submitters jitter under CFS)."""
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    ncpu = os.cpu_count() or 1
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    # Only pin when there are enough cores for every rank's queues.
+    if ncpu >= world * (num_queues + 2):
+        os.environ.setdefault("HIPSTORE_AFFINITY_BASE",
+                              str(local_rank * (num_queues + 2)))
+
 REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO_ROOT)
 
@@ -46,6 +58,7 @@ def start_stack(rank, local_rank, use_gpu, args, tmp):
     from oim_amd.spec import csi_v1 as csi
     from oim_amd.spec.rpc_csi import CSIControllerStub
 
+    _set_affinity_base(args.num_queues)
     daemon_sock = os.path.join(tmp, "hipstored.sock")
     cmd = [os.path.join(REPO_ROOT, "bin", "hipstored"), "-S", daemon_sock,
            "-d", str(local_rank)]

@@ -17,6 +17,9 @@
 //     and gives the CRC32C/verify paths a resident tile to chew on.
 
 #include <hip/hip_runtime.h>
+#include <pthread.h>
+#include <sched.h>
+#include <unistd.h>
 #include <stdlib.h>
 
 #include <algorithm>
@@ -1572,6 +1575,24 @@ struct LatHist {
 
 }  // namespace
 
+namespace {
+
+// Optional queue-thread affinity (HIPSTORE_AFFINITY_BASE=<core>): pin
+// queue q to core base+q, SPDK-reactor style. Spinning submitter
+// threads migrate under CFS otherwise, which adds multi-us jitter.
+void maybe_pin_thread(int queue_index) {
+  const char* env = getenv("HIPSTORE_AFFINITY_BASE");
+  if (env == nullptr) return;
+  const long ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+  const int core = (atoi(env) + queue_index) % static_cast<int>(ncpu);
+  cpu_set_t set;
+  CPU_ZERO(&set);
+  CPU_SET(core, &set);
+  (void)pthread_setaffinity_np(pthread_self(), sizeof(set), &set);
+}
+
+}  // namespace
+
 PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
                         uint32_t io_size, uint32_t queue_depth,
                         int num_queues, double seconds, uint64_t max_ios) {
@@ -1596,6 +1617,7 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
   for (int q = 0; q < num_queues; ++q) {
     threads.emplace_back([&, q] {
       try {
+        maybe_pin_thread(q);
         auto channel = bdev->get_channel();
         uint8_t* buf = static_cast<uint8_t*>(
             alloc_pinned(static_cast<size_t>(io_size) * queue_depth));
@@ -1701,6 +1723,7 @@ struct PerfSession::Impl {
   void worker(int q) {
     using clock = std::chrono::steady_clock;
     try {
+      maybe_pin_thread(q);
       auto channel = bdev->get_channel();
       uint8_t* buf = static_cast<uint8_t*>(
           alloc_pinned(static_cast<size_t>(io_size) * queue_depth));

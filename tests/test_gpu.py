@@ -107,6 +107,25 @@ class TestPersistentEngine:
 
 
 @needs_gpu
+class TestHbmCapacity:
+    def test_quarter_terabyte_bdev(self):
+        """Size for the hardware: a 256 GiB HBM bdev (of 288 GB) with
+        I/O at the far end of the address space."""
+        try:
+            bdev = hs.create_hbm_bdev("huge", 4096, (256 << 30) // 4096,
+                                      device=0)
+        except RuntimeError as exc:
+            pytest.skip(f"insufficient free HBM: {exc}")
+        payload = bytes(random.getrandbits(8) for _ in range(4096))
+        far = bdev.size_bytes - 4096
+        bdev.write(far, payload)
+        bdev.write(0, payload[::-1])
+        assert bdev.read(far, 4096) == payload
+        assert bdev.read(0, 4096) == payload[::-1]
+        del bdev  # free 256 GiB before later tests
+
+
+@needs_gpu
 class TestCrc32c:
     def test_gpu_matches_software(self):
         bdev = hs.create_hbm_bdev("crc-test", 4096, 256, device=0)
