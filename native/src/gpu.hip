@@ -769,6 +769,7 @@ class HbmPersistentChannel : public IoChannel {
     // Device words: [0]=claim counter, [1]=known_tail, [2]=exit_flag.
     HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&claim_ctr_), 24));
     HIP_CHECK(hipMemset(claim_ctr_, 0, 24));
+    HIP_CHECK(hipStreamSynchronize(nullptr));  // memset before launch()
     desc_io_.resize(kRing, nullptr);
     launch();
   }
@@ -972,6 +973,7 @@ class SharedService {
     HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&slots_dev_),
                         kSharedSlots * sizeof(SharedSlot) + 8));
     HIP_CHECK(hipMemset(slots_dev_, 0, kSharedSlots * sizeof(SharedSlot) + 8));
+    HIP_CHECK(hipStreamSynchronize(nullptr));  // memset before config copies
     exit_flag_dev_ = reinterpret_cast<uint32_t*>(slots_dev_ + kSharedSlots);
     void* p_stop = nullptr;
     HIP_CHECK(hipHostMalloc(&p_stop, 4, hipHostMallocMapped));
@@ -1271,6 +1273,12 @@ class HbmBdev : public Bdev {
     HIP_CHECK(hipSetDevice(device));
     HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&base_), size_bytes()));
     HIP_CHECK(hipMemset(base_, 0, size_bytes()));
+    // hipMemset on the null stream is async with respect to the
+    // engine's non-blocking streams: on a 64+ GiB bdev the zeroing
+    // sweep was still running when the first writes landed and wiped
+    // them (far offsets are memset last). Null-stream sync only — a
+    // device-wide sync would block on live persistent service kernels.
+    HIP_CHECK(hipStreamSynchronize(nullptr));
   }
 
   ~HbmBdev() override {
