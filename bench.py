@@ -140,7 +140,7 @@ def main() -> int:
     parser.add_argument("--warmup", type=int, default=5)
     parser.add_argument("--io-size", type=int, default=4096)
     parser.add_argument("--queue-depth", type=int, default=32)
-    parser.add_argument("--num-queues", type=int, default=12)
+    parser.add_argument("--num-queues", type=int, default=14)
     parser.add_argument("--bdev-gb", type=float, default=8.0)
     parser.add_argument("--workload", default="randread")
     parser.add_argument("--engine", default="persistent",
