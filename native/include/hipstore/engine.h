@@ -39,6 +39,10 @@ BdevPtr create_hbm_bdev(const std::string& name, uint64_t block_size,
 // 2=stall queries.
 uint64_t persistent_stat(int which);
 
+// HBM capacity of `device`: (total_bytes, free_bytes) via
+// hipMemGetInfo; (0, 0) without a GPU.
+std::pair<uint64_t, uint64_t> hbm_info(int device);
+
 // Pinned-host buffer helpers (fall back to plain malloc without a GPU).
 void* alloc_pinned(size_t bytes);
 void free_pinned(void* ptr);

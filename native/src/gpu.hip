@@ -35,6 +35,7 @@
 #include <set>
 #include <stdexcept>
 #include <thread>
+#include <utility>
 #include <vector>
 
 #include "hipstore/crc32c.h"
@@ -531,6 +532,17 @@ std::string gpu_pci_address(int device) {
   std::string s(bus_id);
   std::transform(s.begin(), s.end(), s.begin(), ::tolower);
   return s;
+}
+
+std::pair<uint64_t, uint64_t> hbm_info(int device) {
+  if (!gpu_available() || device < 0 || device >= gpu_device_count()) {
+    return {0, 0};
+  }
+  if (hipSetDevice(device) != hipSuccess) return {0, 0};
+  size_t free_bytes = 0;
+  size_t total_bytes = 0;
+  if (hipMemGetInfo(&free_bytes, &total_bytes) != hipSuccess) return {0, 0};
+  return {total_bytes, free_bytes};
 }
 
 void* alloc_pinned(size_t bytes) {

@@ -482,6 +482,21 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         return Json(name);
       });
 
+  server->register_method("get_hbm_info", [device](const Json& p) {
+    // Capacity source for CSI GetCapacity: HBM totals on GPU, host
+    // RAM totals in CPU mode (sysinfo-free approximation via
+    // /proc/meminfo is overkill; report zeros and let callers fall
+    // back when not on a GPU).
+    const int dev = static_cast<int>(p.get_int("device", device));
+    auto [total, free_bytes] = hbm_info(dev);
+    JsonObject o;
+    o["device"] = Json(static_cast<int64_t>(dev));
+    o["total_bytes"] = Json(static_cast<int64_t>(total));
+    o["free_bytes"] = Json(static_cast<int64_t>(free_bytes));
+    o["pci_address"] = Json(gpu_available() ? gpu_pci_address(dev) : "");
+    return Json(std::move(o));
+  });
+
   server->register_method("bdev_copy", [&manager](const Json& p) {
     // Device-side clone/rebuild: HBM->HBM at memory/xGMI rates.
     BdevPtr src = manager.find(p.get_string("src"));

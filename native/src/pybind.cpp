@@ -147,6 +147,11 @@ PYBIND11_MODULE(_hipstore, m) {
       py::arg("num_queues") = 1, py::arg("seconds") = 2.0,
       py::arg("max_ios") = 0);
 
+  m.def("hbm_info", [](int device) {
+    auto [total, free_bytes] = hbm_info(device);
+    return py::make_tuple(total, free_bytes);
+  }, py::arg("device") = 0);
+
   m.def("hbm_copy", [](BdevPtr src, uint64_t src_offset, BdevPtr dst,
                        uint64_t dst_offset, uint64_t length) {
     int status;

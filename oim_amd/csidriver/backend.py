@@ -27,3 +27,8 @@ class OIMBackend:
 
     def delete_device(self, volume_id: str) -> None:
         raise NotImplementedError
+
+    def get_capacity(self):
+        """Available bytes for new volumes, or None when the backend
+        cannot tell (remote mode: the oim.v0 API has no capacity RPC)."""
+        return None

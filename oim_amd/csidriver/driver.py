@@ -147,10 +147,22 @@ class ControllerServer(CSIControllerServicer):
             response.message = "unsupported volume capabilities"
         return response
 
+    def GetCapacity(self, request, context):
+        capacity = self.backend.get_capacity()
+        if capacity is None:
+            context.abort(
+                grpc.StatusCode.UNIMPLEMENTED,
+                "capacity reporting is only available in local mode "
+                "(the oim.v0 controller API has no capacity RPC)")
+        return csi.GetCapacityResponse(available_capacity=capacity)
+
     def ControllerGetCapabilities(self, request, context):
         response = csi.ControllerGetCapabilitiesResponse()
         cap = response.capabilities.add()
         cap.rpc.type = csi.CTRL_CAP_CREATE_DELETE_VOLUME
+        if self.backend.get_capacity() is not None:
+            cap = response.capabilities.add()
+            cap.rpc.type = csi.CTRL_CAP_GET_CAPACITY
         return response
 
 

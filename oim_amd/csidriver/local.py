@@ -102,3 +102,10 @@ class LocalBackend(OIMBackend):
             device = self._find_exported(client, volume_id)
             if device:
                 hipstore.stop_nbd_disk(client, device)
+
+    def get_capacity(self):
+        """Free HBM on the daemon's GPU (None in CPU mode)."""
+        with self._client() as client:
+            info = client.invoke("get_hbm_info")
+        free = int(info.get("free_bytes", 0))
+        return free if free > 0 else None

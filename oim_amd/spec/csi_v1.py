@@ -125,6 +125,15 @@ MESSAGES = [
             Field("message", 2, "string"),
         ],
     ),
+    Message(
+        "GetCapacityRequest",
+        [Field("volume_capabilities", 1, "VolumeCapability", repeated=True)],
+        map_fields=[("parameters", 2, "string", "string")],
+    ),
+    Message(
+        "GetCapacityResponse",
+        [Field("available_capacity", 1, "int64")],
+    ),
     Message("ControllerGetCapabilitiesRequest", []),
     Message(
         "ControllerGetCapabilitiesResponse",
@@ -279,6 +288,7 @@ SERVICES = [
             ("DeleteVolume", "DeleteVolumeRequest", "DeleteVolumeResponse"),
             ("ValidateVolumeCapabilities", "ValidateVolumeCapabilitiesRequest",
              "ValidateVolumeCapabilitiesResponse"),
+            ("GetCapacity", "GetCapacityRequest", "GetCapacityResponse"),
             ("ControllerGetCapabilities", "ControllerGetCapabilitiesRequest",
              "ControllerGetCapabilitiesResponse"),
         ],
@@ -309,7 +319,6 @@ UNIMPLEMENTED_CONTROLLER_METHODS = (
     "ControllerPublishVolume",
     "ControllerUnpublishVolume",
     "ListVolumes",
-    "GetCapacity",
     "CreateSnapshot",
     "DeleteSnapshot",
     "ListSnapshots",

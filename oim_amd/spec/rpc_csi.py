@@ -52,6 +52,9 @@ class CSIControllerServicer:
     def ValidateVolumeCapabilities(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "ValidateVolumeCapabilities")
 
+    def GetCapacity(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "GetCapacity")
+
     def ControllerGetCapabilities(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "ControllerGetCapabilities")
 
@@ -97,6 +100,7 @@ def add_csi_controller_to_server(servicer, server):
         "ValidateVolumeCapabilities": _unary(
             servicer.ValidateVolumeCapabilities,
             csi.ValidateVolumeCapabilitiesRequest),
+        "GetCapacity": _unary(servicer.GetCapacity, csi.GetCapacityRequest),
         "ControllerGetCapabilities": _unary(
             servicer.ControllerGetCapabilities,
             csi.ControllerGetCapabilitiesRequest),
@@ -155,6 +159,7 @@ class CSIControllerStub:
         self.DeleteVolume = method("DeleteVolume", csi.DeleteVolumeResponse)
         self.ValidateVolumeCapabilities = method(
             "ValidateVolumeCapabilities", csi.ValidateVolumeCapabilitiesResponse)
+        self.GetCapacity = method("GetCapacity", csi.GetCapacityResponse)
         self.ControllerGetCapabilities = method(
             "ControllerGetCapabilities", csi.ControllerGetCapabilitiesResponse)
 

@@ -285,3 +285,38 @@ class TestDaemonHbm:
                 hipstore.delete_bdev(client, name)
         finally:
             fixture.stop()
+
+
+@needs_gpu
+class TestCapacityReporting:
+    def test_hbm_info(self):
+        total, free = hs.hbm_info(0)
+        assert total > 200 << 30  # 288 GB class
+        assert 0 < free <= total
+
+    def test_csi_get_capacity_local_mode(self, tmp_path):
+        import grpc
+
+        from oim_amd.common.server import grpc_target
+        from oim_amd.csidriver import FakeExec, LocalBackend, Mounter, OIMDriver
+        from oim_amd.spec import csi_v1 as csi
+        from oim_amd.spec.rpc_csi import CSIControllerStub
+
+        fixture = launch_hipstored(tmp_path, cpu=False)
+        driver = OIMDriver(driver_name="oim-local", node_id="n0",
+                           endpoint=f"unix://{tmp_path}/csi.sock",
+                           backend=LocalBackend(fixture.socket_path),
+                           mounter=Mounter(FakeExec()))
+        driver.start()
+        try:
+            with grpc.insecure_channel(grpc_target(driver.addr())) as ch:
+                stub = CSIControllerStub(ch)
+                response = stub.GetCapacity(csi.GetCapacityRequest(), timeout=30)
+                assert response.available_capacity > 100 << 30
+                caps = stub.ControllerGetCapabilities(
+                    csi.ControllerGetCapabilitiesRequest(), timeout=30)
+                types = {c.rpc.type for c in caps.capabilities}
+                assert csi.CTRL_CAP_GET_CAPACITY in types
+        finally:
+            driver.stop()
+            fixture.stop()

@@ -201,3 +201,30 @@ class TestNodeGetVolumeStats:
                 assert excinfo.value.code() == grpc.StatusCode.NOT_FOUND
         finally:
             driver.stop()
+
+
+class TestGetCapacity:
+    def test_local_mode_cpu_unimplemented(self, hipstored, tmp_path):  # noqa: F811
+        """CPU-mode daemon has no HBM: GetCapacity reports Unimplemented
+        (on a GPU box the local backend reports free HBM bytes)."""
+        backend = LocalBackend(hipstored.socket_path)
+        assert backend.get_capacity() is None  # cpu daemon: no HBM
+        driver = OIMDriver(driver_name="oim-local", node_id="n0",
+                           endpoint=f"unix://{tmp_path}/csi3.sock",
+                           backend=backend, mounter=Mounter(FakeExec()))
+        driver.start()
+        try:
+            with grpc.insecure_channel(grpc_target(driver.addr())) as ch:
+                stub = CSIControllerStub(ch)
+                with pytest.raises(grpc.RpcError) as excinfo:
+                    stub.GetCapacity(csi.GetCapacityRequest(), timeout=30)
+                assert excinfo.value.code() == grpc.StatusCode.UNIMPLEMENTED
+        finally:
+            driver.stop()
+
+    def test_hbm_info_rpc(self, hipstored):  # noqa: F811
+        from oim_amd import hipstore as hsclient
+
+        with hsclient.Client(hipstored.socket_path) as client:
+            info = client.invoke("get_hbm_info")
+            assert "total_bytes" in info and "free_bytes" in info
