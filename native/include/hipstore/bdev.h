@@ -96,6 +96,34 @@ class Bdev {
   // Drive completions; returns number completed.
   virtual int poll(IoChannel* ch) = 0;
 
+  // Cumulative I/O statistics (updated at submit; SPDK's
+  // get_bdevs_iostat shape).
+  struct IoStat {
+    std::atomic<uint64_t> num_read_ops{0};
+    std::atomic<uint64_t> num_write_ops{0};
+    std::atomic<uint64_t> num_unmap_ops{0};
+    std::atomic<uint64_t> bytes_read{0};
+    std::atomic<uint64_t> bytes_written{0};
+  };
+  IoStat& iostat() { return iostat_; }
+  void account(const IoRequest& req) {
+    switch (req.op) {
+      case IoOp::kRead:
+        iostat_.num_read_ops.fetch_add(1, std::memory_order_relaxed);
+        iostat_.bytes_read.fetch_add(req.length, std::memory_order_relaxed);
+        break;
+      case IoOp::kWrite:
+        iostat_.num_write_ops.fetch_add(1, std::memory_order_relaxed);
+        iostat_.bytes_written.fetch_add(req.length, std::memory_order_relaxed);
+        break;
+      case IoOp::kFill:
+        iostat_.num_unmap_ops.fetch_add(1, std::memory_order_relaxed);
+        break;
+      case IoOp::kFlush:
+        break;
+    }
+  }
+
   bool check_bounds(const IoRequest& req) const {
     return req.length > 0 && req.offset % block_size_ == 0 &&
            req.length % block_size_ == 0 &&
@@ -111,6 +139,7 @@ class Bdev {
   uint64_t block_size_;
   uint64_t num_blocks_;
   std::atomic<bool> claimed_{false};
+  IoStat iostat_;
 };
 
 using BdevPtr = std::shared_ptr<Bdev>;

@@ -93,6 +93,7 @@ class MallocBdev : public Bdev {
     if (!check_bounds(req)) {
       status = kIoInvalid;
     } else {
+      account(req);
       std::lock_guard<std::mutex> lock(mutex_);
       switch (req.op) {
         case IoOp::kRead:

@@ -116,6 +116,7 @@ class StripedBdev : public Bdev {
       if (req.on_complete) req.on_complete(kIoInvalid);
       return;
     }
+    account(req);
     // Split [offset, offset+length) at stripe boundaries; unit u maps
     // to child u % N at child offset (u / N) * stripe + intra.
     const uint64_t n = children_.size();
@@ -291,6 +292,7 @@ class ReplicatedBdev : public Bdev {
       if (req.on_complete) req.on_complete(kIoInvalid);
       return;
     }
+    account(req);
     const size_t n = children_.size();
     if (req.op == IoOp::kRead) {
       // Spread reads over replicas by stripe-ish hashing.

@@ -1321,6 +1321,7 @@ class HbmBdev : public Bdev {
         status = kIoInvalid;
       }
     }
+    if (status == kIoOk) account(req);
     if (persistent_ && shared_) {
       static_cast<HbmSharedChannel*>(ch)->enqueue(std::move(req), status);
     } else if (persistent_) {

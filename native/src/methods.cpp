@@ -482,6 +482,33 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         return Json(name);
       });
 
+  server->register_method("get_bdevs_iostat", [&manager](const Json& p) {
+    // SPDK-shaped per-bdev I/O counters.
+    const std::string name = p.get_string("name");
+    JsonArray bdevs;
+    auto one = [](const BdevPtr& bdev) {
+      auto& st = bdev->iostat();
+      JsonObject o;
+      o["name"] = Json(bdev->name());
+      o["num_read_ops"] = Json(static_cast<int64_t>(st.num_read_ops.load()));
+      o["num_write_ops"] = Json(static_cast<int64_t>(st.num_write_ops.load()));
+      o["num_unmap_ops"] = Json(static_cast<int64_t>(st.num_unmap_ops.load()));
+      o["bytes_read"] = Json(static_cast<int64_t>(st.bytes_read.load()));
+      o["bytes_written"] = Json(static_cast<int64_t>(st.bytes_written.load()));
+      return Json(std::move(o));
+    };
+    if (!name.empty()) {
+      BdevPtr bdev = manager.find(name);
+      if (!bdev) not_found("bdev " + name);
+      bdevs.push_back(one(bdev));
+    } else {
+      for (const auto& bdev : manager.list()) bdevs.push_back(one(bdev));
+    }
+    JsonObject o;
+    o["bdevs"] = Json(std::move(bdevs));
+    return Json(std::move(o));
+  });
+
   server->register_method("get_hbm_info", [device](const Json& p) {
     // Capacity source for CSI GetCapacity: HBM totals on GPU, host
     // RAM totals in CPU mode (sysinfo-free approximation via

@@ -252,6 +252,7 @@ class NvmfBdev : public Bdev {
         return;
       }
     }
+    account(req);
     Sqe sqe{};
     sqe.set_nsid(nsid_);
     const uint64_t slba = req.offset / block_size();

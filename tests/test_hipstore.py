@@ -124,3 +124,23 @@ class TestProtocol:
             hipstore.construct_malloc_bdev(a, 1024, 512, name="pipe")
             assert hipstore.get_bdevs(b, "pipe")[0].name == "pipe"
             hipstore.delete_bdev(b, "pipe")
+
+
+class TestIostat:
+    def test_counters_accumulate(self, hipstored):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            hipstore.construct_malloc_bdev(client, 2048, 512, name="stat0")
+            hipstore.perf_run(client, "stat0", workload="randread",
+                              io_size=4096, queue_depth=4, num_queues=1,
+                              seconds=0.1)
+            stats = client.invoke("get_bdevs_iostat", {"name": "stat0"})
+            entry = stats["bdevs"][0]
+            assert entry["num_read_ops"] > 0
+            assert entry["bytes_read"] == entry["num_read_ops"] * 4096
+            assert entry["num_write_ops"] == 0
+            hipstore.delete_bdev(client, "stat0")
+
+    def test_missing_bdev(self, hipstored):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            with pytest.raises(RpcError):
+                client.invoke("get_bdevs_iostat", {"name": "ghost"})
