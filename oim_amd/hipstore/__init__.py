@@ -26,6 +26,8 @@ from .api import (
     remove_vhost_controller,
     get_vhost_controllers,
     perf_run,
+    BDevIostat,
+    get_bdevs_iostat,
 )
 
 __all__ = [
@@ -51,4 +53,6 @@ __all__ = [
     "remove_vhost_controller",
     "get_vhost_controllers",
     "perf_run",
+    "BDevIostat",
+    "get_bdevs_iostat",
 ]

@@ -181,6 +181,32 @@ def get_vhost_controllers(client: Client) -> List[VHostController]:
     return controllers
 
 
+@dataclass
+class BDevIostat:
+    name: str
+    num_read_ops: int
+    num_write_ops: int
+    num_unmap_ops: int
+    bytes_read: int
+    bytes_written: int
+
+
+def get_bdevs_iostat(client: Client, name: str = "") -> List[BDevIostat]:
+    params = {"name": name} if name else {}
+    result = client.invoke("get_bdevs_iostat", params)
+    return [
+        BDevIostat(
+            name=b["name"],
+            num_read_ops=b["num_read_ops"],
+            num_write_ops=b["num_write_ops"],
+            num_unmap_ops=b["num_unmap_ops"],
+            bytes_read=b["bytes_read"],
+            bytes_written=b["bytes_written"],
+        )
+        for b in result["bdevs"]
+    ]
+
+
 def perf_run(
     client: Client,
     bdev_name: str,

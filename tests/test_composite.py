@@ -110,3 +110,21 @@ class TestCompositeRpc:
             assert bdevs[0].product_name == "Replicated Malloc disk"
             assert bdevs[0].num_blocks == 2048
             hipstore.delete_bdev(client, "repl0")
+
+
+class TestPerfEdgeCases:
+    def test_session_zero_and_tiny_steps(self):
+        b = hs.create_malloc_bdev(f"edge-{random.random()}", 512, 8192)
+        s = hs.PerfSession(b, "randread", 4096, 8, 4)
+        r = s.step(0)
+        assert r["io_count"] == 0
+        r = s.step(3)  # fewer IOs than queues
+        assert 0 < r["io_count"] <= 4 * 3
+        r = s.step(1000)
+        assert r["io_count"] >= 1000
+
+    def test_bdevperf_max_ios_cap(self):
+        b = hs.create_malloc_bdev(f"edge2-{random.random()}", 512, 8192)
+        r = hs.run_bdevperf(b, "randread", 4096, 8, 2, 60.0, max_ios=500)
+        assert 500 <= r["io_count"] <= 600  # cap honored, not the 60s
+        assert r["seconds"] < 10
