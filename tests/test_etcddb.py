@@ -71,3 +71,57 @@ class TestEtcdDB:
         assert registry_b.db.lookup(["gpu-3", "address"]) == "tcp://card3:8999"
         db_a.close()
         db_b.close()
+
+
+class TestHaControlPlane:
+    def test_two_registries_one_etcd_full_path(self, fake_etcd, tmp_path):
+        """Config-5 control plane: a controller registers through
+        registry A; a client provisions through registry B's proxy
+        (both share one etcd)."""
+        import grpc
+
+        from oim_amd import spec
+        from oim_amd.common.server import grpc_target
+        from oim_amd.controller import Controller, ControllerServer
+        from fixtures import launch_hipstored
+
+        endpoint, _ = fake_etcd
+        reg_a = Registry(db=EtcdRegistryDB([endpoint]))
+        server_a = RegistryServer(f"unix://{tmp_path}/rega.sock", reg_a)
+        server_a.start()
+        reg_b = Registry(db=EtcdRegistryDB([endpoint]))
+        server_b = RegistryServer(f"unix://{tmp_path}/regb.sock", reg_b)
+        server_b.start()
+        daemon = launch_hipstored(tmp_path, cpu=True)
+        controller = Controller(
+            controller_id="gpu-7",
+            hipstored_socket=daemon.socket_path,
+            controller_address=f"unix://{tmp_path}/ctrl.sock",
+            registry_address=server_a.addr(),  # registers via A
+            registry_delay=3600.0,
+        )
+        ctrl_server = ControllerServer(f"unix://{tmp_path}/ctrl.sock",
+                                       controller)
+        ctrl_server.start()
+        try:
+            controller.register()
+            # B sees the registration through etcd and proxies to it.
+            with grpc.insecure_channel(grpc_target(server_b.addr())) as ch:
+                stub = spec.ControllerStub(ch)
+                metadata = ((spec.CONTROLLER_ID_KEY, "gpu-7"),)
+                stub.ProvisionMallocBDev(
+                    spec.ProvisionMallocBDevRequest(bdev_name="ha-vol",
+                                                    size=1 << 20),
+                    metadata=metadata, timeout=30)
+                stub.CheckMallocBDev(
+                    spec.CheckMallocBDevRequest(bdev_name="ha-vol"),
+                    metadata=metadata, timeout=30)
+                stub.ProvisionMallocBDev(
+                    spec.ProvisionMallocBDevRequest(bdev_name="ha-vol",
+                                                    size=0),
+                    metadata=metadata, timeout=30)
+        finally:
+            ctrl_server.stop()
+            daemon.stop()
+            server_b.stop()
+            server_a.stop()
