@@ -818,11 +818,17 @@ class HbmPersistentChannel : public IoChannel {
 
   int poll() {
     int completed = 0;
-    for (auto& [cb, status] : immediate_) {
-      if (cb) cb(status);
-      ++completed;
+    // Swap out the immediate list before firing: a completion callback
+    // may resubmit, and a failed resubmission appends to immediate_
+    // again — mutating the vector mid-iteration.
+    if (!immediate_.empty()) {
+      auto batch = std::move(immediate_);
+      immediate_.clear();
+      for (auto& [cb, status] : batch) {
+        if (cb) cb(status);
+        ++completed;
+      }
     }
-    immediate_.clear();
     // Retire the contiguous completed prefix of the CQ.
     while (completed_ < tail_) {
       unsigned long long seq = __atomic_load_n(
@@ -1165,11 +1171,17 @@ class HbmSharedChannel : public IoChannel {
 
   int poll() {
     int completed = 0;
-    for (auto& [cb, status] : immediate_) {
-      if (cb) cb(status);
-      ++completed;
+    // Swap out the immediate list before firing: a completion callback
+    // may resubmit, and a failed resubmission appends to immediate_
+    // again — mutating the vector mid-iteration.
+    if (!immediate_.empty()) {
+      auto batch = std::move(immediate_);
+      immediate_.clear();
+      for (auto& [cb, status] : batch) {
+        if (cb) cb(status);
+        ++completed;
+      }
     }
-    immediate_.clear();
     while (completed_ < tail_) {
       unsigned long long seq = __atomic_load_n(
           const_cast<const unsigned long long*>(&cq_[completed_ % kRing]),

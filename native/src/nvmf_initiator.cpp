@@ -288,11 +288,15 @@ class NvmfBdev : public Bdev {
   int poll(IoChannel* ch) override {
     auto* channel = static_cast<NvmfChannel*>(ch);
     int completed = 0;
-    for (auto& [cb, status] : channel->immediate) {
-      if (cb) cb(status);
-      ++completed;
+    if (!channel->immediate.empty()) {
+      // Swapped out first: callbacks may resubmit and append again.
+      auto batch = std::move(channel->immediate);
+      channel->immediate.clear();
+      for (auto& [cb, status] : batch) {
+        if (cb) cb(status);
+        ++completed;
+      }
     }
-    channel->immediate.clear();
     // Drain the socket.
     char chunk[65536];
     while (true) {

@@ -320,3 +320,15 @@ class TestCapacityReporting:
         finally:
             driver.stop()
             fixture.stop()
+
+
+@needs_gpu
+class TestInvalidWorkload:
+    def test_unaligned_io_size_fails_cleanly(self):
+        """io_size below the block size must error out, not hang
+        (regression: failed resubmissions mutated the immediate list
+        mid-iteration)."""
+        bdev = hs.create_hbm_bdev("inv-gpu", 4096, 65536, device=0,
+                                  persistent=True)
+        with pytest.raises(RuntimeError):
+            hs.run_bdevperf(bdev, "randread", 512, 8, 2, 0.5)
