@@ -1,8 +1,13 @@
-"""oimctl: admin CLI for the registry (reference cmd/oimctl/main.go).
+"""oimctl: admin CLI for the registry (reference cmd/oimctl/main.go,
+extended with proxied volume operations).
 
   oimctl --registry tcp://reg:8999 --ca ca.crt --key user.admin.key \
       set host-0/pci 0000:c1:00.0
   oimctl ... get [prefix]
+  oimctl ... provision --controller gpu-0 vol1 64MiB
+  oimctl ... map --controller gpu-0 vol1
+  oimctl ... unmap --controller gpu-0 vol1
+  oimctl ... check --controller gpu-0 vol1
 """
 
 import argparse
@@ -30,6 +35,18 @@ def main(argv=None) -> int:
     get_cmd.add_argument("prefix", nargs="?", default="")
     delete_cmd = sub.add_parser("delete", help="delete a registry value")
     delete_cmd.add_argument("path")
+    for name, help_text in (
+        ("provision", "create (or with size 0 delete) a malloc bdev"),
+        ("map", "map a provisioned volume to a SCSI target"),
+        ("unmap", "unmap a volume"),
+        ("check", "check that a malloc bdev exists"),
+    ):
+        cmd = sub.add_parser(name, help=help_text + " via the registry proxy")
+        cmd.add_argument("--controller", required=True,
+                         help="controller ID to proxy to")
+        cmd.add_argument("volume")
+        if name == "provision":
+            cmd.add_argument("size", help='bytes, or "64MiB"-style; 0 deletes')
     args = parser.parse_args(argv)
     log.init_from_args(args)
 
@@ -48,12 +65,55 @@ def main(argv=None) -> int:
         elif args.command == "delete":
             stub.SetValue(spec.SetValueRequest(
                 value=spec.Value(path=args.path, value="")), timeout=30)
-        else:
+        elif args.command == "get":
             reply = stub.GetValues(
                 spec.GetValuesRequest(path=args.prefix), timeout=30)
             for value in reply.values:
                 print(f"{value.path}: {value.value}")
+        else:
+            controller = spec.ControllerStub(channel)
+            metadata = ((spec.CONTROLLER_ID_KEY, args.controller),)
+            if args.command == "provision":
+                stub_size = parse_size(args.size)
+                controller.ProvisionMallocBDev(
+                    spec.ProvisionMallocBDevRequest(
+                        bdev_name=args.volume, size=stub_size),
+                    metadata=metadata, timeout=60)
+                print(f"provisioned {args.volume} ({stub_size} bytes)"
+                      if stub_size else f"deleted {args.volume}")
+            elif args.command == "map":
+                reply = controller.MapVolume(
+                    spec.MapVolumeRequest(volume_id=args.volume,
+                                          malloc=spec.MallocParams()),
+                    metadata=metadata, timeout=60)
+                pci = reply.pci_address
+                print(f"mapped {args.volume} at "
+                      f"{pci.domain:04x}:{pci.bus:02x}:{pci.device:02x}."
+                      f"{pci.function} target {reply.scsi_disk.target} "
+                      f"lun {reply.scsi_disk.lun}")
+            elif args.command == "unmap":
+                controller.UnmapVolume(
+                    spec.UnmapVolumeRequest(volume_id=args.volume),
+                    metadata=metadata, timeout=60)
+                print(f"unmapped {args.volume}")
+            elif args.command == "check":
+                controller.CheckMallocBDev(
+                    spec.CheckMallocBDevRequest(bdev_name=args.volume),
+                    metadata=metadata, timeout=60)
+                print(f"{args.volume} exists")
     return 0
+
+
+_UNITS = {"": 1, "k": 1 << 10, "m": 1 << 20, "g": 1 << 30, "t": 1 << 40}
+
+
+def parse_size(text: str) -> int:
+    """"64MiB" / "1G" / "4096" -> bytes."""
+    t = text.strip().lower().removesuffix("ib").removesuffix("b")
+    for suffix, mult in _UNITS.items():
+        if suffix and t.endswith(suffix):
+            return int(float(t[: -len(suffix)]) * mult)
+    return int(t)
 
 
 if __name__ == "__main__":

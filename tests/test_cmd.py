@@ -53,3 +53,43 @@ class TestMainModules:
             oim_csi_driver.main([
                 "--nodeid", "n1", "--hipstored-socket", "/x",
                 "--oim-registry-address", "tcp://y:1"])  # both modes
+
+
+class TestOimctlVolumes:
+    def test_provision_map_unmap_via_proxy(self, tmp_path, capsys):
+        import fixtures
+        from oim_amd.controller import Controller, ControllerServer
+
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        registry = Registry(db=MemRegistryDB())
+        reg_server = RegistryServer(f"unix://{tmp_path}/r.sock", registry)
+        reg_server.start()
+        controller = Controller(controller_id="c9",
+                                hipstored_socket=daemon.socket_path)
+        ctrl_server = ControllerServer(f"unix://{tmp_path}/c.sock", controller)
+        ctrl_server.start()
+        registry.db.store(["c9", "address"], f"unix://{tmp_path}/c.sock")
+        endpoint = reg_server.addr()
+        try:
+            assert oimctl.main(["--registry", endpoint, "provision",
+                                "--controller", "c9", "volx", "64MiB"]) == 0
+            assert oimctl.main(["--registry", endpoint, "check",
+                                "--controller", "c9", "volx"]) == 0
+            assert oimctl.main(["--registry", endpoint, "map",
+                                "--controller", "c9", "volx"]) == 0
+            out = capsys.readouterr().out
+            assert "target 0 lun 0" in out
+            assert oimctl.main(["--registry", endpoint, "unmap",
+                                "--controller", "c9", "volx"]) == 0
+            assert oimctl.main(["--registry", endpoint, "provision",
+                                "--controller", "c9", "volx", "0"]) == 0
+        finally:
+            ctrl_server.stop()
+            reg_server.stop()
+            daemon.stop()
+
+    def test_parse_size(self):
+        assert oimctl.parse_size("4096") == 4096
+        assert oimctl.parse_size("64MiB") == 64 << 20
+        assert oimctl.parse_size("1G") == 1 << 30
+        assert oimctl.parse_size("2k") == 2048
