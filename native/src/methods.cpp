@@ -524,6 +524,47 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
     return Json(std::move(o));
   });
 
+  server->register_method(
+      "bdev_clone", [&manager, use_hbm, device, persistent](const Json& p) {
+        // Volume clone: new malloc bdev + device-side range copy (HBM
+        // rates; xGMI when cloning to another GPU via "device").
+        BdevPtr src = manager.find(p.get_string("src"));
+        if (!src) not_found("bdev " + p.get_string("src"));
+        const std::string name = p.get_string("name");
+        if (name.empty()) throw RpcError{kInvalidParams, "name required"};
+        if (manager.find(name)) {
+          throw RpcError{kInvalidParams, "bdev " + name + " already exists"};
+        }
+        BdevPtr clone;
+        if (use_hbm && gpu_available() && src->device_base() != nullptr) {
+          clone = create_hbm_bdev(
+              name, src->block_size(), src->num_blocks(),
+              static_cast<int>(p.get_int("device", src->gpu_device())),
+              p.get_bool("persistent", persistent));
+          int status = hbm_copy_sync(src.get(), 0, clone.get(), 0,
+                                     src->size_bytes());
+          if (status != kIoOk) {
+            throw RpcError{kInternalError, "clone copy failed"};
+          }
+        } else {
+          clone = create_malloc_bdev(name, src->block_size(),
+                                     src->num_blocks());
+          // Host-side copy for CPU bdevs (CI path).
+          const uint64_t chunk = 8 << 20;
+          std::vector<uint8_t> buf(chunk);
+          for (uint64_t off = 0; off < src->size_bytes(); off += chunk) {
+            const uint64_t n = std::min(chunk, src->size_bytes() - off);
+            if (bdev_read_sync(src.get(), off, buf.data(), n) != kIoOk ||
+                bdev_write_sync(clone.get(), off, buf.data(), n) != kIoOk) {
+              throw RpcError{kInternalError, "clone copy failed"};
+            }
+          }
+        }
+        clone->set_product(src->product_name());
+        manager.add(clone);
+        return Json(name);
+      });
+
   server->register_method("bdev_copy", [&manager](const Json& p) {
     // Device-side clone/rebuild: HBM->HBM at memory/xGMI rates.
     BdevPtr src = manager.find(p.get_string("src"));
