@@ -576,6 +576,16 @@ T* device_view(T* host_ptr) {
   return reinterpret_cast<T*>(dev);
 }
 
+// Engine channels tag their kind so HbmBdev can dispatch without
+// RTTI in the hot path (channels of different kinds coexist on one
+// bdev when the auto-fallback kicks in).
+class HbmChannelBase : public IoChannel {
+ public:
+  enum class Kind { kBatched, kPersistent, kShared };
+  explicit HbmChannelBase(Kind kind) : kind(kind) {}
+  const Kind kind;
+};
+
 struct PendingIo {
   IoRequest req;
   int status;
@@ -587,9 +597,10 @@ struct InflightBatch {
   std::vector<std::pair<IoCompletion, int>> completions;
 };
 
-class HbmChannel : public IoChannel {
+class HbmChannel : public HbmChannelBase {
  public:
-  HbmChannel(int device, uint8_t* base, uint64_t size) : base_(base), size_(size) {
+  HbmChannel(int device, uint8_t* base, uint64_t size)
+      : HbmChannelBase(Kind::kBatched), base_(base), size_(size) {
     HIP_CHECK(hipSetDevice(device));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&ring_),
@@ -745,7 +756,14 @@ class HbmChannel : public IoChannel {
 };
 
 // Host side of the persistent service kernel. One instance per channel.
-class HbmPersistentChannel : public IoChannel {
+// Live per-queue service kernels per device: beyond this, new
+// persistent channels fall back to the shared service (each per-queue
+// kernel needs a real hardware queue; GPU_MAX_HW_QUEUES=24 minus
+// headroom for transient/utility streams).
+inline std::atomic<int> g_per_queue_channels[64];
+constexpr int kPerQueueChannelCap = 18;
+
+class HbmPersistentChannel : public HbmChannelBase {
  public:
   static constexpr uint32_t kRing = 32768;      // descriptors (>= 2x max request tiles)
   static constexpr uint32_t kIdleSpins = 500000;  // ~1 s of s_sleep polling
@@ -761,7 +779,9 @@ class HbmPersistentChannel : public IoChannel {
     return w;
   }
 
-  HbmPersistentChannel(int device, uint8_t* base) : base_(base), device_(device) {
+  HbmPersistentChannel(int device, uint8_t* base)
+      : HbmChannelBase(Kind::kPersistent), base_(base), device_(device) {
+    g_per_queue_channels[device].fetch_add(1, std::memory_order_relaxed);
     HIP_CHECK(hipSetDevice(device));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&sq_),
@@ -787,6 +807,7 @@ class HbmPersistentChannel : public IoChannel {
   }
 
   ~HbmPersistentChannel() override {
+    g_per_queue_channels[device_].fetch_sub(1, std::memory_order_relaxed);
     __atomic_store_n(const_cast<uint32_t*>(stop_), 1u, __ATOMIC_RELEASE);
     (void)hipStreamSynchronize(stream_);
     (void)hipHostFree(sq_);
@@ -1109,11 +1130,12 @@ class SharedService {
 // Channel whose ring is serviced by the per-device shared kernel.
 // Ring/completion bookkeeping mirrors HbmPersistentChannel; the
 // difference is who runs the GPU side.
-class HbmSharedChannel : public IoChannel {
+class HbmSharedChannel : public HbmChannelBase {
  public:
   static constexpr uint32_t kRing = 32768;
 
-  HbmSharedChannel(int device, uint8_t* base) : base_(base) {
+  HbmSharedChannel(int device, uint8_t* base)
+      : HbmChannelBase(Kind::kShared), base_(base) {
     service_ = SharedService::instance(device);
     HIP_CHECK(hipSetDevice(device));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&sq_),
@@ -1317,7 +1339,15 @@ class HbmBdev : public Bdev {
   std::shared_ptr<IoChannel> get_channel() override {
     HIP_CHECK(hipSetDevice(device_));
     if (persistent_) {
-      if (shared_) return std::make_shared<HbmSharedChannel>(device_, base_);
+      // Auto-fallback: per-queue service kernels give the best
+      // latency but each needs a hardware queue; past the cap, new
+      // channels multiplex through the shared per-device kernel.
+      const bool over_cap =
+          g_per_queue_channels[device_ & 63].load(
+              std::memory_order_relaxed) >= kPerQueueChannelCap;
+      if (shared_ || over_cap) {
+        return std::make_shared<HbmSharedChannel>(device_, base_);
+      }
       return std::make_shared<HbmPersistentChannel>(device_, base_);
     }
     return std::make_shared<HbmChannel>(device_, base_, size_bytes());
@@ -1334,12 +1364,18 @@ class HbmBdev : public Bdev {
       }
     }
     if (status == kIoOk) account(req);
-    if (persistent_ && shared_) {
-      static_cast<HbmSharedChannel*>(ch)->enqueue(std::move(req), status);
-    } else if (persistent_) {
-      static_cast<HbmPersistentChannel*>(ch)->enqueue(std::move(req), status);
-    } else {
-      static_cast<HbmChannel*>(ch)->enqueue(std::move(req), status);
+    auto* channel = static_cast<HbmChannelBase*>(ch);
+    switch (channel->kind) {
+      case HbmChannelBase::Kind::kShared:
+        static_cast<HbmSharedChannel*>(channel)->enqueue(std::move(req), status);
+        break;
+      case HbmChannelBase::Kind::kPersistent:
+        static_cast<HbmPersistentChannel*>(channel)->enqueue(std::move(req),
+                                                             status);
+        break;
+      case HbmChannelBase::Kind::kBatched:
+        static_cast<HbmChannel*>(channel)->enqueue(std::move(req), status);
+        break;
     }
   }
 
