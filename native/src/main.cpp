@@ -17,6 +17,7 @@
 #include "hipstore/engine.h"
 #include "hipstore/nbd.h"
 #include "hipstore/rpc.h"
+#include "hipstore/vhost.h"
 
 namespace {
 
@@ -39,7 +40,8 @@ void usage(const char* argv0) {
           "  -d N      HIP device index backing malloc bdevs (default 0)\n"
           "  -C        CPU mode: host-RAM bdevs even when a GPU is present\n"
           "  -P        persistent engine: on-GPU polling service kernels (default)\n"
-          "  -B        batched engine: per-poll kernel launches\n",
+          "  -B        batched engine: per-poll kernel launches\n"
+          "  -V DIR    vhost-user socket directory (default: RPC socket dir)\n",
           argv0);
 }
 
@@ -52,15 +54,17 @@ int main(int argc, char** argv) {
   setenv("GPU_MAX_HW_QUEUES", "24", /*overwrite=*/0);
 
   std::string socket_path = "/var/tmp/hipstored.sock";
+  std::string vhost_dir;
   int device = 0;
   bool cpu_only = false;
   bool persistent = true;  // the measured-better engine is the default
 
   int opt;
-  while ((opt = getopt(argc, argv, "S:d:CPBh")) != -1) {
+  while ((opt = getopt(argc, argv, "S:d:V:CPBh")) != -1) {
     switch (opt) {
       case 'S': socket_path = optarg; break;
       case 'd': device = atoi(optarg); break;
+      case 'V': vhost_dir = optarg; break;
       case 'C': cpu_only = true; break;
       case 'P': persistent = true; break;
       case 'B': persistent = false; break;
@@ -68,6 +72,11 @@ int main(int argc, char** argv) {
       default: usage(argv[0]); return 2;
     }
   }
+  if (vhost_dir.empty()) {
+    auto slash = socket_path.rfind('/');
+    vhost_dir = slash == std::string::npos ? "." : socket_path.substr(0, slash);
+  }
+  hipstore::vhost_set_socket_dir(vhost_dir);
 
   const bool use_hbm = !cpu_only && hipstore::gpu_available();
   fprintf(stderr, "hipstored: socket=%s mode=%s device=%d gpus=%d\n",
@@ -92,6 +101,7 @@ int main(int argc, char** argv) {
   }
   fprintf(stderr, "hipstored: shutting down\n");
   hipstore::nbd_stop_all();
+  hipstore::vhost_stop_all();
   server.stop();
   return 0;
 }

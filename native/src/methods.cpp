@@ -16,6 +16,7 @@
 #include "hipstore/nbd.h"
 #include "hipstore/nvmf.h"
 #include "hipstore/rpc.h"
+#include "hipstore/vhost.h"
 
 namespace hipstore {
 
@@ -37,6 +38,7 @@ struct ScsiTarget {
 struct VhostController {
   std::string cpumask;
   ScsiTarget targets[kMaxScsiTargets];
+  VhostDevPtr dev;  // live vhost-user-scsi server for this controller
 };
 
 struct VhostState {
@@ -215,7 +217,27 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
     if (state.controllers.count(name)) {
       throw RpcError{kInvalidParams, "controller " + name + " already exists"};
     }
-    state.controllers[name].cpumask = p.get_string("cpumask", "0x1");
+    // Serve the vhost-user socket immediately (SPDK creates it at
+    // construct time too); the resolver re-reads the target table per
+    // command so add/remove_vhost_scsi_lun is visible without restart.
+    VhostDevPtr dev;
+    try {
+      dev = vhost_start(name, [name](int target) -> BdevPtr {
+        auto& st = vhost_state();
+        std::lock_guard<std::mutex> l(st.mutex);
+        auto it = st.controllers.find(name);
+        if (it == st.controllers.end()) return nullptr;
+        if (target < 0 || target >= kMaxScsiTargets) return nullptr;
+        const ScsiTarget& t = it->second.targets[target];
+        if (!t.used || t.luns.empty()) return nullptr;
+        return BdevManager::instance().find(t.luns[0].bdev_name);
+      });
+    } catch (const std::exception& e) {
+      throw RpcError{kInvalidParams, e.what()};
+    }
+    VhostController& ctrl = state.controllers[name];
+    ctrl.cpumask = p.get_string("cpumask", "0x1");
+    ctrl.dev = dev;
     return Json(JsonObject{});
   });
 
@@ -267,15 +289,22 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   server->register_method("remove_vhost_controller", [&manager](const Json& p) {
     const std::string ctrlr = p.get_string("ctrlr");
     auto& state = vhost_state();
-    std::lock_guard<std::mutex> lock(state.mutex);
-    auto it = state.controllers.find(ctrlr);
-    if (it == state.controllers.end()) not_found("controller " + ctrlr);
-    for (ScsiTarget& target : it->second.targets) {
-      for (const ScsiLun& lun : target.luns) {
-        if (BdevPtr bdev = manager.find(lun.bdev_name)) bdev->release();
+    VhostDevPtr dev;
+    {
+      std::lock_guard<std::mutex> lock(state.mutex);
+      auto it = state.controllers.find(ctrlr);
+      if (it == state.controllers.end()) not_found("controller " + ctrlr);
+      for (ScsiTarget& target : it->second.targets) {
+        for (const ScsiLun& lun : target.luns) {
+          if (BdevPtr bdev = manager.find(lun.bdev_name)) bdev->release();
+        }
       }
+      dev = it->second.dev;
+      state.controllers.erase(it);
     }
-    state.controllers.erase(it);
+    // Outside the lock: ring workers may be blocked in the resolver
+    // (which takes state.mutex) and stop() joins them.
+    vhost_stop(dev);
     return Json(JsonObject{});
   });
 

@@ -1,0 +1,260 @@
+"""vhost-user-scsi target conformance (native/src/vhost.cpp).
+
+An in-process vhost-user master (tests/vhost_client.py) plays QEMU:
+shared guest memory over memfd, virtio split rings, eventfd kick/call.
+Covers the negotiation sequence, the SCSI command set a Linux guest's
+sd driver issues at probe time, data-path reads/writes, error/sense
+paths, indirect descriptors, hot add/remove through the RPC plane, and
+ring stop/teardown. (Reference parity: SPDK vhost_scsi served this
+role behind construct_vhost_scsi_controller, reference
+pkg/oim-controller/controller.go:166-231.)
+"""
+
+import os
+import struct
+import time
+
+import pytest
+
+from oim_amd import hipstore
+
+from fixtures import hipstored  # noqa: F401
+from vhost_client import (FEAT_INDIRECT, VhostUserMaster, GET_QUEUE_NUM)
+
+BLOCK = 512
+NUM_BLOCKS = 8192  # 4 MiB
+
+
+@pytest.fixture
+def vhost_target(hipstored, tmp_path):  # noqa: F811
+    """Daemon + bdev + vhost controller with the bdev at SCSI target 0."""
+    with hipstore.Client(hipstored.socket_path) as client:
+        hipstore.construct_malloc_bdev(
+            client, num_blocks=NUM_BLOCKS, block_size=BLOCK, name="vhb0")
+        client.invoke("construct_vhost_scsi_controller", {"ctrlr": "vh0"})
+        client.invoke("add_vhost_scsi_lun",
+                      {"ctrlr": "vh0", "scsi_target_num": 0,
+                       "bdev_name": "vhb0"})
+        socket_path = os.path.join(os.path.dirname(hipstored.socket_path),
+                                   "vh0")
+        assert os.path.exists(socket_path), "vhost socket not created"
+        master = VhostUserMaster(socket_path)
+        master.negotiate()
+        yield client, master
+        master.close()
+
+
+class TestNegotiation:
+    def test_queue_num_and_features(self, vhost_target):
+        _, master = vhost_target
+        n, = struct.unpack("<Q", master.query(GET_QUEUE_NUM))
+        assert n >= 4  # controlq + eventq + >=2 request queues
+
+    def test_socket_lifecycle(self, hipstored, tmp_path):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            client.invoke("construct_vhost_scsi_controller", {"ctrlr": "vhx"})
+            path = os.path.join(os.path.dirname(hipstored.socket_path), "vhx")
+            assert os.path.exists(path)
+            with pytest.raises(hipstore.RpcError):
+                client.invoke("construct_vhost_scsi_controller",
+                              {"ctrlr": "vhx"})
+            client.invoke("remove_vhost_controller", {"ctrlr": "vhx"})
+            assert not os.path.exists(path)
+
+
+class TestScsiProbe:
+    def test_inquiry(self, vhost_target):
+        _, master = vhost_target
+        result = master.inquiry(0)
+        assert result.status == 0 and result.response == 0
+        assert result.data[0] == 0x00  # direct-access block device
+        assert result.data[8:16] == b"HIPSTORE"
+        assert result.data[16:20] == b"vhb0"
+
+    def test_inquiry_vpd_serial(self, vhost_target):
+        client, master = vhost_target
+        pages = master.inquiry(0, evpd_page=0x00)
+        assert pages.status == 0
+        serial = master.inquiry(0, evpd_page=0x80)
+        assert serial.status == 0
+        n = serial.data[3]
+        uuid = serial.data[4:4 + n].decode()
+        bdev = hipstore.get_bdevs(client, "vhb0")[0]
+        assert bdev.uuid.startswith(uuid[:8])
+
+    def test_inquiry_bad_vpd_page(self, vhost_target):
+        _, master = vhost_target
+        result = master.inquiry(0, evpd_page=0x83)
+        assert result.status == 2  # CHECK CONDITION
+        assert result.sense_key == 5 and result.asc == 0x24
+
+    def test_read_capacity(self, vhost_target):
+        _, master = vhost_target
+        last_lba, block = master.read_capacity10(0)
+        assert (last_lba, block) == (NUM_BLOCKS - 1, BLOCK)
+        last_lba16, block16 = master.read_capacity16(0)
+        assert (last_lba16, block16) == (NUM_BLOCKS - 1, BLOCK)
+
+    def test_test_unit_ready_and_mode_sense(self, vhost_target):
+        _, master = vhost_target
+        assert master.scsi(0, bytes([0x00])).status == 0
+        mode = master.scsi(0, bytes([0x1A, 0, 0, 0, 4, 0]), data_in_len=4)
+        assert mode.status == 0
+
+    def test_report_luns(self, vhost_target):
+        _, master = vhost_target
+        result = master.report_luns(0)
+        assert result.status == 0
+        lun_list_len, = struct.unpack(">I", result.data[:4])
+        assert lun_list_len == 8  # exactly LUN 0
+        assert result.data[8:16] == bytes(8)
+
+
+class TestDataPath:
+    def test_write_read_roundtrip(self, vhost_target):
+        _, master = vhost_target
+        data = os.urandom(8 * BLOCK)
+        assert master.write10(0, 100, data, BLOCK).status == 0
+        result = master.read10(0, 100, 8, BLOCK)
+        assert result.status == 0
+        assert result.data == data
+
+    def test_16_byte_cdbs(self, vhost_target):
+        _, master = vhost_target
+        data = os.urandom(4 * BLOCK)
+        assert master.write16(0, NUM_BLOCKS - 4, data, BLOCK).status == 0
+        result = master.read16(0, NUM_BLOCKS - 4, 4, BLOCK)
+        assert result.status == 0 and result.data == data
+
+    def test_indirect_descriptors(self, vhost_target):
+        _, master = vhost_target
+        data = os.urandom(2 * BLOCK)
+        assert master.write10(0, 7, data, BLOCK, indirect=True).status == 0
+        result = master.read10(0, 7, 2, BLOCK, indirect=True)
+        assert result.status == 0 and result.data == data
+
+    def test_sync_cache(self, vhost_target):
+        _, master = vhost_target
+        assert master.scsi(0, bytes([0x35]) + bytes(9)).status == 0
+
+    def test_many_sequential_commands(self, vhost_target):
+        """More commands than the ring has slots (wrap-around)."""
+        _, master = vhost_target
+        for i in range(40):
+            payload = bytes([i % 256]) * BLOCK
+            assert master.write10(0, i % NUM_BLOCKS, payload,
+                                  BLOCK).status == 0
+        result = master.read10(0, 39, 1, BLOCK)
+        assert result.data == bytes([39]) * BLOCK
+
+
+class TestErrors:
+    def test_bad_target(self, vhost_target):
+        _, master = vhost_target
+        result = master.scsi(3, bytes([0x00]))
+        assert result.response == 3  # VIRTIO_SCSI_S_BAD_TARGET
+
+    def test_bad_lun(self, vhost_target):
+        _, master = vhost_target
+        result = master.scsi(0, bytes([0x00]), lun=5)
+        assert result.status == 2 and result.asc == 0x25
+
+    def test_bad_opcode(self, vhost_target):
+        _, master = vhost_target
+        result = master.scsi(0, bytes([0xFF]))
+        assert result.status == 2
+        assert result.sense_key == 5 and result.asc == 0x20
+
+    def test_lba_out_of_range(self, vhost_target):
+        _, master = vhost_target
+        result = master.read10(0, NUM_BLOCKS, 1, BLOCK)
+        assert result.status == 2 and result.asc == 0x21
+
+
+class TestHotPlug:
+    def test_remove_readd_target(self, vhost_target):
+        client, master = vhost_target
+        assert master.scsi(0, bytes([0x00])).status == 0
+        client.invoke("remove_vhost_scsi_target",
+                      {"ctrlr": "vh0", "scsi_target_num": 0})
+        assert master.scsi(0, bytes([0x00])).response == 3  # gone
+        client.invoke("add_vhost_scsi_lun",
+                      {"ctrlr": "vh0", "scsi_target_num": 0,
+                       "bdev_name": "vhb0"})
+        assert master.scsi(0, bytes([0x00])).status == 0  # back
+
+    def test_second_target(self, vhost_target):
+        client, master = vhost_target
+        hipstore.construct_malloc_bdev(client, num_blocks=1024,
+                                       block_size=BLOCK, name="vhb1")
+        client.invoke("add_vhost_scsi_lun",
+                      {"ctrlr": "vh0", "scsi_target_num": 2,
+                       "bdev_name": "vhb1"})
+        last_lba, _ = master.read_capacity10(2)
+        assert last_lba == 1023
+        data = os.urandom(BLOCK)
+        assert master.write10(2, 0, data, BLOCK).status == 0
+        assert master.read10(2, 0, 1, BLOCK).data == data
+
+
+class TestRingLifecycle:
+    def test_get_vring_base_stops_ring(self, vhost_target):
+        _, master = vhost_target
+        assert master.scsi(0, bytes([0x00])).status == 0
+        base = master.stop_ring()
+        assert base == master.avail_idx  # processed everything submitted
+
+    def test_reconnect(self, hipstored, vhost_target, tmp_path):  # noqa: F811
+        """Master disconnects; a new session works from scratch."""
+        _, master = vhost_target
+        data = os.urandom(BLOCK)
+        assert master.write10(0, 11, data, BLOCK).status == 0
+        path = master.sock.getpeername()
+        master.close()
+        time.sleep(0.1)
+        master2 = VhostUserMaster(path)
+        master2.negotiate()
+        try:
+            result = master2.read10(0, 11, 1, BLOCK)
+            assert result.status == 0 and result.data == data
+        finally:
+            master2.close()
+
+
+@pytest.mark.gpu
+class TestVhostHbm:
+    """Guest I/O lands in MI355X HBM through the engine channels."""
+
+    def test_hbm_roundtrip(self, tmp_path):
+        import fixtures
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=False)
+        try:
+            with hipstore.Client(daemon.socket_path) as client:
+                hipstore.construct_malloc_bdev(
+                    client, num_blocks=262144, block_size=4096, name="vhg0")
+                bdev = hipstore.get_bdevs(client, "vhg0")[0]
+                assert bdev.product_name == "HBM Malloc disk", (
+                    "GPU box must serve the native HBM path")
+                client.invoke("construct_vhost_scsi_controller",
+                              {"ctrlr": "vhgpu"})
+                client.invoke("add_vhost_scsi_lun",
+                              {"ctrlr": "vhgpu", "scsi_target_num": 0,
+                               "bdev_name": "vhg0"})
+                path = os.path.join(os.path.dirname(daemon.socket_path),
+                                    "vhgpu")
+                master = VhostUserMaster(path)
+                master.negotiate()
+                try:
+                    data = os.urandom(64 * 4096)
+                    assert master.write10(0, 512, data, 4096).status == 0
+                    result = master.read10(0, 512, 64, 4096)
+                    assert result.status == 0 and result.data == data
+                    # many commands: exercises the cached engine channel
+                    for i in range(50):
+                        block = os.urandom(4096)
+                        assert master.write10(0, i, block, 4096).status == 0
+                        assert master.read10(0, i, 1, 4096).data == block
+                finally:
+                    master.close()
+        finally:
+            daemon.stop()
