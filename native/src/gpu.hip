@@ -781,7 +781,8 @@ class HbmPersistentChannel : public HbmChannelBase {
 
   HbmPersistentChannel(int device, uint8_t* base)
       : HbmChannelBase(Kind::kPersistent), base_(base), device_(device) {
-    g_per_queue_channels[device].fetch_add(1, std::memory_order_relaxed);
+    // The creator (HbmBdev::get_channel) claimed the per-device slot
+    // with fetch_add BEFORE constructing; release it in the dtor.
     HIP_CHECK(hipSetDevice(device));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&sq_),
@@ -1342,13 +1343,20 @@ class HbmBdev : public Bdev {
       // Auto-fallback: per-queue service kernels give the best
       // latency but each needs a hardware queue; past the cap, new
       // channels multiplex through the shared per-device kernel.
-      const bool over_cap =
-          g_per_queue_channels[device_ & 63].load(
-              std::memory_order_relaxed) >= kPerQueueChannelCap;
-      if (shared_ || over_cap) {
-        return std::make_shared<HbmSharedChannel>(device_, base_);
+      // Claim the slot atomically BEFORE constructing — a
+      // check-then-create race let N concurrent creators all pass the
+      // check and launch N > cap service kernels, which overruns
+      // GPU_MAX_HW_QUEUES and collapses into gang-scheduling.
+      if (!shared_) {
+        int prev = g_per_queue_channels[device_ & 63].fetch_add(
+            1, std::memory_order_relaxed);
+        if (prev < kPerQueueChannelCap) {
+          return std::make_shared<HbmPersistentChannel>(device_, base_);
+        }
+        g_per_queue_channels[device_ & 63].fetch_sub(
+            1, std::memory_order_relaxed);
       }
-      return std::make_shared<HbmPersistentChannel>(device_, base_);
+      return std::make_shared<HbmSharedChannel>(device_, base_);
     }
     return std::make_shared<HbmChannel>(device_, base_, size_bytes());
   }
