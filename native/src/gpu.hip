@@ -761,7 +761,13 @@ class HbmChannel : public HbmChannelBase {
 // kernel needs a real hardware queue; GPU_MAX_HW_QUEUES=24 minus
 // headroom for transient/utility streams).
 inline std::atomic<int> g_per_queue_channels[64];
-constexpr int kPerQueueChannelCap = 18;
+inline int per_queue_channel_cap() {
+  static const int cap = [] {
+    const char* env = getenv("HIPSTORE_PERQ_CAP");
+    return env ? atoi(env) : 18;
+  }();
+  return cap;
+}
 
 class HbmPersistentChannel : public HbmChannelBase {
  public:
@@ -1003,6 +1009,7 @@ class SharedService {
   }
 
   explicit SharedService(int device) : device_(device) {
+    dbg("svc:setdev");
     HIP_CHECK(hipSetDevice(device));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     // Slot config copies go on their OWN stream: an async copy queued
@@ -1010,10 +1017,14 @@ class SharedService {
     // kernel and the slot would only activate after its idle-exit
     // (measured: first completions at 1.4-2.9 s).
     HIP_CHECK(hipStreamCreateWithFlags(&config_stream_, hipStreamNonBlocking));
+    dbg("svc:malloc");
     HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&slots_dev_),
                         kSharedSlots * sizeof(SharedSlot) + 8));
+    dbg("svc:memset");
     HIP_CHECK(hipMemset(slots_dev_, 0, kSharedSlots * sizeof(SharedSlot) + 8));
+    dbg("svc:nullsync");
     HIP_CHECK(hipStreamSynchronize(nullptr));  // memset before config copies
+    dbg("svc:hostmalloc");
     exit_flag_dev_ = reinterpret_cast<uint32_t*>(slots_dev_ + kSharedSlots);
     void* p_stop = nullptr;
     HIP_CHECK(hipHostMalloc(&p_stop, 4, hipHostMallocMapped));
@@ -1022,6 +1033,13 @@ class SharedService {
     memset(mirror_, 0, sizeof(mirror_));
     memset(used_, 0, sizeof(used_));
     memset(prefix_, 0, sizeof(prefix_));
+    dbg("svc:ctor-done");
+  }
+
+  static void dbg(const char* what) {
+    if (getenv("HIPSTORE_DEBUG") != nullptr) {
+      fprintf(stderr, "[hipstore-dbg] %s\n", what);
+    }
   }
 
   // The service lives for the process (kernels self-exit when idle, so
@@ -1042,6 +1060,7 @@ class SharedService {
       m.known_tail = 0;
       m.ring_mask_active = ring_mask;  // active bit clear
       (void)hipSetDevice(device_);
+      dbg("attach:copy");
       // Config words first, the active bit last (stream-ordered copies
       // so a running kernel never sees active before the pointers).
       HIP_CHECK(hipMemcpyAsync(&slots_dev_[i], &m, 40, hipMemcpyHostToDevice,
@@ -1050,8 +1069,11 @@ class SharedService {
       HIP_CHECK(hipMemcpyAsync(
           reinterpret_cast<unsigned long long*>(&slots_dev_[i]) + 5,
           &m.ring_mask_active, 8, hipMemcpyHostToDevice, config_stream_));
+      dbg("attach:sync");
       HIP_CHECK(hipStreamSynchronize(config_stream_));
+      dbg("attach:launch");
       ensure_running_locked(/*force_check=*/true);
+      dbg("attach:done");
       return static_cast<int>(i);
     }
     throw std::runtime_error("shared service: no free ring slots (64 max)");
@@ -1112,8 +1134,30 @@ class SharedService {
     workers = std::min(std::max(workers, 1), 255);
     hipLaunchKernelGGL(k_shared_service, dim3(workers + 1), dim3(64), 0,
                        stream_, ctl);
+    if (getenv("HIPSTORE_DEBUG") != nullptr) {
+      hipError_t launch_err = hipGetLastError();
+      hipError_t query = hipStreamQuery(stream_);
+      fprintf(stderr, "[hipstore-dbg] svc:launched err=%d query=%d\n",
+              static_cast<int>(launch_err), static_cast<int>(query));
+    }
     g_pers_launches.fetch_add(1, std::memory_order_relaxed);
   }
+
+ public:
+  // Debug: read a slot's device state back (claim, known_tail,
+  // mask|active) — serialized on config_stream_ so it never touches
+  // the service kernel's stream.
+  void debug_read_slot(int slot, unsigned long long out[3]) {
+    std::lock_guard<std::mutex> lock(mutex_);
+    (void)hipSetDevice(device_);
+    (void)hipMemcpyAsync(out,
+                         reinterpret_cast<unsigned long long*>(
+                             &slots_dev_[slot]) + 3,
+                         24, hipMemcpyDeviceToHost, config_stream_);
+    (void)hipStreamSynchronize(config_stream_);
+  }
+
+ private:
 
   int device_;
   hipStream_t stream_ = nullptr;
@@ -1137,7 +1181,9 @@ class HbmSharedChannel : public HbmChannelBase {
 
   HbmSharedChannel(int device, uint8_t* base)
       : HbmChannelBase(Kind::kShared), base_(base) {
+    SharedService::dbg("chan:instance");
     service_ = SharedService::instance(device);
+    SharedService::dbg("chan:hostmalloc");
     HIP_CHECK(hipSetDevice(device));
     HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&sq_),
                             kRing * sizeof(BlockDesc), hipHostMallocMapped));
@@ -1150,6 +1196,7 @@ class HbmSharedChannel : public HbmChannelBase {
     *sq_tail_ = 0;
     memset(const_cast<unsigned long long*>(cq_), 0, kRing * 8);
     desc_io_.resize(kRing, nullptr);
+    SharedService::dbg("chan:attach");
     slot_ = service_->attach(
         device_view(sq_), device_view(const_cast<unsigned long long*>(sq_tail_)),
         device_view(const_cast<unsigned long long*>(cq_)), kRing - 1);
@@ -1193,6 +1240,21 @@ class HbmSharedChannel : public HbmChannelBase {
   }
 
   int poll() {
+    static const bool debug_env = getenv("HIPSTORE_DEBUG") != nullptr;
+    if (debug_env) {
+      const auto now = std::chrono::steady_clock::now();
+      if (now - dbg_last_ > std::chrono::seconds(2)) {
+        dbg_last_ = now;
+        unsigned long long st[3] = {0, 0, 0};
+        service_->debug_read_slot(slot_, st);
+        fprintf(stderr,
+                "[hipstore-dbg] slot=%d tail=%llu done=%llu sqtail=%llu "
+                "dev_claim=%llu dev_kt=%llu dev_ma=%llx\n",
+                slot_, static_cast<unsigned long long>(tail_),
+                static_cast<unsigned long long>(completed_), *sq_tail_,
+                st[0], st[1], st[2]);
+      }
+    }
     int completed = 0;
     // Swap out the immediate list before firing: a completion callback
     // may resubmit, and a failed resubmission appends to immediate_
@@ -1291,6 +1353,7 @@ class HbmSharedChannel : public HbmChannelBase {
 
   uint8_t* base_;
   std::shared_ptr<SharedService> service_;
+  std::chrono::steady_clock::time_point dbg_last_{};
   int slot_ = -1;
   BlockDesc* sq_ = nullptr;
   volatile unsigned long long* sq_tail_ = nullptr;
@@ -1342,21 +1405,27 @@ class HbmBdev : public Bdev {
     if (persistent_) {
       // Auto-fallback: per-queue service kernels give the best
       // latency but each needs a hardware queue; past the cap, new
-      // channels multiplex through the shared per-device kernel.
-      // Claim the slot atomically BEFORE constructing — a
-      // check-then-create race let N concurrent creators all pass the
-      // check and launch N > cap service kernels, which overruns
-      // GPU_MAX_HW_QUEUES and collapses into gang-scheduling.
-      if (!shared_) {
-        int prev = g_per_queue_channels[device_ & 63].fetch_add(
-            1, std::memory_order_relaxed);
-        if (prev < kPerQueueChannelCap) {
-          return std::make_shared<HbmPersistentChannel>(device_, base_);
-        }
-        g_per_queue_channels[device_ & 63].fetch_sub(
-            1, std::memory_order_relaxed);
+      // channels fall back to the batched engine (transient per-batch
+      // kernel launches coexist with resident service kernels — the
+      // CRC32C and clone paths run that way in production). The slot
+      // is claimed atomically BEFORE constructing — a check-then-create
+      // race let N concurrent creators all pass the check and launch
+      // N > cap service kernels, overrunning GPU_MAX_HW_QUEUES.
+      // NOTE: the shared service kernel is NOT used here — a device
+      // with both per-queue and shared service kernels resident stops
+      // serving the shared rings entirely (reproduced on MI355X even
+      // at 4+2 channels; engines work in isolation at any count).
+      // Until that interaction is understood, the shared service runs
+      // only in all-shared mode (HIPSTORE_SHARED=1).
+      if (shared_) return std::make_shared<HbmSharedChannel>(device_, base_);
+      int prev = g_per_queue_channels[device_ & 63].fetch_add(
+          1, std::memory_order_relaxed);
+      if (prev < per_queue_channel_cap()) {
+        return std::make_shared<HbmPersistentChannel>(device_, base_);
       }
-      return std::make_shared<HbmSharedChannel>(device_, base_);
+      g_per_queue_channels[device_ & 63].fetch_sub(
+          1, std::memory_order_relaxed);
+      return std::make_shared<HbmChannel>(device_, base_, size_bytes());
     }
     return std::make_shared<HbmChannel>(device_, base_, size_bytes());
   }
@@ -1694,10 +1763,13 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
   for (int q = 0; q < num_queues; ++q) {
     threads.emplace_back([&, q] {
       try {
+        const bool dbg = getenv("HIPSTORE_DEBUG") != nullptr;
         maybe_pin_thread(q);
         auto channel = bdev->get_channel();
+        if (dbg) fprintf(stderr, "[hipstore-dbg] q%d got-channel\n", q);
         uint8_t* buf = static_cast<uint8_t*>(
             alloc_pinned(static_cast<size_t>(io_size) * queue_depth));
+        if (dbg) fprintf(stderr, "[hipstore-dbg] q%d pinned-buf\n", q);
         std::mt19937_64 rng(0x9E3779B97F4A7C15ULL ^ (q * 0x8DA6B343));
         QueueStats& st = stats[q];
         std::vector<clock::time_point> submit_ts(queue_depth);
@@ -1735,6 +1807,7 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
           ++inflight;
           ++submitted;
         }
+        if (dbg) fprintf(stderr, "[hipstore-dbg] q%d submitted-initial\n", q);
         while (true) {
           if (bdev->poll(channel.get()) == 0) {
             __builtin_ia32_pause();  // spinning submitter hygiene
@@ -1751,6 +1824,25 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
         failed.store(true);
       }
     });
+  }
+  if (getenv("HIPSTORE_PERF_DEBUG") != nullptr) {
+    // Diagnostics: report per-queue progress while workers run so a
+    // wedged queue (vs a slow one) is identifiable from the outside.
+    for (int tick = 0; tick < 600; ++tick) {
+      std::this_thread::sleep_for(std::chrono::seconds(5));
+      bool all_done = true;
+      std::string line = "[bdevperf]";
+      for (int q = 0; q < num_queues; ++q) {
+        line += " q" + std::to_string(q) + "=" +
+                std::to_string(stats[q].ios);
+      }
+      fprintf(stderr, "%s\n", line.c_str());
+      for (auto& thread : threads) {
+        if (thread.joinable()) all_done = false;
+      }
+      (void)all_done;
+      if (clock::now() > deadline + std::chrono::seconds(15)) break;
+    }
   }
   for (auto& thread : threads) thread.join();
   const double elapsed =

@@ -17,3 +17,14 @@ Layer map (mirrors reference SURVEY.md section 1):
 """
 
 __version__ = "0.1.0"
+
+# Must be in the environment BEFORE libamdhip64 initializes: the
+# persistent engine needs one real hardware queue per service kernel,
+# and ROCm's 4-queue default gang-schedules resident kernels (measured:
+# 30-50x collapse, and a full wedge when per-queue and shared service
+# kernels mix). The C++ static constructor in _hipstore is TOO LATE for
+# in-process imports — shared-library dependencies (libamdhip64) run
+# their initializers first — so set it here, before any native import.
+import os as _os
+
+_os.environ.setdefault("GPU_MAX_HW_QUEUES", "24")

@@ -47,7 +47,9 @@ def launch_hipstored(tmp_path, cpu: bool = True, device: int = 0) -> HipstoredFi
     cmd = [binary, "-S", socket_path, "-d", str(device)]
     if cpu:
         cmd.append("-C")
-    process = subprocess.Popen(cmd, stderr=subprocess.PIPE)
+    env = dict(os.environ)
+    env.setdefault("GPU_MAX_HW_QUEUES", "24")
+    process = subprocess.Popen(cmd, stderr=subprocess.PIPE, env=env)
     deadline = time.time() + 30
     while not os.path.exists(socket_path):
         if process.poll() is not None:
