@@ -233,8 +233,9 @@ class TestVhostHbm:
                 hipstore.construct_malloc_bdev(
                     client, num_blocks=262144, block_size=4096, name="vhg0")
                 bdev = hipstore.get_bdevs(client, "vhg0")[0]
-                assert bdev.product_name == "HBM Malloc disk", (
-                    "GPU box must serve the native HBM path")
+                assert "hbm" in bdev.driver_specific, (
+                    "GPU box must serve the native HBM path, got %r"
+                    % (bdev.driver_specific,))
                 client.invoke("construct_vhost_scsi_controller",
                               {"ctrlr": "vhgpu"})
                 client.invoke("add_vhost_scsi_lun",
