@@ -15,6 +15,8 @@
 #include <sys/un.h>
 #include <unistd.h>
 
+#include <hip/hip_runtime.h>
+
 #include <cstring>
 #include <mutex>
 #include <set>
@@ -434,6 +436,7 @@ class VhostUserScsiDev {
     uint8_t* map = nullptr;   // mmap base (includes mmap_offset slack)
     size_t maplen = 0;
     uint8_t* base = nullptr;  // map + mmap_offset
+    bool hip_registered = false;
   };
 
   bool set_mem_table(Msg& msg) {
@@ -464,13 +467,25 @@ class VhostUserScsiDev {
       region.map = static_cast<uint8_t*>(map);
       region.maplen = maplen;
       region.base = region.map + w.mmap_offset;
+      // Register guest memory with the GPU so the HBM engine's copy
+      // kernels DMA guest<->HBM directly (SPDK requires DMA-able guest
+      // memory for vhost the same way). Without this, the engine's
+      // hipHostGetDevicePointer on a guest address fails.
+      if (gpu_available()) {
+        region.hip_registered =
+            hipHostRegister(region.base, region.size,
+                            hipHostRegisterMapped) == hipSuccess;
+      }
       regions_.push_back(region);
     }
     return true;
   }
 
   void unmap_regions() {
-    for (const Region& r : regions_) ::munmap(r.map, r.maplen);
+    for (const Region& r : regions_) {
+      if (r.hip_registered) (void)hipHostUnregister(r.base);
+      ::munmap(r.map, r.maplen);
+    }
     regions_.clear();
   }
 
@@ -618,7 +633,15 @@ class VhostUserScsiDev {
       if (ring.last_avail == avail_idx) break;
       uint16_t head = ring.avail->ring[ring.last_avail % ring.num];
       ring.last_avail++;
-      uint32_t written = process_chain(index, head);
+      uint32_t written = 0;
+      try {
+        written = process_chain(index, head);
+      } catch (const std::exception& e) {
+        // A bad buffer (e.g. unregisterable guest memory) must fail
+        // the one command, never kill the ring worker / daemon.
+        fprintf(stderr, "vhost %s: command failed: %s\n", name_.c_str(),
+                e.what());
+      }
       ring.used->ring[ring.used_idx % ring.num] =
           VringUsedElem{head, written};
       __atomic_store_n(&ring.used->idx, ++ring.used_idx, __ATOMIC_RELEASE);
@@ -902,13 +925,17 @@ class VhostUserScsiDev {
             build_sense(resp, 0x05, 0x24, 0x00);
             return 0;
           }
-          std::vector<uint8_t> bounce(bytes);
-          if (is_write) gather(bounce.data(), iovs, bytes);
+          // Pinned bounce: the HBM engine DMAs from host memory, so a
+          // plain vector is not eligible. alloc_pinned falls back to
+          // malloc on CPU-only daemons.
+          uint8_t* bounce = static_cast<uint8_t*>(alloc_pinned(bytes));
+          if (is_write) gather(bounce, iovs, bytes);
           status = bdev_io(bdev, is_write ? IoOp::kWrite : IoOp::kRead,
-                           lba * block, bounce.data(), bytes);
+                           lba * block, bounce, bytes);
           if (!is_write && status == kIoOk) {
-            scatter(iovs, bounce.data(), bytes);
+            scatter(iovs, bounce, bytes);
           }
+          free_pinned(bounce);
         }
         if (status != kIoOk) {
           resp->response = kRespFailure;
