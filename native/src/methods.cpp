@@ -7,6 +7,8 @@
 // (reference local.go:53-57, controller.go:76,204,239).
 
 #include <cinttypes>
+#include <cstdio>
+#include <cstring>
 #include <mutex>
 
 #include "hipstore/bdev.h"
@@ -539,12 +541,22 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   });
 
   server->register_method("get_hbm_info", [device](const Json& p) {
-    // Capacity source for CSI GetCapacity: HBM totals on GPU, host
-    // RAM totals in CPU mode (sysinfo-free approximation via
-    // /proc/meminfo is overkill; report zeros and let callers fall
-    // back when not on a GPU).
+    // Capacity source for CSI GetCapacity: HBM totals on GPU; in CPU
+    // mode malloc bdevs consume host RAM, so report MemTotal /
+    // MemAvailable instead (config-1 semantics).
     const int dev = static_cast<int>(p.get_int("device", device));
     auto [total, free_bytes] = hbm_info(dev);
+    if (total == 0) {
+      if (FILE* f = fopen("/proc/meminfo", "r")) {
+        char key[64];
+        unsigned long long kb = 0;
+        while (fscanf(f, "%63s %llu kB\n", key, &kb) == 2) {
+          if (strcmp(key, "MemTotal:") == 0) total = kb * 1024;
+          if (strcmp(key, "MemAvailable:") == 0) free_bytes = kb * 1024;
+        }
+        fclose(f);
+      }
+    }
     JsonObject o;
     o["device"] = Json(static_cast<int64_t>(dev));
     o["total_bytes"] = Json(static_cast<int64_t>(total));

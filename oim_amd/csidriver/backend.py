@@ -8,6 +8,11 @@ from __future__ import annotations
 from typing import Dict, Optional, Tuple
 
 
+class VolumeExistsError(ValueError):
+    """A volume with this name exists with incompatible parameters
+    (CSI CreateVolume must answer ALREADY_EXISTS, spec.md csi v1)."""
+
+
 class OIMBackend:
     def create_volume(self, name: str, size: int) -> Tuple[str, Dict[str, str]]:
         """Returns (volume_id, volume_context)."""

@@ -34,7 +34,7 @@ from ..spec.rpc_csi import (
     add_csi_identity_to_server,
     add_csi_node_to_server,
 )
-from .backend import OIMBackend
+from .backend import OIMBackend, VolumeExistsError
 from .cephemu import EMULATIONS
 from .mount import Mounter
 
@@ -100,6 +100,9 @@ class ControllerServer(CSIControllerServicer):
             try:
                 volume_id, volume_context = self.backend.create_volume(
                     request.name, size)
+            except VolumeExistsError as exc:
+                # Same name, incompatible parameters (CSI spec: 6)
+                context.abort(grpc.StatusCode.ALREADY_EXISTS, str(exc))
             except ValueError as exc:
                 context.abort(grpc.StatusCode.INVALID_ARGUMENT, str(exc))
             except grpc.RpcError as exc:
@@ -180,6 +183,9 @@ class NodeServer(CSINodeServicer):
         if not request.staging_target_path:
             context.abort(grpc.StatusCode.INVALID_ARGUMENT,
                           "missing staging target path")
+        if request.volume_capability.WhichOneof("access_type") is None:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume capability")
         if request.volume_capability.WhichOneof("access_type") == "block":
             context.abort(grpc.StatusCode.INVALID_ARGUMENT,
                           "block volumes are not supported")
@@ -233,6 +239,9 @@ class NodeServer(CSINodeServicer):
                 or not request.target_path):
             context.abort(grpc.StatusCode.INVALID_ARGUMENT,
                           "missing volume id or paths")
+        if request.volume_capability.WhichOneof("access_type") is None:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume capability")
         with self._volume_mutex.get(request.volume_id):
             if self.mounter.is_mount_point(request.target_path):
                 return csi.NodePublishVolumeResponse()

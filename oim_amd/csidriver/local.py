@@ -11,7 +11,7 @@ from typing import Dict, Tuple
 
 from .. import hipstore
 from ..log import from_context
-from .backend import OIMBackend
+from .backend import OIMBackend, VolumeExistsError
 
 MIN_VOLUME_SIZE = 1 << 20
 MAX_VOLUME_SIZE = 1 << 40
@@ -43,7 +43,7 @@ class LocalBackend(OIMBackend):
                 bdevs = []
             if bdevs:
                 if bdevs[0].size_bytes != size:
-                    raise ValueError(
+                    raise VolumeExistsError(
                         f"volume {name} exists with different size")
                 return name, {}
             hipstore.construct_malloc_bdev(
