@@ -141,7 +141,7 @@ class TestCSISanity:
             self._create(stub, "sanity-2", 1 << 20)
             with pytest.raises(grpc.RpcError) as excinfo:
                 self._create(stub, "sanity-2", 2 << 20)
-            assert excinfo.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+            assert excinfo.value.code() == grpc.StatusCode.ALREADY_EXISTS
             stub.DeleteVolume(csi.DeleteVolumeRequest(volume_id="sanity-2"),
                               timeout=30)
 
@@ -204,11 +204,12 @@ class TestNodeGetVolumeStats:
 
 
 class TestGetCapacity:
-    def test_local_mode_cpu_unimplemented(self, hipstored, tmp_path):  # noqa: F811
-        """CPU-mode daemon has no HBM: GetCapacity reports Unimplemented
-        (on a GPU box the local backend reports free HBM bytes)."""
+    def test_local_mode_cpu_host_memory(self, hipstored, tmp_path):  # noqa: F811
+        """CPU-mode daemon: malloc bdevs consume host RAM, so capacity
+        is host MemAvailable (on a GPU box: free HBM bytes)."""
         backend = LocalBackend(hipstored.socket_path)
-        assert backend.get_capacity() is None  # cpu daemon: no HBM
+        capacity = backend.get_capacity()
+        assert capacity is not None and capacity > 0
         driver = OIMDriver(driver_name="oim-local", node_id="n0",
                            endpoint=f"unix://{tmp_path}/csi3.sock",
                            backend=backend, mounter=Mounter(FakeExec()))
@@ -216,9 +217,8 @@ class TestGetCapacity:
         try:
             with grpc.insecure_channel(grpc_target(driver.addr())) as ch:
                 stub = CSIControllerStub(ch)
-                with pytest.raises(grpc.RpcError) as excinfo:
-                    stub.GetCapacity(csi.GetCapacityRequest(), timeout=30)
-                assert excinfo.value.code() == grpc.StatusCode.UNIMPLEMENTED
+                reply = stub.GetCapacity(csi.GetCapacityRequest(), timeout=30)
+                assert reply.available_capacity == capacity
         finally:
             driver.stop()
 
