@@ -37,3 +37,27 @@ class OIMBackend:
         """Available bytes for new volumes, or None when the backend
         cannot tell (remote mode: the oim.v0 API has no capacity RPC)."""
         return None
+
+    # --- snapshots (optional; local mode only — backed by hipstored's
+    # bdev_clone, an HBM-rate device-side copy) ---------------------------
+
+    def supports_snapshots(self) -> bool:
+        return False
+
+    def create_snapshot(self, name, source_volume_id):
+        """Returns (snapshot_id, size_bytes, creation_unix_seconds).
+        Raises VolumeExistsError on a name conflict with a different
+        source, LookupError if the source volume does not exist."""
+        raise NotImplementedError
+
+    def delete_snapshot(self, snapshot_id) -> None:
+        raise NotImplementedError
+
+    def list_snapshots(self):
+        """[(snapshot_id, source_volume_id, size_bytes, ctime), ...]"""
+        raise NotImplementedError
+
+    def restore_snapshot(self, snapshot_id, volume_name):
+        """Clone a snapshot into a new volume; returns (volume_id,
+        size_bytes). Raises LookupError for unknown snapshots."""
+        raise NotImplementedError

@@ -96,6 +96,12 @@ class ControllerServer(CSIControllerServicer):
                           "missing volume capabilities")
         _validate_capabilities(request.volume_capabilities, context)
         size = request.capacity_range.required_bytes or (1 << 20)
+        source = request.volume_content_source
+        if source.WhichOneof("type") == "snapshot":
+            return self._create_from_snapshot(request, context)
+        if source.WhichOneof("type") == "volume":
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "volume cloning source is not supported")
         with self._name_mutex.get(request.name):
             try:
                 volume_id, volume_context = self.backend.create_volume(
@@ -112,6 +118,33 @@ class ControllerServer(CSIControllerServicer):
         response.volume.capacity_bytes = size
         for key, value in volume_context.items():
             response.volume.volume_context[key] = value
+        return response
+
+    def _create_from_snapshot(self, request, context):
+        if not self.backend.supports_snapshots():
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "snapshot restore is not supported in this mode")
+        snapshot_id = request.volume_content_source.snapshot.snapshot_id
+        if not snapshot_id:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing snapshot id in content source")
+        with self._name_mutex.get(request.name):
+            try:
+                volume_id, size = self.backend.restore_snapshot(
+                    snapshot_id, request.name)
+            except LookupError as exc:
+                context.abort(grpc.StatusCode.NOT_FOUND, str(exc))
+            except grpc.RpcError as exc:
+                context.abort(exc.code(), exc.details())
+        required = request.capacity_range.required_bytes
+        if required and required > size:
+            context.abort(
+                grpc.StatusCode.OUT_OF_RANGE,
+                f"snapshot is {size} bytes; cannot satisfy {required}")
+        response = csi.CreateVolumeResponse()
+        response.volume.volume_id = volume_id
+        response.volume.capacity_bytes = size
+        response.volume.content_source.snapshot.snapshot_id = snapshot_id
         return response
 
     def DeleteVolume(self, request, context):
@@ -166,6 +199,72 @@ class ControllerServer(CSIControllerServicer):
         if self.backend.get_capacity() is not None:
             cap = response.capabilities.add()
             cap.rpc.type = csi.CTRL_CAP_GET_CAPACITY
+        if self.backend.supports_snapshots():
+            cap = response.capabilities.add()
+            cap.rpc.type = csi.CTRL_CAP_CREATE_DELETE_SNAPSHOT
+            cap = response.capabilities.add()
+            cap.rpc.type = csi.CTRL_CAP_LIST_SNAPSHOTS
+        return response
+
+    def CreateSnapshot(self, request, context):
+        """Snapshot = hipstored bdev_clone (device-side HBM-rate copy);
+        instantly ready_to_use (no background upload)."""
+        if not self.backend.supports_snapshots():
+            context.abort(grpc.StatusCode.UNIMPLEMENTED,
+                          "snapshots are not supported in this mode")
+        if not request.source_volume_id:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing source volume id")
+        if not request.name:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT, "missing name")
+        with self._name_mutex.get("snap-" + request.name):
+            try:
+                snapshot_id, size, ctime = self.backend.create_snapshot(
+                    request.name, request.source_volume_id)
+            except VolumeExistsError as exc:
+                context.abort(grpc.StatusCode.ALREADY_EXISTS, str(exc))
+            except LookupError as exc:
+                context.abort(grpc.StatusCode.NOT_FOUND, str(exc))
+            except grpc.RpcError as exc:
+                context.abort(exc.code(), exc.details())
+        response = csi.CreateSnapshotResponse()
+        response.snapshot.snapshot_id = snapshot_id
+        response.snapshot.source_volume_id = request.source_volume_id
+        response.snapshot.size_bytes = size
+        response.snapshot.creation_time.seconds = ctime
+        response.snapshot.ready_to_use = True
+        return response
+
+    def DeleteSnapshot(self, request, context):
+        if not self.backend.supports_snapshots():
+            context.abort(grpc.StatusCode.UNIMPLEMENTED,
+                          "snapshots are not supported in this mode")
+        if not request.snapshot_id:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing snapshot id")
+        with self._name_mutex.get("snap-" + request.snapshot_id):
+            try:
+                self.backend.delete_snapshot(request.snapshot_id)
+            except grpc.RpcError as exc:
+                context.abort(exc.code(), exc.details())
+        return csi.DeleteSnapshotResponse()
+
+    def ListSnapshots(self, request, context):
+        if not self.backend.supports_snapshots():
+            context.abort(grpc.StatusCode.UNIMPLEMENTED,
+                          "snapshots are not supported in this mode")
+        response = csi.ListSnapshotsResponse()
+        for snap_id, source, size, ctime in self.backend.list_snapshots():
+            if request.snapshot_id and snap_id != request.snapshot_id:
+                continue
+            if request.source_volume_id and source != request.source_volume_id:
+                continue
+            entry = response.entries.add()
+            entry.snapshot.snapshot_id = snap_id
+            entry.snapshot.source_volume_id = source
+            entry.snapshot.size_bytes = size
+            entry.snapshot.creation_time.seconds = ctime
+            entry.snapshot.ready_to_use = True
         return response
 
 

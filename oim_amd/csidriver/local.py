@@ -26,6 +26,9 @@ class LocalBackend(OIMBackend):
     def __init__(self, hipstored_socket: str, nbd_prefix: str = "/dev/nbd"):
         self.socket = hipstored_socket
         self.nbd_prefix = nbd_prefix
+        # snapshot id -> (source volume, creation time); hipstored owns
+        # the data, this is presentation metadata only
+        self._snap_meta = {}
 
     def _client(self) -> hipstore.Client:
         return hipstore.Client(self.socket)
@@ -104,8 +107,89 @@ class LocalBackend(OIMBackend):
                 hipstore.stop_nbd_disk(client, device)
 
     def get_capacity(self):
-        """Free HBM on the daemon's GPU (None in CPU mode)."""
+        """Free HBM on the daemon's GPU (host MemAvailable in CPU
+        mode — malloc bdevs consume host RAM there)."""
         with self._client() as client:
             info = client.invoke("get_hbm_info")
         free = int(info.get("free_bytes", 0))
         return free if free > 0 else None
+
+    # --- snapshots (bdev_clone-backed) --------------------------------------
+
+    SNAP_PREFIX = "csi-snap-"
+
+    def supports_snapshots(self) -> bool:
+        return True
+
+    def create_snapshot(self, name, source_volume_id):
+        import time as _time
+
+        snap_id = self.SNAP_PREFIX + name
+        with self._client() as client:
+            try:
+                existing = hipstore.get_bdevs(client, snap_id)
+            except hipstore.RpcError as err:
+                if not err.is_not_found():
+                    raise
+                existing = []
+            if existing:
+                meta = self._snap_meta.get(snap_id)
+                if meta and meta[0] != source_volume_id:
+                    raise VolumeExistsError(
+                        f"snapshot {name} exists for volume {meta[0]}")
+                return (snap_id, existing[0].size_bytes,
+                        meta[1] if meta else 0)
+            try:
+                hipstore.get_bdevs(client, source_volume_id)
+            except hipstore.RpcError as err:
+                if err.is_not_found():
+                    raise LookupError(
+                        f"volume {source_volume_id} not found") from None
+                raise
+            client.invoke("bdev_clone",
+                          {"src": source_volume_id, "name": snap_id})
+            bdev = hipstore.get_bdevs(client, snap_id)[0]
+        ctime = int(_time.time())
+        self._snap_meta[snap_id] = (source_volume_id, ctime)
+        return snap_id, bdev.size_bytes, ctime
+
+    def delete_snapshot(self, snapshot_id) -> None:
+        with self._client() as client:
+            try:
+                hipstore.delete_bdev(client, snapshot_id)
+            except hipstore.RpcError as err:
+                if not err.is_not_found():
+                    raise
+        self._snap_meta.pop(snapshot_id, None)
+
+    def list_snapshots(self):
+        out = []
+        with self._client() as client:
+            for bdev in hipstore.get_bdevs(client):
+                if not bdev.name.startswith(self.SNAP_PREFIX):
+                    continue
+                source, ctime = self._snap_meta.get(bdev.name, ("", 0))
+                out.append((bdev.name, source, bdev.size_bytes, ctime))
+        return out
+
+    def restore_snapshot(self, snapshot_id, volume_name):
+        with self._client() as client:
+            try:
+                existing = hipstore.get_bdevs(client, volume_name)
+            except hipstore.RpcError as err:
+                if not err.is_not_found():
+                    raise
+                existing = []
+            if existing:  # idempotent re-create from the same snapshot
+                return volume_name, existing[0].size_bytes
+            try:
+                hipstore.get_bdevs(client, snapshot_id)
+            except hipstore.RpcError as err:
+                if err.is_not_found():
+                    raise LookupError(
+                        f"snapshot {snapshot_id} not found") from None
+                raise
+            client.invoke("bdev_clone",
+                          {"src": snapshot_id, "name": volume_name})
+            bdev = hipstore.get_bdevs(client, volume_name)[0]
+        return volume_name, bdev.size_bytes

@@ -38,6 +38,8 @@ CTRL_CAP_CREATE_DELETE_VOLUME = 1
 CTRL_CAP_PUBLISH_UNPUBLISH_VOLUME = 2
 CTRL_CAP_LIST_VOLUMES = 3
 CTRL_CAP_GET_CAPACITY = 4
+CTRL_CAP_CREATE_DELETE_SNAPSHOT = 5
+CTRL_CAP_LIST_SNAPSHOTS = 6
 
 # NodeServiceCapability.RPC.Type
 NODE_CAP_UNKNOWN = 0
@@ -77,6 +79,7 @@ MESSAGES = [
             Field("name", 1, "string"),
             Field("capacity_range", 2, "CapacityRange"),
             Field("volume_capabilities", 3, "VolumeCapability", repeated=True),
+            Field("volume_content_source", 6, "VolumeContentSource"),
         ],
         map_fields=[("parameters", 4, "string", "string"),
                     ("secrets", 5, "string", "string")],
@@ -87,8 +90,62 @@ MESSAGES = [
         [
             Field("capacity_bytes", 1, "int64"),
             Field("volume_id", 2, "string"),
+            Field("content_source", 4, "VolumeContentSource"),
         ],
         map_fields=[("volume_context", 3, "string", "string")],
+    ),
+    Message(
+        "VolumeContentSource",
+        [
+            Field("snapshot", 1, "VolumeContentSource.SnapshotSource",
+                  oneof="type"),
+            Field("volume", 2, "VolumeContentSource.VolumeSource",
+                  oneof="type"),
+        ],
+    ),
+    # Snapshots (CSI v1 csi.proto field numbers; backed by hipstored's
+    # bdev_clone — HBM-rate device-side copies)
+    Message(
+        "CreateSnapshotRequest",
+        [
+            Field("source_volume_id", 1, "string"),
+            Field("name", 2, "string"),
+        ],
+        map_fields=[("secrets", 3, "string", "string"),
+                    ("parameters", 4, "string", "string")],
+    ),
+    Message("CreateSnapshotResponse", [Field("snapshot", 1, "Snapshot")]),
+    Message(
+        "Snapshot",
+        [
+            Field("size_bytes", 1, "int64"),
+            Field("snapshot_id", 2, "string"),
+            Field("source_volume_id", 3, "string"),
+            Field("creation_time", 4, ".google.protobuf.Timestamp"),
+            Field("ready_to_use", 5, "bool"),
+        ],
+    ),
+    Message(
+        "DeleteSnapshotRequest",
+        [Field("snapshot_id", 1, "string")],
+        map_fields=[("secrets", 2, "string", "string")],
+    ),
+    Message("DeleteSnapshotResponse", []),
+    Message(
+        "ListSnapshotsRequest",
+        [
+            Field("max_entries", 1, "int32"),
+            Field("starting_token", 2, "string"),
+            Field("source_volume_id", 3, "string"),
+            Field("snapshot_id", 4, "string"),
+        ],
+    ),
+    Message(
+        "ListSnapshotsResponse",
+        [
+            Field("entries", 1, "ListSnapshotsResponse.Entry", repeated=True),
+            Field("next_token", 2, "string"),
+        ],
     ),
     Message(
         "CapacityRange",
@@ -234,6 +291,9 @@ MESSAGES = [
 # patch the type references:
 _NESTED = {
     "PluginCapability.Service": "PluginCapabilityService",
+    "VolumeContentSource.SnapshotSource": "VolumeContentSourceSnapshotSource",
+    "VolumeContentSource.VolumeSource": "VolumeContentSourceVolumeSource",
+    "ListSnapshotsResponse.Entry": "ListSnapshotsResponseEntry",
     "VolumeCapability.BlockVolume": "VolumeCapabilityBlockVolume",
     "VolumeCapability.MountVolume": "VolumeCapabilityMountVolume",
     "VolumeCapability.AccessMode": "VolumeCapabilityAccessMode",
@@ -269,6 +329,11 @@ MESSAGES += [
     ),
     Message("ControllerServiceCapabilityRPC", [Field("type", 1, "int32")]),
     Message("NodeServiceCapabilityRPC", [Field("type", 1, "int32")]),
+    Message("VolumeContentSourceSnapshotSource",
+            [Field("snapshot_id", 1, "string")]),
+    Message("VolumeContentSourceVolumeSource",
+            [Field("volume_id", 1, "string")]),
+    Message("ListSnapshotsResponseEntry", [Field("snapshot", 1, "Snapshot")]),
 ]
 
 SERVICES = [
@@ -291,6 +356,12 @@ SERVICES = [
             ("GetCapacity", "GetCapacityRequest", "GetCapacityResponse"),
             ("ControllerGetCapabilities", "ControllerGetCapabilitiesRequest",
              "ControllerGetCapabilitiesResponse"),
+            ("CreateSnapshot", "CreateSnapshotRequest",
+             "CreateSnapshotResponse"),
+            ("DeleteSnapshot", "DeleteSnapshotRequest",
+             "DeleteSnapshotResponse"),
+            ("ListSnapshots", "ListSnapshotsRequest",
+             "ListSnapshotsResponse"),
         ],
     ),
     Service(
@@ -319,15 +390,13 @@ UNIMPLEMENTED_CONTROLLER_METHODS = (
     "ControllerPublishVolume",
     "ControllerUnpublishVolume",
     "ListVolumes",
-    "CreateSnapshot",
-    "DeleteSnapshot",
-    "ListSnapshots",
     "ControllerExpandVolume",
 )
 UNIMPLEMENTED_NODE_METHODS = (
     "NodeExpandVolume",
 )
 
+import google.protobuf.timestamp_pb2  # noqa: E402  (registers timestamp.proto)
 import google.protobuf.wrappers_pb2  # noqa: E402  (registers wrappers.proto)
 
 _classes = build_file(
@@ -335,7 +404,8 @@ _classes = build_file(
     package=PACKAGE,
     messages=MESSAGES,
     services=SERVICES,
-    dependencies=["google/protobuf/wrappers.proto"],
+    dependencies=["google/protobuf/wrappers.proto",
+                  "google/protobuf/timestamp.proto"],
 )
 
 globals().update(_classes)

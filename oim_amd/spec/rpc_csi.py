@@ -58,6 +58,15 @@ class CSIControllerServicer:
     def ControllerGetCapabilities(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "ControllerGetCapabilities")
 
+    def CreateSnapshot(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "CreateSnapshot")
+
+    def DeleteSnapshot(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "DeleteSnapshot")
+
+    def ListSnapshots(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "ListSnapshots")
+
 
 class CSINodeServicer:
     def NodeGetVolumeStats(self, request, context):
@@ -104,6 +113,12 @@ def add_csi_controller_to_server(servicer, server):
         "ControllerGetCapabilities": _unary(
             servicer.ControllerGetCapabilities,
             csi.ControllerGetCapabilitiesRequest),
+        "CreateSnapshot": _unary(servicer.CreateSnapshot,
+                                 csi.CreateSnapshotRequest),
+        "DeleteSnapshot": _unary(servicer.DeleteSnapshot,
+                                 csi.DeleteSnapshotRequest),
+        "ListSnapshots": _unary(servicer.ListSnapshots,
+                                csi.ListSnapshotsRequest),
     }
     for name in csi.UNIMPLEMENTED_CONTROLLER_METHODS:
         handlers[name] = _unimplemented_handler(name)
@@ -162,6 +177,11 @@ class CSIControllerStub:
         self.GetCapacity = method("GetCapacity", csi.GetCapacityResponse)
         self.ControllerGetCapabilities = method(
             "ControllerGetCapabilities", csi.ControllerGetCapabilitiesResponse)
+        self.CreateSnapshot = method("CreateSnapshot",
+                                     csi.CreateSnapshotResponse)
+        self.DeleteSnapshot = method("DeleteSnapshot",
+                                     csi.DeleteSnapshotResponse)
+        self.ListSnapshots = method("ListSnapshots", csi.ListSnapshotsResponse)
 
 
 class CSINodeStub:

@@ -273,3 +273,125 @@ class TestNodeSanity:
         byte_usage = [u for u in reply.usage
                       if u.unit == csi.USAGE_UNIT_BYTES][0]
         assert byte_usage.total > 0
+
+
+class TestSnapshotSanity:
+    """CSI snapshot RPCs (backed by hipstored bdev_clone; the sanity
+    suite's snapshot behavior matrix)."""
+
+    def _create_volume(self, env, name, size=1 << 20):
+        request = csi.CreateVolumeRequest(name=name)
+        request.capacity_range.required_bytes = size
+        request.volume_capabilities.add().CopyFrom(mount_cap())
+        return env["controller"].CreateVolume(request, timeout=10)
+
+    def test_capabilities_include_snapshots(self, sanity_env):
+        caps = sanity_env["controller"].ControllerGetCapabilities(
+            csi.ControllerGetCapabilitiesRequest(), timeout=10)
+        types = {cap.rpc.type for cap in caps.capabilities}
+        assert csi.CTRL_CAP_CREATE_DELETE_SNAPSHOT in types
+        assert csi.CTRL_CAP_LIST_SNAPSHOTS in types
+
+    def test_create_snapshot_missing_fields(self, sanity_env):
+        ctrl = sanity_env["controller"]
+        with pytest.raises(grpc.RpcError) as excinfo:
+            ctrl.CreateSnapshot(
+                csi.CreateSnapshotRequest(name="s"), timeout=10)
+        assert_code(excinfo, grpc.StatusCode.INVALID_ARGUMENT)
+        with pytest.raises(grpc.RpcError) as excinfo:
+            ctrl.CreateSnapshot(
+                csi.CreateSnapshotRequest(source_volume_id="v"), timeout=10)
+        assert_code(excinfo, grpc.StatusCode.INVALID_ARGUMENT)
+
+    def test_snapshot_of_unknown_volume(self, sanity_env):
+        with pytest.raises(grpc.RpcError) as excinfo:
+            sanity_env["controller"].CreateSnapshot(
+                csi.CreateSnapshotRequest(source_volume_id="ghost",
+                                          name="snap-x"), timeout=10)
+        assert_code(excinfo, grpc.StatusCode.NOT_FOUND)
+
+    def test_snapshot_lifecycle(self, sanity_env):
+        ctrl = sanity_env["controller"]
+        self._create_volume(sanity_env, "snap-src")
+        try:
+            created = ctrl.CreateSnapshot(
+                csi.CreateSnapshotRequest(source_volume_id="snap-src",
+                                          name="nightly"), timeout=10)
+            snap = created.snapshot
+            assert snap.snapshot_id
+            assert snap.source_volume_id == "snap-src"
+            assert snap.size_bytes == 1 << 20
+            assert snap.ready_to_use
+            assert snap.creation_time.seconds > 0
+            # idempotent: same name + same source -> same snapshot
+            again = ctrl.CreateSnapshot(
+                csi.CreateSnapshotRequest(source_volume_id="snap-src",
+                                          name="nightly"), timeout=10)
+            assert again.snapshot.snapshot_id == snap.snapshot_id
+            # conflict: same name, different source -> ALREADY_EXISTS
+            self._create_volume(sanity_env, "snap-src-2")
+            with pytest.raises(grpc.RpcError) as excinfo:
+                ctrl.CreateSnapshot(
+                    csi.CreateSnapshotRequest(source_volume_id="snap-src-2",
+                                              name="nightly"), timeout=10)
+            assert_code(excinfo, grpc.StatusCode.ALREADY_EXISTS)
+            # list: filter by source
+            listed = ctrl.ListSnapshots(
+                csi.ListSnapshotsRequest(source_volume_id="snap-src"),
+                timeout=10)
+            assert [e.snapshot.snapshot_id for e in listed.entries] == \
+                [snap.snapshot_id]
+            # delete; repeat delete is idempotent
+            ctrl.DeleteSnapshot(csi.DeleteSnapshotRequest(
+                snapshot_id=snap.snapshot_id), timeout=10)
+            ctrl.DeleteSnapshot(csi.DeleteSnapshotRequest(
+                snapshot_id=snap.snapshot_id), timeout=10)
+            listed = ctrl.ListSnapshots(csi.ListSnapshotsRequest(), timeout=10)
+            assert snap.snapshot_id not in [
+                e.snapshot.snapshot_id for e in listed.entries]
+        finally:
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id="snap-src"),
+                              timeout=10)
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id="snap-src-2"),
+                              timeout=10)
+
+    def test_restore_volume_from_snapshot(self, sanity_env):
+        ctrl = sanity_env["controller"]
+        self._create_volume(sanity_env, "restore-src")
+        try:
+            snap = ctrl.CreateSnapshot(
+                csi.CreateSnapshotRequest(source_volume_id="restore-src",
+                                          name="base"), timeout=10).snapshot
+            request = csi.CreateVolumeRequest(name="restored-vol")
+            request.capacity_range.required_bytes = 1 << 20
+            request.volume_capabilities.add().CopyFrom(mount_cap())
+            request.volume_content_source.snapshot.snapshot_id = \
+                snap.snapshot_id
+            restored = ctrl.CreateVolume(request, timeout=10)
+            assert restored.volume.volume_id == "restored-vol"
+            assert restored.volume.capacity_bytes == 1 << 20
+            assert (restored.volume.content_source.snapshot.snapshot_id
+                    == snap.snapshot_id)
+            # restore asking for more than the snapshot -> OUT_OF_RANGE
+            request2 = csi.CreateVolumeRequest(name="restored-too-big")
+            request2.capacity_range.required_bytes = 4 << 20
+            request2.volume_capabilities.add().CopyFrom(mount_cap())
+            request2.volume_content_source.snapshot.snapshot_id = \
+                snap.snapshot_id
+            with pytest.raises(grpc.RpcError) as excinfo:
+                ctrl.CreateVolume(request2, timeout=10)
+            assert excinfo.value.code() in (grpc.StatusCode.OUT_OF_RANGE,)
+            # restore from unknown snapshot -> NOT_FOUND
+            request3 = csi.CreateVolumeRequest(name="restored-ghost")
+            request3.volume_capabilities.add().CopyFrom(mount_cap())
+            request3.volume_content_source.snapshot.snapshot_id = "ghost"
+            with pytest.raises(grpc.RpcError) as excinfo:
+                ctrl.CreateVolume(request3, timeout=10)
+            assert_code(excinfo, grpc.StatusCode.NOT_FOUND)
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(
+                volume_id="restored-vol"), timeout=10)
+            ctrl.DeleteSnapshot(csi.DeleteSnapshotRequest(
+                snapshot_id=snap.snapshot_id), timeout=10)
+        finally:
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(
+                volume_id="restore-src"), timeout=10)
