@@ -259,3 +259,93 @@ class TestVhostHbm:
                     master.close()
         finally:
             daemon.stop()
+
+
+class TestVhostRobustness:
+    """Hostile-guest input: malformed descriptors and protocol misuse
+    must fail the command or drop the session — never crash hipstored
+    (a guest can hand the device any bytes it likes)."""
+
+    def test_bad_gpa_in_descriptor(self, vhost_target):
+        client, master = vhost_target
+        master.tag += 1
+        req = (bytes([1, 0, 0x40, 0, 0, 0, 0, 0])
+               + struct.pack("<Q", master.tag) + bytes(3)
+               + bytes([0x00]).ljust(32, b"\0"))
+        from vhost_client import GPA_BASE, REQ_OFF, RESP_OFF, DESC_WRITE
+        master.mem[REQ_OFF:REQ_OFF + 51] = req
+        # request desc points outside the registered region
+        master._write_desc(0, 0xDEAD0000000, 51, 1, 1)  # NEXT->1
+        master._write_desc(1, GPA_BASE + RESP_OFF, 108, DESC_WRITE)
+        master._submit(0)
+        used_len = master._wait_used()
+        assert used_len == 0  # chain rejected, nothing written
+        # daemon and session still healthy
+        assert master.scsi(0, bytes([0x00])).status == 0
+        assert hipstore.get_bdevs(client, "vhb0")
+
+    def test_self_referencing_chain_bounded(self, vhost_target):
+        client, master = vhost_target
+        from vhost_client import GPA_BASE, REQ_OFF
+        master._write_desc(0, GPA_BASE + REQ_OFF, 51, 1, 0)  # NEXT -> itself
+        master._submit(0)
+        used_len = master._wait_used()
+        assert used_len == 0  # hop bound kicked in
+        assert master.scsi(0, bytes([0x00])).status == 0
+
+    def test_chain_without_writable_resp(self, vhost_target):
+        client, master = vhost_target
+        from vhost_client import GPA_BASE, REQ_OFF
+        master.tag += 1
+        req = (bytes([1, 0, 0x40, 0, 0, 0, 0, 0])
+               + struct.pack("<Q", master.tag) + bytes(3)
+               + bytes([0x00]).ljust(32, b"\0"))
+        master.mem[REQ_OFF:REQ_OFF + 51] = req
+        master._write_desc(0, GPA_BASE + REQ_OFF, 51, 0)  # no resp desc
+        master._submit(0)
+        assert master._wait_used() == 0
+        assert master.scsi(0, bytes([0x00])).status == 0
+
+    def test_bad_protocol_version_drops_session(self, hipstored, tmp_path):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            client.invoke("construct_vhost_scsi_controller", {"ctrlr": "vhbad"})
+            path = os.path.join(os.path.dirname(hipstored.socket_path),
+                                "vhbad")
+            import socket as socketmod
+            sock = socketmod.socket(socketmod.AF_UNIX, socketmod.SOCK_STREAM)
+            sock.settimeout(5)
+            sock.connect(path)
+            # version field 0x3 is invalid (spec: 0x1)
+            sock.sendall(struct.pack("<III", 1, 0x3, 0))
+            assert sock.recv(16) == b""  # slave hangs up
+            sock.close()
+            # fresh, correct session still works
+            master = VhostUserMaster(path)
+            master.negotiate()
+            master.close()
+            client.invoke("remove_vhost_controller", {"ctrlr": "vhbad"})
+
+    def test_oversized_payload_drops_session(self, hipstored, tmp_path):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            client.invoke("construct_vhost_scsi_controller", {"ctrlr": "vhsz"})
+            path = os.path.join(os.path.dirname(hipstored.socket_path),
+                                "vhsz")
+            import socket as socketmod
+            sock = socketmod.socket(socketmod.AF_UNIX, socketmod.SOCK_STREAM)
+            sock.settimeout(5)
+            sock.connect(path)
+            sock.sendall(struct.pack("<III", 5, 0x1, 1 << 20))  # 1MiB claim
+            assert sock.recv(16) == b""  # slave hangs up, no allocation
+            sock.close()
+            client.invoke("remove_vhost_controller", {"ctrlr": "vhsz"})
+
+    def test_huge_read_rejected_with_sense(self, vhost_target):
+        """A READ beyond the device (or the per-command byte cap) gets
+        CHECK CONDITION before any guest-sized allocation happens."""
+        _, master = vhost_target
+        cdb = bytearray(10)
+        cdb[0] = 0x28
+        cdb[7:9] = struct.pack(">H", 16384)  # 8 MiB at 512-byte blocks
+        result = master.scsi(0, bytes(cdb), data_in_len=4096)
+        assert result.status == 2
+        assert result.asc == 0x21
