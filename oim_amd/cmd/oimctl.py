@@ -40,6 +40,7 @@ def main(argv=None) -> int:
         ("map", "map a provisioned volume to a SCSI target"),
         ("unmap", "unmap a volume"),
         ("check", "check that a malloc bdev exists"),
+        ("clone", "clone a malloc bdev (device-side HBM-rate copy)"),
     ):
         cmd = sub.add_parser(name, help=help_text + " via the registry proxy")
         cmd.add_argument("--controller", required=True,
@@ -47,6 +48,8 @@ def main(argv=None) -> int:
         cmd.add_argument("volume")
         if name == "provision":
             cmd.add_argument("size", help='bytes, or "64MiB"-style; 0 deletes')
+        if name == "clone":
+            cmd.add_argument("dest", help="name of the clone to create")
     args = parser.parse_args(argv)
     log.init_from_args(args)
 
@@ -101,6 +104,12 @@ def main(argv=None) -> int:
                     spec.CheckMallocBDevRequest(bdev_name=args.volume),
                     metadata=metadata, timeout=60)
                 print(f"{args.volume} exists")
+            elif args.command == "clone":
+                controller.CloneMallocBDev(
+                    spec.CloneMallocBDevRequest(source=args.volume,
+                                                dest=args.dest),
+                    metadata=metadata, timeout=120)
+                print(f"cloned {args.volume} -> {args.dest}")
     return 0
 
 

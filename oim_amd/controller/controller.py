@@ -306,6 +306,44 @@ class Controller(spec.ControllerServicer):
                                   f"BDev {name} not found")
         return spec.CheckMallocBDevReply()
 
+    def CloneMallocBDev(self, request, context):
+        """oim-amd extension (docs/spec.md): device-side bdev clone —
+        backs CSI snapshots in remote mode. Idempotent on an existing
+        destination of the same geometry."""
+        source, dest = request.source, request.dest
+        if not source or not dest:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "source and dest are required")
+        with self._volume_mutex.locked(dest):
+            with self._client() as client:
+                try:
+                    src_bdevs = hipstore.get_bdevs(client, source)
+                except hipstore.RpcError as err:
+                    if err.is_not_found():
+                        context.abort(grpc.StatusCode.NOT_FOUND,
+                                      f"BDev {source} not found")
+                    context.abort(grpc.StatusCode.INTERNAL, str(err))
+                try:
+                    existing = hipstore.get_bdevs(client, dest)
+                except hipstore.RpcError as err:
+                    if not err.is_not_found():
+                        context.abort(grpc.StatusCode.INTERNAL, str(err))
+                    existing = []
+                if existing:
+                    same = (existing[0].block_size == src_bdevs[0].block_size
+                            and existing[0].num_blocks
+                            == src_bdevs[0].num_blocks)
+                    if not same:
+                        context.abort(
+                            grpc.StatusCode.ALREADY_EXISTS,
+                            f"BDev {dest} exists with different geometry")
+                    return spec.CloneMallocBDevReply()
+                try:
+                    client.invoke("bdev_clone", {"src": source, "name": dest})
+                except hipstore.RpcError as err:
+                    context.abort(grpc.StatusCode.INTERNAL, str(err))
+        return spec.CloneMallocBDevReply()
+
     # --- self-registration --------------------------------------------------
 
     def register(self) -> None:

@@ -137,7 +137,7 @@ class ControllerServer(CSIControllerServicer):
             except grpc.RpcError as exc:
                 context.abort(exc.code(), exc.details())
         required = request.capacity_range.required_bytes
-        if required and required > size:
+        if size and required and required > size:
             context.abort(
                 grpc.StatusCode.OUT_OF_RANGE,
                 f"snapshot is {size} bytes; cannot satisfy {required}")

@@ -65,6 +65,13 @@ class RemoteBackend(OIMBackend):
         self.sysfs_block_dir = sysfs_block_dir
         self.dev_dir = dev_dir
         self.device_timeout = device_timeout
+        # Presentation metadata for snapshots created through this
+        # driver: snap id -> (source volume, ctime, size hint). The
+        # oim.v0 API has no stat/list RPC, so sizes come from the
+        # create_volume calls this process made (0 = unknown, which
+        # CSI allows).
+        self._snap_meta: Dict[str, tuple] = {}
+        self._volume_sizes: Dict[str, int] = {}
 
     # --- registry plumbing --------------------------------------------------
 
@@ -91,6 +98,7 @@ class RemoteBackend(OIMBackend):
             stub.ProvisionMallocBDev(
                 spec.ProvisionMallocBDevRequest(bdev_name=name, size=size),
                 metadata=self._metadata(), timeout=30)
+        self._volume_sizes[name] = size
         return name, {}
 
     def delete_volume(self, volume_id: str) -> None:
@@ -112,6 +120,74 @@ class RemoteBackend(OIMBackend):
                 if err.code() == grpc.StatusCode.NOT_FOUND:
                     return False
                 raise
+
+    # --- snapshots (CloneMallocBDev extension, docs/spec.md) ----------------
+
+    SNAP_PREFIX = "csi-snap-"
+
+    def supports_snapshots(self) -> bool:
+        return True
+
+    def create_snapshot(self, name, source_volume_id):
+        import time as _time
+
+        from .backend import VolumeExistsError
+
+        snap_id = self.SNAP_PREFIX + name
+        existing = self._snap_meta.get(snap_id)
+        if existing and existing[0] != source_volume_id:
+            raise VolumeExistsError(
+                f"snapshot {name} exists for volume {existing[0]}")
+        with self._dial_registry() as channel:
+            stub = spec.ControllerStub(channel)
+            try:
+                stub.CloneMallocBDev(
+                    spec.CloneMallocBDevRequest(source=source_volume_id,
+                                                dest=snap_id),
+                    metadata=self._metadata(), timeout=60)
+            except grpc.RpcError as err:
+                if err.code() == grpc.StatusCode.NOT_FOUND:
+                    raise LookupError(
+                        f"volume {source_volume_id} not found") from None
+                raise
+        if existing:
+            return snap_id, existing[2], existing[1]
+        ctime = int(_time.time())
+        size = self._volume_sizes.get(source_volume_id, 0)
+        self._snap_meta[snap_id] = (source_volume_id, ctime, size)
+        return snap_id, size, ctime
+
+    def delete_snapshot(self, snapshot_id) -> None:
+        with self._dial_registry() as channel:
+            stub = spec.ControllerStub(channel)
+            stub.ProvisionMallocBDev(
+                spec.ProvisionMallocBDevRequest(bdev_name=snapshot_id,
+                                                size=0),
+                metadata=self._metadata(), timeout=30)
+        self._snap_meta.pop(snapshot_id, None)
+
+    def list_snapshots(self):
+        return [(snap_id, meta[0], meta[2], meta[1])
+                for snap_id, meta in sorted(self._snap_meta.items())]
+
+    def restore_snapshot(self, snapshot_id, volume_name):
+        with self._dial_registry() as channel:
+            stub = spec.ControllerStub(channel)
+            try:
+                stub.CloneMallocBDev(
+                    spec.CloneMallocBDevRequest(source=snapshot_id,
+                                                dest=volume_name),
+                    metadata=self._metadata(), timeout=60)
+            except grpc.RpcError as err:
+                if err.code() == grpc.StatusCode.NOT_FOUND:
+                    raise LookupError(
+                        f"snapshot {snapshot_id} not found") from None
+                raise
+        meta = self._snap_meta.get(snapshot_id)
+        size = meta[2] if meta else 0
+        if size:
+            self._volume_sizes[volume_name] = size
+        return volume_name, size
 
     # --- device lifecycle ---------------------------------------------------
 

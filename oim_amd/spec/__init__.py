@@ -18,6 +18,8 @@ from .oim_v0 import (  # noqa: F401
     ProvisionMallocBDevReply,
     CheckMallocBDevRequest,
     CheckMallocBDevReply,
+    CloneMallocBDevRequest,
+    CloneMallocBDevReply,
 )
 from .rpc import (  # noqa: F401
     CONTROLLER_ID_KEY,

@@ -65,6 +65,9 @@ MESSAGES = [
     Message("ProvisionMallocBDevReply", []),
     Message("CheckMallocBDevRequest", [Field("bdev_name", 1, "string")]),
     Message("CheckMallocBDevReply", []),
+    Message("CloneMallocBDevRequest",
+            [Field("source", 1, "string"), Field("dest", 2, "string")]),
+    Message("CloneMallocBDevReply", []),
 ]
 
 SERVICES = [
@@ -82,6 +85,7 @@ SERVICES = [
             ("UnmapVolume", "UnmapVolumeRequest", "UnmapVolumeReply"),
             ("ProvisionMallocBDev", "ProvisionMallocBDevRequest", "ProvisionMallocBDevReply"),
             ("CheckMallocBDev", "CheckMallocBDevRequest", "CheckMallocBDevReply"),
+            ("CloneMallocBDev", "CloneMallocBDevRequest", "CloneMallocBDevReply"),
         ],
     ),
 ]
@@ -107,3 +111,5 @@ ProvisionMallocBDevRequest = _classes["ProvisionMallocBDevRequest"]
 ProvisionMallocBDevReply = _classes["ProvisionMallocBDevReply"]
 CheckMallocBDevRequest = _classes["CheckMallocBDevRequest"]
 CheckMallocBDevReply = _classes["CheckMallocBDevReply"]
+CloneMallocBDevRequest = _classes["CloneMallocBDevRequest"]
+CloneMallocBDevReply = _classes["CloneMallocBDevReply"]

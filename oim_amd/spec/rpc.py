@@ -48,6 +48,7 @@ class ControllerStub:
             "ProvisionMallocBDev", pb.ProvisionMallocBDevReply
         )
         self.CheckMallocBDev = method("CheckMallocBDev", pb.CheckMallocBDevReply)
+        self.CloneMallocBDev = method("CloneMallocBDev", pb.CloneMallocBDevReply)
 
 
 class RegistryServicer:
@@ -74,6 +75,9 @@ class ControllerServicer:
 
     def CheckMallocBDev(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "CheckMallocBDev not implemented")
+
+    def CloneMallocBDev(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "CloneMallocBDev not implemented")
 
 
 def _unary(fn, request_class):
@@ -102,6 +106,7 @@ def add_controller_to_server(servicer: ControllerServicer, server: grpc.Server) 
             servicer.ProvisionMallocBDev, pb.ProvisionMallocBDevRequest
         ),
         "CheckMallocBDev": _unary(servicer.CheckMallocBDev, pb.CheckMallocBDevRequest),
+        "CloneMallocBDev": _unary(servicer.CloneMallocBDev, pb.CloneMallocBDevRequest),
     }
     server.add_generic_rpc_handlers(
         (grpc.method_handlers_generic_handler(CONTROLLER_SERVICE, handlers),)

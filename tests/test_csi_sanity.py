@@ -395,3 +395,76 @@ class TestSnapshotSanity:
         finally:
             ctrl.DeleteVolume(csi.DeleteVolumeRequest(
                 volume_id="restore-src"), timeout=10)
+
+
+class TestRemoteSnapshots:
+    """Snapshots in remote mode ride the CloneMallocBDev oim.v0
+    extension through the registry proxy (docs/spec.md)."""
+
+    def test_snapshot_via_proxy(self, hipstored, tmp_path):  # noqa: F811
+        from oim_amd.controller import Controller, ControllerServer
+        from oim_amd.csidriver import RemoteBackend
+        from oim_amd.registry import MemRegistryDB, Registry, RegistryServer
+
+        registry = Registry(db=MemRegistryDB())
+        reg_server = RegistryServer(f"unix://{tmp_path}/reg.sock", registry)
+        reg_server.start()
+        controller = Controller(controller_id="snap-host",
+                                hipstored_socket=hipstored.socket_path)
+        ctrl_server = ControllerServer(f"unix://{tmp_path}/ctrl.sock",
+                                       controller)
+        ctrl_server.start()
+        registry.db.store(["snap-host", "address"],
+                          f"unix://{tmp_path}/ctrl.sock")
+        backend = RemoteBackend(registry_address=reg_server.addr(),
+                                controller_id="snap-host")
+        driver = OIMDriver(driver_name="remote.oim-amd.test", node_id="n0",
+                           endpoint=f"unix://{tmp_path}/csi.sock",
+                           backend=backend, mounter=Mounter(FakeExec()))
+        driver.start()
+        channel = grpc.insecure_channel(
+            grpc_target(f"unix://{tmp_path}/csi.sock"))
+        ctrl = CSIControllerStub(channel)
+        try:
+            caps = ctrl.ControllerGetCapabilities(
+                csi.ControllerGetCapabilitiesRequest(), timeout=10)
+            types = {cap.rpc.type for cap in caps.capabilities}
+            assert csi.CTRL_CAP_CREATE_DELETE_SNAPSHOT in types
+
+            create = csi.CreateVolumeRequest(name="rvol")
+            create.capacity_range.required_bytes = 1 << 20
+            create.volume_capabilities.add().CopyFrom(mount_cap())
+            ctrl.CreateVolume(create, timeout=10)
+            snap = ctrl.CreateSnapshot(
+                csi.CreateSnapshotRequest(source_volume_id="rvol",
+                                          name="r1"), timeout=10).snapshot
+            assert snap.snapshot_id == "csi-snap-r1"
+            assert snap.size_bytes == 1 << 20
+            # the clone exists on the daemon
+            from oim_amd import hipstore
+            with hipstore.Client(hipstored.socket_path) as client:
+                assert hipstore.get_bdevs(client, "csi-snap-r1")
+            # snapshot of unknown volume -> NOT_FOUND through the proxy
+            with pytest.raises(grpc.RpcError) as excinfo:
+                ctrl.CreateSnapshot(
+                    csi.CreateSnapshotRequest(source_volume_id="ghost",
+                                              name="r2"), timeout=10)
+            assert_code(excinfo, grpc.StatusCode.NOT_FOUND)
+            # restore + cleanup
+            request = csi.CreateVolumeRequest(name="rvol-restored")
+            request.volume_capabilities.add().CopyFrom(mount_cap())
+            request.volume_content_source.snapshot.snapshot_id = \
+                snap.snapshot_id
+            restored = ctrl.CreateVolume(request, timeout=10)
+            assert restored.volume.volume_id == "rvol-restored"
+            ctrl.DeleteSnapshot(csi.DeleteSnapshotRequest(
+                snapshot_id=snap.snapshot_id), timeout=10)
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(
+                volume_id="rvol-restored"), timeout=10)
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id="rvol"),
+                              timeout=10)
+        finally:
+            channel.close()
+            driver.stop()
+            ctrl_server.stop()
+            reg_server.stop()
