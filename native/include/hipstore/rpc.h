@@ -42,6 +42,7 @@ class RpcServer {
 
   void register_method(const std::string& name, RpcMethod fn);
   bool has_method(const std::string& name) const;
+  std::vector<std::string> method_names() const;
 
   void start();  // binds + spawns accept loop
   void stop();

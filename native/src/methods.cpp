@@ -346,16 +346,11 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
 
   // --- native extras -------------------------------------------------------
 
-  server->register_method("get_rpc_methods", [](const Json&) {
+  server->register_method("get_rpc_methods", [server](const Json&) {
+    // Live registry, not a hand-kept list (SPDK semantics: the actual
+    // dispatch table). Includes methods registered after this one.
     JsonArray out;
-    for (const char* name :
-         {"get_bdevs", "delete_bdev", "construct_malloc_bdev",
-          "construct_rbd_bdev", "start_nbd_disk", "get_nbd_disks",
-          "stop_nbd_disk", "construct_vhost_scsi_controller",
-          "add_vhost_scsi_lun", "remove_vhost_scsi_target",
-          "remove_vhost_controller", "get_vhost_controllers",
-          "get_rpc_methods", "perf_run", "perf_session_start",
-          "perf_session_step", "perf_session_stop"}) {
+    for (const std::string& name : server->method_names()) {
       out.push_back(Json(name));
     }
     return Json(std::move(out));

@@ -24,6 +24,15 @@ bool RpcServer::has_method(const std::string& name) const {
   return methods_.count(name) > 0;
 }
 
+std::vector<std::string> RpcServer::method_names() const {
+  std::lock_guard<std::mutex> lock(mutex_);
+  std::vector<std::string> names;
+  names.reserve(methods_.size());
+  for (const auto& [name, fn] : methods_) names.push_back(name);
+  return names;
+}
+
+
 void RpcServer::start() {
   listen_fd_ = socket(AF_UNIX, SOCK_STREAM | SOCK_CLOEXEC, 0);
   if (listen_fd_ < 0) throw std::runtime_error("rpc: socket() failed");
