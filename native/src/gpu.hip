@@ -1033,6 +1033,16 @@ class SharedService {
     memset(mirror_, 0, sizeof(mirror_));
     memset(used_, 0, sizeof(used_));
     memset(prefix_, 0, sizeof(prefix_));
+    // Warm both streams now so their hardware queues are allocated
+    // HERE (bdev creation / first attach) and not while many channel
+    // threads are concurrently creating queues and launching service
+    // kernels — the mixed-engine wedge (profiles/README.md) is
+    // consistent with a queue brought up in that storm never
+    // dispatching its kernel.
+    HIP_CHECK(hipMemsetAsync(exit_flag_dev_, 0, 4, stream_));
+    HIP_CHECK(hipMemsetAsync(exit_flag_dev_, 0, 4, config_stream_));
+    HIP_CHECK(hipStreamSynchronize(stream_));
+    HIP_CHECK(hipStreamSynchronize(config_stream_));
     dbg("svc:ctor-done");
   }
 
@@ -1495,6 +1505,13 @@ BdevPtr create_hbm_bdev(const std::string& name, uint64_t block_size,
   }
   if (device < 0 || device >= gpu_device_count()) {
     throw std::runtime_error("hipstore: bad device index");
+  }
+  if (persistent) {
+    // Bring the per-device shared service (and its hardware queues)
+    // up NOW, in this single-threaded moment, rather than lazily from
+    // whichever channel thread first crosses the per-queue cap — see
+    // the stream-warm note in the SharedService constructor.
+    SharedService::instance(device);
   }
   return std::make_shared<HbmBdev>(name, block_size, num_blocks, device,
                                    persistent);
