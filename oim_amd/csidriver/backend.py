@@ -38,6 +38,11 @@ class OIMBackend:
         cannot tell (remote mode: the oim.v0 API has no capacity RPC)."""
         return None
 
+    def list_volumes(self):
+        """[(volume_id, size_bytes), ...] or None when the backend
+        cannot enumerate (remote mode: no list RPC in oim.v0)."""
+        return None
+
     # --- snapshots (optional; local mode only — backed by hipstored's
     # bdev_clone, an HBM-rate device-side copy) ---------------------------
 

@@ -23,7 +23,7 @@ import grpc
 
 from .. import __version__
 from ..common.server import NonBlockingGRPCServer
-from ..common.tracing import LogServerInterceptor
+from ..common.tracing import LogServerInterceptor, strip_secrets_formatter
 from ..log import from_context
 from ..spec import csi_v1 as csi
 from ..spec.rpc_csi import (
@@ -204,6 +204,40 @@ class ControllerServer(CSIControllerServicer):
             cap.rpc.type = csi.CTRL_CAP_CREATE_DELETE_SNAPSHOT
             cap = response.capabilities.add()
             cap.rpc.type = csi.CTRL_CAP_LIST_SNAPSHOTS
+        if self.backend.list_volumes() is not None:
+            cap = response.capabilities.add()
+            cap.rpc.type = csi.CTRL_CAP_LIST_VOLUMES
+        return response
+
+    def ListVolumes(self, request, context):
+        volumes = self.backend.list_volumes()
+        if volumes is None:
+            context.abort(grpc.StatusCode.UNIMPLEMENTED,
+                          "volume listing is not available in this mode")
+        if request.max_entries < 0:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "max_entries must be non-negative")
+        start = 0
+        if request.starting_token:
+            try:
+                start = int(request.starting_token)
+            except ValueError:
+                context.abort(grpc.StatusCode.ABORTED,
+                              f"bad starting_token "
+                              f"{request.starting_token!r}")
+            if start < 0 or start > len(volumes):
+                context.abort(grpc.StatusCode.ABORTED,
+                              "starting_token out of range")
+        end = len(volumes)
+        if request.max_entries:
+            end = min(end, start + request.max_entries)
+        response = csi.ListVolumesResponse()
+        for volume_id, size in volumes[start:end]:
+            entry = response.entries.add()
+            entry.volume.volume_id = volume_id
+            entry.volume.capacity_bytes = size
+        if end < len(volumes):
+            response.next_token = str(end)
         return response
 
     def CreateSnapshot(self, request, context):
@@ -407,8 +441,12 @@ class OIMDriver:
         self.identity = IdentityServer(driver_name)
         self.controller = ControllerServer(backend)
         self.node = NodeServer(node_id, backend, mounter)
+        # CSI requests carry `secrets` maps (NodeStage, CreateVolume):
+        # payload logging must redact them (the reference used
+        # protosanitizer StripSecrets the same way).
         self.server = NonBlockingGRPCServer(
-            endpoint=endpoint, interceptors=[LogServerInterceptor()])
+            endpoint=endpoint,
+            interceptors=[LogServerInterceptor(strip_secrets_formatter)])
 
     def start(self) -> None:
         def register(server):

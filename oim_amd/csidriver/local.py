@@ -114,6 +114,15 @@ class LocalBackend(OIMBackend):
         free = int(info.get("free_bytes", 0))
         return free if free > 0 else None
 
+    def list_volumes(self):
+        out = []
+        with self._client() as client:
+            for bdev in hipstore.get_bdevs(client):
+                if bdev.name.startswith(self.SNAP_PREFIX):
+                    continue  # snapshots are listed via ListSnapshots
+                out.append((bdev.name, bdev.size_bytes))
+        return out
+
     # --- snapshots (bdev_clone-backed) --------------------------------------
 
     SNAP_PREFIX = "csi-snap-"

@@ -191,6 +191,18 @@ MESSAGES = [
         "GetCapacityResponse",
         [Field("available_capacity", 1, "int64")],
     ),
+    Message(
+        "ListVolumesRequest",
+        [Field("max_entries", 1, "int32"),
+         Field("starting_token", 2, "string")],
+    ),
+    Message(
+        "ListVolumesResponse",
+        [
+            Field("entries", 1, "ListVolumesResponse.Entry", repeated=True),
+            Field("next_token", 2, "string"),
+        ],
+    ),
     Message("ControllerGetCapabilitiesRequest", []),
     Message(
         "ControllerGetCapabilitiesResponse",
@@ -294,6 +306,7 @@ _NESTED = {
     "VolumeContentSource.SnapshotSource": "VolumeContentSourceSnapshotSource",
     "VolumeContentSource.VolumeSource": "VolumeContentSourceVolumeSource",
     "ListSnapshotsResponse.Entry": "ListSnapshotsResponseEntry",
+    "ListVolumesResponse.Entry": "ListVolumesResponseEntry",
     "VolumeCapability.BlockVolume": "VolumeCapabilityBlockVolume",
     "VolumeCapability.MountVolume": "VolumeCapabilityMountVolume",
     "VolumeCapability.AccessMode": "VolumeCapabilityAccessMode",
@@ -334,6 +347,7 @@ MESSAGES += [
     Message("VolumeContentSourceVolumeSource",
             [Field("volume_id", 1, "string")]),
     Message("ListSnapshotsResponseEntry", [Field("snapshot", 1, "Snapshot")]),
+    Message("ListVolumesResponseEntry", [Field("volume", 1, "Volume")]),
 ]
 
 SERVICES = [
@@ -362,6 +376,7 @@ SERVICES = [
              "DeleteSnapshotResponse"),
             ("ListSnapshots", "ListSnapshotsRequest",
              "ListSnapshotsResponse"),
+            ("ListVolumes", "ListVolumesRequest", "ListVolumesResponse"),
         ],
     ),
     Service(
@@ -389,7 +404,6 @@ SERVICES = [
 UNIMPLEMENTED_CONTROLLER_METHODS = (
     "ControllerPublishVolume",
     "ControllerUnpublishVolume",
-    "ListVolumes",
     "ControllerExpandVolume",
 )
 UNIMPLEMENTED_NODE_METHODS = (

@@ -67,6 +67,9 @@ class CSIControllerServicer:
     def ListSnapshots(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "ListSnapshots")
 
+    def ListVolumes(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "ListVolumes")
+
 
 class CSINodeServicer:
     def NodeGetVolumeStats(self, request, context):
@@ -119,6 +122,7 @@ def add_csi_controller_to_server(servicer, server):
                                  csi.DeleteSnapshotRequest),
         "ListSnapshots": _unary(servicer.ListSnapshots,
                                 csi.ListSnapshotsRequest),
+        "ListVolumes": _unary(servicer.ListVolumes, csi.ListVolumesRequest),
     }
     for name in csi.UNIMPLEMENTED_CONTROLLER_METHODS:
         handlers[name] = _unimplemented_handler(name)
@@ -182,6 +186,7 @@ class CSIControllerStub:
         self.DeleteSnapshot = method("DeleteSnapshot",
                                      csi.DeleteSnapshotResponse)
         self.ListSnapshots = method("ListSnapshots", csi.ListSnapshotsResponse)
+        self.ListVolumes = method("ListVolumes", csi.ListVolumesResponse)
 
 
 class CSINodeStub:

@@ -468,3 +468,43 @@ class TestRemoteSnapshots:
             driver.stop()
             ctrl_server.stop()
             reg_server.stop()
+
+
+class TestListVolumes:
+    """ListVolumes with CSI paging semantics (sanity: token paging,
+    ABORTED on a bad token)."""
+
+    def test_paged_listing(self, sanity_env):
+        ctrl = sanity_env["controller"]
+        names = [f"lv-{i}" for i in range(5)]
+        for name in names:
+            request = csi.CreateVolumeRequest(name=name)
+            request.capacity_range.required_bytes = 1 << 20
+            request.volume_capabilities.add().CopyFrom(mount_cap())
+            ctrl.CreateVolume(request, timeout=10)
+        try:
+            caps = ctrl.ControllerGetCapabilities(
+                csi.ControllerGetCapabilitiesRequest(), timeout=10)
+            assert csi.CTRL_CAP_LIST_VOLUMES in {
+                cap.rpc.type for cap in caps.capabilities}
+            seen = []
+            token = ""
+            while True:
+                reply = ctrl.ListVolumes(
+                    csi.ListVolumesRequest(max_entries=2,
+                                           starting_token=token), timeout=10)
+                seen.extend(e.volume.volume_id for e in reply.entries)
+                if not reply.next_token:
+                    break
+                token = reply.next_token
+            assert set(names) <= set(seen)
+            assert len(seen) == len(set(seen))  # no duplicates across pages
+            with pytest.raises(grpc.RpcError) as excinfo:
+                ctrl.ListVolumes(
+                    csi.ListVolumesRequest(starting_token="bogus"),
+                    timeout=10)
+            assert_code(excinfo, grpc.StatusCode.ABORTED)
+        finally:
+            for name in names:
+                ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id=name),
+                                  timeout=10)
