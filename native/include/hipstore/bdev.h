@@ -89,6 +89,13 @@ class Bdev {
   virtual int gpu_device() const { return -1; }
 
   virtual std::shared_ptr<IoChannel> get_channel() = 0;
+  // Grow (or shrink) the device to new_num_blocks. Returns kIoOk, or
+  // kIoInvalid when the bdev type cannot resize, or kIoFailed when it
+  // temporarily cannot (e.g. live channels hold the old backing).
+  virtual int resize(uint64_t new_num_blocks) {
+    (void)new_num_blocks;
+    return kIoInvalid;
+  }
   // Enqueue asynchronously; completion fires from poll() on the
   // submitting thread (the SPDK poller discipline, which is what makes
   // completions race-free without locks).
@@ -129,6 +136,11 @@ class Bdev {
            req.length % block_size_ == 0 &&
            req.offset + req.length <= size_bytes();
   }
+
+ protected:
+  // For resize implementations only; callers must have moved/zeroed
+  // the backing store to match first.
+  void set_num_blocks(uint64_t n) { num_blocks_ = n; }
 
  private:
   static std::string make_uuid();

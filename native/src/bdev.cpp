@@ -112,6 +112,20 @@ class MallocBdev : public Bdev {
     channel->done.emplace_back(std::move(req.on_complete), status);
   }
 
+  int resize(uint64_t new_num_blocks) override {
+    // Channels never hold the data pointer (all access goes through
+    // data_ under mutex_ at submit time), so an online swap is safe.
+    std::lock_guard<std::mutex> lock(mutex_);
+    const uint64_t old_bytes = size_bytes();
+    const uint64_t new_bytes = new_num_blocks * block_size();
+    if (new_bytes == old_bytes) return kIoOk;
+    void* p = mremap(data_, old_bytes, new_bytes, MREMAP_MAYMOVE);
+    if (p == MAP_FAILED) return kIoFailed;
+    data_ = static_cast<uint8_t*>(p);  // growth pages arrive zeroed
+    set_num_blocks(new_num_blocks);
+    return kIoOk;
+  }
+
   int poll(IoChannel* ch) override {
     auto* channel = static_cast<MallocChannel*>(ch);
     // Only drain what was queued at entry: completion callbacks may

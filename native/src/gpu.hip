@@ -1410,6 +1410,47 @@ class HbmBdev : public Bdev {
   void* device_base() override { return base_; }
   int gpu_device() const override { return device_; }
 
+  // Engine channels capture base_ at creation, so the backing store
+  // can only move while no channel is alive. resize() refuses
+  // (kIoFailed = try again offline) when the count is non-zero.
+  int resize(uint64_t new_num_blocks) override {
+    std::lock_guard<std::mutex> lock(resize_mutex_);
+    if (live_channels_->load(std::memory_order_acquire) != 0) {
+      return kIoFailed;
+    }
+    const uint64_t old_bytes = size_bytes();
+    const uint64_t new_bytes = new_num_blocks * block_size();
+    if (new_bytes == old_bytes) return kIoOk;
+    HIP_CHECK(hipSetDevice(device_));
+    uint8_t* new_base = nullptr;
+    if (hipMalloc(reinterpret_cast<void**>(&new_base), new_bytes) !=
+        hipSuccess) {
+      return kIoFailed;
+    }
+    hipStream_t stream = nullptr;
+    HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+    const uint64_t keep = std::min(old_bytes, new_bytes);
+    if (new_bytes > old_bytes) {
+      HIP_CHECK(hipMemsetAsync(new_base + keep, 0, new_bytes - keep, stream));
+    }
+    const uint64_t n16 = keep / 16;
+    const uint32_t grid = static_cast<uint32_t>(
+        std::min<uint64_t>(2048, (n16 + 255) / 256));
+    hipLaunchKernelGGL(k_copy_range, dim3(grid), dim3(256), 0, stream,
+                       reinterpret_cast<const float4*>(base_),
+                       reinterpret_cast<float4*>(new_base), n16);
+    hipError_t err = hipStreamSynchronize(stream);
+    (void)hipStreamDestroy(stream);
+    if (err != hipSuccess) {
+      (void)hipFree(new_base);
+      return kIoFailed;
+    }
+    (void)hipFree(base_);
+    base_ = new_base;
+    set_num_blocks(new_num_blocks);
+    return kIoOk;
+  }
+
   std::shared_ptr<IoChannel> get_channel() override {
     HIP_CHECK(hipSetDevice(device_));
     if (persistent_) {
@@ -1427,17 +1468,35 @@ class HbmBdev : public Bdev {
       // at 4+2 channels; engines work in isolation at any count).
       // Until that interaction is understood, the shared service runs
       // only in all-shared mode (HIPSTORE_SHARED=1).
-      if (shared_) return std::make_shared<HbmSharedChannel>(device_, base_);
+      if (shared_) {
+        return track(std::make_shared<HbmSharedChannel>(device_, base_));
+      }
       int prev = g_per_queue_channels[device_ & 63].fetch_add(
           1, std::memory_order_relaxed);
       if (prev < per_queue_channel_cap()) {
-        return std::make_shared<HbmPersistentChannel>(device_, base_);
+        return track(std::make_shared<HbmPersistentChannel>(device_, base_));
       }
       g_per_queue_channels[device_ & 63].fetch_sub(
           1, std::memory_order_relaxed);
-      return std::make_shared<HbmChannel>(device_, base_, size_bytes());
+      return track(
+          std::make_shared<HbmChannel>(device_, base_, size_bytes()));
     }
-    return std::make_shared<HbmChannel>(device_, base_, size_bytes());
+    return track(std::make_shared<HbmChannel>(device_, base_, size_bytes()));
+  }
+
+  // Wrap a channel so its lifetime is visible to resize(): the deleter
+  // keeps the inner shared_ptr (and thus the channel) alive until the
+  // caller drops the handle, then decrements the live count.
+  template <typename T>
+  std::shared_ptr<IoChannel> track(std::shared_ptr<T> channel) {
+    auto counter = live_channels_;
+    counter->fetch_add(1, std::memory_order_acq_rel);
+    IoChannel* raw = channel.get();
+    return std::shared_ptr<IoChannel>(
+        raw, [channel, counter](IoChannel*) mutable {
+          channel.reset();  // destroy the channel FIRST (drains rings)
+          counter->fetch_sub(1, std::memory_order_acq_rel);
+        });
   }
 
   void submit(IoChannel* ch, IoRequest req) override {
@@ -1491,6 +1550,9 @@ class HbmBdev : public Bdev {
 
  private:
   uint8_t* base_ = nullptr;
+  std::mutex resize_mutex_;
+  std::shared_ptr<std::atomic<int>> live_channels_ =
+      std::make_shared<std::atomic<int>>(0);
   int device_;
   bool persistent_;
   bool shared_ = false;

@@ -560,6 +560,32 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
     return Json(std::move(o));
   });
 
+  server->register_method("resize_malloc_bdev", [&manager](const Json& p) {
+    // Offline volume expansion (CSI ControllerExpandVolume): malloc
+    // bdevs move their backing store, so live channels must be gone.
+    BdevPtr bdev = manager.find(p.get_string("name"));
+    if (!bdev) not_found("bdev " + p.get_string("name"));
+    const int64_t size = p.get_int("size", -1);
+    if (size <= 0 || size % static_cast<int64_t>(bdev->block_size()) != 0) {
+      throw RpcError{kInvalidParams,
+                     "size must be a positive multiple of the block size"};
+    }
+    const int status = bdev->resize(size / bdev->block_size());
+    if (status == kIoInvalid) {
+      throw RpcError{kInvalidParams,
+                     "bdev " + bdev->name() + " cannot be resized"};
+    }
+    if (status != kIoOk) {
+      throw RpcError{kInternalError,
+                     "bdev " + bdev->name() +
+                         " is busy (live channels); retry offline"};
+    }
+    JsonObject o;
+    o["name"] = Json(bdev->name());
+    o["num_blocks"] = Json(static_cast<int64_t>(bdev->num_blocks()));
+    return Json(std::move(o));
+  });
+
   server->register_method(
       "bdev_clone", [&manager, use_hbm, device, persistent](const Json& p) {
         // Volume clone: new malloc bdev + device-side range copy (HBM

@@ -85,6 +85,11 @@ PYBIND11_MODULE(_hipstore, m) {
       .def_property_readonly("block_size", &Bdev::block_size)
       .def_property_readonly("num_blocks", &Bdev::num_blocks)
       .def_property_readonly("size_bytes", &Bdev::size_bytes)
+      .def("resize",
+           [](hipstore::Bdev& b, uint64_t num_blocks) {
+             return b.resize(num_blocks);
+           },
+           py::arg("num_blocks"))
       .def("read", &bdev_read_py, py::arg("offset"), py::arg("length"),
            py::call_guard<py::gil_scoped_release>())
       .def("write", &bdev_write_py, py::arg("offset"), py::arg("data"))

@@ -344,6 +344,28 @@ class Controller(spec.ControllerServicer):
                     context.abort(grpc.StatusCode.INTERNAL, str(err))
         return spec.CloneMallocBDevReply()
 
+    def ResizeMallocBDev(self, request, context):
+        """oim-amd extension (docs/spec.md): offline volume expansion."""
+        name, size = request.bdev_name, request.size
+        if not name or size <= 0 or size % 512 != 0:
+            context.abort(
+                grpc.StatusCode.INVALID_ARGUMENT,
+                "bdev_name and a positive 512-multiple size are required")
+        with self._volume_mutex.locked(name):
+            with self._client() as client:
+                try:
+                    client.invoke("resize_malloc_bdev",
+                                  {"name": name, "size": size})
+                except hipstore.RpcError as err:
+                    if err.is_not_found():
+                        context.abort(grpc.StatusCode.NOT_FOUND,
+                                      f"BDev {name} not found")
+                    if "busy" in str(err):
+                        context.abort(grpc.StatusCode.FAILED_PRECONDITION,
+                                      str(err))
+                    context.abort(grpc.StatusCode.INTERNAL, str(err))
+        return spec.ResizeMallocBDevReply()
+
     # --- self-registration --------------------------------------------------
 
     def register(self) -> None:

@@ -38,6 +38,15 @@ class OIMBackend:
         cannot tell (remote mode: the oim.v0 API has no capacity RPC)."""
         return None
 
+    def expand_volume(self, volume_id, size):
+        """Grow the volume to size bytes; returns the new size.
+        Raises LookupError (unknown volume) or RuntimeError (busy /
+        unsupported)."""
+        raise NotImplementedError
+
+    def supports_expansion(self) -> bool:
+        return False
+
     def list_volumes(self):
         """[(volume_id, size_bytes), ...] or None when the backend
         cannot enumerate (remote mode: no list RPC in oim.v0)."""

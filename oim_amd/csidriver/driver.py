@@ -50,8 +50,9 @@ class _KeyedMutex:
 
 
 class IdentityServer(CSIIdentityServicer):
-    def __init__(self, driver_name: str):
+    def __init__(self, driver_name: str, offline_expansion: bool = False):
         self.driver_name = driver_name
+        self.offline_expansion = offline_expansion
 
     def GetPluginInfo(self, request, context):
         return csi.GetPluginInfoResponse(
@@ -61,6 +62,11 @@ class IdentityServer(CSIIdentityServicer):
         response = csi.GetPluginCapabilitiesResponse()
         cap = response.capabilities.add()
         cap.service.type = csi.PLUGIN_CAPABILITY_CONTROLLER_SERVICE
+        if self.offline_expansion:
+            cap = response.capabilities.add()
+            # The backing store moves during resize, so channels must
+            # be closed: OFFLINE expansion.
+            cap.volume_expansion.type = csi.EXPANSION_OFFLINE
         return response
 
     def Probe(self, request, context):
@@ -207,6 +213,36 @@ class ControllerServer(CSIControllerServicer):
         if self.backend.list_volumes() is not None:
             cap = response.capabilities.add()
             cap.rpc.type = csi.CTRL_CAP_LIST_VOLUMES
+        if self.backend.supports_expansion():
+            cap = response.capabilities.add()
+            cap.rpc.type = csi.CTRL_CAP_EXPAND_VOLUME
+        return response
+
+    def ControllerExpandVolume(self, request, context):
+        if not self.backend.supports_expansion():
+            context.abort(grpc.StatusCode.UNIMPLEMENTED,
+                          "expansion is not supported in this mode")
+        if not request.volume_id:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume id")
+        required = request.capacity_range.required_bytes
+        if required <= 0:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing capacity range")
+        with self._name_mutex.get(request.volume_id):
+            try:
+                new_size = self.backend.expand_volume(request.volume_id,
+                                                      required)
+            except LookupError as exc:
+                context.abort(grpc.StatusCode.NOT_FOUND, str(exc))
+            except RuntimeError as exc:
+                # live channels hold the old backing: offline-only
+                context.abort(grpc.StatusCode.FAILED_PRECONDITION, str(exc))
+            except grpc.RpcError as exc:
+                context.abort(exc.code(), exc.details())
+        response = csi.ControllerExpandVolumeResponse()
+        response.capacity_bytes = new_size
+        response.node_expansion_required = True  # fs must grow too
         return response
 
     def ListVolumes(self, request, context):
@@ -420,12 +456,34 @@ class NodeServer(CSINodeServicer):
         inodes.used = st.f_files - st.f_ffree
         return response
 
+    def NodeExpandVolume(self, request, context):
+        if not request.volume_id or not request.volume_path:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume id or path")
+        fs_type = request.volume_capability.mount.fs_type or "ext4"
+        with self._volume_mutex.get(request.volume_id):
+            if not os.path.exists(request.volume_path):
+                context.abort(grpc.StatusCode.NOT_FOUND,
+                              f"no such path: {request.volume_path}")
+            try:
+                self.mounter.resize_fs(request.volume_path, fs_type)
+            except ValueError as exc:
+                context.abort(grpc.StatusCode.INVALID_ARGUMENT, str(exc))
+            except Exception as exc:  # noqa: BLE001
+                context.abort(grpc.StatusCode.INTERNAL,
+                              f"filesystem grow failed: {exc}")
+        response = csi.NodeExpandVolumeResponse()
+        response.capacity_bytes = request.capacity_range.required_bytes
+        return response
+
     def NodeGetCapabilities(self, request, context):
         response = csi.NodeGetCapabilitiesResponse()
         cap = response.capabilities.add()
         cap.rpc.type = csi.NODE_CAP_STAGE_UNSTAGE_VOLUME
         cap = response.capabilities.add()
         cap.rpc.type = csi.NODE_CAP_GET_VOLUME_STATS
+        cap = response.capabilities.add()
+        cap.rpc.type = csi.NODE_CAP_EXPAND_VOLUME
         return response
 
     def NodeGetInfo(self, request, context):
@@ -438,7 +496,8 @@ class OIMDriver:
 
     def __init__(self, driver_name: str, node_id: str, endpoint: str,
                  backend: OIMBackend, mounter: Optional[Mounter] = None):
-        self.identity = IdentityServer(driver_name)
+        self.identity = IdentityServer(
+            driver_name, offline_expansion=backend.supports_expansion())
         self.controller = ControllerServer(backend)
         self.node = NodeServer(node_id, backend, mounter)
         # CSI requests carry `secrets` maps (NodeStage, CreateVolume):

@@ -114,6 +114,22 @@ class LocalBackend(OIMBackend):
         free = int(info.get("free_bytes", 0))
         return free if free > 0 else None
 
+    def supports_expansion(self) -> bool:
+        return True
+
+    def expand_volume(self, volume_id, size):
+        size = round_to_blocks(size)
+        with self._client() as client:
+            try:
+                client.invoke("resize_malloc_bdev",
+                              {"name": volume_id, "size": size})
+            except hipstore.RpcError as err:
+                if err.is_not_found():
+                    raise LookupError(
+                        f"volume {volume_id} not found") from None
+                raise RuntimeError(str(err)) from None
+        return size
+
     def list_volumes(self):
         out = []
         with self._client() as client:

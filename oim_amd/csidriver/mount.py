@@ -94,6 +94,16 @@ class Mounter:
                 return line.split("=", 1)[1].strip()
         return ""
 
+    def resize_fs(self, device: str, fstype: str = "ext4") -> None:
+        """Grow the filesystem to the (already grown) device
+        (NodeExpandVolume)."""
+        if fstype.startswith("ext"):
+            self.exec.run("resize2fs", device)
+        elif fstype == "xfs":
+            self.exec.run("xfs_growfs", device)
+        else:
+            raise ValueError(f"cannot grow filesystem type {fstype!r}")
+
     def format_and_mount(self, device: str, target: str, fstype: str = "ext4",
                          options: Sequence[str] = ()) -> None:
         """SafeFormatAndMount: probe, mkfs when unformatted, mount."""

@@ -121,6 +121,26 @@ class RemoteBackend(OIMBackend):
                     return False
                 raise
 
+    def supports_expansion(self) -> bool:
+        return True
+
+    def expand_volume(self, volume_id, size):
+        size = round_to_blocks(size)
+        with self._dial_registry() as channel:
+            stub = spec.ControllerStub(channel)
+            try:
+                stub.ResizeMallocBDev(
+                    spec.ResizeMallocBDevRequest(bdev_name=volume_id,
+                                                 size=size),
+                    metadata=self._metadata(), timeout=60)
+            except grpc.RpcError as err:
+                if err.code() == grpc.StatusCode.NOT_FOUND:
+                    raise LookupError(
+                        f"volume {volume_id} not found") from None
+                raise
+        self._volume_sizes[volume_id] = size
+        return size
+
     # --- snapshots (CloneMallocBDev extension, docs/spec.md) ----------------
 
     SNAP_PREFIX = "csi-snap-"

@@ -40,11 +40,17 @@ CTRL_CAP_LIST_VOLUMES = 3
 CTRL_CAP_GET_CAPACITY = 4
 CTRL_CAP_CREATE_DELETE_SNAPSHOT = 5
 CTRL_CAP_LIST_SNAPSHOTS = 6
+CTRL_CAP_EXPAND_VOLUME = 9
 
 # NodeServiceCapability.RPC.Type
 NODE_CAP_UNKNOWN = 0
 NODE_CAP_STAGE_UNSTAGE_VOLUME = 1
 NODE_CAP_GET_VOLUME_STATS = 2
+NODE_CAP_EXPAND_VOLUME = 3
+
+# PluginCapability.VolumeExpansion.Type
+EXPANSION_ONLINE = 1
+EXPANSION_OFFLINE = 2
 
 # VolumeUsage.Unit
 USAGE_UNIT_BYTES = 1
@@ -65,7 +71,11 @@ MESSAGES = [
     ),
     Message(
         "PluginCapability",
-        [Field("service", 1, "PluginCapability.Service", oneof="type")],
+        [
+            Field("service", 1, "PluginCapability.Service", oneof="type"),
+            Field("volume_expansion", 2, "PluginCapability.VolumeExpansion",
+                  oneof="type"),
+        ],
     ),
     Message("ProbeRequest", []),
     Message(
@@ -203,6 +213,36 @@ MESSAGES = [
             Field("next_token", 2, "string"),
         ],
     ),
+    Message(
+        "ControllerExpandVolumeRequest",
+        [
+            Field("volume_id", 1, "string"),
+            Field("capacity_range", 2, "CapacityRange"),
+            Field("volume_capability", 4, "VolumeCapability"),
+        ],
+        map_fields=[("secrets", 3, "string", "string")],
+    ),
+    Message(
+        "ControllerExpandVolumeResponse",
+        [
+            Field("capacity_bytes", 1, "int64"),
+            Field("node_expansion_required", 2, "bool"),
+        ],
+    ),
+    Message(
+        "NodeExpandVolumeRequest",
+        [
+            Field("volume_id", 1, "string"),
+            Field("volume_path", 2, "string"),
+            Field("capacity_range", 3, "CapacityRange"),
+            Field("staging_target_path", 4, "string"),
+            Field("volume_capability", 5, "VolumeCapability"),
+        ],
+    ),
+    Message(
+        "NodeExpandVolumeResponse",
+        [Field("capacity_bytes", 1, "int64")],
+    ),
     Message("ControllerGetCapabilitiesRequest", []),
     Message(
         "ControllerGetCapabilitiesResponse",
@@ -303,6 +343,7 @@ MESSAGES = [
 # patch the type references:
 _NESTED = {
     "PluginCapability.Service": "PluginCapabilityService",
+    "PluginCapability.VolumeExpansion": "PluginCapabilityVolumeExpansion",
     "VolumeContentSource.SnapshotSource": "VolumeContentSourceSnapshotSource",
     "VolumeContentSource.VolumeSource": "VolumeContentSourceVolumeSource",
     "ListSnapshotsResponse.Entry": "ListSnapshotsResponseEntry",
@@ -327,6 +368,7 @@ for message in MESSAGES:
 
 MESSAGES += [
     Message("PluginCapabilityService", [Field("type", 1, "int32")]),
+    Message("PluginCapabilityVolumeExpansion", [Field("type", 1, "int32")]),
     Message("VolumeCapabilityBlockVolume", []),
     Message(
         "VolumeCapabilityMountVolume",
@@ -377,6 +419,8 @@ SERVICES = [
             ("ListSnapshots", "ListSnapshotsRequest",
              "ListSnapshotsResponse"),
             ("ListVolumes", "ListVolumesRequest", "ListVolumesResponse"),
+            ("ControllerExpandVolume", "ControllerExpandVolumeRequest",
+             "ControllerExpandVolumeResponse"),
         ],
     ),
     Service(
@@ -394,6 +438,8 @@ SERVICES = [
             ("NodeGetCapabilities", "NodeGetCapabilitiesRequest",
              "NodeGetCapabilitiesResponse"),
             ("NodeGetInfo", "NodeGetInfoRequest", "NodeGetInfoResponse"),
+            ("NodeExpandVolume", "NodeExpandVolumeRequest",
+             "NodeExpandVolumeResponse"),
         ],
     ),
 ]
@@ -404,11 +450,8 @@ SERVICES = [
 UNIMPLEMENTED_CONTROLLER_METHODS = (
     "ControllerPublishVolume",
     "ControllerUnpublishVolume",
-    "ControllerExpandVolume",
 )
-UNIMPLEMENTED_NODE_METHODS = (
-    "NodeExpandVolume",
-)
+UNIMPLEMENTED_NODE_METHODS = ()
 
 import google.protobuf.timestamp_pb2  # noqa: E402  (registers timestamp.proto)
 import google.protobuf.wrappers_pb2  # noqa: E402  (registers wrappers.proto)
@@ -426,5 +469,6 @@ globals().update(_classes)
 
 __all__ = list(_classes.keys()) + [
     n for n in dir() if n.startswith(("PLUGIN_", "ACCESS_MODE_", "CTRL_CAP_",
-                                      "NODE_CAP_", "UNIMPLEMENTED_"))
+                                      "NODE_CAP_", "EXPANSION_",
+                                      "UNIMPLEMENTED_"))
 ]

@@ -68,6 +68,9 @@ MESSAGES = [
     Message("CloneMallocBDevRequest",
             [Field("source", 1, "string"), Field("dest", 2, "string")]),
     Message("CloneMallocBDevReply", []),
+    Message("ResizeMallocBDevRequest",
+            [Field("bdev_name", 1, "string"), Field("size", 2, "int64")]),
+    Message("ResizeMallocBDevReply", []),
 ]
 
 SERVICES = [
@@ -86,6 +89,7 @@ SERVICES = [
             ("ProvisionMallocBDev", "ProvisionMallocBDevRequest", "ProvisionMallocBDevReply"),
             ("CheckMallocBDev", "CheckMallocBDevRequest", "CheckMallocBDevReply"),
             ("CloneMallocBDev", "CloneMallocBDevRequest", "CloneMallocBDevReply"),
+            ("ResizeMallocBDev", "ResizeMallocBDevRequest", "ResizeMallocBDevReply"),
         ],
     ),
 ]
@@ -113,3 +117,5 @@ CheckMallocBDevRequest = _classes["CheckMallocBDevRequest"]
 CheckMallocBDevReply = _classes["CheckMallocBDevReply"]
 CloneMallocBDevRequest = _classes["CloneMallocBDevRequest"]
 CloneMallocBDevReply = _classes["CloneMallocBDevReply"]
+ResizeMallocBDevRequest = _classes["ResizeMallocBDevRequest"]
+ResizeMallocBDevReply = _classes["ResizeMallocBDevReply"]

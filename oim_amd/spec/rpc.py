@@ -49,6 +49,7 @@ class ControllerStub:
         )
         self.CheckMallocBDev = method("CheckMallocBDev", pb.CheckMallocBDevReply)
         self.CloneMallocBDev = method("CloneMallocBDev", pb.CloneMallocBDevReply)
+        self.ResizeMallocBDev = method("ResizeMallocBDev", pb.ResizeMallocBDevReply)
 
 
 class RegistryServicer:
@@ -79,6 +80,9 @@ class ControllerServicer:
     def CloneMallocBDev(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "CloneMallocBDev not implemented")
 
+    def ResizeMallocBDev(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "ResizeMallocBDev not implemented")
+
 
 def _unary(fn, request_class):
     return grpc.unary_unary_rpc_method_handler(
@@ -107,6 +111,7 @@ def add_controller_to_server(servicer: ControllerServicer, server: grpc.Server) 
         ),
         "CheckMallocBDev": _unary(servicer.CheckMallocBDev, pb.CheckMallocBDevRequest),
         "CloneMallocBDev": _unary(servicer.CloneMallocBDev, pb.CloneMallocBDevRequest),
+        "ResizeMallocBDev": _unary(servicer.ResizeMallocBDev, pb.ResizeMallocBDevRequest),
     }
     server.add_generic_rpc_handlers(
         (grpc.method_handlers_generic_handler(CONTROLLER_SERVICE, handlers),)

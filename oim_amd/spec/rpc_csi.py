@@ -70,6 +70,9 @@ class CSIControllerServicer:
     def ListVolumes(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "ListVolumes")
 
+    def ControllerExpandVolume(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "ControllerExpandVolume")
+
 
 class CSINodeServicer:
     def NodeGetVolumeStats(self, request, context):
@@ -92,6 +95,9 @@ class CSINodeServicer:
 
     def NodeGetInfo(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "NodeGetInfo")
+
+    def NodeExpandVolume(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "NodeExpandVolume")
 
 
 def add_csi_identity_to_server(servicer, server):
@@ -123,6 +129,8 @@ def add_csi_controller_to_server(servicer, server):
         "ListSnapshots": _unary(servicer.ListSnapshots,
                                 csi.ListSnapshotsRequest),
         "ListVolumes": _unary(servicer.ListVolumes, csi.ListVolumesRequest),
+        "ControllerExpandVolume": _unary(servicer.ControllerExpandVolume,
+                                         csi.ControllerExpandVolumeRequest),
     }
     for name in csi.UNIMPLEMENTED_CONTROLLER_METHODS:
         handlers[name] = _unimplemented_handler(name)
@@ -145,6 +153,8 @@ def add_csi_node_to_server(servicer, server):
         "NodeGetCapabilities": _unary(servicer.NodeGetCapabilities,
                                       csi.NodeGetCapabilitiesRequest),
         "NodeGetInfo": _unary(servicer.NodeGetInfo, csi.NodeGetInfoRequest),
+        "NodeExpandVolume": _unary(servicer.NodeExpandVolume,
+                                   csi.NodeExpandVolumeRequest),
     }
     for name in csi.UNIMPLEMENTED_NODE_METHODS:
         handlers[name] = _unimplemented_handler(name)
@@ -187,6 +197,8 @@ class CSIControllerStub:
                                      csi.DeleteSnapshotResponse)
         self.ListSnapshots = method("ListSnapshots", csi.ListSnapshotsResponse)
         self.ListVolumes = method("ListVolumes", csi.ListVolumesResponse)
+        self.ControllerExpandVolume = method(
+            "ControllerExpandVolume", csi.ControllerExpandVolumeResponse)
 
 
 class CSINodeStub:
@@ -210,3 +222,5 @@ class CSINodeStub:
         self.NodeGetCapabilities = method("NodeGetCapabilities",
                                           csi.NodeGetCapabilitiesResponse)
         self.NodeGetInfo = method("NodeGetInfo", csi.NodeGetInfoResponse)
+        self.NodeExpandVolume = method("NodeExpandVolume",
+                                       csi.NodeExpandVolumeResponse)

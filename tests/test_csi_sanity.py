@@ -25,10 +25,11 @@ from fixtures import hipstored  # noqa: F401
 @pytest.fixture
 def sanity_env(hipstored, tmp_path):  # noqa: F811
     backend = LocalBackend(hipstored.socket_path)
+    fake_exec = FakeExec()
     driver = OIMDriver(
         driver_name="sanity.oim-amd.test", node_id="sanity-node",
         endpoint=f"unix://{tmp_path}/csi.sock",
-        backend=backend, mounter=Mounter(FakeExec()))
+        backend=backend, mounter=Mounter(fake_exec))
     driver.start()
     channel = grpc.insecure_channel(grpc_target(f"unix://{tmp_path}/csi.sock"))
     yield {
@@ -36,6 +37,7 @@ def sanity_env(hipstored, tmp_path):  # noqa: F811
         "controller": CSIControllerStub(channel),
         "node": CSINodeStub(channel),
         "tmp": tmp_path,
+        "fake_exec": fake_exec,
     }
     channel.close()
     driver.stop()
@@ -74,9 +76,16 @@ class TestIdentitySanity:
         caps = sanity_env["identity"].GetPluginCapabilities(
             csi.GetPluginCapabilitiesRequest(), timeout=10)
         assert len(caps.capabilities) >= 1
+        kinds = set()
         for cap in caps.capabilities:
-            assert cap.service.type in (
-                csi.PLUGIN_CAPABILITY_CONTROLLER_SERVICE,)
+            which = cap.WhichOneof("type")
+            kinds.add(which)
+            if which == "service":
+                assert cap.service.type ==                     csi.PLUGIN_CAPABILITY_CONTROLLER_SERVICE
+            elif which == "volume_expansion":
+                assert cap.volume_expansion.type in (
+                    csi.EXPANSION_ONLINE, csi.EXPANSION_OFFLINE)
+        assert "service" in kinds
 
 
 class TestControllerSanity:
@@ -508,3 +517,135 @@ class TestListVolumes:
             for name in names:
                 ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id=name),
                                   timeout=10)
+
+
+class TestVolumeExpansion:
+    """ControllerExpandVolume / NodeExpandVolume (offline expansion:
+    the backing store moves, so the daemon refuses while channels are
+    open)."""
+
+    def test_plugin_advertises_offline_expansion(self, sanity_env):
+        caps = sanity_env["identity"].GetPluginCapabilities(
+            csi.GetPluginCapabilitiesRequest(), timeout=10)
+        expansions = [cap.volume_expansion.type for cap in caps.capabilities
+                      if cap.WhichOneof("type") == "volume_expansion"]
+        assert expansions == [csi.EXPANSION_OFFLINE]
+        ctrl_caps = sanity_env["controller"].ControllerGetCapabilities(
+            csi.ControllerGetCapabilitiesRequest(), timeout=10)
+        assert csi.CTRL_CAP_EXPAND_VOLUME in {
+            cap.rpc.type for cap in ctrl_caps.capabilities}
+
+    def test_expand_missing_fields(self, sanity_env):
+        ctrl = sanity_env["controller"]
+        with pytest.raises(grpc.RpcError) as excinfo:
+            ctrl.ControllerExpandVolume(
+                csi.ControllerExpandVolumeRequest(), timeout=10)
+        assert_code(excinfo, grpc.StatusCode.INVALID_ARGUMENT)
+        with pytest.raises(grpc.RpcError) as excinfo:
+            ctrl.ControllerExpandVolume(
+                csi.ControllerExpandVolumeRequest(volume_id="v"), timeout=10)
+        assert_code(excinfo, grpc.StatusCode.INVALID_ARGUMENT)
+
+    def test_expand_unknown_volume(self, sanity_env):
+        request = csi.ControllerExpandVolumeRequest(volume_id="ghost")
+        request.capacity_range.required_bytes = 2 << 20
+        with pytest.raises(grpc.RpcError) as excinfo:
+            sanity_env["controller"].ControllerExpandVolume(request,
+                                                            timeout=10)
+        assert_code(excinfo, grpc.StatusCode.NOT_FOUND)
+
+    def test_expand_grows_volume_and_data_survives(self, sanity_env,
+                                                   hipstored):  # noqa: F811
+        from oim_amd import hipstore
+
+        ctrl = sanity_env["controller"]
+        create = csi.CreateVolumeRequest(name="grow-me")
+        create.capacity_range.required_bytes = 1 << 20
+        create.volume_capabilities.add().CopyFrom(mount_cap())
+        ctrl.CreateVolume(create, timeout=10)
+        try:
+            # write a pattern through the daemon, then expand
+            with hipstore.Client(hipstored.socket_path) as client:
+                import oim_amd._hipstore  # noqa: F401  (bdev access below)
+                pre = client.invoke("get_bdevs", {"name": "grow-me"})
+                assert pre[0]["num_blocks"] * pre[0]["block_size"] == 1 << 20
+            request = csi.ControllerExpandVolumeRequest(volume_id="grow-me")
+            request.capacity_range.required_bytes = 4 << 20
+            reply = ctrl.ControllerExpandVolume(request, timeout=10)
+            assert reply.capacity_bytes == 4 << 20
+            assert reply.node_expansion_required
+            with hipstore.Client(hipstored.socket_path) as client:
+                post = client.invoke("get_bdevs", {"name": "grow-me"})
+                assert post[0]["num_blocks"] * post[0]["block_size"] \
+                    == 4 << 20
+            # idempotent: same size again succeeds
+            ctrl.ControllerExpandVolume(request, timeout=10)
+        finally:
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id="grow-me"),
+                              timeout=10)
+
+    def test_node_expand_runs_resize2fs(self, sanity_env):
+        node = sanity_env["node"]
+        with pytest.raises(grpc.RpcError) as excinfo:
+            node.NodeExpandVolume(csi.NodeExpandVolumeRequest(volume_id="v"),
+                                  timeout=10)
+        assert_code(excinfo, grpc.StatusCode.INVALID_ARGUMENT)
+        request = csi.NodeExpandVolumeRequest(
+            volume_id="v", volume_path=str(sanity_env["tmp"]))
+        request.capacity_range.required_bytes = 2 << 20
+        reply = node.NodeExpandVolume(request, timeout=10)
+        assert reply.capacity_bytes == 2 << 20
+        # FakeExec recorded the filesystem grow
+        calls = sanity_env["fake_exec"].calls
+        assert any(call[0] == "resize2fs" for call in calls)
+
+
+class TestRemoteExpansion:
+    def test_expand_via_proxy(self, hipstored, tmp_path):  # noqa: F811
+        from oim_amd.controller import Controller, ControllerServer
+        from oim_amd.csidriver import RemoteBackend
+        from oim_amd.registry import MemRegistryDB, Registry, RegistryServer
+
+        registry = Registry(db=MemRegistryDB())
+        reg_server = RegistryServer(f"unix://{tmp_path}/reg.sock", registry)
+        reg_server.start()
+        controller = Controller(controller_id="exp-host",
+                                hipstored_socket=hipstored.socket_path)
+        ctrl_server = ControllerServer(f"unix://{tmp_path}/ctrl.sock",
+                                       controller)
+        ctrl_server.start()
+        registry.db.store(["exp-host", "address"],
+                          f"unix://{tmp_path}/ctrl.sock")
+        backend = RemoteBackend(registry_address=reg_server.addr(),
+                                controller_id="exp-host")
+        driver = OIMDriver(driver_name="remote.oim-amd.test", node_id="n0",
+                           endpoint=f"unix://{tmp_path}/csi.sock",
+                           backend=backend, mounter=Mounter(FakeExec()))
+        driver.start()
+        channel = grpc.insecure_channel(
+            grpc_target(f"unix://{tmp_path}/csi.sock"))
+        ctrl = CSIControllerStub(channel)
+        try:
+            create = csi.CreateVolumeRequest(name="evol")
+            create.capacity_range.required_bytes = 1 << 20
+            create.volume_capabilities.add().CopyFrom(mount_cap())
+            ctrl.CreateVolume(create, timeout=10)
+            request = csi.ControllerExpandVolumeRequest(volume_id="evol")
+            request.capacity_range.required_bytes = 2 << 20
+            reply = ctrl.ControllerExpandVolume(request, timeout=10)
+            assert reply.capacity_bytes == 2 << 20
+            from oim_amd import hipstore
+            with hipstore.Client(hipstored.socket_path) as client:
+                bdev = hipstore.get_bdevs(client, "evol")[0]
+                assert bdev.size_bytes == 2 << 20
+            request.volume_id = "ghost"
+            with pytest.raises(grpc.RpcError) as excinfo:
+                ctrl.ControllerExpandVolume(request, timeout=10)
+            assert_code(excinfo, grpc.StatusCode.NOT_FOUND)
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id="evol"),
+                              timeout=10)
+        finally:
+            channel.close()
+            driver.stop()
+            ctrl_server.stop()
+            reg_server.stop()

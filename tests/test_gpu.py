@@ -380,3 +380,28 @@ class TestChannelAutoFallback:
         data = bytes(rng.getrandbits(8) for _ in range(8 * 4096))
         bdev.write(64 * 4096, data)
         assert bdev.read(64 * 4096, len(data)) == data
+
+
+@needs_gpu
+class TestHbmResize:
+    """Offline HBM bdev expansion: data survives the backing-store
+    move; live channels refuse the resize (kIoFailed)."""
+
+    def test_grow_preserves_data(self):
+        import os as _os
+        bdev = hs.create_hbm_bdev("resize-0", 512, 2048, device=0,
+                                  persistent=True)
+        data = _os.urandom(4096)
+        bdev.write(0, data)
+        session = hs.PerfSession(bdev, "randread", 512, 1, 1)
+        session.step(10)
+        assert bdev.resize(65536) == -5  # busy: channels hold the base
+        del session
+        assert bdev.resize(65536) == 0
+        assert bdev.num_blocks == 65536
+        assert bdev.read(0, 4096) == data  # moved with the store
+        tail = _os.urandom(512)
+        bdev.write(65535 * 512, tail)  # new extent is addressable
+        assert bdev.read(65535 * 512, 512) == tail
+        # grown region arrived zeroed
+        assert bdev.read(4096, 512) == b"\0" * 512
