@@ -335,6 +335,13 @@ class VhostUserScsiDev {
     (void)!::send(conn, buf, 12 + size, MSG_NOSIGNAL);
   }
 
+  static void close_unconsumed(Msg* msg) {
+    for (int& fd : msg->fds) {
+      if (fd >= 0) ::close(fd);
+      fd = -1;
+    }
+  }
+
   void serve(int conn) {
     Msg msg;
     while (!stopping_.load(std::memory_order_relaxed)) {
@@ -343,10 +350,12 @@ class VhostUserScsiDev {
       if (fds[1].revents) return;
       if (!fds[0].revents) continue;
       msg = Msg{};
-      if (!recv_msg(conn, &msg)) return;
-      if (!handle(conn, msg)) return;
-      for (int fd : msg.fds)
-        if (fd >= 0) ::close(fd);  // only unconsumed fds remain
+      bool ok = recv_msg(conn, &msg);
+      if (ok) ok = handle(conn, msg);
+      // Close fds the handler did not consume — ALSO on failure paths
+      // (a malformed message with fds attached must not leak them).
+      close_unconsumed(&msg);
+      if (!ok) return;
     }
   }
 
