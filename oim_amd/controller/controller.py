@@ -354,12 +354,20 @@ class Controller(spec.ControllerServicer):
         with self._volume_mutex.locked(name):
             with self._client() as client:
                 try:
-                    client.invoke("resize_malloc_bdev",
-                                  {"name": name, "size": size})
+                    bdevs = hipstore.get_bdevs(client, name)
                 except hipstore.RpcError as err:
                     if err.is_not_found():
                         context.abort(grpc.StatusCode.NOT_FOUND,
                                       f"BDev {name} not found")
+                    context.abort(grpc.StatusCode.INTERNAL, str(err))
+                if size <= bdevs[0].size_bytes:
+                    # grow-only (CSI expansion must never shrink);
+                    # idempotent for the current size and below
+                    return spec.ResizeMallocBDevReply()
+                try:
+                    client.invoke("resize_malloc_bdev",
+                                  {"name": name, "size": size})
+                except hipstore.RpcError as err:
                     if "busy" in str(err):
                         context.abort(grpc.StatusCode.FAILED_PRECONDITION,
                                       str(err))

@@ -121,12 +121,18 @@ class LocalBackend(OIMBackend):
         size = round_to_blocks(size)
         with self._client() as client:
             try:
-                client.invoke("resize_malloc_bdev",
-                              {"name": volume_id, "size": size})
+                current = hipstore.get_bdevs(client, volume_id)[0].size_bytes
             except hipstore.RpcError as err:
                 if err.is_not_found():
                     raise LookupError(
                         f"volume {volume_id} not found") from None
+                raise RuntimeError(str(err)) from None
+            if size <= current:
+                return current  # CSI expansion never shrinks
+            try:
+                client.invoke("resize_malloc_bdev",
+                              {"name": volume_id, "size": size})
+            except hipstore.RpcError as err:
                 raise RuntimeError(str(err)) from None
         return size
 

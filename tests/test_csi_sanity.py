@@ -649,3 +649,25 @@ class TestRemoteExpansion:
             driver.stop()
             ctrl_server.stop()
             reg_server.stop()
+
+
+class TestExpansionNeverShrinks:
+    def test_shrink_request_is_noop(self, sanity_env, hipstored):  # noqa: F811
+        from oim_amd import hipstore
+
+        ctrl = sanity_env["controller"]
+        create = csi.CreateVolumeRequest(name="no-shrink")
+        create.capacity_range.required_bytes = 4 << 20
+        create.volume_capabilities.add().CopyFrom(mount_cap())
+        ctrl.CreateVolume(create, timeout=10)
+        try:
+            request = csi.ControllerExpandVolumeRequest(volume_id="no-shrink")
+            request.capacity_range.required_bytes = 1 << 20  # smaller
+            reply = ctrl.ControllerExpandVolume(request, timeout=10)
+            assert reply.capacity_bytes == 4 << 20  # unchanged
+            with hipstore.Client(hipstored.socket_path) as client:
+                assert hipstore.get_bdevs(
+                    client, "no-shrink")[0].size_bytes == 4 << 20
+        finally:
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id="no-shrink"),
+                              timeout=10)
