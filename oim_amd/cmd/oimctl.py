@@ -41,6 +41,7 @@ def main(argv=None) -> int:
         ("unmap", "unmap a volume"),
         ("check", "check that a malloc bdev exists"),
         ("clone", "clone a malloc bdev (device-side HBM-rate copy)"),
+        ("resize", "grow a malloc bdev (offline)"),
     ):
         cmd = sub.add_parser(name, help=help_text + " via the registry proxy")
         cmd.add_argument("--controller", required=True,
@@ -50,6 +51,8 @@ def main(argv=None) -> int:
             cmd.add_argument("size", help='bytes, or "64MiB"-style; 0 deletes')
         if name == "clone":
             cmd.add_argument("dest", help="name of the clone to create")
+        if name == "resize":
+            cmd.add_argument("size", help='new size: bytes or "8GiB"-style')
     args = parser.parse_args(argv)
     log.init_from_args(args)
 
@@ -110,6 +113,13 @@ def main(argv=None) -> int:
                                                 dest=args.dest),
                     metadata=metadata, timeout=120)
                 print(f"cloned {args.volume} -> {args.dest}")
+            elif args.command == "resize":
+                new_size = parse_size(args.size)
+                controller.ResizeMallocBDev(
+                    spec.ResizeMallocBDevRequest(bdev_name=args.volume,
+                                                 size=new_size),
+                    metadata=metadata, timeout=120)
+                print(f"resized {args.volume} to {new_size} bytes")
     return 0
 
 

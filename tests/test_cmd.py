@@ -81,6 +81,12 @@ class TestOimctlVolumes:
             assert "target 0 lun 0" in out
             assert oimctl.main(["--registry", endpoint, "unmap",
                                 "--controller", "c9", "volx"]) == 0
+            assert oimctl.main(["--registry", endpoint, "clone",
+                                "--controller", "c9", "volx", "volx-c"]) == 0
+            assert oimctl.main(["--registry", endpoint, "resize",
+                                "--controller", "c9", "volx", "128MiB"]) == 0
+            assert oimctl.main(["--registry", endpoint, "provision",
+                                "--controller", "c9", "volx-c", "0"]) == 0
             assert oimctl.main(["--registry", endpoint, "provision",
                                 "--controller", "c9", "volx", "0"]) == 0
         finally:
