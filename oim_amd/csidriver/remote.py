@@ -76,13 +76,19 @@ class RemoteBackend(OIMBackend):
     # --- registry plumbing --------------------------------------------------
 
     def _dial_registry(self) -> grpc.Channel:
-        """Per-operation dial (remote.go:101-114): no held connections."""
+        """Per-operation dial (remote.go:101-114): no held connections.
+        Outgoing payloads are logged with secrets stripped, like the
+        reference's ChooseDialOpts interceptor chain (grpc.go:56)."""
+        from ..common.tracing import LogClientInterceptor
+
         target = grpc_target(self.registry_address)
         if self.tls is not None:
             creds = load_tls_channel_credentials(self.tls)
             options = channel_options_for_peer("component.registry")
-            return grpc.secure_channel(target, creds, options=options)
-        return grpc.insecure_channel(target)
+            channel = grpc.secure_channel(target, creds, options=options)
+        else:
+            channel = grpc.insecure_channel(target)
+        return grpc.intercept_channel(channel, LogClientInterceptor())
 
     def _metadata(self):
         return ((spec.CONTROLLER_ID_KEY, self.controller_id),)
