@@ -74,6 +74,15 @@ with grpc.insecure_channel(grpc_target(f"unix://{work}/registry.sock")) as ch:
           f"{reply.scsi_disk.lun}")
 EOF
 
+echo "--- snapshot / clone / resize via oimctl ---"
+oimctl_cmd() {
+    python -m oim_amd.cmd.oimctl --registry "unix://$WORK/registry.sock" "$@"
+}
+oimctl_cmd clone  --controller demo-gpu0 demo-vol demo-vol-snap
+oimctl_cmd resize --controller demo-gpu0 demo-vol 128MiB
+oimctl_cmd check  --controller demo-gpu0 demo-vol-snap
+oimctl_cmd provision --controller demo-gpu0 demo-vol-snap 0
+
 echo "--- fio-shaped benchmark on the mapped volume ---"
 python -m oim_amd.bench.fio_harness --socket "$WORK/hipstored.sock" \
     --bdev demo-vol --rw randread --bs 4096 --iodepth 32 --numjobs 4 \
