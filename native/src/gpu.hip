@@ -769,6 +769,17 @@ inline int per_queue_channel_cap() {
   return cap;
 }
 
+// Which engine serves channels past the cap: "batched" (default; see
+// the mixed-engine wedge note in get_channel) or "shared" (for the
+// round-2 wedge experiments, tools/wedge_experiments.sh).
+inline bool fallback_is_shared() {
+  static const bool shared = [] {
+    const char* env = getenv("HIPSTORE_FALLBACK");
+    return env != nullptr && strcmp(env, "shared") == 0;
+  }();
+  return shared;
+}
+
 class HbmPersistentChannel : public HbmChannelBase {
  public:
   static constexpr uint32_t kRing = 32768;      // descriptors (>= 2x max request tiles)
@@ -1478,6 +1489,9 @@ class HbmBdev : public Bdev {
       }
       g_per_queue_channels[device_ & 63].fetch_sub(
           1, std::memory_order_relaxed);
+      if (fallback_is_shared()) {  // wedge experiments only
+        return track(std::make_shared<HbmSharedChannel>(device_, base_));
+      }
       return track(
           std::make_shared<HbmChannel>(device_, base_, size_bytes()));
     }
