@@ -46,6 +46,27 @@ bin/hipstored: $(CORE_OBJS) $(B)/main.o
 	mkdir -p bin
 	$(HIPCC) $(HIPFLAGS) $^ $(LDLIBS) -o $@
 
+# AddressSanitizer build of the daemon: run the protocol-heavy tests
+# against it for leak/UB checking:
+#   make asan && TEST_HIPSTORED_BINARY=bin/hipstored-asan \
+#     python -m pytest tests/test_vhost.py tests/test_nvmf.py -m "not gpu"
+ASAN_FLAGS := -O1 -g -fsanitize=address -fno-omit-frame-pointer
+ASAN_OBJS := $(patsubst $(B)/%.o,$(B)/asan/%.o,$(CORE_OBJS)) $(B)/asan/main.o
+
+$(B)/asan:
+	mkdir -p $(B)/asan
+
+$(B)/asan/%.o: native/src/%.cpp $(HDRS) | $(B)/asan
+	$(HIPCC) $(CXXFLAGS) $(ASAN_FLAGS) -c $< -o $@
+
+$(B)/asan/gpu.o: native/src/gpu.hip $(HDRS) | $(B)/asan
+	$(HIPCC) $(CXXFLAGS) $(ASAN_FLAGS) $(HIPFLAGS) -c $< -o $@
+
+.PHONY: asan
+asan: $(ASAN_OBJS)
+	mkdir -p bin
+	$(HIPCC) $(HIPFLAGS) -fsanitize=address $^ $(LDLIBS) -o bin/hipstored-asan
+
 clean:
 	rm -rf $(B) bin oim_amd/_hipstore*.so
 
