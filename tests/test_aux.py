@@ -8,7 +8,6 @@ import sys
 import threading
 import time
 
-import pytest
 
 from oim_amd import log
 from oim_amd.bench.perftype import (

@@ -2,7 +2,6 @@
 
 import subprocess
 import sys
-import threading
 
 import pytest
 

@@ -1,6 +1,5 @@
 """Unit tests for oim_amd.common (PCI, paths, endpoints, server)."""
 
-import threading
 
 import grpc
 import pytest

@@ -7,7 +7,7 @@ from concurrent import futures
 import grpc
 import pytest
 
-from oim_amd.registry import MemRegistryDB, Registry, RegistryServer
+from oim_amd.registry import Registry, RegistryServer
 from oim_amd.registry.etcddb import EtcdRegistryDB, _range_end
 from oim_amd.registry.etcdpb import FakeEtcdServicer, add_fake_etcd_to_server
 

@@ -5,7 +5,6 @@ Counterpart of the reference's pkg/oim-registry/registry_test.go
 """
 
 import os
-import threading
 
 import grpc
 import pytest
@@ -252,8 +251,6 @@ class TestTLS:
     def test_evil_ca_client_rejected(self, tls_registry, cas):
         registry, server, trusted = tls_registry
         _, evil = cas
-        from oim_amd.common import TLSConfig
-
         # evil client presents a cert from the wrong CA but trusts the
         # real server CA (so the failure is the server rejecting us).
         creds = grpc.ssl_channel_credentials(
