@@ -124,7 +124,10 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
       [&manager, use_hbm, device, persistent](const Json& p) {
         const int64_t num_blocks = p.get_int("num_blocks");
         const int64_t block_size = p.get_int("block_size");
-        if (num_blocks <= 0 || block_size <= 0 || block_size % 512 != 0) {
+        if (num_blocks <= 0 || block_size <= 0 || block_size % 512 != 0 ||
+            num_blocks > (int64_t{1} << 62) / block_size) {
+          // the product must not wrap uint64 (a wrapped size would
+          // report a consistent-but-lying geometry)
           throw RpcError{kInvalidParams, "invalid num_blocks/block_size"};
         }
         std::string name = p.get_string("name");
