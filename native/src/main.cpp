@@ -15,6 +15,7 @@
 #include <mutex>
 
 #include "hipstore/engine.h"
+#include "hipstore/json.h"
 #include "hipstore/nbd.h"
 #include "hipstore/rpc.h"
 #include "hipstore/vhost.h"
@@ -41,7 +42,8 @@ void usage(const char* argv0) {
           "  -C        CPU mode: host-RAM bdevs even when a GPU is present\n"
           "  -P        persistent engine: on-GPU polling service kernels (default)\n"
           "  -B        batched engine: per-poll kernel launches\n"
-          "  -V DIR    vhost-user socket directory (default: RPC socket dir)\n",
+          "  -V DIR    vhost-user socket directory (default: RPC socket dir)\n"
+          "  -c FILE   apply a save_config snapshot at startup\n",
           argv0);
 }
 
@@ -55,16 +57,18 @@ int main(int argc, char** argv) {
 
   std::string socket_path = "/var/tmp/hipstored.sock";
   std::string vhost_dir;
+  std::string config_file;
   int device = 0;
   bool cpu_only = false;
   bool persistent = true;  // the measured-better engine is the default
 
   int opt;
-  while ((opt = getopt(argc, argv, "S:d:V:CPBh")) != -1) {
+  while ((opt = getopt(argc, argv, "S:d:V:c:CPBh")) != -1) {
     switch (opt) {
       case 'S': socket_path = optarg; break;
       case 'd': device = atoi(optarg); break;
       case 'V': vhost_dir = optarg; break;
+      case 'c': config_file = optarg; break;
       case 'C': cpu_only = true; break;
       case 'P': persistent = true; break;
       case 'B': persistent = false; break;
@@ -85,6 +89,34 @@ int main(int argc, char** argv) {
 
   hipstore::RpcServer server(socket_path);
   hipstore::register_storage_methods(&server, use_hbm, device, persistent);
+  if (!config_file.empty()) {
+    // Recreate the saved topology BEFORE serving clients, by
+    // replaying the snapshot through the normal dispatch path.
+    try {
+      FILE* f = fopen(config_file.c_str(), "r");
+      if (f == nullptr) throw std::runtime_error("cannot open " + config_file);
+      std::string text;
+      char buf[4096];
+      size_t n;
+      while ((n = fread(buf, 1, sizeof(buf), f)) > 0) text.append(buf, n);
+      fclose(f);
+      hipstore::JsonObject request;
+      request["jsonrpc"] = hipstore::Json(std::string("2.0"));
+      request["id"] = hipstore::Json(int64_t{0});
+      request["method"] = hipstore::Json(std::string("load_config"));
+      request["params"] = hipstore::Json::parse(text);
+      hipstore::Json reply =
+          server.dispatch(hipstore::Json(std::move(request)));
+      if (reply.get("error") != nullptr) {
+        throw std::runtime_error(
+            reply.get("error")->get_string("message"));
+      }
+      fprintf(stderr, "hipstored: applied config %s\n", config_file.c_str());
+    } catch (const std::exception& e) {
+      fprintf(stderr, "hipstored: config load failed: %s\n", e.what());
+      return 1;
+    }
+  }
   try {
     server.start();
   } catch (const std::exception& e) {

@@ -363,9 +363,15 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   // nvmf_create_target exports existing bdevs as namespaces (nsid =
   // position in `bdevs` + 1); construct_nvme_tcp_bdev connects an
   // initiator bdev to any NVMe/TCP target (ours or a foreign one).
+  struct TargetEntry {
+    std::shared_ptr<NvmfTcpTarget> target;
+    std::string listen_addr;
+    bool digests;
+    std::vector<std::string> ns_bdevs;  // nsid = index + 1
+  };
   struct Targets {
     std::mutex mutex;
-    std::map<std::string, std::shared_ptr<NvmfTcpTarget>> by_nqn;
+    std::map<std::string, TargetEntry> by_nqn;
   };
   auto targets = std::make_shared<Targets>();
 
@@ -394,7 +400,16 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         JsonObject o;
         o["port"] = Json(static_cast<int64_t>(target->port()));
         o["subnqn"] = Json(subnqn);
-        targets->by_nqn[subnqn] = std::move(target);
+        TargetEntry entry;
+        entry.listen_addr = p.get_string("listen_addr");
+        entry.digests = p.get_bool("digests", true);
+        if (const Json* bdevs = p.get("bdevs")) {
+          for (const Json& name : bdevs->as_array()) {
+            entry.ns_bdevs.push_back(name.as_string());
+          }
+        }
+        entry.target = std::move(target);
+        targets->by_nqn[subnqn] = std::move(entry);
         return Json(std::move(o));
       });
 
@@ -402,7 +417,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
     std::lock_guard<std::mutex> lock(targets->mutex);
     auto it = targets->by_nqn.find(p.get_string("subnqn"));
     if (it == targets->by_nqn.end()) not_found("nvmf target");
-    it->second->stop();
+    it->second.target->stop();
     targets->by_nqn.erase(it);
     return Json(JsonObject{});
   });
@@ -727,6 +742,118 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
     o["lat_p999_us"] = Json(r.lat_p999_us);
     o["lat_max_us"] = Json(r.lat_max_us);
     return Json(std::move(o));
+  });
+
+  // --- configuration snapshot (SPDK save_config/load_config shape) -------
+  // Emits {"subsystems":[{"subsystem":..., "config":[{"method":...,
+  // "params":...}]}]} so a daemon restart (or `hipstored -c FILE`)
+  // recreates the control-plane topology. Data does NOT survive —
+  // malloc bdevs are RAM/HBM by contract (reference spec.md:116-119) —
+  // this is the SPDK-compatible control-state checkpoint. Composite
+  // and RBD bdevs are not yet emitted (construction params are not
+  // retained for them).
+  server->register_method("save_config", [&manager, targets](const Json&) {
+    JsonArray bdev_cfg;
+    for (const auto& bdev : manager.list()) {
+      if (bdev->product_name() != "Malloc disk") continue;
+      JsonObject params;
+      params["name"] = Json(bdev->name());
+      params["num_blocks"] = Json(static_cast<int64_t>(bdev->num_blocks()));
+      params["block_size"] = Json(static_cast<int64_t>(bdev->block_size()));
+      JsonObject entry;
+      entry["method"] = Json(std::string("construct_malloc_bdev"));
+      entry["params"] = Json(std::move(params));
+      bdev_cfg.push_back(Json(std::move(entry)));
+    }
+    JsonArray vhost_cfg;
+    {
+      auto& state = vhost_state();
+      std::lock_guard<std::mutex> lock(state.mutex);
+      for (const auto& [name, ctrl] : state.controllers) {
+        JsonObject params;
+        params["ctrlr"] = Json(name);
+        params["cpumask"] = Json(ctrl.cpumask);
+        JsonObject entry;
+        entry["method"] = Json(std::string("construct_vhost_scsi_controller"));
+        entry["params"] = Json(std::move(params));
+        vhost_cfg.push_back(Json(std::move(entry)));
+        for (int t = 0; t < kMaxScsiTargets; ++t) {
+          const ScsiTarget& target = ctrl.targets[t];
+          if (!target.used || target.luns.empty()) continue;
+          JsonObject lp;
+          lp["ctrlr"] = Json(name);
+          lp["scsi_target_num"] = Json(static_cast<int64_t>(t));
+          lp["bdev_name"] = Json(target.luns[0].bdev_name);
+          JsonObject le;
+          le["method"] = Json(std::string("add_vhost_scsi_lun"));
+          le["params"] = Json(std::move(lp));
+          vhost_cfg.push_back(Json(std::move(le)));
+        }
+      }
+    }
+    JsonArray nvmf_cfg;
+    {
+      std::lock_guard<std::mutex> lock(targets->mutex);
+      for (const auto& [subnqn, entry] : targets->by_nqn) {
+        JsonObject params;
+        params["subnqn"] = Json(subnqn);
+        params["listen_addr"] = Json(entry.listen_addr);
+        params["port"] = Json(static_cast<int64_t>(entry.target->port()));
+        params["digests"] = Json(entry.digests);
+        JsonArray ns;
+        for (const std::string& name : entry.ns_bdevs) {
+          ns.push_back(Json(name));
+        }
+        params["bdevs"] = Json(std::move(ns));
+        JsonObject e;
+        e["method"] = Json(std::string("nvmf_create_target"));
+        e["params"] = Json(std::move(params));
+        nvmf_cfg.push_back(Json(std::move(e)));
+      }
+    }
+    auto subsystem = [](const char* name, JsonArray cfg) {
+      JsonObject o;
+      o["subsystem"] = Json(std::string(name));
+      o["config"] = Json(std::move(cfg));
+      return Json(std::move(o));
+    };
+    JsonArray subsystems;
+    subsystems.push_back(subsystem("bdev", std::move(bdev_cfg)));
+    subsystems.push_back(subsystem("vhost", std::move(vhost_cfg)));
+    subsystems.push_back(subsystem("nvmf", std::move(nvmf_cfg)));
+    JsonObject out;
+    out["subsystems"] = Json(std::move(subsystems));
+    return Json(std::move(out));
+  });
+
+  server->register_method("load_config", [server](const Json& p) {
+    const Json* subsystems = p.get("subsystems");
+    if (subsystems == nullptr) {
+      throw RpcError{kInvalidParams, "subsystems required"};
+    }
+    int64_t applied = 0;
+    for (const Json& subsystem : subsystems->as_array()) {
+      const Json* config = subsystem.get("config");
+      if (config == nullptr) continue;
+      for (const Json& entry : config->as_array()) {
+        JsonObject request;
+        request["jsonrpc"] = Json(std::string("2.0"));
+        request["id"] = Json(int64_t{0});
+        request["method"] = Json(entry.get_string("method"));
+        if (const Json* params = entry.get("params")) {
+          request["params"] = *params;
+        }
+        Json reply = server->dispatch(Json(std::move(request)));
+        if (reply.get("error") != nullptr) {
+          throw RpcError{kInvalidParams,
+                         "load_config failed at " +
+                             entry.get_string("method") + ": " +
+                             reply.get("error")->get_string("message")};
+        }
+        ++applied;
+      }
+    }
+    return Json(applied);
   });
 }
 

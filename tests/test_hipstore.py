@@ -181,3 +181,78 @@ class TestClone:
             raise AssertionError("expected failure for CPU bdevs")
         except RuntimeError:
             pass
+
+
+class TestConfigSnapshot:
+    """save_config / load_config / `hipstored -c` (SPDK's config
+    snapshot shape): the control-plane topology survives a daemon
+    restart; data does not (RAM/HBM by contract, spec.md:116-119)."""
+
+    def test_save_restart_with_config(self, tmp_path):
+        import json as jsonmod
+        import subprocess
+        import time
+
+        import fixtures
+
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        with hipstore.Client(daemon.socket_path) as client:
+            hipstore.construct_malloc_bdev(
+                client, num_blocks=2048, block_size=512, name="cfg0")
+            client.invoke("construct_vhost_scsi_controller",
+                          {"ctrlr": "cfgv"})
+            client.invoke("add_vhost_scsi_lun",
+                          {"ctrlr": "cfgv", "scsi_target_num": 1,
+                           "bdev_name": "cfg0"})
+            target = client.invoke("nvmf_create_target",
+                                   {"listen_addr": "127.0.0.1", "port": 0,
+                                    "bdevs": ["cfg0"]})
+            config = client.invoke("save_config")
+        daemon.stop()
+        assert {s["subsystem"] for s in config["subsystems"]} == \
+            {"bdev", "vhost", "nvmf"}
+
+        config_path = tmp_path / "cfg.json"
+        config_path.write_text(jsonmod.dumps(config))
+        sock2 = str(tmp_path / "h2.sock")
+        proc = subprocess.Popen(
+            [fixtures.DEFAULT_BINARY, "-S", sock2, "-C",
+             "-c", str(config_path)], stderr=subprocess.PIPE)
+        deadline = time.time() + 30
+        import os as osmod
+        while not osmod.path.exists(sock2):
+            assert proc.poll() is None, proc.stderr.read().decode()
+            assert time.time() < deadline
+            time.sleep(0.05)
+        try:
+            with hipstore.Client(sock2) as client:
+                bdev = hipstore.get_bdevs(client, "cfg0")[0]
+                assert (bdev.num_blocks, bdev.block_size) == (2048, 512)
+                controllers = hipstore.get_vhost_controllers(client)
+                assert controllers[0].controller == "cfgv"
+                assert controllers[0].scsi_targets[0].id == 1
+                # the nvmf target came back on its saved port
+                replayed = client.invoke("save_config")
+                nvmf = [entry for sub in replayed["subsystems"]
+                        if sub["subsystem"] == "nvmf"
+                        for entry in sub["config"]]
+                assert nvmf and nvmf[0]["params"]["bdevs"] == ["cfg0"]
+                assert nvmf[0]["params"]["port"] == target["port"]
+        finally:
+            proc.terminate()
+            proc.wait(timeout=10)
+
+    def test_load_config_rpc(self, hipstored):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            applied = client.invoke("load_config", {"subsystems": [
+                {"subsystem": "bdev", "config": [
+                    {"method": "construct_malloc_bdev",
+                     "params": {"name": "lc0", "num_blocks": 1024,
+                                "block_size": 512}}]}]})
+            assert applied == 1
+            assert hipstore.get_bdevs(client, "lc0")
+            with pytest.raises(hipstore.RpcError):
+                client.invoke("load_config", {"subsystems": [
+                    {"subsystem": "bdev", "config": [
+                        {"method": "no_such_method", "params": {}}]}]})
+            hipstore.delete_bdev(client, "lc0")
