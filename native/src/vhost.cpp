@@ -18,6 +18,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstring>
+#include <map>
 #include <mutex>
 #include <set>
 #include <stdexcept>
@@ -259,6 +260,14 @@ class VhostUserScsiDev {
   const std::string& socket_path() const { return socket_path_; }
 
  private:
+  // One channel per (ring worker, bdev): channels are single-threaded
+  // by contract, and per-worker ownership lets the request queues run
+  // their I/O in parallel instead of convoying on one mutex. The map
+  // lives in the worker thread and dies with it (teardown joins the
+  // workers, so no channel outlives the session).
+  using ChannelCache =
+      std::map<Bdev*, std::pair<BdevPtr, std::shared_ptr<IoChannel>>>;
+
   // -- connection handling ---------------------------------------------
 
   void accept_loop() {
@@ -613,11 +622,11 @@ class VhostUserScsiDev {
       ring.enabled.store(false);
     }
     unmap_regions();
-    channels_.clear();
     negotiated_features_ = 0;
   }
 
   void ring_worker(unsigned index) {
+    ChannelCache channels;  // this worker's engine channels
     Vring& ring = rings_[index];
     while (ring.running.load(std::memory_order_relaxed)) {
       pollfd fds[2] = {{ring.kick, POLLIN, 0}, {ring.stop_pipe[0], POLLIN, 0}};
@@ -629,11 +638,11 @@ class VhostUserScsiDev {
         (void)!read(ring.kick, &n, 8);
       }
       if (!ring.enabled.load(std::memory_order_acquire)) continue;
-      drain_ring(index);
+      drain_ring(index, &channels);
     }
   }
 
-  void drain_ring(unsigned index) {
+  void drain_ring(unsigned index, ChannelCache* channels) {
     Vring& ring = rings_[index];
     bool did_work = false;
     while (true) {
@@ -644,7 +653,7 @@ class VhostUserScsiDev {
       ring.last_avail++;
       uint32_t written = 0;
       try {
-        written = process_chain(index, head);
+        written = process_chain(index, head, channels);
       } catch (const std::exception& e) {
         // A bad buffer (e.g. unregisterable guest memory) must fail
         // the one command, never kill the ring worker / daemon.
@@ -693,11 +702,12 @@ class VhostUserScsiDev {
 
   // Returns bytes written into device-writable descriptors (the value
   // published in the used ring).
-  uint32_t process_chain(unsigned ring_index, uint16_t head) {
+  uint32_t process_chain(unsigned ring_index, uint16_t head,
+                         ChannelCache* channels) {
     Vring& ring = rings_[ring_index];
     std::vector<Iov> out, in;
     if (!collect_iovs(ring, head, &out, &in)) return 0;
-    if (ring_index >= 2) return handle_scsi_cmd(out, in);
+    if (ring_index >= 2) return handle_scsi_cmd(out, in, channels);
     // Control / event queue: acknowledge TMFs and AN queries with an
     // all-zero response (FUNCTION_COMPLETE / S_OK).
     size_t n = iov_total(in);
@@ -707,19 +717,18 @@ class VhostUserScsiDev {
 
   // -- SCSI command execution --------------------------------------------
 
-  std::shared_ptr<IoChannel> channel_for(const BdevPtr& bdev) {
-    std::lock_guard<std::mutex> lock(channels_mutex_);
-    auto it = channels_.find(bdev.get());
-    if (it != channels_.end()) return it->second.second;
+  static std::shared_ptr<IoChannel> channel_for(ChannelCache* cache,
+                                                const BdevPtr& bdev) {
+    auto it = cache->find(bdev.get());
+    if (it != cache->end()) return it->second.second;
     auto ch = bdev->get_channel();
-    channels_[bdev.get()] = {bdev, ch};  // pins the bdev alive with it
+    (*cache)[bdev.get()] = {bdev, ch};  // pins the bdev alive with it
     return ch;
   }
 
-  int bdev_io(const BdevPtr& bdev, IoOp op, uint64_t offset, void* buf,
-              uint64_t len) {
-    auto ch = channel_for(bdev);
-    std::lock_guard<std::mutex> lock(io_mutex_);  // channels are 1-thread
+  static int bdev_io(ChannelCache* cache, const BdevPtr& bdev, IoOp op,
+                     uint64_t offset, void* buf, uint64_t len) {
+    auto ch = channel_for(cache, bdev);
     int status = 1;
     bool done = false;
     IoRequest req;
@@ -750,7 +759,8 @@ class VhostUserScsiDev {
   }
 
   uint32_t handle_scsi_cmd(const std::vector<Iov>& out,
-                           const std::vector<Iov>& in) {
+                           const std::vector<Iov>& in,
+                           ChannelCache* channels) {
     if (out.empty() || in.empty()) return 0;
     ScsiCmdReq req{};
     if (gather(reinterpret_cast<uint8_t*>(&req), out, sizeof(req)) <
@@ -785,7 +795,8 @@ class VhostUserScsiDev {
       }
     }
 
-    size_t data_in_written = execute(req, data_out, data_in, &resp);
+    size_t data_in_written =
+        execute(req, data_out, data_in, &resp, channels);
 
     // Response goes into the first sizeof(resp) writable bytes.
     size_t resp_written =
@@ -796,7 +807,8 @@ class VhostUserScsiDev {
   // Returns the number of data-in bytes produced (for the used-ring
   // "written" field).
   size_t execute(const ScsiCmdReq& req, const std::vector<Iov>& data_out,
-                 const std::vector<Iov>& data_in, ScsiCmdResp* resp) {
+                 const std::vector<Iov>& data_in, ScsiCmdResp* resp,
+                 ChannelCache* channels) {
     // Single-level LUN addressing: lun[0]==1 selects the target by
     // lun[1]; bytes 2-3 carry the LUN (flat-space). Only LUN 0 exists.
     if (req.lun[0] != 1) {
@@ -927,7 +939,8 @@ class VhostUserScsiDev {
         const std::vector<Iov>& iovs = is_write ? data_out : data_in;
         int status;
         if (iovs.size() == 1 && iovs[0].len >= bytes) {
-          status = bdev_io(bdev, is_write ? IoOp::kWrite : IoOp::kRead,
+          status = bdev_io(channels, bdev,
+                           is_write ? IoOp::kWrite : IoOp::kRead,
                            lba * block, iovs[0].base, bytes);
         } else {
           if (iov_total(iovs) < bytes) {
@@ -939,7 +952,8 @@ class VhostUserScsiDev {
           // malloc on CPU-only daemons.
           uint8_t* bounce = static_cast<uint8_t*>(alloc_pinned(bytes));
           if (is_write) gather(bounce, iovs, bytes);
-          status = bdev_io(bdev, is_write ? IoOp::kWrite : IoOp::kRead,
+          status = bdev_io(channels, bdev,
+                           is_write ? IoOp::kWrite : IoOp::kRead,
                            lba * block, bounce, bytes);
           if (!is_write && status == kIoOk) {
             scatter(iovs, bounce, bytes);
@@ -972,9 +986,6 @@ class VhostUserScsiDev {
   std::vector<Region> regions_;
   Vring rings_[kMaxVrings];
 
-  std::mutex channels_mutex_;
-  std::mutex io_mutex_;
-  std::map<Bdev*, std::pair<BdevPtr, std::shared_ptr<IoChannel>>> channels_;
 };
 
 // ---- module state ------------------------------------------------------
