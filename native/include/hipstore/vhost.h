@@ -43,6 +43,13 @@ std::string vhost_socket_path(const std::string& ctrlr_name);
 VhostDevPtr vhost_start(const std::string& name,
                         std::function<BdevPtr(int target)> resolver);
 
+// virtio-blk personality (SPDK's vhost-user-blk twin): one bdev, all
+// rings are request queues, 16-byte virtio_blk_outhdr + trailing
+// status byte. resolver(0) supplies the bdev.
+VhostDevPtr vhost_start_blk(const std::string& name,
+                            std::function<BdevPtr(int)> resolver,
+                            bool readonly);
+
 // Stop the accept loop and all ring workers, unlink the socket. Safe
 // to call twice. Must NOT be called while holding locks the resolver
 // takes (workers may be blocked in it).

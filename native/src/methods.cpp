@@ -40,7 +40,9 @@ struct ScsiTarget {
 struct VhostController {
   std::string cpumask;
   ScsiTarget targets[kMaxScsiTargets];
-  VhostDevPtr dev;  // live vhost-user-scsi server for this controller
+  std::string blk_bdev;   // non-empty => virtio-blk personality
+  bool blk_readonly = false;
+  VhostDevPtr dev;  // live vhost-user server for this controller
 };
 
 struct VhostState {
@@ -246,6 +248,50 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
     return Json(JsonObject{});
   });
 
+  server->register_method(
+      "construct_vhost_blk_controller", [&manager](const Json& p) {
+        // SPDK vhost-user-blk twin: one bdev, claimed for the
+        // controller's lifetime; rings serve virtio-blk requests.
+        const std::string name = p.get_string("ctrlr");
+        const std::string dev_name = p.get_string("dev_name");
+        if (name.empty() || dev_name.empty()) {
+          throw RpcError{kInvalidParams, "ctrlr and dev_name required"};
+        }
+        auto& state = vhost_state();
+        std::lock_guard<std::mutex> lock(state.mutex);
+        if (state.controllers.count(name)) {
+          throw RpcError{kInvalidParams,
+                         "controller " + name + " already exists"};
+        }
+        BdevPtr bdev = manager.find(dev_name);
+        if (!bdev) not_found("bdev " + dev_name);
+        if (!bdev->claim()) {
+          throw RpcError{kInvalidParams, "bdev " + dev_name + " is claimed"};
+        }
+        VhostDevPtr dev;
+        try {
+          dev = vhost_start_blk(
+              name,
+              [name](int) -> BdevPtr {
+                auto& st = vhost_state();
+                std::lock_guard<std::mutex> l(st.mutex);
+                auto it = st.controllers.find(name);
+                if (it == st.controllers.end()) return nullptr;
+                return BdevManager::instance().find(it->second.blk_bdev);
+              },
+              p.get_bool("readonly", false));
+        } catch (const std::exception& e) {
+          bdev->release();
+          throw RpcError{kInvalidParams, e.what()};
+        }
+        VhostController& ctrl = state.controllers[name];
+        ctrl.cpumask = p.get_string("cpumask", "0x1");
+        ctrl.blk_bdev = dev_name;
+        ctrl.blk_readonly = p.get_bool("readonly", false);
+        ctrl.dev = dev;
+        return Json(JsonObject{});
+      });
+
   server->register_method("add_vhost_scsi_lun", [&manager](const Json& p) {
     const std::string ctrlr = p.get_string("ctrlr");
     const int64_t target_num = p.get_int("scsi_target_num", -1);
@@ -254,6 +300,10 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
     std::lock_guard<std::mutex> lock(state.mutex);
     auto it = state.controllers.find(ctrlr);
     if (it == state.controllers.end()) not_found("controller " + ctrlr);
+    if (!it->second.blk_bdev.empty()) {
+      throw RpcError{kInvalidParams,
+                     "controller " + ctrlr + " is a blk controller"};
+    }
     if (target_num < 0 || target_num >= kMaxScsiTargets) {
       throw RpcError{kInvalidParams, "scsi_target_num out of range"};
     }
@@ -304,6 +354,11 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
           if (BdevPtr bdev = manager.find(lun.bdev_name)) bdev->release();
         }
       }
+      if (!it->second.blk_bdev.empty()) {
+        if (BdevPtr bdev = manager.find(it->second.blk_bdev)) {
+          bdev->release();
+        }
+      }
       dev = it->second.dev;
       state.controllers.erase(it);
     }
@@ -337,7 +392,14 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         scsi.push_back(Json(std::move(to)));
       }
       JsonObject backend;
-      backend["scsi"] = Json(std::move(scsi));
+      if (!ctrl.blk_bdev.empty()) {
+        JsonObject blk;
+        blk["bdev"] = Json(ctrl.blk_bdev);
+        blk["readonly"] = Json(ctrl.blk_readonly);
+        backend["block"] = Json(std::move(blk));
+      } else {
+        backend["scsi"] = Json(std::move(scsi));
+      }
       JsonObject o;
       o["ctrlr"] = Json(name);
       o["cpumask"] = Json(ctrl.cpumask);
@@ -774,6 +836,15 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         params["ctrlr"] = Json(name);
         params["cpumask"] = Json(ctrl.cpumask);
         JsonObject entry;
+        if (!ctrl.blk_bdev.empty()) {
+          params["dev_name"] = Json(ctrl.blk_bdev);
+          params["readonly"] = Json(ctrl.blk_readonly);
+          entry["method"] =
+              Json(std::string("construct_vhost_blk_controller"));
+          entry["params"] = Json(std::move(params));
+          vhost_cfg.push_back(Json(std::move(entry)));
+          continue;
+        }
         entry["method"] = Json(std::string("construct_vhost_scsi_controller"));
         entry["params"] = Json(std::move(params));
         vhost_cfg.push_back(Json(std::move(entry)));

@@ -212,11 +212,17 @@ size_t gather(uint8_t* dst, const std::vector<Iov>& iovs, size_t len) {
 
 class VhostUserScsiDev {
  public:
+  enum class Personality { kScsi, kBlk };
+
   VhostUserScsiDev(std::string name, std::string socket_path,
-                   std::function<BdevPtr(int)> resolver)
+                   std::function<BdevPtr(int)> resolver,
+                   Personality personality = Personality::kScsi,
+                   bool readonly = false)
       : name_(std::move(name)),
         socket_path_(std::move(socket_path)),
-        resolver_(std::move(resolver)) {}
+        resolver_(std::move(resolver)),
+        personality_(personality),
+        readonly_(readonly) {}
 
   ~VhostUserScsiDev() { stop(); }
 
@@ -707,12 +713,104 @@ class VhostUserScsiDev {
     Vring& ring = rings_[ring_index];
     std::vector<Iov> out, in;
     if (!collect_iovs(ring, head, &out, &in)) return 0;
+    if (personality_ == Personality::kBlk) {
+      return handle_blk_cmd(out, in, channels);
+    }
     if (ring_index >= 2) return handle_scsi_cmd(out, in, channels);
     // Control / event queue: acknowledge TMFs and AN queries with an
     // all-zero response (FUNCTION_COMPLETE / S_OK).
     size_t n = iov_total(in);
     for (Iov& v : in) memset(v.base, 0, v.len);
     return static_cast<uint32_t>(n);
+  }
+
+  // -- virtio-blk command execution --------------------------------------
+
+  // Request: virtio_blk_outhdr {u32 type; u32 ioprio; u64 sector} in
+  // device-readable descriptors, then data; the LAST device-writable
+  // byte is the status (OK=0, IOERR=1, UNSUPP=2). Sectors are always
+  // 512-byte units regardless of bdev block size.
+  uint32_t handle_blk_cmd(const std::vector<Iov>& out,
+                          const std::vector<Iov>& in,
+                          ChannelCache* channels) {
+    constexpr uint32_t kTypeIn = 0, kTypeOut = 1, kTypeFlush = 4,
+                       kTypeGetId = 8;
+    constexpr uint8_t kStatusOk = 0, kStatusIoErr = 1, kStatusUnsupp = 2;
+    if (out.empty() || in.empty()) return 0;
+    uint8_t header[16];
+    if (gather(header, out, sizeof(header)) < sizeof(header)) return 0;
+    uint32_t type;
+    uint64_t sector;
+    memcpy(&type, header, 4);
+    memcpy(&sector, header + 8, 8);
+
+    // status byte = last writable byte; data-in = everything before it
+    std::vector<Iov> data_in = in;
+    uint8_t* status_ptr;
+    {
+      Iov& last = data_in.back();
+      status_ptr = last.base + last.len - 1;
+      if (--last.len == 0) data_in.pop_back();
+    }
+    // data-out = readable bytes after the header
+    std::vector<Iov> data_out;
+    {
+      size_t skip = sizeof(header);
+      for (Iov v : out) {
+        if (skip >= v.len) {
+          skip -= v.len;
+          continue;
+        }
+        data_out.push_back(Iov{v.base + skip, v.len - skip});
+        skip = 0;
+      }
+    }
+    uint8_t status = kStatusOk;
+    BdevPtr bdev = resolver_(0);
+    if (!bdev) {
+      status = kStatusIoErr;
+    } else if (type == kTypeFlush) {
+      status = kStatusOk;
+    } else if (type == kTypeGetId) {
+      char id[20] = {0};
+      strncpy(id, bdev->name().c_str(), sizeof(id) - 1);
+      scatter(data_in, reinterpret_cast<uint8_t*>(id),
+              std::min<size_t>(sizeof(id), iov_total(data_in)));
+      status = kStatusOk;
+    } else if (type == kTypeIn || type == kTypeOut) {
+      const bool is_write = type == kTypeOut;
+      const std::vector<Iov>& iovs = is_write ? data_out : data_in;
+      const uint64_t bytes = iov_total(iovs);
+      const uint64_t offset = sector * 512;
+      if ((is_write && readonly_) || bytes == 0 ||
+          bytes % bdev->block_size() != 0 ||
+          offset % bdev->block_size() != 0 ||
+          offset + bytes > bdev->size_bytes() || bytes > kMaxIoBytes) {
+        status = kStatusIoErr;
+      } else if (iovs.size() == 1) {
+        if (bdev_io(channels, bdev, is_write ? IoOp::kWrite : IoOp::kRead,
+                    offset, iovs[0].base, bytes) != kIoOk) {
+          status = kStatusIoErr;
+        }
+      } else {
+        uint8_t* bounce = static_cast<uint8_t*>(alloc_pinned(bytes));
+        if (is_write) gather(bounce, iovs, bytes);
+        int io_status =
+            bdev_io(channels, bdev, is_write ? IoOp::kWrite : IoOp::kRead,
+                    offset, bounce, bytes);
+        if (io_status == kIoOk && !is_write) scatter(iovs, bounce, bytes);
+        free_pinned(bounce);
+        if (io_status != kIoOk) status = kStatusIoErr;
+      }
+    } else {
+      status = kStatusUnsupp;
+    }
+    *status_ptr = status;
+    const uint32_t data_written =
+        (type == kTypeIn && status == kStatusOk)
+            ? static_cast<uint32_t>(iov_total(data_in))
+            : (type == kTypeGetId ? 20u : 0u);
+    return data_written + 1;  // +1 for the status byte
   }
 
   // -- SCSI command execution --------------------------------------------
@@ -975,6 +1073,8 @@ class VhostUserScsiDev {
   const std::string name_;
   const std::string socket_path_;
   const std::function<BdevPtr(int)> resolver_;
+  const Personality personality_ = Personality::kScsi;
+  const bool readonly_ = false;
 
   int listen_fd_ = -1;
   int stop_pipe_[2] = {-1, -1};
@@ -1010,6 +1110,18 @@ VhostDevPtr vhost_start(const std::string& name,
                         std::function<BdevPtr(int)> resolver) {
   auto dev = std::make_shared<VhostUserScsiDev>(name, vhost_socket_path(name),
                                                 std::move(resolver));
+  dev->start();
+  std::lock_guard<std::mutex> lock(g_vhost_mutex);
+  g_devices.insert(dev);
+  return dev;
+}
+
+VhostDevPtr vhost_start_blk(const std::string& name,
+                            std::function<BdevPtr(int)> resolver,
+                            bool readonly) {
+  auto dev = std::make_shared<VhostUserScsiDev>(
+      name, vhost_socket_path(name), std::move(resolver),
+      VhostUserScsiDev::Personality::kBlk, readonly);
   dev->start();
   std::lock_guard<std::mutex> lock(g_vhost_mutex);
   g_devices.insert(dev);

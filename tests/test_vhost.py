@@ -349,3 +349,70 @@ class TestVhostRobustness:
         result = master.scsi(0, bytes(cdb), data_in_len=4096)
         assert result.status == 2
         assert result.asc == 0x21
+
+
+class TestVhostBlk:
+    """virtio-blk personality (SPDK's vhost-user-blk twin): 16-byte
+    outhdr + trailing status byte; queue 0 is a request queue."""
+
+    @pytest.fixture
+    def blk_target(self, hipstored, tmp_path):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            hipstore.construct_malloc_bdev(
+                client, num_blocks=NUM_BLOCKS, block_size=BLOCK, name="blk0")
+            client.invoke("construct_vhost_blk_controller",
+                          {"ctrlr": "vb0", "dev_name": "blk0"})
+            path = os.path.join(os.path.dirname(hipstored.socket_path), "vb0")
+            master = VhostUserMaster(path, queue=0)
+            master.negotiate()
+            yield client, master
+            master.close()
+
+    def test_rw_roundtrip_and_get_id(self, blk_target):
+        _, master = blk_target
+        data = os.urandom(4 * BLOCK)
+        status, _ = master.blk_write(8, data)
+        assert status == 0
+        status, read_back = master.blk_read(8, len(data))
+        assert status == 0 and read_back == data
+        status, ident = master.blk(8, 0, data_in_len=20)  # GET_ID
+        assert status == 0
+        assert ident.rstrip(b"\0") == b"blk0"
+        status, _ = master.blk(4, 0)  # FLUSH
+        assert status == 0
+
+    def test_errors(self, blk_target):
+        client, master = blk_target
+        status, _ = master.blk_read(NUM_BLOCKS, BLOCK)  # beyond the end
+        assert status == 1  # IOERR
+        status, _ = master.blk(99, 0, data_in_len=BLOCK)  # unknown type
+        assert status == 2  # UNSUPP
+        # bdev claimed by the controller; scsi lun add refused
+        with pytest.raises(hipstore.RpcError):
+            client.invoke("add_vhost_scsi_lun",
+                          {"ctrlr": "vb0", "scsi_target_num": 0,
+                           "bdev_name": "blk0"})
+        listing = hipstore.get_vhost_controllers(client)
+        assert any(c.controller == "vb0" for c in listing)
+
+    def test_readonly_controller(self, hipstored, tmp_path):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            hipstore.construct_malloc_bdev(
+                client, num_blocks=1024, block_size=512, name="blkro")
+            client.invoke("construct_vhost_blk_controller",
+                          {"ctrlr": "vbro", "dev_name": "blkro",
+                           "readonly": True})
+            path = os.path.join(os.path.dirname(hipstored.socket_path),
+                                "vbro")
+            master = VhostUserMaster(path, queue=0)
+            master.negotiate()
+            try:
+                status, _ = master.blk_write(0, b"\xEE" * 512)
+                assert status == 1  # IOERR: write to readonly device
+                status, _ = master.blk_read(0, 512)
+                assert status == 0
+            finally:
+                master.close()
+            client.invoke("remove_vhost_controller", {"ctrlr": "vbro"})
+            # claim released: the bdev is deletable again
+            hipstore.delete_bdev(client, "blkro")

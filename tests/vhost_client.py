@@ -74,7 +74,9 @@ class ScsiResult:
 
 
 class VhostUserMaster:
-    def __init__(self, path: str, mem_size: int = 4 << 20):
+    def __init__(self, path: str, mem_size: int = 4 << 20,
+                 queue: int = QUEUE):
+        self.queue = queue
         self.sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
         self.sock.settimeout(10)
         self.sock.connect(path)
@@ -127,18 +129,19 @@ class VhostUserMaster:
         region = struct.pack("<II", 1, 0) + struct.pack(
             "<QQQQ", GPA_BASE, self.mem_size, UADDR_BASE, 0)
         self._send(SET_MEM_TABLE, region, fds=[self.memfd])
-        self._send(SET_VRING_NUM, struct.pack("<II", QUEUE, QSIZE))
-        self._send(SET_VRING_BASE, struct.pack("<II", QUEUE, 0))
+        q = self.queue
+        self._send(SET_VRING_NUM, struct.pack("<II", q, QSIZE))
+        self._send(SET_VRING_BASE, struct.pack("<II", q, 0))
         self._send(SET_VRING_ADDR, struct.pack(
-            "<IIQQQQ", QUEUE, 0, UADDR_BASE + DESC_OFF, UADDR_BASE + USED_OFF,
+            "<IIQQQQ", q, 0, UADDR_BASE + DESC_OFF, UADDR_BASE + USED_OFF,
             UADDR_BASE + AVAIL_OFF, 0))
-        self._send(SET_VRING_CALL, struct.pack("<Q", QUEUE), fds=[self.call])
-        self._send(SET_VRING_KICK, struct.pack("<Q", QUEUE), fds=[self.kick])
-        self._send(SET_VRING_ENABLE, struct.pack("<II", QUEUE, 1))
+        self._send(SET_VRING_CALL, struct.pack("<Q", q), fds=[self.call])
+        self._send(SET_VRING_KICK, struct.pack("<Q", q), fds=[self.kick])
+        self._send(SET_VRING_ENABLE, struct.pack("<II", q, 1))
 
     def stop_ring(self) -> int:
         """GET_VRING_BASE stops the ring; returns last_avail."""
-        reply = self.query(GET_VRING_BASE, struct.pack("<II", QUEUE, 0))
+        reply = self.query(GET_VRING_BASE, struct.pack("<II", self.queue, 0))
         _, base = struct.unpack("<II", reply)
         return base
 
@@ -231,6 +234,41 @@ class VhostUserMaster:
         data = (bytes(self.mem[DATA_IN_OFF:DATA_IN_OFF + data_in_len])
                 if data_in_len else b"")
         return ScsiResult(raw_resp, data)
+
+    # -- virtio-blk ---------------------------------------------------------
+
+    def blk(self, blk_type: int, sector: int, data_out: bytes = b"",
+            data_in_len: int = 0):
+        """Submit a virtio-blk request; returns (status_byte, data)."""
+        header = struct.pack("<IIQ", blk_type, 0, sector)
+        self.mem[REQ_OFF:REQ_OFF + 16] = header
+        if data_out:
+            self.mem[DATA_OUT_OFF:DATA_OUT_OFF + len(data_out)] = data_out
+        status_off = RESP_OFF
+        self.mem[status_off] = 0xAA  # sentinel
+        chain = [(GPA_BASE + REQ_OFF, 16, 0)]
+        if data_out:
+            chain.append((GPA_BASE + DATA_OUT_OFF, len(data_out), 0))
+        if data_in_len:
+            chain.append((GPA_BASE + DATA_IN_OFF, data_in_len, DESC_WRITE))
+        chain.append((GPA_BASE + status_off, 1, DESC_WRITE))
+        for i, (gpa, length, flags) in enumerate(chain):
+            last = i == len(chain) - 1
+            self._write_desc(i, gpa, length,
+                             flags | (0 if last else DESC_NEXT),
+                             0 if last else i + 1)
+        self._submit(0)
+        self._wait_used()
+        status = self.mem[status_off]
+        data = (bytes(self.mem[DATA_IN_OFF:DATA_IN_OFF + data_in_len])
+                if data_in_len else b"")
+        return status, data
+
+    def blk_read(self, sector: int, length: int):
+        return self.blk(0, sector, data_in_len=length)
+
+    def blk_write(self, sector: int, data: bytes):
+        return self.blk(1, sector, data_out=data)
 
     # convenience wrappers
 
