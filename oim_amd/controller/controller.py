@@ -374,6 +374,20 @@ class Controller(spec.ControllerServicer):
                     context.abort(grpc.StatusCode.INTERNAL, str(err))
         return spec.ResizeMallocBDevReply()
 
+    def ListMallocBDevs(self, request, context):
+        """oim-amd extension (docs/spec.md): enumerate malloc bdevs."""
+        reply = spec.ListMallocBDevsReply()
+        with self._client() as client:
+            for bdev in hipstore.get_bdevs(client):
+                if request.prefix and not bdev.name.startswith(request.prefix):
+                    continue
+                info = reply.bdevs.add()
+                info.name = bdev.name
+                info.size = bdev.size_bytes
+                info.block_size = bdev.block_size
+                info.product_name = bdev.product_name
+        return reply
+
     # --- self-registration --------------------------------------------------
 
     def register(self) -> None:

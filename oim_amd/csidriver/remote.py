@@ -193,8 +193,31 @@ class RemoteBackend(OIMBackend):
         self._snap_meta.pop(snapshot_id, None)
 
     def list_snapshots(self):
-        return [(snap_id, meta[0], meta[2], meta[1])
-                for snap_id, meta in sorted(self._snap_meta.items())]
+        """Daemon truth via ListMallocBDevs; source/ctime come from
+        this driver's metadata when it created the snapshot (the
+        oim.v0 API carries no provenance), else blank/0."""
+        out = []
+        with self._dial_registry() as channel:
+            stub = spec.ControllerStub(channel)
+            reply = stub.ListMallocBDevs(
+                spec.ListMallocBDevsRequest(prefix=self.SNAP_PREFIX),
+                metadata=self._metadata(), timeout=30)
+        for info in reply.bdevs:
+            source, ctime = "", 0
+            meta = self._snap_meta.get(info.name)
+            if meta:
+                source, ctime = meta[0], meta[1]
+            out.append((info.name, source, info.size, ctime))
+        return out
+
+    def list_volumes(self):
+        with self._dial_registry() as channel:
+            stub = spec.ControllerStub(channel)
+            reply = stub.ListMallocBDevs(
+                spec.ListMallocBDevsRequest(),
+                metadata=self._metadata(), timeout=30)
+        return [(info.name, info.size) for info in reply.bdevs
+                if not info.name.startswith(self.SNAP_PREFIX)]
 
     def restore_snapshot(self, snapshot_id, volume_name):
         with self._dial_registry() as channel:

@@ -71,6 +71,13 @@ MESSAGES = [
     Message("ResizeMallocBDevRequest",
             [Field("bdev_name", 1, "string"), Field("size", 2, "int64")]),
     Message("ResizeMallocBDevReply", []),
+    Message("ListMallocBDevsRequest", [Field("prefix", 1, "string")]),
+    Message("BDevInfo",
+            [Field("name", 1, "string"), Field("size", 2, "int64"),
+             Field("block_size", 3, "int64"),
+             Field("product_name", 4, "string")]),
+    Message("ListMallocBDevsReply",
+            [Field("bdevs", 1, "BDevInfo", repeated=True)]),
 ]
 
 SERVICES = [
@@ -90,6 +97,7 @@ SERVICES = [
             ("CheckMallocBDev", "CheckMallocBDevRequest", "CheckMallocBDevReply"),
             ("CloneMallocBDev", "CloneMallocBDevRequest", "CloneMallocBDevReply"),
             ("ResizeMallocBDev", "ResizeMallocBDevRequest", "ResizeMallocBDevReply"),
+            ("ListMallocBDevs", "ListMallocBDevsRequest", "ListMallocBDevsReply"),
         ],
     ),
 ]
@@ -119,3 +127,6 @@ CloneMallocBDevRequest = _classes["CloneMallocBDevRequest"]
 CloneMallocBDevReply = _classes["CloneMallocBDevReply"]
 ResizeMallocBDevRequest = _classes["ResizeMallocBDevRequest"]
 ResizeMallocBDevReply = _classes["ResizeMallocBDevReply"]
+ListMallocBDevsRequest = _classes["ListMallocBDevsRequest"]
+BDevInfo = _classes["BDevInfo"]
+ListMallocBDevsReply = _classes["ListMallocBDevsReply"]

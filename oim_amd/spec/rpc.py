@@ -50,6 +50,7 @@ class ControllerStub:
         self.CheckMallocBDev = method("CheckMallocBDev", pb.CheckMallocBDevReply)
         self.CloneMallocBDev = method("CloneMallocBDev", pb.CloneMallocBDevReply)
         self.ResizeMallocBDev = method("ResizeMallocBDev", pb.ResizeMallocBDevReply)
+        self.ListMallocBDevs = method("ListMallocBDevs", pb.ListMallocBDevsReply)
 
 
 class RegistryServicer:
@@ -83,6 +84,9 @@ class ControllerServicer:
     def ResizeMallocBDev(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "ResizeMallocBDev not implemented")
 
+    def ListMallocBDevs(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "ListMallocBDevs not implemented")
+
 
 def _unary(fn, request_class):
     return grpc.unary_unary_rpc_method_handler(
@@ -112,6 +116,7 @@ def add_controller_to_server(servicer: ControllerServicer, server: grpc.Server) 
         "CheckMallocBDev": _unary(servicer.CheckMallocBDev, pb.CheckMallocBDevRequest),
         "CloneMallocBDev": _unary(servicer.CloneMallocBDev, pb.CloneMallocBDevRequest),
         "ResizeMallocBDev": _unary(servicer.ResizeMallocBDev, pb.ResizeMallocBDevRequest),
+        "ListMallocBDevs": _unary(servicer.ListMallocBDevs, pb.ListMallocBDevsRequest),
     }
     server.add_generic_rpc_handlers(
         (grpc.method_handlers_generic_handler(CONTROLLER_SERVICE, handlers),)

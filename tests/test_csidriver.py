@@ -318,9 +318,24 @@ class TestCSIEndToEnd:
     def test_unimplemented_methods(self, control_plane):
         with csi_channel(control_plane) as channel:
             method = channel.unary_unary(
-                "/csi.v1.Controller/ListVolumes",
+                "/csi.v1.Controller/ControllerPublishVolume",
                 request_serializer=lambda b: b,
                 response_deserializer=lambda b: b)
             with pytest.raises(grpc.RpcError) as excinfo:
                 method(b"", timeout=5)
             assert excinfo.value.code() == grpc.StatusCode.UNIMPLEMENTED
+
+    def test_remote_list_volumes(self, control_plane):
+        """Remote mode enumerates via the ListMallocBDevs extension."""
+        from oim_amd.spec.rpc_csi import CSIControllerStub
+
+        with csi_channel(control_plane) as channel:
+            ctrl = CSIControllerStub(channel)
+            create = csi.CreateVolumeRequest(name="lv-remote")
+            create.capacity_range.required_bytes = 1 << 20
+            create.volume_capabilities.add().CopyFrom(single_writer_cap())
+            ctrl.CreateVolume(create, timeout=10)
+            reply = ctrl.ListVolumes(csi.ListVolumesRequest(), timeout=10)
+            assert "lv-remote" in [e.volume.volume_id for e in reply.entries]
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id="lv-remote"),
+                              timeout=10)
