@@ -42,11 +42,16 @@ def main(argv=None) -> int:
         ("check", "check that a malloc bdev exists"),
         ("clone", "clone a malloc bdev (device-side HBM-rate copy)"),
         ("resize", "grow a malloc bdev (offline)"),
+        ("list", "list malloc bdevs"),
     ):
         cmd = sub.add_parser(name, help=help_text + " via the registry proxy")
         cmd.add_argument("--controller", required=True,
                          help="controller ID to proxy to")
-        cmd.add_argument("volume")
+        if name == "list":
+            cmd.add_argument("volume", nargs="?", default="",
+                             help="name prefix filter")
+        else:
+            cmd.add_argument("volume")
         if name == "provision":
             cmd.add_argument("size", help='bytes, or "64MiB"-style; 0 deletes')
         if name == "clone":
@@ -113,6 +118,13 @@ def main(argv=None) -> int:
                                                 dest=args.dest),
                     metadata=metadata, timeout=120)
                 print(f"cloned {args.volume} -> {args.dest}")
+            elif args.command == "list":
+                reply = controller.ListMallocBDevs(
+                    spec.ListMallocBDevsRequest(prefix=args.volume),
+                    metadata=metadata, timeout=60)
+                for info in reply.bdevs:
+                    print(f"{info.name}  {info.size}  bs={info.block_size}  "
+                          f"{info.product_name}")
             elif args.command == "resize":
                 new_size = parse_size(args.size)
                 controller.ResizeMallocBDev(

@@ -82,6 +82,10 @@ class TestOimctlVolumes:
                                 "--controller", "c9", "volx"]) == 0
             assert oimctl.main(["--registry", endpoint, "clone",
                                 "--controller", "c9", "volx", "volx-c"]) == 0
+            assert oimctl.main(["--registry", endpoint, "list",
+                                "--controller", "c9"]) == 0
+            out = capsys.readouterr().out
+            assert "volx-c" in out
             assert oimctl.main(["--registry", endpoint, "resize",
                                 "--controller", "c9", "volx", "128MiB"]) == 0
             assert oimctl.main(["--registry", endpoint, "provision",
