@@ -75,3 +75,8 @@ class OIMBackend:
         """Clone a snapshot into a new volume; returns (volume_id,
         size_bytes). Raises LookupError for unknown snapshots."""
         raise NotImplementedError
+
+    def clone_volume(self, source_volume_id, volume_name):
+        """CSI volume cloning (CreateVolume from a volume source);
+        same contract as restore_snapshot."""
+        raise NotImplementedError

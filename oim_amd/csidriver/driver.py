@@ -106,8 +106,7 @@ class ControllerServer(CSIControllerServicer):
         if source.WhichOneof("type") == "snapshot":
             return self._create_from_snapshot(request, context)
         if source.WhichOneof("type") == "volume":
-            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
-                          "volume cloning source is not supported")
+            return self._create_from_volume(request, context)
         with self._name_mutex.get(request.name):
             try:
                 volume_id, volume_context = self.backend.create_volume(
@@ -151,6 +150,35 @@ class ControllerServer(CSIControllerServicer):
         response.volume.volume_id = volume_id
         response.volume.capacity_bytes = size
         response.volume.content_source.snapshot.snapshot_id = snapshot_id
+        return response
+
+    def _create_from_volume(self, request, context):
+        """CSI volume cloning: CreateVolume with a volume content
+        source — a bdev_clone of the source volume."""
+        if not self.backend.supports_snapshots():
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "volume cloning is not supported in this mode")
+        source_id = request.volume_content_source.volume.volume_id
+        if not source_id:
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT,
+                          "missing volume id in content source")
+        with self._name_mutex.get(request.name):
+            try:
+                volume_id, size = self.backend.clone_volume(
+                    source_id, request.name)
+            except LookupError as exc:
+                context.abort(grpc.StatusCode.NOT_FOUND, str(exc))
+            except grpc.RpcError as exc:
+                context.abort(exc.code(), exc.details())
+        required = request.capacity_range.required_bytes
+        if size and required and required > size:
+            context.abort(
+                grpc.StatusCode.OUT_OF_RANGE,
+                f"source volume is {size} bytes; cannot satisfy {required}")
+        response = csi.CreateVolumeResponse()
+        response.volume.volume_id = volume_id
+        response.volume.capacity_bytes = size
+        response.volume.content_source.volume.volume_id = source_id
         return response
 
     def DeleteVolume(self, request, context):
@@ -210,6 +238,8 @@ class ControllerServer(CSIControllerServicer):
             cap.rpc.type = csi.CTRL_CAP_CREATE_DELETE_SNAPSHOT
             cap = response.capabilities.add()
             cap.rpc.type = csi.CTRL_CAP_LIST_SNAPSHOTS
+            cap = response.capabilities.add()
+            cap.rpc.type = csi.CTRL_CAP_CLONE_VOLUME
         if self.backend.list_volumes() is not None:
             cap = response.capabilities.add()
             cap.rpc.type = csi.CTRL_CAP_LIST_VOLUMES

@@ -203,6 +203,9 @@ class LocalBackend(OIMBackend):
                 out.append((bdev.name, source, bdev.size_bytes, ctime))
         return out
 
+    def clone_volume(self, source_volume_id, volume_name):
+        return self.restore_snapshot(source_volume_id, volume_name)
+
     def restore_snapshot(self, snapshot_id, volume_name):
         with self._client() as client:
             try:

@@ -219,6 +219,15 @@ class RemoteBackend(OIMBackend):
         return [(info.name, info.size) for info in reply.bdevs
                 if not info.name.startswith(self.SNAP_PREFIX)]
 
+    def clone_volume(self, source_volume_id, volume_name):
+        volume_id, size = self.restore_snapshot(source_volume_id,
+                                                volume_name)
+        if not size:
+            size = self._volume_sizes.get(source_volume_id, 0)
+            if size:
+                self._volume_sizes[volume_name] = size
+        return volume_id, size
+
     def restore_snapshot(self, snapshot_id, volume_name):
         with self._dial_registry() as channel:
             stub = spec.ControllerStub(channel)

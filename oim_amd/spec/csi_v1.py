@@ -40,6 +40,7 @@ CTRL_CAP_LIST_VOLUMES = 3
 CTRL_CAP_GET_CAPACITY = 4
 CTRL_CAP_CREATE_DELETE_SNAPSHOT = 5
 CTRL_CAP_LIST_SNAPSHOTS = 6
+CTRL_CAP_CLONE_VOLUME = 7
 CTRL_CAP_EXPAND_VOLUME = 9
 
 # NodeServiceCapability.RPC.Type

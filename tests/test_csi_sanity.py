@@ -671,3 +671,42 @@ class TestExpansionNeverShrinks:
         finally:
             ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id="no-shrink"),
                               timeout=10)
+
+
+class TestVolumeCloning:
+    """CreateVolume from a volume content source (CLONE_VOLUME)."""
+
+    def test_clone_volume(self, sanity_env):
+        ctrl = sanity_env["controller"]
+        caps = ctrl.ControllerGetCapabilities(
+            csi.ControllerGetCapabilitiesRequest(), timeout=10)
+        assert csi.CTRL_CAP_CLONE_VOLUME in {
+            cap.rpc.type for cap in caps.capabilities}
+        create = csi.CreateVolumeRequest(name="clone-src")
+        create.capacity_range.required_bytes = 1 << 20
+        create.volume_capabilities.add().CopyFrom(mount_cap())
+        ctrl.CreateVolume(create, timeout=10)
+        try:
+            request = csi.CreateVolumeRequest(name="clone-dst")
+            request.volume_capabilities.add().CopyFrom(mount_cap())
+            request.volume_content_source.volume.volume_id = "clone-src"
+            reply = ctrl.CreateVolume(request, timeout=10)
+            assert reply.volume.volume_id == "clone-dst"
+            assert reply.volume.capacity_bytes == 1 << 20
+            assert (reply.volume.content_source.volume.volume_id
+                    == "clone-src")
+            # idempotent retry
+            again = ctrl.CreateVolume(request, timeout=10)
+            assert again.volume.volume_id == "clone-dst"
+            # unknown source
+            request2 = csi.CreateVolumeRequest(name="clone-ghost")
+            request2.volume_capabilities.add().CopyFrom(mount_cap())
+            request2.volume_content_source.volume.volume_id = "nope"
+            with pytest.raises(grpc.RpcError) as excinfo:
+                ctrl.CreateVolume(request2, timeout=10)
+            assert_code(excinfo, grpc.StatusCode.NOT_FOUND)
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id="clone-dst"),
+                              timeout=10)
+        finally:
+            ctrl.DeleteVolume(csi.DeleteVolumeRequest(volume_id="clone-src"),
+                              timeout=10)
