@@ -257,6 +257,24 @@ class TestVhostHbm:
                         assert master.read10(0, i, 1, 4096).data == block
                 finally:
                     master.close()
+                # blk personality over the same HBM engine
+                client.invoke("construct_malloc_bdev",
+                              {"name": "vhgb", "num_blocks": 65536,
+                               "block_size": 4096})
+                client.invoke("construct_vhost_blk_controller",
+                              {"ctrlr": "vbgpu", "dev_name": "vhgb"})
+                blk_path = os.path.join(
+                    os.path.dirname(daemon.socket_path), "vbgpu")
+                blk = VhostUserMaster(blk_path, queue=0)
+                blk.negotiate()
+                try:
+                    payload = os.urandom(8 * 4096)
+                    status, _ = blk.blk_write(64, payload)  # sector units
+                    assert status == 0
+                    status, back = blk.blk_read(64, len(payload))
+                    assert status == 0 and back == payload
+                finally:
+                    blk.close()
         finally:
             daemon.stop()
 
