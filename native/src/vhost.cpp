@@ -51,6 +51,8 @@ enum : uint32_t {
   kSetProtocolFeatures = 16,
   kGetQueueNum = 17,
   kSetVringEnable = 18,
+  kGetConfig = 24,
+  kSetConfig = 25,
 };
 
 constexpr uint32_t kVersionMask = 0x3;
@@ -65,11 +67,14 @@ constexpr uint64_t kFeatIndirect = 1ull << 28;
 constexpr uint64_t kFeatProtocol = 1ull << 30;
 constexpr uint64_t kDeviceFeatures = kFeatVersion1 | kFeatIndirect |
                                      kFeatProtocol;
-constexpr uint64_t kProtocolFeatures = 1ull << 0;  // MQ
+constexpr uint64_t kProtocolFeatureMq = 1ull << 0;
+constexpr uint64_t kProtocolFeatureConfig = 1ull << 9;  // GET/SET_CONFIG
 
 constexpr int kMaxVrings = 8;      // controlq + eventq + 6 request queues
 constexpr int kMaxRegions = 8;
-constexpr size_t kMaxPayload = 8 + kMaxRegions * 32;  // SET_MEM_TABLE
+// Largest legal payloads: SET_MEM_TABLE (8 + 32*regions) and
+// GET/SET_CONFIG (12-byte header + up to 256 config bytes).
+constexpr size_t kMaxPayload = 12 + 256;
 
 struct VringStateWire {
   uint32_t index;
@@ -339,6 +344,30 @@ class VhostUserScsiDev {
     return (msg->flags & kVersionMask) == kVersion1;
   }
 
+  // virtio_blk_config: capacity in 512-byte sectors at offset 0,
+  // blk_size (u32) at offset 20; everything else zero (no geometry /
+  // topology hints).
+  void fill_virtio_config(uint8_t* config, size_t size) {
+    if (personality_ != Personality::kBlk || size < 24) return;
+    BdevPtr bdev = resolver_(0);
+    if (!bdev) return;
+    const uint64_t sectors = bdev->size_bytes() / 512;
+    memcpy(config, &sectors, 8);
+    const uint32_t blk_size = static_cast<uint32_t>(bdev->block_size());
+    memcpy(config + 20, &blk_size, 4);
+  }
+
+  void send_reply_big(int conn, uint32_t request, const void* payload,
+                      uint32_t size) {
+    uint8_t buf[12 + 12 + 256];
+    uint32_t flags = kVersion1 | kFlagReply;
+    memcpy(buf, &request, 4);
+    memcpy(buf + 4, &flags, 4);
+    memcpy(buf + 8, &size, 4);
+    memcpy(buf + 12, payload, size);
+    (void)!::send(conn, buf, 12 + size, MSG_NOSIGNAL);
+  }
+
   void send_reply(int conn, uint32_t request, const void* payload,
                   uint32_t size) {
     uint8_t buf[12 + 64];
@@ -392,7 +421,10 @@ class VhostUserScsiDev {
         read_payload(msg, &negotiated_features_);
         return true;
       case kGetProtocolFeatures: {
-        uint64_t f = kProtocolFeatures;
+        // blk devices carry their geometry in virtio config space
+        // (QEMU reads capacity via GET_CONFIG); scsi needs only MQ.
+        uint64_t f = kProtocolFeatureMq;
+        if (personality_ == Personality::kBlk) f |= kProtocolFeatureConfig;
         send_reply(conn, msg.request, &f, 8);
         return true;
       }
@@ -436,6 +468,29 @@ class VhostUserScsiDev {
         rings_[s.index].enabled.store(s.num != 0, std::memory_order_release);
         return true;
       }
+      case kGetConfig: {
+        // VhostUserConfig {u32 offset; u32 size; u32 flags; u8 region[]}
+        uint32_t offset = 0, size = 0, flags = 0;
+        if (msg.payload.size() < 12) return false;
+        memcpy(&offset, msg.payload.data(), 4);
+        memcpy(&size, msg.payload.data() + 4, 4);
+        memcpy(&flags, msg.payload.data() + 8, 4);
+        if (size > 256) return false;
+        uint8_t config[256] = {0};
+        fill_virtio_config(config, sizeof(config));
+        uint8_t reply[12 + 256];
+        memcpy(reply, &offset, 4);
+        memcpy(reply + 4, &size, 4);
+        memcpy(reply + 8, &flags, 4);
+        for (uint32_t i = 0; i < size; ++i) {
+          reply[12 + i] =
+              (offset + i < sizeof(config)) ? config[offset + i] : 0;
+        }
+        send_reply_big(conn, msg.request, reply, 12 + size);
+        return true;
+      }
+      case kSetConfig:
+        return true;  // no writable config fields
       case kGetVringBase: {
         VringStateWire s;
         if (!read_payload(msg, &s) || s.index >= kMaxVrings) return false;

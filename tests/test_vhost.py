@@ -386,6 +386,15 @@ class TestVhostBlk:
             yield client, master
             master.close()
 
+    def test_virtio_config_capacity(self, blk_target):
+        """QEMU reads the device size via GET_CONFIG (virtio config
+        space); capacity is in 512-byte sectors."""
+        from vhost_client import read_blk_config
+        _, master = blk_target
+        config = read_blk_config(master)
+        assert config["capacity_sectors"] == NUM_BLOCKS * BLOCK // 512
+        assert config["blk_size"] == BLOCK
+
     def test_rw_roundtrip_and_get_id(self, blk_target):
         _, master = blk_target
         data = os.urandom(4 * BLOCK)

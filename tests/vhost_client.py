@@ -332,3 +332,15 @@ class VhostUserMaster:
         cdb[0] = 0xA0
         cdb[6:10] = struct.pack(">I", 256)
         return self.scsi(target, bytes(cdb), data_in_len=256)
+
+GET_CONFIG = 24
+
+
+def read_blk_config(master) -> dict:
+    """VHOST_USER_GET_CONFIG -> virtio_blk_config fields."""
+    payload = struct.pack("<III", 0, 24, 0) + bytes(24)
+    reply = master.query(GET_CONFIG, payload)
+    region = reply[12:]
+    capacity, = struct.unpack("<Q", region[0:8])
+    blk_size, = struct.unpack("<I", region[20:24])
+    return {"capacity_sectors": capacity, "blk_size": blk_size}
