@@ -67,6 +67,8 @@ constexpr uint64_t kFeatIndirect = 1ull << 28;
 constexpr uint64_t kFeatProtocol = 1ull << 30;
 constexpr uint64_t kDeviceFeatures = kFeatVersion1 | kFeatIndirect |
                                      kFeatProtocol;
+constexpr uint64_t kFeatBlkDiscard = 1ull << 13;
+constexpr uint64_t kFeatBlkWriteZeroes = 1ull << 14;
 constexpr uint64_t kProtocolFeatureMq = 1ull << 0;
 constexpr uint64_t kProtocolFeatureConfig = 1ull << 9;  // GET/SET_CONFIG
 
@@ -355,6 +357,17 @@ class VhostUserScsiDev {
     memcpy(config, &sectors, 8);
     const uint32_t blk_size = static_cast<uint32_t>(bdev->block_size());
     memcpy(config + 20, &blk_size, 4);
+    if (size >= 56) {
+      // discard/write-zeroes geometry (virtio_blk_config offsets)
+      const uint32_t max_sectors = kMaxIoBytes / 512;
+      const uint32_t one_seg = 256;
+      const uint32_t align = blk_size / 512;
+      memcpy(config + 36, &max_sectors, 4);   // max_discard_sectors
+      memcpy(config + 40, &one_seg, 4);       // max_discard_seg
+      memcpy(config + 44, &align, 4);         // discard_sector_alignment
+      memcpy(config + 48, &max_sectors, 4);   // max_write_zeroes_sectors
+      memcpy(config + 52, &one_seg, 4);       // max_write_zeroes_seg
+    }
   }
 
   void send_reply_big(int conn, uint32_t request, const void* payload,
@@ -414,6 +427,9 @@ class VhostUserScsiDev {
     switch (msg.request) {
       case kGetFeatures: {
         uint64_t f = kDeviceFeatures;
+        if (personality_ == Personality::kBlk) {
+          f |= kFeatBlkDiscard | kFeatBlkWriteZeroes;
+        }
         send_reply(conn, msg.request, &f, 8);
         return true;
       }
@@ -789,7 +805,8 @@ class VhostUserScsiDev {
                           const std::vector<Iov>& in,
                           ChannelCache* channels) {
     constexpr uint32_t kTypeIn = 0, kTypeOut = 1, kTypeFlush = 4,
-                       kTypeGetId = 8;
+                       kTypeGetId = 8, kTypeDiscard = 11,
+                       kTypeWriteZeroes = 13;
     constexpr uint8_t kStatusOk = 0, kStatusIoErr = 1, kStatusUnsupp = 2;
     if (out.empty() || in.empty()) return 0;
     uint8_t header[16];
@@ -832,6 +849,48 @@ class VhostUserScsiDev {
       scatter(data_in, reinterpret_cast<uint8_t*>(id),
               std::min<size_t>(sizeof(id), iov_total(data_in)));
       status = kStatusOk;
+    } else if (type == kTypeDiscard || type == kTypeWriteZeroes) {
+      // data-out = N x 16-byte segments {sector u64; num u32; flags u32}
+      const uint64_t seg_bytes = iov_total(data_out);
+      status = kStatusOk;
+      if (readonly_ || seg_bytes % 16 != 0 || seg_bytes == 0 ||
+          seg_bytes / 16 > 256) {
+        status = kStatusIoErr;
+      } else {
+        std::vector<uint8_t> segs(seg_bytes);
+        gather(segs.data(), data_out, seg_bytes);
+        for (size_t off = 0; off < seg_bytes && status == kStatusOk;
+             off += 16) {
+          uint64_t seg_sector;
+          uint32_t num;
+          memcpy(&seg_sector, segs.data() + off, 8);
+          memcpy(&num, segs.data() + off + 8, 4);
+          if (num == 0) continue;
+          const uint64_t z_off = seg_sector * 512;
+          const uint64_t z_len = static_cast<uint64_t>(num) * 512;
+          if (z_off % bdev->block_size() != 0 ||
+              z_len % bdev->block_size() != 0 ||
+              z_off + z_len > bdev->size_bytes() || z_len > kMaxIoBytes) {
+            status = kStatusIoErr;
+            break;
+          }
+          auto ch = channel_for(channels, bdev);
+          int io_status = 1;
+          bool done = false;
+          IoRequest zero;
+          zero.op = IoOp::kFill;
+          zero.offset = z_off;
+          zero.length = z_len;
+          zero.fill = 0;
+          zero.on_complete = [&](int st) {
+            io_status = st;
+            done = true;
+          };
+          bdev->submit(ch.get(), std::move(zero));
+          while (!done) bdev->poll(ch.get());
+          if (io_status != kIoOk) status = kStatusIoErr;
+        }
+      }
     } else if (type == kTypeIn || type == kTypeOut) {
       const bool is_write = type == kTypeOut;
       const std::vector<Iov>& iovs = is_write ? data_out : data_in;
@@ -991,6 +1050,47 @@ class VhostUserScsiDev {
       case 0x35:  // SYNCHRONIZE CACHE(10)
       case 0x91:  // SYNCHRONIZE CACHE(16)
         return 0;
+      case 0x42: {  // UNMAP: zero the listed extents (thin-provision trim)
+        const uint32_t param_len = rbe16(cdb + 7);
+        if (param_len < 8) return 0;  // empty list: success
+        std::vector<uint8_t> list(param_len);
+        if (gather(list.data(), data_out, param_len) < param_len) {
+          build_sense(resp, 0x05, 0x24, 0x00);
+          return 0;
+        }
+        const uint16_t desc_bytes = rbe16(list.data() + 2);
+        for (uint32_t off = 8; off + 16 <= 8u + desc_bytes &&
+                               off + 16 <= param_len;
+             off += 16) {
+          const uint64_t lba = rbe64(list.data() + off);
+          const uint32_t nlb = rbe32(list.data() + off + 8);
+          if (nlb == 0) continue;
+          if (lba + nlb > nblocks ||
+              static_cast<uint64_t>(nlb) * block > kMaxIoBytes) {
+            build_sense(resp, 0x05, 0x21, 0x00);
+            return 0;
+          }
+          IoRequest zero;
+          auto ch = channel_for(channels, bdev);
+          int status = 1;
+          bool done = false;
+          zero.op = IoOp::kFill;
+          zero.offset = lba * block;
+          zero.length = static_cast<uint64_t>(nlb) * block;
+          zero.fill = 0;
+          zero.on_complete = [&](int st) {
+            status = st;
+            done = true;
+          };
+          bdev->submit(ch.get(), std::move(zero));
+          while (!done) bdev->poll(ch.get());
+          if (status != kIoOk) {
+            resp->response = kRespFailure;
+            return 0;
+          }
+        }
+        return 0;
+      }
       case 0x12: {  // INQUIRY
         uint8_t buf[96] = {};
         size_t len;
@@ -998,10 +1098,17 @@ class VhostUserScsiDev {
           switch (cdb[2]) {
             case 0x00:  // supported VPD pages
               buf[1] = 0x00;
-              buf[3] = 2;
+              buf[3] = 3;
               buf[4] = 0x00;
               buf[5] = 0x80;
-              len = 6;
+              buf[6] = 0xB2;
+              len = 7;
+              break;
+            case 0xB2:  // logical block provisioning: LBPU (UNMAP)
+              buf[1] = 0xB2;
+              buf[3] = 4;
+              buf[5] = 0x80;  // LBPU
+              len = 8;
               break;
             case 0x80: {  // unit serial number
               const std::string& uuid = bdev->uuid();
@@ -1047,6 +1154,7 @@ class VhostUserScsiDev {
         uint8_t buf[32] = {};
         be64(buf, nblocks - 1);
         be32(buf + 8, static_cast<uint32_t>(block));
+        buf[14] = 0x80;  // LBPME: logical block provisioning (UNMAP ok)
         uint32_t alloc = rbe32(cdb + 10);
         return scatter(data_in, buf, std::min<size_t>(sizeof(buf), alloc));
       }

@@ -443,3 +443,69 @@ class TestVhostBlk:
             client.invoke("remove_vhost_controller", {"ctrlr": "vbro"})
             # claim released: the bdev is deletable again
             hipstore.delete_bdev(client, "blkro")
+
+
+class TestTrim:
+    """Thin-provisioning trim: SCSI UNMAP and virtio-blk DISCARD zero
+    the addressed extents."""
+
+    def test_scsi_unmap(self, vhost_target):
+        _, master = vhost_target
+        data = os.urandom(8 * BLOCK)
+        assert master.write10(0, 16, data, BLOCK).status == 0
+        # VPD B2 advertises LBPU; READ CAPACITY(16) sets LBPME
+        vpd = master.inquiry(0, evpd_page=0xB2)
+        assert vpd.status == 0 and vpd.data[5] & 0x80
+        # UNMAP blocks 16..19 (first 4 of the 8 written)
+        descriptors = struct.pack(">QII", 16, 4, 0)
+        param = struct.pack(">HH", 2 + 4 + len(descriptors),
+                            len(descriptors)) + bytes(4) + descriptors
+        cdb = bytearray(10)
+        cdb[0] = 0x42
+        cdb[7:9] = struct.pack(">H", len(param))
+        result = master.scsi(0, bytes(cdb), data_out=param)
+        assert result.status == 0, (result.status, result.sense)
+        zeroed = master.read10(0, 16, 4, BLOCK)
+        assert zeroed.data == bytes(4 * BLOCK)
+        kept = master.read10(0, 20, 4, BLOCK)
+        assert kept.data == data[4 * BLOCK:]
+
+    def test_scsi_unmap_out_of_range(self, vhost_target):
+        _, master = vhost_target
+        descriptors = struct.pack(">QII", NUM_BLOCKS, 4, 0)
+        param = struct.pack(">HH", 6 + len(descriptors),
+                            len(descriptors)) + bytes(4) + descriptors
+        cdb = bytearray(10)
+        cdb[0] = 0x42
+        cdb[7:9] = struct.pack(">H", len(param))
+        result = master.scsi(0, bytes(cdb), data_out=param)
+        assert result.status == 2 and result.asc == 0x21
+
+    def test_blk_discard(self, hipstored, tmp_path):  # noqa: F811
+        from vhost_client import FEAT_VERSION_1
+        with hipstore.Client(hipstored.socket_path) as client:
+            hipstore.construct_malloc_bdev(
+                client, num_blocks=2048, block_size=512, name="blktrim")
+            client.invoke("construct_vhost_blk_controller",
+                          {"ctrlr": "vbt", "dev_name": "blktrim"})
+            path = os.path.join(os.path.dirname(hipstored.socket_path),
+                                "vbt")
+            master = VhostUserMaster(path, queue=0)
+            master.negotiate(features_extra=(1 << 13) | (1 << 14))
+            try:
+                data = os.urandom(4 * 512)
+                assert master.blk_write(32, data)[0] == 0
+                segment = struct.pack("<QII", 32, 2, 0)  # first 2 sectors
+                status, _ = master.blk(11, 0, data_out=segment)  # DISCARD
+                assert status == 0
+                status, back = master.blk_read(32, 4 * 512)
+                assert status == 0
+                assert back[:1024] == bytes(1024)
+                assert back[1024:] == data[1024:]
+                # WRITE_ZEROES with a bad segment -> IOERR
+                bad = struct.pack("<QII", 4096, 2, 0)
+                assert master.blk(13, 0, data_out=bad)[0] == 1
+            finally:
+                master.close()
+            client.invoke("remove_vhost_controller", {"ctrlr": "vbt"})
+            hipstore.delete_bdev(client, "blktrim")
