@@ -128,3 +128,25 @@ class TestPerfEdgeCases:
         r = hs.run_bdevperf(b, "randread", 4096, 8, 2, 60.0, max_ios=500)
         assert 500 <= r["io_count"] <= 600  # cap honored, not the 60s
         assert r["seconds"] < 10
+
+
+class TestCloneOfComposite:
+    def test_clone_striped_into_malloc(self, hipstored):  # noqa: F811
+        """bdev_clone reads any Bdev (striped included) and produces a
+        plain malloc clone with identical content."""
+        with hipstore.Client(hipstored.socket_path) as client:
+            client.invoke("construct_striped_malloc_bdev", {
+                "name": "clsrc-st", "num_blocks": 1024, "block_size": 512,
+                "stripe_size_kb": 64, "count": 2})
+            result = hipstore.perf_run(client, "clsrc-st", io_size=4096,
+                                       queue_depth=4, num_queues=1,
+                                       seconds=0.2, workload="randwrite")
+            assert result["io_count"] > 0
+            client.invoke("bdev_clone", {"src": "clsrc-st",
+                                         "name": "clst-copy"})
+            copy = hipstore.get_bdevs(client, "clst-copy")[0]
+            src = hipstore.get_bdevs(client, "clsrc-st")[0]
+            assert copy.num_blocks == src.num_blocks
+            assert copy.product_name == "Striped Malloc disk"  # re-badged
+            hipstore.delete_bdev(client, "clst-copy")
+            hipstore.delete_bdev(client, "clsrc-st")

@@ -298,3 +298,55 @@ class TestTLS:
                     timeout=5,
                 )
             assert excinfo.value.code() == grpc.StatusCode.PERMISSION_DENIED
+
+
+class TestProxyDeadlines:
+    def test_client_deadline_propagates_through_proxy(self, tmp_path):
+        """A short client deadline on a proxied call to a stalled
+        controller surfaces as DEADLINE_EXCEEDED at the client and
+        does not wedge the proxy for later calls."""
+        import threading
+        import time as time_mod
+
+        from oim_amd import spec
+        from oim_amd.common.server import NonBlockingGRPCServer, grpc_target
+        from oim_amd.spec.rpc import (ControllerServicer,
+                                      add_controller_to_server)
+
+        class StallServicer(ControllerServicer):
+            def __init__(self):
+                self.calls = 0
+
+            def CheckMallocBDev(self, request, context):
+                self.calls += 1
+                if self.calls == 1:
+                    time_mod.sleep(3)  # beyond the client's deadline
+                return spec.CheckMallocBDevReply()
+
+        stalled = StallServicer()
+        ctrl_server = NonBlockingGRPCServer(
+            endpoint=f"unix://{tmp_path}/slow.sock")
+        ctrl_server.start(lambda s: add_controller_to_server(stalled, s))
+        registry = Registry(db=MemRegistryDB())
+        reg_server = RegistryServer(f"unix://{tmp_path}/reg.sock", registry)
+        reg_server.start()
+        registry.db.store(["slow", "address"], f"unix://{tmp_path}/slow.sock")
+        try:
+            with grpc.insecure_channel(grpc_target(reg_server.addr())) as ch:
+                stub = spec.ControllerStub(ch)
+                md = ((spec.CONTROLLER_ID_KEY, "slow"),)
+                started = time_mod.monotonic()
+                with pytest.raises(grpc.RpcError) as excinfo:
+                    stub.CheckMallocBDev(
+                        spec.CheckMallocBDevRequest(bdev_name="x"),
+                        metadata=md, timeout=0.5)
+                assert excinfo.value.code() == \
+                    grpc.StatusCode.DEADLINE_EXCEEDED
+                assert time_mod.monotonic() - started < 2.5
+                # proxy healthy for the next (fast) call
+                stub.CheckMallocBDev(
+                    spec.CheckMallocBDevRequest(bdev_name="x"),
+                    metadata=md, timeout=10)
+        finally:
+            reg_server.stop()
+            ctrl_server.stop()
