@@ -234,7 +234,7 @@ __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
   while (true) {
     // One lane claims the next descriptor index; the wave follows
     // (64-bit broadcast as two 32-bit halves).
-    unsigned long long claim;
+    unsigned long long claim = 0;
     if (lane == 0) {
       claim = __hip_atomic_fetch_add(ctl.claim_counter, 1ull,
                                      __ATOMIC_RELAXED,
@@ -428,7 +428,7 @@ __global__ __launch_bounds__(64) void k_shared_service(SharedCtl ctl) {
       unsigned long long c = slot_ld(slot, 3);
       unsigned long long claim = ~0ull;
       while (c < kt) {
-        unsigned long long witnessed;
+        unsigned long long witnessed = 0;
         if (lane == 0) {
           witnessed = atomicCAS(&slot->claim, c, c + 1);
         }
@@ -743,7 +743,7 @@ class HbmChannel : public HbmChannelBase {
   }
 
   uint8_t* base_;
-  uint64_t size_;
+  [[maybe_unused]] uint64_t size_;  // geometry echo; bounds live in HbmBdev
   int device_ = 0;
   hipStream_t stream_ = nullptr;
   BlockDesc* ring_ = nullptr;      // pinned host

@@ -667,7 +667,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   });
 
   server->register_method(
-      "bdev_clone", [&manager, use_hbm, device, persistent](const Json& p) {
+      "bdev_clone", [&manager, use_hbm, persistent](const Json& p) {
         // Volume clone: new malloc bdev + device-side range copy (HBM
         // rates; xGMI when cloning to another GPU via "device").
         BdevPtr src = manager.find(p.get_string("src"));

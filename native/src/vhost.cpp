@@ -162,7 +162,6 @@ constexpr uint8_t kStatusCheckCondition = 0x02;
 
 constexpr uint64_t kMaxIoBytes = 4ull << 20;  // per-command cap
 
-void be16(uint8_t* p, uint16_t v) { p[0] = v >> 8; p[1] = v; }
 void be32(uint8_t* p, uint32_t v) {
   p[0] = v >> 24; p[1] = v >> 16; p[2] = v >> 8; p[3] = v;
 }
