@@ -139,6 +139,8 @@ class ControllerServer(CSIControllerServicer):
                     snapshot_id, request.name)
             except LookupError as exc:
                 context.abort(grpc.StatusCode.NOT_FOUND, str(exc))
+            except RuntimeError as exc:
+                context.abort(grpc.StatusCode.INTERNAL, str(exc))
             except grpc.RpcError as exc:
                 context.abort(exc.code(), exc.details())
         required = request.capacity_range.required_bytes
@@ -168,6 +170,8 @@ class ControllerServer(CSIControllerServicer):
                     source_id, request.name)
             except LookupError as exc:
                 context.abort(grpc.StatusCode.NOT_FOUND, str(exc))
+            except RuntimeError as exc:
+                context.abort(grpc.StatusCode.INTERNAL, str(exc))
             except grpc.RpcError as exc:
                 context.abort(exc.code(), exc.details())
         required = request.capacity_range.required_bytes
@@ -325,6 +329,8 @@ class ControllerServer(CSIControllerServicer):
                 context.abort(grpc.StatusCode.ALREADY_EXISTS, str(exc))
             except LookupError as exc:
                 context.abort(grpc.StatusCode.NOT_FOUND, str(exc))
+            except RuntimeError as exc:
+                context.abort(grpc.StatusCode.INTERNAL, str(exc))
             except grpc.RpcError as exc:
                 context.abort(exc.code(), exc.details())
         response = csi.CreateSnapshotResponse()
@@ -345,6 +351,8 @@ class ControllerServer(CSIControllerServicer):
         with self._name_mutex.get("snap-" + request.snapshot_id):
             try:
                 self.backend.delete_snapshot(request.snapshot_id)
+            except RuntimeError as exc:
+                context.abort(grpc.StatusCode.INTERNAL, str(exc))
             except grpc.RpcError as exc:
                 context.abort(exc.code(), exc.details())
         return csi.DeleteSnapshotResponse()

@@ -177,8 +177,11 @@ class LocalBackend(OIMBackend):
                     raise LookupError(
                         f"volume {source_volume_id} not found") from None
                 raise
-            client.invoke("bdev_clone",
-                          {"src": source_volume_id, "name": snap_id})
+            try:
+                client.invoke("bdev_clone",
+                              {"src": source_volume_id, "name": snap_id})
+            except hipstore.RpcError as err:
+                raise RuntimeError(str(err)) from None
             bdev = hipstore.get_bdevs(client, snap_id)[0]
         ctime = int(_time.time())
         self._snap_meta[snap_id] = (source_volume_id, ctime)
@@ -190,7 +193,7 @@ class LocalBackend(OIMBackend):
                 hipstore.delete_bdev(client, snapshot_id)
             except hipstore.RpcError as err:
                 if not err.is_not_found():
-                    raise
+                    raise RuntimeError(str(err)) from None
         self._snap_meta.pop(snapshot_id, None)
 
     def list_snapshots(self):
@@ -223,7 +226,10 @@ class LocalBackend(OIMBackend):
                     raise LookupError(
                         f"snapshot {snapshot_id} not found") from None
                 raise
-            client.invoke("bdev_clone",
-                          {"src": snapshot_id, "name": volume_name})
+            try:
+                client.invoke("bdev_clone",
+                              {"src": snapshot_id, "name": volume_name})
+            except hipstore.RpcError as err:
+                raise RuntimeError(str(err)) from None
             bdev = hipstore.get_bdevs(client, volume_name)[0]
         return volume_name, bdev.size_bytes
