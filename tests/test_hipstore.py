@@ -256,3 +256,26 @@ class TestConfigSnapshot:
                     {"subsystem": "bdev", "config": [
                         {"method": "no_such_method", "params": {}}]}]})
             hipstore.delete_bdev(client, "lc0")
+
+    def test_blk_controller_in_config(self, hipstored):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            hipstore.construct_malloc_bdev(
+                client, num_blocks=1024, block_size=512, name="cblk")
+            client.invoke("construct_vhost_blk_controller",
+                          {"ctrlr": "cfgblk", "dev_name": "cblk",
+                           "readonly": True})
+            config = client.invoke("save_config")
+            vhost = [e for sub in config["subsystems"]
+                     if sub["subsystem"] == "vhost" for e in sub["config"]]
+            blk = [e for e in vhost
+                   if e["method"] == "construct_vhost_blk_controller"]
+            assert blk and blk[0]["params"]["dev_name"] == "cblk"
+            assert blk[0]["params"]["readonly"] is True
+            client.invoke("remove_vhost_controller", {"ctrlr": "cfgblk"})
+            # replaying the saved vhost entry recreates the controller
+            client.invoke("load_config", {"subsystems": [
+                {"subsystem": "vhost", "config": blk}]})
+            listing = hipstore.get_vhost_controllers(client)
+            assert any(c.controller == "cfgblk" for c in listing)
+            client.invoke("remove_vhost_controller", {"ctrlr": "cfgblk"})
+            hipstore.delete_bdev(client, "cblk")
