@@ -59,9 +59,15 @@ class RpcServer {
   int listen_fd_ = -1;
   std::atomic<bool> running_{false};
   std::thread accept_thread_;
-  // (thread, finished flag) so the accept loop can reap exited ones.
-  std::vector<std::pair<std::thread, std::shared_ptr<std::atomic<bool>>>>
-      connections_;
+  // Connection bookkeeping: the accept loop reaps finished threads,
+  // and stop() shuts the live fds down so blocked reads unblock
+  // (otherwise shutdown waits on connected-but-idle clients).
+  struct Connection {
+    std::thread thread;
+    std::shared_ptr<std::atomic<bool>> done;
+    int fd;
+  };
+  std::vector<Connection> connections_;
   std::map<std::string, RpcMethod> methods_;
   mutable std::mutex mutex_;
 };

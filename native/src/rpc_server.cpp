@@ -56,8 +56,16 @@ void RpcServer::stop() {
   ::shutdown(listen_fd_, SHUT_RDWR);
   ::close(listen_fd_);
   if (accept_thread_.joinable()) accept_thread_.join();
-  for (auto& [t, done] : connections_) {
-    if (t.joinable()) t.join();
+  {
+    // Unblock connection threads parked in read() on live clients;
+    // without this, shutdown waits for every client to hang up.
+    std::lock_guard<std::mutex> lock(mutex_);
+    for (auto& conn : connections_) {
+      if (!conn.done->load()) ::shutdown(conn.fd, SHUT_RDWR);
+    }
+  }
+  for (auto& conn : connections_) {
+    if (conn.thread.joinable()) conn.thread.join();
   }
   connections_.clear();
   unlink(socket_path_.c_str());
@@ -74,20 +82,20 @@ void RpcServer::accept_loop() {
     // Reap finished connection threads so a long-lived daemon does not
     // accumulate joinable handles (one per past client).
     for (auto it = connections_.begin(); it != connections_.end();) {
-      if (it->second->load()) {
-        it->first.join();
+      if (it->done->load()) {
+        it->thread.join();
         it = connections_.erase(it);
       } else {
         ++it;
       }
     }
     auto done = std::make_shared<std::atomic<bool>>(false);
-    connections_.emplace_back(
+    connections_.push_back(Connection{
         std::thread([this, fd, done] {
           serve_connection(fd);
           done->store(true);
         }),
-        done);
+        done, fd});
   }
 }
 

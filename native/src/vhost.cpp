@@ -701,8 +701,20 @@ class VhostUserScsiDev {
     negotiated_features_ = 0;
   }
 
+  static bool pipeline_enabled() {
+    static const bool on = [] {
+      const char* env = getenv("HIPSTORE_VHOST_PIPELINE");
+      return env != nullptr && atoi(env) != 0;
+    }();
+    return on;
+  }
+
   void ring_worker(unsigned index) {
     ChannelCache channels;  // this worker's engine channels
+    if (pipeline_enabled()) {
+      ring_worker_pipelined(index, &channels);
+      return;
+    }
     Vring& ring = rings_[index];
     while (ring.running.load(std::memory_order_relaxed)) {
       pollfd fds[2] = {{ring.kick, POLLIN, 0}, {ring.stop_pipe[0], POLLIN, 0}};
@@ -716,6 +728,233 @@ class VhostUserScsiDev {
       if (!ring.enabled.load(std::memory_order_acquire)) continue;
       drain_ring(index, &channels);
     }
+  }
+
+  // -- pipelined worker (HIPSTORE_VHOST_PIPELINE=1) ----------------------
+  //
+  // Keeps up to kMaxInflight commands outstanding per ring instead of
+  // one sync I/O per command: simple single-extent READ/WRITE chains
+  // (the entire hot path of a guest) are submitted asynchronously and
+  // their used entries published out of order as the engine completes
+  // them (virtio permits this — the id field identifies the chain).
+  // Everything else falls back to the synchronous per-command path.
+  static constexpr uint32_t kMaxInflight = 64;
+
+  struct InflightIo {
+    uint16_t head;
+    std::vector<Iov> resp_iovs;   // SCSI response location (empty = blk)
+    uint8_t* blk_status = nullptr;
+    uint32_t data_in_bytes = 0;   // reported in the used entry
+  };
+
+  void ring_worker_pipelined(unsigned index, ChannelCache* channels) {
+    Vring& ring = rings_[index];
+    uint32_t inflight = 0;
+    std::chrono::steady_clock::time_point stop_seen{};
+    while (ring.running.load(std::memory_order_relaxed) ||
+           inflight > 0) {
+      if (!ring.running.load(std::memory_order_relaxed)) {
+        // Bounded drain: give outstanding I/O 10 s, then abandon (the
+        // channel dtor will reap; a dead engine must not wedge join).
+        const auto now = std::chrono::steady_clock::now();
+        if (stop_seen == std::chrono::steady_clock::time_point{}) {
+          stop_seen = now;
+        } else if (now - stop_seen > std::chrono::seconds(10)) {
+          return;
+        }
+      }
+      // Sleep on the kick only when idle; otherwise poll channels hot.
+      const int timeout_ms = inflight > 0 ? 0 : 100;
+      pollfd fds[2] = {{ring.kick, POLLIN, 0}, {ring.stop_pipe[0], POLLIN, 0}};
+      if (::poll(fds, 2, timeout_ms) < 0 && errno != EINTR) return;
+      if (fds[1].revents && inflight == 0) return;
+      if (fds[0].revents) {
+        uint64_t n;
+        (void)!read(ring.kick, &n, 8);
+      }
+      if (ring.enabled.load(std::memory_order_acquire)) {
+        bool published = false;
+        while (inflight < kMaxInflight) {
+          uint16_t avail_idx =
+              __atomic_load_n(&ring.avail->idx, __ATOMIC_ACQUIRE);
+          if (ring.last_avail == avail_idx) break;
+          uint16_t head = ring.avail->ring[ring.last_avail % ring.num];
+          ring.last_avail++;
+          if (submit_async(index, head, channels, &inflight)) continue;
+          // Fallback: synchronous command (probe, trim, errors...)
+          uint32_t written = 0;
+          try {
+            written = process_chain(index, head, channels);
+          } catch (const std::exception& e) {
+            fprintf(stderr, "vhost %s: command failed: %s\n",
+                    name_.c_str(), e.what());
+          }
+          publish_used(ring, head, written);
+          published = true;
+        }
+        if (published && ring.call >= 0) {
+          uint64_t one = 1;
+          (void)!write(ring.call, &one, 8);
+        }
+      }
+      for (auto& [bdev, entry] : *channels) {
+        entry.first->poll(entry.second.get());
+      }
+    }
+  }
+
+  void publish_used(Vring& ring, uint16_t head, uint32_t written) {
+    ring.used->ring[ring.used_idx % ring.num] = VringUsedElem{head, written};
+    __atomic_store_n(&ring.used->idx, ++ring.used_idx, __ATOMIC_RELEASE);
+  }
+
+  // Try the async fast path; returns false to use the sync fallback.
+  bool submit_async(unsigned ring_index, uint16_t head,
+                    ChannelCache* channels, uint32_t* inflight) {
+    Vring& ring = rings_[ring_index];
+    std::vector<Iov> out, in;
+    if (!collect_iovs(ring, head, &out, &in)) return false;
+    const bool blk = personality_ == Personality::kBlk;
+    if (!blk && ring_index < 2) return false;  // scsi control/event queue
+
+    BdevPtr bdev;
+    bool is_write = false;
+    uint64_t offset = 0;
+    Iov data{nullptr, 0};
+    auto done = std::make_shared<InflightIo>();
+    done->head = head;
+
+    if (blk) {
+      if (out.empty() || in.empty()) return false;
+      uint8_t header[16];
+      if (gather(header, out, sizeof(header)) < sizeof(header)) return false;
+      uint32_t type;
+      uint64_t sector;
+      memcpy(&type, header, 4);
+      memcpy(&sector, header + 8, 8);
+      if (type != 0 && type != 1) return false;  // only IN/OUT
+      is_write = type == 1;
+      if (is_write && readonly_) return false;
+      // single data extent + 1-byte status, nothing scattered
+      std::vector<Iov> data_io = is_write ? out : in;
+      if (is_write) {
+        size_t skip = sizeof(header);
+        std::vector<Iov> trimmed;
+        for (Iov v : data_io) {
+          if (skip >= v.len) { skip -= v.len; continue; }
+          trimmed.push_back(Iov{v.base + skip, v.len - skip});
+          skip = 0;
+        }
+        data_io.swap(trimmed);
+        if (in.size() != 1 || in[0].len != 1) return false;
+        done->blk_status = in[0].base;
+      } else {
+        if (data_io.size() != 2 || data_io[1].len != 1) return false;
+        done->blk_status = data_io[1].base;
+        data_io.pop_back();
+      }
+      if (data_io.size() != 1) return false;
+      data = data_io[0];
+      bdev = resolver_(0);
+      offset = sector * 512;
+      if (!is_write) done->data_in_bytes = static_cast<uint32_t>(data.len);
+    } else {
+      ScsiCmdReq req{};
+      if (out.empty() || in.empty()) return false;
+      if (gather(reinterpret_cast<uint8_t*>(&req), out, sizeof(req)) <
+          sizeof(req)) {
+        return false;
+      }
+      const uint8_t opc = req.cdb[0];
+      if (opc != 0x28 && opc != 0x2a && opc != 0x88 && opc != 0x8a) {
+        return false;  // only READ/WRITE 10/16
+      }
+      is_write = opc == 0x2a || opc == 0x8a;
+      if (req.lun[0] != 1 ||
+          (((uint16_t(req.lun[2]) << 8) | req.lun[3]) & 0x3fff) != 0) {
+        return false;
+      }
+      bdev = resolver_(req.lun[1]);
+      if (!bdev) return false;
+      uint64_t lba, count;
+      if (opc == 0x28 || opc == 0x2a) {
+        lba = rbe32(req.cdb + 2);
+        count = rbe16(req.cdb + 7);
+      } else {
+        lba = rbe64(req.cdb + 2);
+        count = rbe32(req.cdb + 10);
+      }
+      const uint64_t bytes = count * bdev->block_size();
+      if (bytes == 0 || lba + count > bdev->num_blocks() ||
+          bytes > kMaxIoBytes) {
+        return false;
+      }
+      offset = lba * bdev->block_size();
+      if (is_write) {
+        // out = [req][data]; single contiguous extent only
+        size_t skip = sizeof(ScsiCmdReq);
+        std::vector<Iov> trimmed;
+        for (Iov v : out) {
+          if (skip >= v.len) { skip -= v.len; continue; }
+          trimmed.push_back(Iov{v.base + skip, v.len - skip});
+          skip = 0;
+        }
+        if (trimmed.size() != 1 || trimmed[0].len < bytes) return false;
+        data = Iov{trimmed[0].base, bytes};
+        done->resp_iovs = in;
+      } else {
+        // in = [resp][data]
+        size_t skip = sizeof(ScsiCmdResp);
+        std::vector<Iov> trimmed;
+        for (Iov v : in) {
+          if (skip >= v.len) { skip -= v.len; continue; }
+          trimmed.push_back(Iov{v.base + skip, v.len - skip});
+          skip = 0;
+        }
+        if (trimmed.size() != 1 || trimmed[0].len < bytes) return false;
+        data = Iov{trimmed[0].base, bytes};
+        done->resp_iovs = in;
+        done->data_in_bytes = static_cast<uint32_t>(bytes);
+      }
+    }
+    if (!bdev || data.base == nullptr) return false;
+    if (offset % bdev->block_size() != 0 ||
+        data.len % bdev->block_size() != 0 ||
+        offset + data.len > bdev->size_bytes()) {
+      return false;
+    }
+
+    auto ch = channel_for(channels, bdev);
+    Vring* ring_ptr = &ring;
+    IoRequest io;
+    io.op = is_write ? IoOp::kWrite : IoOp::kRead;
+    io.offset = offset;
+    io.length = data.len;
+    io.buffer = data.base;
+    io.on_complete = [this, ring_ptr, done, inflight, blk](int status) {
+      uint32_t written = done->data_in_bytes;
+      if (blk) {
+        *done->blk_status = status == kIoOk ? 0 : 1;
+        written = (status == kIoOk ? written : 0) + 1;
+      } else {
+        ScsiCmdResp resp{};
+        if (status != kIoOk) resp.response = kRespFailure;
+        size_t resp_written = scatter(
+            done->resp_iovs, reinterpret_cast<const uint8_t*>(&resp),
+            sizeof(resp));
+        written = static_cast<uint32_t>(resp_written) +
+                  (status == kIoOk ? written : 0);
+      }
+      publish_used(*ring_ptr, done->head, written);
+      if (ring_ptr->call >= 0) {
+        uint64_t one = 1;
+        (void)!write(ring_ptr->call, &one, 8);
+      }
+      --*inflight;
+    };
+    bdev->submit(ch.get(), std::move(io));
+    ++*inflight;
+    return true;
   }
 
   void drain_ring(unsigned index, ChannelCache* channels) {

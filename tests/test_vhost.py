@@ -509,3 +509,84 @@ class TestTrim:
                 master.close()
             client.invoke("remove_vhost_controller", {"ctrlr": "vbt"})
             hipstore.delete_bdev(client, "blktrim")
+
+
+class TestPipelinedWorker:
+    """HIPSTORE_VHOST_PIPELINE=1: up to 64 outstanding commands per
+    ring, used entries possibly out of order (id identifies the
+    chain). Same wire behavior as the sync worker otherwise."""
+
+    @pytest.fixture
+    def pipelined(self, tmp_path, monkeypatch):
+        import fixtures
+        monkeypatch.setenv("HIPSTORE_VHOST_PIPELINE", "1")
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        with hipstore.Client(daemon.socket_path) as client:
+            hipstore.construct_malloc_bdev(
+                client, num_blocks=NUM_BLOCKS, block_size=BLOCK, name="pp0")
+            client.invoke("construct_vhost_scsi_controller", {"ctrlr": "vp0"})
+            client.invoke("add_vhost_scsi_lun",
+                          {"ctrlr": "vp0", "scsi_target_num": 0,
+                           "bdev_name": "pp0"})
+            master = VhostUserMaster(
+                os.path.join(os.path.dirname(daemon.socket_path), "vp0"))
+            master.negotiate()
+            yield client, master
+            master.close()
+            daemon.stop()
+
+    def test_roundtrip_and_probe_fallback(self, pipelined):
+        _, master = pipelined
+        # fast path (READ/WRITE) and sync fallback (INQUIRY) interleave
+        data = os.urandom(4 * BLOCK)
+        assert master.write10(0, 5, data, BLOCK).status == 0
+        assert master.inquiry(0).data[8:16] == b"HIPSTORE"
+        result = master.read10(0, 5, 4, BLOCK)
+        assert result.status == 0 and result.data == data
+
+    def test_multiple_outstanding(self, pipelined):
+        """Submit a burst of reads before consuming any completion;
+        every chain completes exactly once with its own id."""
+        _, master = pipelined
+        import struct as structmod
+        from vhost_client import (DATA_IN_OFF, DESC_NEXT, DESC_WRITE,
+                                  GPA_BASE, REQ_OFF, RESP_OFF)
+        pattern = {}
+        for i in range(4):
+            block = bytes([0x30 + i]) * BLOCK
+            assert master.write10(0, 100 + i, block, BLOCK).status == 0
+            pattern[i] = block
+        # 4 outstanding READ(10)s, separate descriptor slots per chain
+        for i in range(4):
+            master.tag += 1
+            req = (bytes([1, 0, 0x40, 0, 0, 0, 0, 0])
+                   + structmod.pack("<Q", master.tag) + bytes(3))
+            cdb = bytearray(10)
+            cdb[0] = 0x28
+            cdb[2:6] = structmod.pack(">I", 100 + i)
+            cdb[7:9] = structmod.pack(">H", 1)
+            req += bytes(cdb).ljust(32, b"\0")
+            req_gpa = REQ_OFF + 0x100 * i
+            resp_gpa = RESP_OFF + 0x100 * i
+            data_gpa = DATA_IN_OFF + BLOCK * i
+            master.mem[req_gpa:req_gpa + 51] = req
+            master.mem[resp_gpa:resp_gpa + 108] = bytes(108)
+            base_slot = i * 3
+            master._write_desc(base_slot, GPA_BASE + req_gpa, 51,
+                               DESC_NEXT, base_slot + 1)
+            master._write_desc(base_slot + 1, GPA_BASE + resp_gpa, 108,
+                               DESC_WRITE | DESC_NEXT, base_slot + 2)
+            master._write_desc(base_slot + 2, GPA_BASE + data_gpa, BLOCK,
+                               DESC_WRITE)
+            master._submit(base_slot)
+        heads = set()
+        for _ in range(4):
+            master._wait_used()
+            slot = 0x3000 + 4 + 8 * ((master.used_idx - 1) % 16)
+            head, _ = structmod.unpack("<II", master.mem[slot:slot + 8])
+            heads.add(head)
+        assert heads == {0, 3, 6, 9}
+        for i in range(4):
+            data = bytes(master.mem[DATA_IN_OFF + BLOCK * i:
+                                    DATA_IN_OFF + BLOCK * (i + 1)])
+            assert data == pattern[i], f"chain {i}"
