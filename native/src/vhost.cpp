@@ -749,7 +749,10 @@ class VhostUserScsiDev {
 
   void ring_worker_pipelined(unsigned index, ChannelCache* channels) {
     Vring& ring = rings_[index];
-    uint32_t inflight = 0;
+    // shared (not stack) so a completion fired from the channel
+    // destructor after an abandoned drain cannot touch a dead frame
+    auto inflight_box = std::make_shared<uint32_t>(0);
+    uint32_t& inflight = *inflight_box;
     std::chrono::steady_clock::time_point stop_seen{};
     while (ring.running.load(std::memory_order_relaxed) ||
            inflight > 0) {
@@ -780,7 +783,7 @@ class VhostUserScsiDev {
           if (ring.last_avail == avail_idx) break;
           uint16_t head = ring.avail->ring[ring.last_avail % ring.num];
           ring.last_avail++;
-          if (submit_async(index, head, channels, &inflight)) continue;
+          if (submit_async(index, head, channels, inflight_box)) continue;
           // Fallback: synchronous command (probe, trim, errors...)
           uint32_t written = 0;
           try {
@@ -810,7 +813,8 @@ class VhostUserScsiDev {
 
   // Try the async fast path; returns false to use the sync fallback.
   bool submit_async(unsigned ring_index, uint16_t head,
-                    ChannelCache* channels, uint32_t* inflight) {
+                    ChannelCache* channels,
+                    std::shared_ptr<uint32_t> inflight) {
     Vring& ring = rings_[ring_index];
     std::vector<Iov> out, in;
     if (!collect_iovs(ring, head, &out, &in)) return false;
@@ -932,6 +936,8 @@ class VhostUserScsiDev {
     io.length = data.len;
     io.buffer = data.base;
     io.on_complete = [this, ring_ptr, done, inflight, blk](int status) {
+      // (inflight shared_ptr keeps the counter alive past an
+      // abandoned drain; see ring_worker_pipelined)
       uint32_t written = done->data_in_bytes;
       if (blk) {
         *done->blk_status = status == kIoOk ? 0 : 1;
