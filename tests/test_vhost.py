@@ -590,3 +590,70 @@ class TestPipelinedWorker:
             data = bytes(master.mem[DATA_IN_OFF + BLOCK * i:
                                     DATA_IN_OFF + BLOCK * (i + 1)])
             assert data == pattern[i], f"chain {i}"
+
+
+class TestVhostProtocolFuzz:
+    """Random bytes at the vhost-user socket must never kill the
+    daemon (binary protocol parser hardening)."""
+
+    def test_random_message_streams(self, hipstored, tmp_path):  # noqa: F811
+        import socket as socketmod
+
+        from hypothesis import HealthCheck, given, settings
+        from hypothesis import strategies as st
+
+        with hipstore.Client(hipstored.socket_path) as client:
+            client.invoke("construct_vhost_scsi_controller",
+                          {"ctrlr": "vhfz"})
+            path = os.path.join(os.path.dirname(hipstored.socket_path),
+                                "vhfz")
+
+            @settings(max_examples=80, deadline=None,
+                      suppress_health_check=[
+                          HealthCheck.function_scoped_fixture])
+            @given(st.binary(min_size=0, max_size=300))
+            def fuzz(blob):
+                sock = socketmod.socket(socketmod.AF_UNIX,
+                                        socketmod.SOCK_STREAM)
+                sock.settimeout(2)
+                try:
+                    sock.connect(path)
+                    sock.sendall(blob)
+                    sock.shutdown(socketmod.SHUT_WR)
+                    while sock.recv(4096):
+                        pass
+                except (TimeoutError, ConnectionError, OSError):
+                    pass
+                finally:
+                    sock.close()
+
+            fuzz()
+            # structured: valid header, random request id and payload
+            @settings(max_examples=80, deadline=None,
+                      suppress_health_check=[
+                          HealthCheck.function_scoped_fixture])
+            @given(st.integers(min_value=0, max_value=40),
+                   st.binary(max_size=268))
+            def fuzz2(request, payload):
+                sock = socketmod.socket(socketmod.AF_UNIX,
+                                        socketmod.SOCK_STREAM)
+                sock.settimeout(2)
+                try:
+                    sock.connect(path)
+                    sock.sendall(struct.pack("<III", request, 0x1,
+                                             len(payload)) + payload)
+                    sock.shutdown(socketmod.SHUT_WR)
+                    while sock.recv(4096):
+                        pass
+                except (TimeoutError, ConnectionError, OSError):
+                    pass
+                finally:
+                    sock.close()
+
+            fuzz2()
+            # controller + daemon still healthy; a fresh master works
+            master = VhostUserMaster(path)
+            master.negotiate()
+            master.close()
+            client.invoke("remove_vhost_controller", {"ctrlr": "vhfz"})
+            assert isinstance(client.invoke("get_rpc_methods"), list)
