@@ -55,3 +55,33 @@ class TestBenchContract:
         assert result["n_gpus"] == 2
         # aggregate over both ranks: 2 ranks x steps x STEP_IOS I/Os
         assert result["value"] > 0
+
+
+class TestVhostHarness:
+    def test_cli_run(self, tmp_path):
+        import fixtures
+        from oim_amd import hipstore
+        from oim_amd.bench import vhost_harness
+
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        try:
+            with hipstore.Client(daemon.socket_path) as client:
+                hipstore.construct_malloc_bdev(
+                    client, num_blocks=16384, block_size=512, name="vhb")
+            import contextlib, io
+            out = io.StringIO()
+            with contextlib.redirect_stdout(out):
+                rc = vhost_harness.main([
+                    "--socket", daemon.socket_path, "--bdev", "vhb",
+                    "--rw", "randrw", "--bs", "4096", "--iodepth", "8",
+                    "--runtime", "0.3", "--perfdash"])
+            assert rc == 0
+            text = out.getvalue()
+            assert "IOPS=" in text
+            assert "[Finished:Performance]" in text
+            # controller cleaned up: socket name free for a rerun
+            with hipstore.Client(daemon.socket_path) as client:
+                assert not [c for c in hipstore.get_vhost_controllers(client)
+                            if c.controller == "vhost-bench"]
+        finally:
+            daemon.stop()
