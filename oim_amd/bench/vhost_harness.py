@@ -36,10 +36,14 @@ from .vhost_client import (
 
 class OutstandingScsiRing:
     """Drives up to `iodepth` 3-descriptor READ/WRITE chains through
-    one vhost ring, reusing a fixed slot layout per in-flight tag."""
+    one vhost ring, reusing a fixed slot layout per in-flight tag.
+    personality "scsi" (virtio-scsi CDBs) or "blk" (virtio-blk
+    outhdr + status byte)."""
 
     def __init__(self, master: VhostUserMaster, block_size: int,
-                 io_size: int, num_blocks: int, iodepth: int, rw: str):
+                 io_size: int, num_blocks: int, iodepth: int, rw: str,
+                 personality: str = "scsi"):
+        self.personality = personality
         self.m = master
         self.block = block_size
         self.io_size = io_size
@@ -61,6 +65,25 @@ class OutstandingScsiRing:
         req_gpa = REQ_OFF + 0x100 * tag
         resp_gpa = RESP_OFF + 0x100 * tag
         data_gpa = DATA_IN_OFF + self.io_size * tag
+        base = tag * 3
+        if self.personality == "blk":
+            sector = lba * self.block // 512
+            header = struct.pack("<IIQ", 1 if write else 0, 0, sector)
+            self.m.mem[req_gpa:req_gpa + 16] = header
+            self.m._write_desc(base, GPA_BASE + req_gpa, 16,
+                               DESC_NEXT, base + 1)
+            if write:
+                self.m._write_desc(base + 1, GPA_BASE + data_gpa,
+                                   self.io_size, DESC_NEXT, base + 2)
+            else:
+                self.m._write_desc(base + 1, GPA_BASE + data_gpa,
+                                   self.io_size, DESC_WRITE | DESC_NEXT,
+                                   base + 2)
+            self.m._write_desc(base + 2, GPA_BASE + resp_gpa, 1,
+                               DESC_WRITE)
+            self.submit_ts[base] = time.perf_counter()
+            self.m._submit(base)
+            return
         cdb = bytearray(10)
         cdb[0] = 0x2A if write else 0x28
         cdb[2:6] = struct.pack(">I", lba)
@@ -69,7 +92,6 @@ class OutstandingScsiRing:
                + struct.pack("<Q", tag + 1) + bytes(3)
                + bytes(cdb).ljust(32, b"\0"))
         self.m.mem[req_gpa:req_gpa + 51] = req
-        base = tag * 3
         if write:
             self.m._write_desc(base, GPA_BASE + req_gpa, 51,
                                DESC_NEXT, base + 1)
@@ -120,29 +142,36 @@ def main(argv=None) -> int:
     parser.add_argument("--iodepth", type=int, default=16)
     parser.add_argument("--runtime", type=float, default=5.0)
     parser.add_argument("--ctrlr", default="vhost-bench")
+    parser.add_argument("--personality", default="scsi",
+                        choices=["scsi", "blk"])
     parser.add_argument("--perfdash", action="store_true")
     args = parser.parse_args(argv)
 
     with hipstore.Client(args.socket) as client:
         bdevs = hipstore.get_bdevs(client, args.bdev)
         bdev = bdevs[0]
-        client.invoke("construct_vhost_scsi_controller",
-                      {"ctrlr": args.ctrlr})
-        client.invoke("add_vhost_scsi_lun",
-                      {"ctrlr": args.ctrlr, "scsi_target_num": 0,
-                       "bdev_name": args.bdev})
+        if args.personality == "blk":
+            client.invoke("construct_vhost_blk_controller",
+                          {"ctrlr": args.ctrlr, "dev_name": args.bdev})
+        else:
+            client.invoke("construct_vhost_scsi_controller",
+                          {"ctrlr": args.ctrlr})
+            client.invoke("add_vhost_scsi_lun",
+                          {"ctrlr": args.ctrlr, "scsi_target_num": 0,
+                           "bdev_name": args.bdev})
         vhost_path = os.path.join(os.path.dirname(args.socket), args.ctrlr)
         qsize = 1
         while qsize < args.iodepth * 3 + 1:
             qsize *= 2
         mem = (1 << 20) + DATA_IN_OFF + args.iodepth * args.bs
-        master = VhostUserMaster(vhost_path, mem_size=max(mem, 8 << 20),
-                                 qsize=max(qsize, 16))
+        master = VhostUserMaster(
+            vhost_path, mem_size=max(mem, 8 << 20), qsize=max(qsize, 16),
+            queue=0 if args.personality == "blk" else 2)
         master.negotiate()
         try:
             ring = OutstandingScsiRing(
                 master, bdev.block_size, args.bs, bdev.num_blocks,
-                args.iodepth, args.rw)
+                args.iodepth, args.rw, personality=args.personality)
             start = time.perf_counter()
             ring.run(args.runtime)
             elapsed = time.perf_counter() - start
@@ -175,7 +204,7 @@ def main(argv=None) -> int:
                 "labels": {"bdev": args.bdev, "rw": args.rw,
                            "bs": str(args.bs),
                            "iodepth": str(args.iodepth),
-                           "path": "vhost-user-scsi"},
+                           "path": f"vhost-user-{args.personality}"},
             }],
             "labels": {"suite": "hipstored-vhost"},
         }

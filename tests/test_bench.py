@@ -83,5 +83,13 @@ class TestVhostHarness:
             with hipstore.Client(daemon.socket_path) as client:
                 assert not [c for c in hipstore.get_vhost_controllers(client)
                             if c.controller == "vhost-bench"]
+            # blk personality path
+            out = io.StringIO()
+            with contextlib.redirect_stdout(out):
+                rc = vhost_harness.main([
+                    "--socket", daemon.socket_path, "--bdev", "vhb",
+                    "--personality", "blk", "--bs", "4096",
+                    "--iodepth", "8", "--runtime", "0.3"])
+            assert rc == 0 and "IOPS=" in out.getvalue()
         finally:
             daemon.stop()
