@@ -6,6 +6,7 @@
 // kInvalidParams like SPDK does; all reference callers tolerate that
 // (reference local.go:53-57, controller.go:76,204,239).
 
+#include <algorithm>
 #include <cinttypes>
 #include <cstdio>
 #include <cstring>
@@ -97,6 +98,28 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
                               bool persistent) {
   auto& manager = BdevManager::instance();
 
+  // Construction-parameter registry for save_config: composite, RBD
+  // and clone bdevs cannot be introspected back into construct params
+  // the way plain malloc bdevs can, so record the creating call.
+  struct Creation {
+    uint64_t seq;
+    std::string method;
+    Json params;
+  };
+  struct Creations {
+    std::mutex mutex;
+    uint64_t next_seq = 0;
+    std::map<std::string, Creation> by_name;
+  };
+  auto creations = std::make_shared<Creations>();
+  auto record_creation = [creations](const std::string& name,
+                                     const std::string& method,
+                                     const Json& params) {
+    std::lock_guard<std::mutex> lock(creations->mutex);
+    creations->by_name[name] = Creation{creations->next_seq++, method, params};
+  };
+
+
   server->register_method("get_bdevs", [&manager](const Json& p) {
     const std::string name = p.get_string("name");
     JsonArray out;
@@ -110,7 +133,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
     return Json(std::move(out));
   });
 
-  server->register_method("delete_bdev", [&manager](const Json& p) {
+  server->register_method("delete_bdev", [&manager, creations](const Json& p) {
     const std::string name = p.get_string("name");
     BdevPtr bdev = manager.find(name);
     if (!bdev) not_found("bdev " + name);
@@ -118,6 +141,10 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
       throw RpcError{kInvalidParams, "bdev " + name + " is claimed"};
     }
     manager.remove(name);
+    {
+      std::lock_guard<std::mutex> lock(creations->mutex);
+      creations->by_name.erase(name);
+    }
     return Json(JsonObject{});
   });
 
@@ -152,7 +179,8 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
       });
 
   server->register_method(
-      "construct_rbd_bdev", [&manager, use_hbm, device](const Json& p) {
+      "construct_rbd_bdev",
+      [&manager, use_hbm, device, record_creation](const Json& p) {
         // Ceph RBD path. Without a reachable cluster (this environment
         // has no network) the image is emulated by a local backing
         // store of `config.emu_size_mb` (default 1 GiB) — the
@@ -189,6 +217,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         // (product_name != "Malloc disk", reference controller.go:205).
         inner->set_product("Ceph Rbd Disk");
         manager.add(inner);
+        record_creation(name, "construct_rbd_bdev", p);
         return Json(name);
       });
 
@@ -437,6 +466,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   };
   auto targets = std::make_shared<Targets>();
 
+
   server->register_method(
       "nvmf_create_target", [&manager, targets](const Json& p) {
         const std::string subnqn =
@@ -540,7 +570,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
 
   server->register_method(
       "construct_striped_malloc_bdev",
-      [&manager, make_children](const Json& p) {
+      [&manager, make_children, record_creation](const Json& p) {
         const std::string name = p.get_string("name");
         const int64_t num_blocks = p.get_int("num_blocks");   // per child
         const int64_t block_size = p.get_int("block_size", 512);
@@ -561,12 +591,13 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         } catch (const std::exception& e) {
           throw RpcError{kInvalidParams, e.what()};
         }
+        record_creation(name, "construct_striped_malloc_bdev", p);
         return Json(name);
       });
 
   server->register_method(
       "construct_replicated_malloc_bdev",
-      [&manager, make_children](const Json& p) {
+      [&manager, make_children, record_creation](const Json& p) {
         const std::string name = p.get_string("name");
         const int64_t num_blocks = p.get_int("num_blocks");
         const int64_t block_size = p.get_int("block_size", 512);
@@ -585,6 +616,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         } catch (const std::exception& e) {
           throw RpcError{kInvalidParams, e.what()};
         }
+        record_creation(name, "construct_replicated_malloc_bdev", p);
         return Json(name);
       });
 
@@ -814,10 +846,25 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   // this is the SPDK-compatible control-state checkpoint. Composite
   // and RBD bdevs are not yet emitted (construction params are not
   // retained for them).
-  server->register_method("save_config", [&manager, targets](const Json&) {
+  server->register_method("save_config", [&manager, targets,
+                                          creations](const Json&) {
     JsonArray bdev_cfg;
+    std::vector<std::pair<uint64_t, Json>> recorded;
+    {
+      std::lock_guard<std::mutex> lock(creations->mutex);
+      for (const auto& [name, creation] : creations->by_name) {
+        JsonObject entry;
+        entry["method"] = Json(creation.method);
+        entry["params"] = creation.params;
+        recorded.emplace_back(creation.seq, Json(std::move(entry)));
+      }
+    }
     for (const auto& bdev : manager.list()) {
       if (bdev->product_name() != "Malloc disk") continue;
+      {
+        std::lock_guard<std::mutex> lock(creations->mutex);
+        if (creations->by_name.count(bdev->name())) continue;
+      }
       JsonObject params;
       params["name"] = Json(bdev->name());
       params["num_blocks"] = Json(static_cast<int64_t>(bdev->num_blocks()));
@@ -826,6 +873,13 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
       entry["method"] = Json(std::string("construct_malloc_bdev"));
       entry["params"] = Json(std::move(params));
       bdev_cfg.push_back(Json(std::move(entry)));
+    }
+    // Recorded creations (composites, RBD) in creation order so
+    // dependencies replay before dependents.
+    std::sort(recorded.begin(), recorded.end(),
+              [](const auto& a, const auto& b) { return a.first < b.first; });
+    for (auto& [seq, entry] : recorded) {
+      bdev_cfg.push_back(std::move(entry));
     }
     JsonArray vhost_cfg;
     {

@@ -279,3 +279,51 @@ class TestConfigSnapshot:
             assert any(c.controller == "cfgblk" for c in listing)
             client.invoke("remove_vhost_controller", {"ctrlr": "cfgblk"})
             hipstore.delete_bdev(client, "cblk")
+
+    def test_composites_in_config(self, tmp_path):
+        """Striped/replicated/RBD bdevs replay through their recorded
+        construction params (creation order preserved)."""
+        import json as jsonmod
+        import os as osmod
+        import subprocess
+        import time
+
+        import fixtures
+
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        with hipstore.Client(daemon.socket_path) as client:
+            client.invoke("construct_striped_malloc_bdev", {
+                "name": "cfgstripe", "num_blocks": 1024, "block_size": 512,
+                "stripe_size_kb": 64, "count": 2})
+            client.invoke("construct_replicated_malloc_bdev", {
+                "name": "cfgrepl", "num_blocks": 512, "block_size": 512,
+                "count": 2})
+            config = client.invoke("save_config")
+        daemon.stop()
+        bdev_cfg = [e for sub in config["subsystems"]
+                    if sub["subsystem"] == "bdev" for e in sub["config"]]
+        methods = [e["method"] for e in bdev_cfg]
+        assert "construct_striped_malloc_bdev" in methods
+        assert "construct_replicated_malloc_bdev" in methods
+
+        config_path = tmp_path / "cfg2.json"
+        config_path.write_text(jsonmod.dumps(config))
+        sock2 = str(tmp_path / "h3.sock")
+        proc = subprocess.Popen(
+            [fixtures.DEFAULT_BINARY, "-S", sock2, "-C",
+             "-c", str(config_path)], stderr=subprocess.PIPE)
+        deadline = time.time() + 30
+        while not osmod.path.exists(sock2):
+            assert proc.poll() is None, proc.stderr.read().decode()
+            assert time.time() < deadline
+            time.sleep(0.05)
+        try:
+            with hipstore.Client(sock2) as client:
+                stripe = hipstore.get_bdevs(client, "cfgstripe")[0]
+                assert stripe.product_name == "Striped Malloc disk"
+                assert stripe.num_blocks == 2 * 1024
+                repl = hipstore.get_bdevs(client, "cfgrepl")[0]
+                assert repl.product_name == "Replicated Malloc disk"
+        finally:
+            proc.terminate()
+            proc.wait(timeout=10)
