@@ -179,4 +179,9 @@ class BdevManager {
 BdevPtr create_malloc_bdev(const std::string& name, uint64_t block_size,
                            uint64_t num_blocks);
 
+// File-backed bdev (SPDK aio bdev): geometry from the file's current
+// size; data persists across daemon restarts.
+BdevPtr create_file_bdev(const std::string& name, const std::string& path,
+                         uint64_t block_size);
+
 }  // namespace hipstore

@@ -1,6 +1,9 @@
 #include "hipstore/bdev.h"
 
+#include <fcntl.h>
 #include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
 
 #include <cstring>
 #include <deque>
@@ -145,11 +148,147 @@ class MallocBdev : public Bdev {
   uint8_t* data_ = nullptr;
 };
 
+// File-backed bdev (SPDK's aio bdev): pread/pwrite against a regular
+// file or block device. The only bdev type whose DATA survives daemon
+// restarts — malloc bdevs are RAM/HBM by contract.
+class FileBdev : public Bdev {
+ public:
+  FileBdev(const std::string& name, const std::string& path,
+           uint64_t block_size, uint64_t num_blocks, int fd)
+      : Bdev(name, "AIO disk", block_size, num_blocks),
+        path_(path),
+        fd_(fd) {}
+
+  ~FileBdev() override { ::close(fd_); }
+
+  const std::string& path() const { return path_; }
+
+  std::shared_ptr<IoChannel> get_channel() override {
+    return std::make_shared<MallocChannel>();
+  }
+
+  void submit(IoChannel* ch, IoRequest req) override {
+    auto* channel = static_cast<MallocChannel*>(ch);
+    int status = kIoOk;
+    if (!check_bounds(req)) {
+      status = kIoInvalid;
+    } else {
+      account(req);
+      switch (req.op) {
+        case IoOp::kRead:
+          if (full_pread(req.buffer, req.length, req.offset) != 0) {
+            status = kIoFailed;
+          }
+          break;
+        case IoOp::kWrite:
+          if (full_pwrite(req.buffer, req.length, req.offset) != 0) {
+            status = kIoFailed;
+          }
+          break;
+        case IoOp::kFill: {
+          std::vector<uint8_t> zeros(
+              std::min<uint64_t>(req.length, 1 << 20), req.fill);
+          uint64_t done = 0;
+          while (done < req.length && status == kIoOk) {
+            uint64_t n = std::min<uint64_t>(zeros.size(), req.length - done);
+            if (full_pwrite(zeros.data(), n, req.offset + done) != 0) {
+              status = kIoFailed;
+            }
+            done += n;
+          }
+          break;
+        }
+        case IoOp::kFlush:
+          if (fdatasync(fd_) != 0) status = kIoFailed;
+          break;
+      }
+    }
+    channel->done.emplace_back(std::move(req.on_complete), status);
+  }
+
+  int poll(IoChannel* ch) override {
+    auto* channel = static_cast<MallocChannel*>(ch);
+    size_t n = channel->done.size();
+    for (size_t i = 0; i < n; ++i) {
+      auto [cb, status] = std::move(channel->done.front());
+      channel->done.pop_front();
+      if (cb) cb(status);
+    }
+    return static_cast<int>(n);
+  }
+
+  int resize(uint64_t new_num_blocks) override {
+    const uint64_t new_bytes = new_num_blocks * block_size();
+    if (ftruncate(fd_, static_cast<off_t>(new_bytes)) != 0) return kIoFailed;
+    set_num_blocks(new_num_blocks);
+    return kIoOk;
+  }
+
+ private:
+  int full_pread(void* buf, uint64_t len, uint64_t off) {
+    uint8_t* p = static_cast<uint8_t*>(buf);
+    while (len > 0) {
+      ssize_t n = pread(fd_, p, len, static_cast<off_t>(off));
+      if (n < 0) return -1;
+      if (n == 0) {  // short file (sparse tail): reads as zeros
+        memset(p, 0, len);
+        return 0;
+      }
+      p += n;
+      off += n;
+      len -= n;
+    }
+    return 0;
+  }
+
+  int full_pwrite(const void* buf, uint64_t len, uint64_t off) {
+    const uint8_t* p = static_cast<const uint8_t*>(buf);
+    while (len > 0) {
+      ssize_t n = pwrite(fd_, p, len, static_cast<off_t>(off));
+      if (n <= 0) return -1;
+      p += n;
+      off += n;
+      len -= n;
+    }
+    return 0;
+  }
+
+  const std::string path_;
+  const int fd_;
+};
+
 }  // namespace
 
 BdevPtr create_malloc_bdev(const std::string& name, uint64_t block_size,
                            uint64_t num_blocks) {
   return std::make_shared<MallocBdev>(name, block_size, num_blocks);
+}
+
+BdevPtr create_file_bdev(const std::string& name, const std::string& path,
+                         uint64_t block_size) {
+  int fd = ::open(path.c_str(), O_RDWR | O_CREAT | O_CLOEXEC, 0600);
+  if (fd < 0) {
+    throw std::runtime_error("aio bdev: cannot open " + path);
+  }
+  struct stat st {};
+  if (fstat(fd, &st) != 0) {
+    ::close(fd);
+    throw std::runtime_error("aio bdev: cannot stat " + path);
+  }
+  uint64_t bytes = static_cast<uint64_t>(st.st_size);
+  if (S_ISBLK(st.st_mode)) {
+    ::close(fd);
+    throw std::runtime_error(
+        "aio bdev: raw block devices are not supported here; use a file");
+  }
+  if (bytes == 0 || bytes % block_size != 0) {
+    ::close(fd);
+    throw std::runtime_error(
+        "aio bdev: file size must be a non-zero multiple of block_size "
+        "(create/truncate the file first)");
+  }
+  return std::make_shared<FileBdev>(name, path, block_size,
+                                    bytes / block_size, fd);
 }
 
 }  // namespace hipstore

@@ -179,6 +179,32 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
       });
 
   server->register_method(
+      "construct_aio_bdev", [&manager, record_creation](const Json& p) {
+        // SPDK aio bdev: file-backed, geometry from file size; the
+        // only bdev whose data survives restarts.
+        const std::string name = p.get_string("name");
+        const std::string filename = p.get_string("filename");
+        const int64_t block_size = p.get_int("block_size", 512);
+        if (name.empty() || filename.empty() || block_size <= 0 ||
+            block_size % 512 != 0) {
+          throw RpcError{kInvalidParams,
+                         "name/filename/block_size required"};
+        }
+        if (manager.find(name)) {
+          throw RpcError{kInvalidParams, "bdev " + name + " already exists"};
+        }
+        BdevPtr bdev;
+        try {
+          bdev = create_file_bdev(name, filename, block_size);
+        } catch (const std::exception& e) {
+          throw RpcError{kInvalidParams, e.what()};
+        }
+        manager.add(bdev);
+        record_creation(name, "construct_aio_bdev", p);
+        return Json(name);
+      });
+
+  server->register_method(
       "construct_rbd_bdev",
       [&manager, use_hbm, device, record_creation](const Json& p) {
         // Ceph RBD path. Without a reachable cluster (this environment
