@@ -14,7 +14,9 @@ class VolumeExistsError(ValueError):
 
 
 class OIMBackend:
-    def create_volume(self, name: str, size: int) -> Tuple[str, Dict[str, str]]:
+    def create_volume(self, name: str, size: int,
+                      parameters: Dict[str, str] = None
+                      ) -> Tuple[str, Dict[str, str]]:
         """Returns (volume_id, volume_context)."""
         raise NotImplementedError
 

@@ -110,7 +110,7 @@ class ControllerServer(CSIControllerServicer):
         with self._name_mutex.get(request.name):
             try:
                 volume_id, volume_context = self.backend.create_volume(
-                    request.name, size)
+                    request.name, size, dict(request.parameters))
             except VolumeExistsError as exc:
                 # Same name, incompatible parameters (CSI spec: 6)
                 context.abort(grpc.StatusCode.ALREADY_EXISTS, str(exc))

@@ -23,9 +23,13 @@ def round_to_blocks(size: int) -> int:
 
 
 class LocalBackend(OIMBackend):
-    def __init__(self, hipstored_socket: str, nbd_prefix: str = "/dev/nbd"):
+    def __init__(self, hipstored_socket: str, nbd_prefix: str = "/dev/nbd",
+                 aio_dir: str = "/var/lib/oim-aio"):
         self.socket = hipstored_socket
         self.nbd_prefix = nbd_prefix
+        # Backing directory for `backing: aio` StorageClass volumes
+        # (file-backed, data survives daemon restarts).
+        self.aio_dir = aio_dir
         # snapshot id -> (source volume, creation time); hipstored owns
         # the data, this is presentation metadata only
         self._snap_meta = {}
@@ -33,10 +37,15 @@ class LocalBackend(OIMBackend):
     def _client(self) -> hipstore.Client:
         return hipstore.Client(self.socket)
 
-    def create_volume(self, name: str, size: int) -> Tuple[str, Dict[str, str]]:
+    def create_volume(self, name: str, size: int,
+                      parameters: Dict[str, str] = None
+                      ) -> Tuple[str, Dict[str, str]]:
         if size > MAX_VOLUME_SIZE:
             raise ValueError(f"volume too large: {size}")
         size = round_to_blocks(size)
+        backing = (parameters or {}).get("backing", "malloc")
+        if backing not in ("malloc", "aio"):
+            raise ValueError(f"unknown backing {backing!r}")
         with self._client() as client:
             try:
                 bdevs = hipstore.get_bdevs(client, name)
@@ -49,6 +58,17 @@ class LocalBackend(OIMBackend):
                     raise VolumeExistsError(
                         f"volume {name} exists with different size")
                 return name, {}
+            if backing == "aio":
+                # File-backed (persists across daemon restarts); the
+                # file IS the volume, sized here, removed on delete.
+                os.makedirs(self.aio_dir, exist_ok=True)
+                path = os.path.join(self.aio_dir, f"{name}.img")
+                with open(path, "ab") as f:
+                    f.truncate(size)
+                client.invoke("construct_aio_bdev",
+                              {"name": name, "filename": path,
+                               "block_size": 512})
+                return name, {"backing": "aio"}
             hipstore.construct_malloc_bdev(
                 client, num_blocks=size // 512, block_size=512, name=name)
         return name, {}
@@ -56,10 +76,20 @@ class LocalBackend(OIMBackend):
     def delete_volume(self, volume_id: str) -> None:
         with self._client() as client:
             try:
+                bdevs = hipstore.get_bdevs(client, volume_id)
+                is_aio = bdevs and bdevs[0].product_name == "AIO disk"
                 hipstore.delete_bdev(client, volume_id)
             except hipstore.RpcError as err:
                 if not err.is_not_found():
                     raise
+                return
+        if is_aio:
+            # CSI DeleteVolume destroys the data: remove the backing file
+            path = os.path.join(self.aio_dir, f"{volume_id}.img")
+            try:
+                os.unlink(path)
+            except FileNotFoundError:
+                pass
 
     def check_volume_exists(self, volume_id: str) -> bool:
         with self._client() as client:

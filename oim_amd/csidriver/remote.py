@@ -95,7 +95,13 @@ class RemoteBackend(OIMBackend):
 
     # --- volume lifecycle ---------------------------------------------------
 
-    def create_volume(self, name: str, size: int) -> Tuple[str, Dict[str, str]]:
+    def create_volume(self, name: str, size: int,
+                      parameters: Dict[str, str] = None
+                      ) -> Tuple[str, Dict[str, str]]:
+        if (parameters or {}).get("backing", "malloc") != "malloc":
+            raise ValueError(
+                "remote mode provisions malloc bdevs only (oim.v0 "
+                "ProvisionMallocBDev)")
         if size > MAX_VOLUME_SIZE:
             raise ValueError(f"volume too large: {size}")
         size = round_to_blocks(size)
