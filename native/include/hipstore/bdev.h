@@ -87,6 +87,9 @@ class Bdev {
   // compute (CRC32C, verify, striping); others return nullptr/-1.
   virtual void* device_base() { return nullptr; }
   virtual int gpu_device() const { return -1; }
+  // File-backed bdevs report their backing path (get_bdevs
+  // driver_specific, SPDK aio shape); others return "".
+  virtual std::string backing_path() const { return {}; }
 
   virtual std::shared_ptr<IoChannel> get_channel() = 0;
   // Grow (or shrink) the device to new_num_blocks. Returns kIoOk, or

@@ -162,6 +162,7 @@ class FileBdev : public Bdev {
   ~FileBdev() override { ::close(fd_); }
 
   const std::string& path() const { return path_; }
+  std::string backing_path() const override { return path_; }
 
   std::shared_ptr<IoChannel> get_channel() override {
     return std::make_shared<MallocChannel>();

@@ -84,6 +84,12 @@ Json bdev_to_json(const BdevPtr& bdev) {
     JsonObject ds;
     ds["hbm"] = Json(std::move(hbm));
     o["driver_specific"] = Json(std::move(ds));
+  } else if (!bdev->backing_path().empty()) {
+    JsonObject aio;
+    aio["filename"] = Json(bdev->backing_path());
+    JsonObject ds;
+    ds["aio"] = Json(std::move(aio));
+    o["driver_specific"] = Json(std::move(ds));
   }
   return Json(std::move(o));
 }

@@ -756,3 +756,22 @@ class TestAioBackedVolumes:
         finally:
             channel.close()
             driver.stop()
+
+    def test_aio_volume_expand(self, hipstored, tmp_path):  # noqa: F811
+        """Expansion works for file-backed volumes too (ftruncate)."""
+        backend = LocalBackend(hipstored.socket_path,
+                               aio_dir=str(tmp_path / "aio"))
+        _, ctx = backend.create_volume("grow-aio", 1 << 20,
+                                       {"backing": "aio"})
+        assert ctx["backing"] == "aio"
+        new_size = backend.expand_volume("grow-aio", 4 << 20)
+        assert new_size == 4 << 20
+        backing = tmp_path / "aio" / "grow-aio.img"
+        assert backing.stat().st_size == 4 << 20
+        from oim_amd import hipstore
+        with hipstore.Client(hipstored.socket_path) as client:
+            bdev = hipstore.get_bdevs(client, "grow-aio")[0]
+            assert bdev.size_bytes == 4 << 20
+            assert bdev.driver_specific["aio"]["filename"] == str(backing)
+        backend.delete_volume("grow-aio")
+        assert not backing.exists()
