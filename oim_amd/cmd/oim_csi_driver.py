@@ -14,6 +14,8 @@ def main(argv=None) -> int:
     parser.add_argument("--drivername", default="oim-malloc")
     parser.add_argument("--nodeid", required=True)
     # local mode
+    parser.add_argument("--aio-dir", default="/var/lib/oim-aio",
+                        help="backing dir for `backing: aio` volumes")
     parser.add_argument("--hipstored-socket", default="",
                         help="local mode: hipstored JSON-RPC socket")
     # remote mode
@@ -37,7 +39,8 @@ def main(argv=None) -> int:
     if local:
         if args.emulate:
             parser.error("--emulate requires remote mode")
-        backend = LocalBackend(args.hipstored_socket)
+        backend = LocalBackend(args.hipstored_socket,
+                               aio_dir=args.aio_dir)
     else:
         if not args.controller_id:
             parser.error("remote mode requires --controller-id")
