@@ -93,3 +93,28 @@ class TestVhostHarness:
             assert rc == 0 and "IOPS=" in out.getvalue()
         finally:
             daemon.stop()
+
+    def test_pipelined_daemon_qd32(self, tmp_path, monkeypatch):
+        """Deep-queue run against the pipelined ring worker."""
+        import contextlib
+        import io
+
+        import fixtures
+        from oim_amd import hipstore
+        from oim_amd.bench import vhost_harness
+
+        monkeypatch.setenv("HIPSTORE_VHOST_PIPELINE", "1")
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        try:
+            with hipstore.Client(daemon.socket_path) as client:
+                hipstore.construct_malloc_bdev(
+                    client, num_blocks=32768, block_size=512, name="vhp")
+            out = io.StringIO()
+            with contextlib.redirect_stdout(out):
+                rc = vhost_harness.main([
+                    "--socket", daemon.socket_path, "--bdev", "vhp",
+                    "--rw", "randrw", "--bs", "4096", "--iodepth", "32",
+                    "--runtime", "0.5"])
+            assert rc == 0 and "IOPS=" in out.getvalue()
+        finally:
+            daemon.stop()
