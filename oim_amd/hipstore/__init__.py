@@ -16,6 +16,7 @@ from .api import (
     get_bdevs,
     delete_bdev,
     construct_malloc_bdev,
+    construct_aio_bdev,
     construct_rbd_bdev,
     start_nbd_disk,
     get_nbd_disks,
@@ -43,6 +44,7 @@ __all__ = [
     "get_bdevs",
     "delete_bdev",
     "construct_malloc_bdev",
+    "construct_aio_bdev",
     "construct_rbd_bdev",
     "start_nbd_disk",
     "get_nbd_disks",

@@ -87,6 +87,13 @@ def construct_malloc_bdev(
     return client.invoke("construct_malloc_bdev", params)
 
 
+def construct_aio_bdev(client: Client, name: str, filename: str,
+                       block_size: int = 512) -> str:
+    """File-backed bdev (SPDK aio): data survives daemon restarts."""
+    return client.invoke("construct_aio_bdev", {
+        "name": name, "filename": filename, "block_size": block_size})
+
+
 def construct_rbd_bdev(
     client: Client,
     pool_name: str,
