@@ -350,3 +350,48 @@ class TestProxyDeadlines:
         finally:
             reg_server.stop()
             ctrl_server.stop()
+
+
+class TestProxyGarbagePayload:
+    def test_malformed_protobuf_through_proxy(self, tmp_path):
+        """The proxy splices raw bytes; a garbage payload must surface
+        as a gRPC error from the backend's deserializer, and both
+        proxy and controller stay healthy."""
+        import fixtures
+        from oim_amd.common.server import grpc_target
+        from oim_amd.controller import Controller, ControllerServer
+
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        registry = Registry(db=MemRegistryDB())
+        reg_server = RegistryServer(f"unix://{tmp_path}/r.sock", registry)
+        reg_server.start()
+        controller = Controller(controller_id="fz",
+                                hipstored_socket=daemon.socket_path)
+        ctrl_server = ControllerServer(f"unix://{tmp_path}/c.sock", controller)
+        ctrl_server.start()
+        registry.db.store(["fz", "address"], f"unix://{tmp_path}/c.sock")
+        try:
+            with grpc.insecure_channel(grpc_target(reg_server.addr())) as ch:
+                raw = ch.unary_unary(
+                    "/oim.v0.Controller/ProvisionMallocBDev",
+                    request_serializer=lambda b: b,
+                    response_deserializer=lambda b: b)
+                md = ((spec.CONTROLLER_ID_KEY, "fz"),)
+                for blob in (b"\xff" * 64, b"\x0a", os.urandom(120)):
+                    try:
+                        raw(blob, metadata=md, timeout=10)
+                    except grpc.RpcError:
+                        pass  # any clean gRPC error is acceptable
+                # healthy afterwards
+                stub = spec.ControllerStub(ch)
+                stub.ProvisionMallocBDev(
+                    spec.ProvisionMallocBDevRequest(bdev_name="ok",
+                                                    size=1 << 20),
+                    metadata=md, timeout=30)
+                stub.ProvisionMallocBDev(
+                    spec.ProvisionMallocBDevRequest(bdev_name="ok", size=0),
+                    metadata=md, timeout=30)
+        finally:
+            ctrl_server.stop()
+            reg_server.stop()
+            daemon.stop()
