@@ -1342,12 +1342,27 @@ class VhostUserScsiDev {
           switch (cdb[2]) {
             case 0x00:  // supported VPD pages
               buf[1] = 0x00;
-              buf[3] = 3;
+              buf[3] = 4;
               buf[4] = 0x00;
               buf[5] = 0x80;
-              buf[6] = 0xB2;
-              len = 7;
+              buf[6] = 0x83;
+              buf[7] = 0xB2;
+              len = 8;
               break;
+            case 0x83: {  // device identification: T10 vendor-id
+              // designator from the bdev uuid (/dev/disk/by-id source)
+              const std::string& uuid = bdev->uuid();
+              const size_t id_len = std::min<size_t>(uuid.size(), 36);
+              buf[1] = 0x83;
+              buf[4] = 0x02;  // codeset: ASCII
+              buf[5] = 0x01;  // assoc LUN, designator type: T10 vendor id
+              buf[7] = static_cast<uint8_t>(8 + id_len);
+              memcpy(buf + 8, "HIPSTORE", 8);
+              memcpy(buf + 16, uuid.data(), id_len);
+              len = 16 + id_len;
+              buf[3] = static_cast<uint8_t>(len - 4);
+              break;
+            }
             case 0xB2:  // logical block provisioning: LBPU (UNMAP)
               buf[1] = 0xB2;
               buf[3] = 4;

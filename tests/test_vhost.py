@@ -82,9 +82,22 @@ class TestScsiProbe:
         bdev = hipstore.get_bdevs(client, "vhb0")[0]
         assert bdev.uuid.startswith(uuid[:8])
 
+    def test_inquiry_vpd_device_id(self, vhost_target):
+        """VPD 0x83: T10 designator carrying the bdev uuid (the
+        /dev/disk/by-id source)."""
+        client, master = vhost_target
+        result = master.inquiry(0, evpd_page=0x83)
+        assert result.status == 0
+        assert result.data[1] == 0x83
+        designator_len = result.data[7]
+        ident = result.data[8:8 + designator_len]
+        assert ident[:8] == b"HIPSTORE"
+        bdev = hipstore.get_bdevs(client, "vhb0")[0]
+        assert ident[8:].decode() == bdev.uuid[:designator_len - 8]
+
     def test_inquiry_bad_vpd_page(self, vhost_target):
         _, master = vhost_target
-        result = master.inquiry(0, evpd_page=0x83)
+        result = master.inquiry(0, evpd_page=0x77)
         assert result.status == 2  # CHECK CONDITION
         assert result.sense_key == 5 and result.asc == 0x24
 
