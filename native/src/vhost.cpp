@@ -1059,6 +1059,7 @@ class VhostUserScsiDev {
     uint64_t sector;
     memcpy(&type, header, 4);
     memcpy(&sector, header + 8, 8);
+    size_t id_bytes = 0;
 
     // status byte = last writable byte; data-in = everything before it
     std::vector<Iov> data_in = in;
@@ -1090,8 +1091,9 @@ class VhostUserScsiDev {
     } else if (type == kTypeGetId) {
       char id[20] = {0};
       strncpy(id, bdev->name().c_str(), sizeof(id) - 1);
-      scatter(data_in, reinterpret_cast<uint8_t*>(id),
-              std::min<size_t>(sizeof(id), iov_total(data_in)));
+      id_bytes = scatter(
+          data_in, reinterpret_cast<uint8_t*>(id),
+          std::min<size_t>(sizeof(id), iov_total(data_in)));
       status = kStatusOk;
     } else if (type == kTypeDiscard || type == kTypeWriteZeroes) {
       // data-out = N x 16-byte segments {sector u64; num u32; flags u32}
@@ -1167,7 +1169,8 @@ class VhostUserScsiDev {
     const uint32_t data_written =
         (type == kTypeIn && status == kStatusOk)
             ? static_cast<uint32_t>(iov_total(data_in))
-            : (type == kTypeGetId ? 20u : 0u);
+            : (type == kTypeGetId ? static_cast<uint32_t>(id_bytes)
+                                  : 0u);
     return data_written + 1;  // +1 for the status byte
   }
 
