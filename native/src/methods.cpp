@@ -472,6 +472,19 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
 
   // --- native extras -------------------------------------------------------
 
+  server->register_method("spdk_get_version", [](const Json&) {
+    // SPDK wire parity: tools probe this to fingerprint the daemon.
+    JsonObject version;
+    version["version"] = Json(std::string("oim-amd 0.1.0 (hipstored)"));
+    JsonObject fields;
+    fields["major"] = Json(int64_t{0});
+    fields["minor"] = Json(int64_t{1});
+    fields["patch"] = Json(int64_t{0});
+    fields["suffix"] = Json(std::string("-hipstored"));
+    version["fields"] = Json(std::move(fields));
+    return Json(std::move(version));
+  });
+
   server->register_method("get_rpc_methods", [server](const Json&) {
     // Live registry, not a hand-kept list (SPDK semantics: the actual
     // dispatch table). Includes methods registered after this one.

@@ -183,6 +183,14 @@ class TestClone:
             pass
 
 
+class TestVersion:
+    def test_spdk_get_version(self, hipstored):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            version = client.invoke("spdk_get_version")
+            assert "oim-amd" in version["version"]
+            assert version["fields"]["major"] == 0
+
+
 class TestConfigSnapshot:
     """save_config / load_config / `hipstored -c` (SPDK's config
     snapshot shape): the control-plane topology survives a daemon
