@@ -121,7 +121,8 @@ class VhostUserMaster:
         self._send(request, payload)
         return self._recv_reply()
 
-    def negotiate(self, features_extra: int = 0):
+    def handshake(self, features_extra: int = 0):
+        """Feature negotiation + guest memory; no rings yet."""
         features, = struct.unpack("<Q", self.query(GET_FEATURES))
         assert features & FEAT_VERSION_1, hex(features)
         self._send(SET_FEATURES, struct.pack(
@@ -132,15 +133,32 @@ class VhostUserMaster:
         region = struct.pack("<II", 1, 0) + struct.pack(
             "<QQQQ", GPA_BASE, self.mem_size, UADDR_BASE, 0)
         self._send(SET_MEM_TABLE, region, fds=[self.memfd])
-        q = self.queue
-        self._send(SET_VRING_NUM, struct.pack("<II", q, self.qsize))
-        self._send(SET_VRING_BASE, struct.pack("<II", q, 0))
+
+    def setup_ring(self, queue: int, desc_off: int, avail_off: int,
+                   used_off: int, qsize: int, kick: int, call: int):
+        """Configure + enable one vring at the given guest offsets."""
+        self._send(SET_VRING_NUM, struct.pack("<II", queue, qsize))
+        self._send(SET_VRING_BASE, struct.pack("<II", queue, 0))
         self._send(SET_VRING_ADDR, struct.pack(
-            "<IIQQQQ", q, 0, UADDR_BASE + DESC_OFF, UADDR_BASE + USED_OFF,
-            UADDR_BASE + AVAIL_OFF, 0))
-        self._send(SET_VRING_CALL, struct.pack("<Q", q), fds=[self.call])
-        self._send(SET_VRING_KICK, struct.pack("<Q", q), fds=[self.kick])
-        self._send(SET_VRING_ENABLE, struct.pack("<II", q, 1))
+            "<IIQQQQ", queue, 0, UADDR_BASE + desc_off,
+            UADDR_BASE + used_off, UADDR_BASE + avail_off, 0))
+        self._send(SET_VRING_CALL, struct.pack("<Q", queue), fds=[call])
+        self._send(SET_VRING_KICK, struct.pack("<Q", queue), fds=[kick])
+        self._send(SET_VRING_ENABLE, struct.pack("<II", queue, 1))
+
+    def add_ring(self, queue: int, region_base: int,
+                 qsize: int) -> "VhostRing":
+        """Independent ring with its own eventfds and memory region
+        (multi-queue benchmarking)."""
+        ring = VhostRing(self, queue, region_base, qsize)
+        self.setup_ring(queue, ring.desc_off, ring.avail_off,
+                        ring.used_off, qsize, ring.kick, ring.call)
+        return ring
+
+    def negotiate(self, features_extra: int = 0):
+        self.handshake(features_extra)
+        self.setup_ring(self.queue, DESC_OFF, AVAIL_OFF, USED_OFF,
+                        self.qsize, self.kick, self.call)
 
     def stop_ring(self) -> int:
         """GET_VRING_BASE stops the ring; returns last_avail."""
@@ -348,3 +366,70 @@ def read_blk_config(master) -> dict:
     capacity, = struct.unpack("<Q", region[0:8])
     blk_size, = struct.unpack("<I", region[20:24])
     return {"capacity_sectors": capacity, "blk_size": blk_size}
+
+
+class VhostRing:
+    """One vring with private eventfds inside a master's guest memory
+    (multi-queue benchmarking; the legacy single-ring helpers on
+    VhostUserMaster stay as-is for the conformance tests).
+
+    Layout relative to region_base: descriptor table, avail ring, then
+    the used ring page-aligned; everything past header_bytes is the
+    caller's for request/data buffers."""
+
+    def __init__(self, master: VhostUserMaster, queue: int,
+                 region_base: int, qsize: int):
+        self.m = master
+        self.queue = queue
+        self.base = region_base
+        self.qsize = qsize
+        self.desc_off = region_base
+        self.avail_off = region_base + 16 * qsize
+        used_unaligned = self.avail_off + 4 + 2 * qsize
+        self.used_off = (used_unaligned + 0xFFF) & ~0xFFF
+        self.header_bytes = self.used_off + 4 + 8 * qsize - region_base
+        self.kick = os.eventfd(0)
+        self.call = os.eventfd(0, os.EFD_NONBLOCK)
+        self.avail_idx = 0
+        self.used_idx = 0
+        self.last_used_head = 0
+
+    def close(self):
+        os.close(self.kick)
+        os.close(self.call)
+
+    def write_desc(self, slot: int, gpa: int, length: int, flags: int,
+                   nxt: int = 0):
+        off = self.desc_off + 16 * slot
+        self.m.mem[off:off + 16] = struct.pack("<QIHH", gpa, length,
+                                               flags, nxt)
+
+    def submit(self, head: int):
+        slot = self.avail_off + 4 + 2 * (self.avail_idx % self.qsize)
+        self.m.mem[slot:slot + 2] = struct.pack("<H", head)
+        self.avail_idx += 1
+        self.m.mem[self.avail_off + 2:self.avail_off + 4] = struct.pack(
+            "<H", self.avail_idx & 0xFFFF)
+        os.eventfd_write(self.kick, 1)
+
+    def wait_used(self, timeout: float = 10.0) -> int:
+        import time
+        deadline = time.time() + timeout
+        while True:
+            idx, = struct.unpack(
+                "<H", self.m.mem[self.used_off + 2:self.used_off + 4])
+            if idx != self.used_idx & 0xFFFF:
+                break
+            remaining = deadline - time.time()
+            if remaining <= 0:
+                raise TimeoutError("no completion from vhost target")
+            select.select([self.call], [], [], min(remaining, 0.5))
+            try:
+                os.eventfd_read(self.call)
+            except BlockingIOError:
+                pass
+        slot = self.used_off + 4 + 8 * (self.used_idx % self.qsize)
+        head, used_len = struct.unpack("<II", self.m.mem[slot:slot + 8])
+        self.used_idx += 1
+        self.last_used_head = head
+        return used_len

@@ -24,90 +24,83 @@ import time
 
 from .. import hipstore
 from .vhost_client import (
-    DATA_IN_OFF,
     DESC_NEXT,
     DESC_WRITE,
     GPA_BASE,
-    REQ_OFF,
-    RESP_OFF,
     VhostUserMaster,
 )
 
 
-class OutstandingScsiRing:
+class OutstandingRing:
     """Drives up to `iodepth` 3-descriptor READ/WRITE chains through
-    one vhost ring, reusing a fixed slot layout per in-flight tag.
-    personality "scsi" (virtio-scsi CDBs) or "blk" (virtio-blk
-    outhdr + status byte)."""
+    one VhostRing, a fixed slot layout per in-flight tag. personality
+    "scsi" (virtio-scsi CDBs) or "blk" (virtio-blk outhdr + status).
+    req_base/data_base are guest offsets owned by this ring."""
 
-    def __init__(self, master: VhostUserMaster, block_size: int,
-                 io_size: int, num_blocks: int, iodepth: int, rw: str,
-                 personality: str = "scsi"):
+    def __init__(self, ring, req_base: int, data_base: int,
+                 block_size: int, io_size: int, num_blocks: int,
+                 iodepth: int, rw: str, personality: str = "scsi",
+                 seed: int = 0x5EED):
+        self.ring = ring
+        self.mem = ring.m.mem
+        self.req_base = req_base
+        self.data_base = data_base
         self.personality = personality
-        self.m = master
         self.block = block_size
         self.io_size = io_size
         self.blocks_per_io = io_size // block_size
         self.units = num_blocks // self.blocks_per_io
         self.iodepth = iodepth
         self.rw = rw
-        self.rng = random.Random(0x5EED)
+        self.rng = random.Random(seed)
         self.submit_ts = {}
         self.lat_us = []
         self.completed = 0
-        if iodepth * 3 > master.qsize:
+        if iodepth * 3 > ring.qsize:
             raise ValueError("iodepth needs a bigger ring (qsize)")
 
     def _chain(self, tag: int):
         lba = (self.rng.randrange(self.units)) * self.blocks_per_io
         write = self.rw == "randwrite" or (
             self.rw == "randrw" and self.rng.random() < 0.5)
-        req_gpa = REQ_OFF + 0x100 * tag
-        resp_gpa = RESP_OFF + 0x100 * tag
-        data_gpa = DATA_IN_OFF + self.io_size * tag
+        req_gpa = self.req_base + 0x100 * tag       # request header
+        resp_gpa = req_gpa + 0x80                   # response / status
+        data_gpa = self.data_base + self.io_size * tag
         base = tag * 3
         if self.personality == "blk":
             sector = lba * self.block // 512
             header = struct.pack("<IIQ", 1 if write else 0, 0, sector)
-            self.m.mem[req_gpa:req_gpa + 16] = header
-            self.m._write_desc(base, GPA_BASE + req_gpa, 16,
-                               DESC_NEXT, base + 1)
-            if write:
-                self.m._write_desc(base + 1, GPA_BASE + data_gpa,
-                                   self.io_size, DESC_NEXT, base + 2)
-            else:
-                self.m._write_desc(base + 1, GPA_BASE + data_gpa,
-                                   self.io_size, DESC_WRITE | DESC_NEXT,
-                                   base + 2)
-            self.m._write_desc(base + 2, GPA_BASE + resp_gpa, 1,
-                               DESC_WRITE)
-            self.submit_ts[base] = time.perf_counter()
-            self.m._submit(base)
-            return
-        cdb = bytearray(10)
-        cdb[0] = 0x2A if write else 0x28
-        cdb[2:6] = struct.pack(">I", lba)
-        cdb[7:9] = struct.pack(">H", self.blocks_per_io)
-        req = (bytes([1, 0, 0x40, 0, 0, 0, 0, 0])
-               + struct.pack("<Q", tag + 1) + bytes(3)
-               + bytes(cdb).ljust(32, b"\0"))
-        self.m.mem[req_gpa:req_gpa + 51] = req
-        if write:
-            self.m._write_desc(base, GPA_BASE + req_gpa, 51,
-                               DESC_NEXT, base + 1)
-            self.m._write_desc(base + 1, GPA_BASE + data_gpa, self.io_size,
-                               DESC_NEXT, base + 2)
-            self.m._write_desc(base + 2, GPA_BASE + resp_gpa, 108,
-                               DESC_WRITE)
+            self.mem[req_gpa:req_gpa + 16] = header
+            self.ring.write_desc(base, GPA_BASE + req_gpa, 16,
+                                 DESC_NEXT, base + 1)
+            flags = DESC_NEXT if write else (DESC_WRITE | DESC_NEXT)
+            self.ring.write_desc(base + 1, GPA_BASE + data_gpa,
+                                 self.io_size, flags, base + 2)
+            self.ring.write_desc(base + 2, GPA_BASE + resp_gpa, 1,
+                                 DESC_WRITE)
         else:
-            self.m._write_desc(base, GPA_BASE + req_gpa, 51,
-                               DESC_NEXT, base + 1)
-            self.m._write_desc(base + 1, GPA_BASE + resp_gpa, 108,
-                               DESC_WRITE | DESC_NEXT, base + 2)
-            self.m._write_desc(base + 2, GPA_BASE + data_gpa, self.io_size,
-                               DESC_WRITE)
+            cdb = bytearray(10)
+            cdb[0] = 0x2A if write else 0x28
+            cdb[2:6] = struct.pack(">I", lba)
+            cdb[7:9] = struct.pack(">H", self.blocks_per_io)
+            req = (bytes([1, 0, 0x40, 0, 0, 0, 0, 0])
+                   + struct.pack("<Q", tag + 1) + bytes(3)
+                   + bytes(cdb).ljust(32, b"\0"))
+            self.mem[req_gpa:req_gpa + 51] = req
+            self.ring.write_desc(base, GPA_BASE + req_gpa, 51,
+                                 DESC_NEXT, base + 1)
+            if write:
+                self.ring.write_desc(base + 1, GPA_BASE + data_gpa,
+                                     self.io_size, DESC_NEXT, base + 2)
+                self.ring.write_desc(base + 2, GPA_BASE + resp_gpa, 108,
+                                     DESC_WRITE)
+            else:
+                self.ring.write_desc(base + 1, GPA_BASE + resp_gpa, 108,
+                                     DESC_WRITE | DESC_NEXT, base + 2)
+                self.ring.write_desc(base + 2, GPA_BASE + data_gpa,
+                                     self.io_size, DESC_WRITE)
         self.submit_ts[base] = time.perf_counter()
-        self.m._submit(base)
+        self.ring.submit(base)
 
     def run(self, seconds: float) -> None:
         deadline = time.perf_counter() + seconds
@@ -116,8 +109,8 @@ class OutstandingScsiRing:
         stopping = False
         inflight = self.iodepth
         while inflight:
-            self.m._wait_used()
-            head = self.m.last_used_head
+            self.ring.wait_used()
+            head = self.ring.last_used_head
             now = time.perf_counter()
             started = self.submit_ts.pop(head)
             self.lat_us.append((now - started) * 1e6)
@@ -139,7 +132,11 @@ def main(argv=None) -> int:
     parser.add_argument("--rw", default="randread",
                         choices=["randread", "randwrite", "randrw"])
     parser.add_argument("--bs", type=int, default=4096)
-    parser.add_argument("--iodepth", type=int, default=16)
+    parser.add_argument("--iodepth", type=int, default=16,
+                        help="outstanding chains PER ring")
+    parser.add_argument("--numjobs", type=int, default=1,
+                        help="request rings driven in parallel "
+                             "(scsi: up to 6, blk: up to 8)")
     parser.add_argument("--runtime", type=float, default=5.0)
     parser.add_argument("--ctrlr", default="vhost-bench")
     parser.add_argument("--personality", default="scsi",
@@ -160,32 +157,60 @@ def main(argv=None) -> int:
                           {"ctrlr": args.ctrlr, "scsi_target_num": 0,
                            "bdev_name": args.bdev})
         vhost_path = os.path.join(os.path.dirname(args.socket), args.ctrlr)
-        qsize = 1
+        max_jobs = 8 if args.personality == "blk" else 6
+        if not 1 <= args.numjobs <= max_jobs:
+            raise SystemExit(f"--numjobs must be 1..{max_jobs} "
+                             f"for {args.personality}")
+        first_queue = 0 if args.personality == "blk" else 2
+        qsize = 16
         while qsize < args.iodepth * 3 + 1:
             qsize *= 2
-        mem = (1 << 20) + DATA_IN_OFF + args.iodepth * args.bs
+        # per-ring slab: ring header + 0x100/request slot + data buffers
+        header_guess = ((16 + 2) * qsize + 0x2000 + 8 * qsize + 0xFFF) \
+            & ~0xFFF
+        slab = header_guess + args.iodepth * 0x100 + args.iodepth * args.bs
+        slab = (slab + 0xFFFF) & ~0xFFFF
+        mem = 0x10000 + args.numjobs * slab
         master = VhostUserMaster(
-            vhost_path, mem_size=max(mem, 8 << 20), qsize=max(qsize, 16),
-            queue=0 if args.personality == "blk" else 2)
-        master.negotiate()
+            vhost_path, mem_size=max(mem, 8 << 20), qsize=qsize,
+            queue=first_queue)
+        master.handshake()
+        rings = []
+        jobs = []
         try:
-            ring = OutstandingScsiRing(
-                master, bdev.block_size, args.bs, bdev.num_blocks,
-                args.iodepth, args.rw, personality=args.personality)
+            for i in range(args.numjobs):
+                base = 0x10000 + i * slab
+                ring = master.add_ring(first_queue + i, base, qsize)
+                req_base = base + ((ring.header_bytes + 0xFFF) & ~0xFFF)
+                data_base = req_base + args.iodepth * 0x100
+                rings.append(ring)
+                jobs.append(OutstandingRing(
+                    ring, req_base, data_base, bdev.block_size, args.bs,
+                    bdev.num_blocks, args.iodepth, args.rw,
+                    personality=args.personality, seed=0x5EED + i))
+            import threading
+            threads = [threading.Thread(target=j.run, args=(args.runtime,))
+                       for j in jobs]
             start = time.perf_counter()
-            ring.run(args.runtime)
+            for t in threads:
+                t.start()
+            for t in threads:
+                t.join()
             elapsed = time.perf_counter() - start
         finally:
+            for ring in rings:
+                ring.close()
             master.close()
             client.invoke("remove_vhost_controller", {"ctrlr": args.ctrlr})
 
-    iops = ring.completed / elapsed
-    lat = sorted(ring.lat_us)
+    completed = sum(j.completed for j in jobs)
+    iops = completed / elapsed
+    lat = sorted(x for j in jobs for x in j.lat_us)
     pct = lambda p: lat[min(len(lat) - 1, int(len(lat) * p))] if lat else 0
-    print(f"{args.bdev} via vhost: rw={args.rw}, bs={args.bs}, "
-          f"iodepth={args.iodepth}")
+    print(f"{args.bdev} via vhost ({args.personality}): rw={args.rw}, "
+          f"bs={args.bs}, iodepth={args.iodepth}, numjobs={args.numjobs}")
     print(f"  IOPS={iops:,.0f}, BW={iops * args.bs / 1e6:,.1f} MB/s "
-          f"({ring.completed} ios in {elapsed:.2f}s)")
+          f"({completed} ios in {elapsed:.2f}s)")
     if lat:
         print(f"  lat (usec): avg={statistics.fmean(lat):.1f}, "
               f"p50={pct(0.5):.0f}, p99={pct(0.99):.0f}, "
@@ -204,6 +229,7 @@ def main(argv=None) -> int:
                 "labels": {"bdev": args.bdev, "rw": args.rw,
                            "bs": str(args.bs),
                            "iodepth": str(args.iodepth),
+                           "numjobs": str(args.numjobs),
                            "path": f"vhost-user-{args.personality}"},
             }],
             "labels": {"suite": "hipstored-vhost"},

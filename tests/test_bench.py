@@ -114,7 +114,7 @@ class TestVhostHarness:
                 rc = vhost_harness.main([
                     "--socket", daemon.socket_path, "--bdev", "vhp",
                     "--rw", "randrw", "--bs", "4096", "--iodepth", "32",
-                    "--runtime", "0.5"])
+                    "--numjobs", "3", "--runtime", "0.5"])
             assert rc == 0 and "IOPS=" in out.getvalue()
         finally:
             daemon.stop()
