@@ -187,6 +187,12 @@ class FileBdev : public Bdev {
           }
           break;
         case IoOp::kFill: {
+          if (req.fill == 0 &&
+              fallocate(fd_, FALLOC_FL_PUNCH_HOLE | FALLOC_FL_KEEP_SIZE,
+                        static_cast<off_t>(req.offset),
+                        static_cast<off_t>(req.length)) == 0) {
+            break;  // real discard: the extent becomes a hole
+          }
           std::vector<uint8_t> zeros(
               std::min<uint64_t>(req.length, 1 << 20), req.fill);
           uint64_t done = 0;

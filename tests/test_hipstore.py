@@ -401,3 +401,43 @@ class TestAioBdev:
                               {"name": "aiox",
                                "filename": str(tmp_path / "empty.img"),
                                "block_size": 512})  # zero-size file
+
+    def test_trim_punches_holes(self, hipstored, tmp_path):  # noqa: F811
+        """Zero-fill on an AIO bdev punches a hole (sparse backing)."""
+        backing = tmp_path / "sparse.img"
+        backing.write_bytes(b"\xff" * (1 << 20))
+        with hipstore.Client(hipstored.socket_path) as client:
+            hipstore.construct_aio_bdev(client, "aio-trim", str(backing))
+            # trim the middle 512 KiB through the vhost SCSI UNMAP path
+            client.invoke("construct_vhost_scsi_controller",
+                          {"ctrlr": "vtrim"})
+            client.invoke("add_vhost_scsi_lun",
+                          {"ctrlr": "vtrim", "scsi_target_num": 0,
+                           "bdev_name": "aio-trim"})
+            import struct as structmod
+
+            from vhost_client import VhostUserMaster
+            path = __import__("os").path.join(
+                __import__("os").path.dirname(hipstored.socket_path),
+                "vtrim")
+            master = VhostUserMaster(path)
+            master.negotiate()
+            try:
+                descriptors = structmod.pack(">QII", 256, 1024, 0)
+                param = structmod.pack(">HH", 6 + 16, 16) + bytes(4) \
+                    + descriptors
+                cdb = bytearray(10)
+                cdb[0] = 0x42
+                cdb[7:9] = structmod.pack(">H", len(param))
+                result = master.scsi(0, bytes(cdb), data_out=param)
+                assert result.status == 0
+            finally:
+                master.close()
+            client.invoke("remove_vhost_controller", {"ctrlr": "vtrim"})
+            hipstore.delete_bdev(client, "aio-trim")
+        data = backing.read_bytes()
+        assert data[256 * 512:(256 + 1024) * 512] == bytes(1024 * 512)
+        assert data[:256 * 512] == b"\xff" * (256 * 512)
+        # the file is now sparse (fewer blocks than its size implies)
+        st = backing.stat()
+        assert st.st_blocks * 512 < 1 << 20
