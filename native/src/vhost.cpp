@@ -356,6 +356,8 @@ class VhostUserScsiDev {
     memcpy(config, &sectors, 8);
     const uint32_t blk_size = static_cast<uint32_t>(bdev->block_size());
     memcpy(config + 20, &blk_size, 4);
+    const uint16_t num_queues = 8;  // MQ-aware guests read this
+    memcpy(config + 34, &num_queues, 2);
     if (size >= 56) {
       // discard/write-zeroes geometry (virtio_blk_config offsets)
       const uint32_t max_sectors = kMaxIoBytes / 512;

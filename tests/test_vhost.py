@@ -407,6 +407,17 @@ class TestVhostBlk:
         config = read_blk_config(master)
         assert config["capacity_sectors"] == NUM_BLOCKS * BLOCK // 512
         assert config["blk_size"] == BLOCK
+        # offset reads work (QEMU reads fields piecemeal)
+        reply = master.query(
+            __import__("vhost_client").GET_CONFIG,
+            struct.pack("<III", 20, 4, 0) + bytes(4))
+        blk_size, = struct.unpack("<I", reply[12:16])
+        assert blk_size == BLOCK
+        num_queues, = struct.unpack(
+            "<H", master.query(
+                __import__("vhost_client").GET_CONFIG,
+                struct.pack("<III", 34, 2, 0) + bytes(2))[12:14])
+        assert num_queues == 8
 
     def test_rw_roundtrip_and_get_id(self, blk_target):
         _, master = blk_target
