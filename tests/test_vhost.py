@@ -621,6 +621,20 @@ class TestVhostProtocolFuzz:
     daemon (binary protocol parser hardening)."""
 
     def test_random_message_streams(self, hipstored, tmp_path):  # noqa: F811
+        self._fuzz(hipstored)
+
+    def test_random_streams_pipelined_worker(self, tmp_path, monkeypatch):
+        """Same fuzz against the pipelined worker's descriptor parser
+        (submit_async walks guest-controlled chains)."""
+        import fixtures
+        monkeypatch.setenv("HIPSTORE_VHOST_PIPELINE", "1")
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        try:
+            self._fuzz(daemon)
+        finally:
+            daemon.stop()
+
+    def _fuzz(self, hipstored):  # noqa: F811
         import socket as socketmod
 
         from hypothesis import HealthCheck, given, settings
