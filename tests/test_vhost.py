@@ -770,3 +770,48 @@ class TestMultiQueue:
                 ring_b.close()
                 master.close()
             client.invoke("remove_vhost_controller", {"ctrlr": "vmq"})
+
+
+class TestPipelinedRobustness:
+    """Malformed descriptor chains against the pipelined worker
+    (submit_async parses guest-controlled chains before the sync
+    fallback does)."""
+
+    @pytest.fixture
+    def pipelined_target(self, tmp_path, monkeypatch):
+        import fixtures
+        monkeypatch.setenv("HIPSTORE_VHOST_PIPELINE", "1")
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        with hipstore.Client(daemon.socket_path) as client:
+            hipstore.construct_malloc_bdev(
+                client, num_blocks=NUM_BLOCKS, block_size=BLOCK, name="ppr0")
+            client.invoke("construct_vhost_scsi_controller", {"ctrlr": "vpr"})
+            client.invoke("add_vhost_scsi_lun",
+                          {"ctrlr": "vpr", "scsi_target_num": 0,
+                           "bdev_name": "ppr0"})
+            master = VhostUserMaster(
+                os.path.join(os.path.dirname(daemon.socket_path), "vpr"))
+            master.negotiate()
+            yield client, master
+            master.close()
+            daemon.stop()
+
+    def test_bad_gpa_and_loops(self, pipelined_target):
+        from vhost_client import DESC_WRITE, GPA_BASE, REQ_OFF, RESP_OFF
+        client, master = pipelined_target
+        # out-of-range GPA
+        master._write_desc(0, 0xDEAD0000000, 51, 1, 1)
+        master._write_desc(1, GPA_BASE + RESP_OFF, 108, DESC_WRITE)
+        master._submit(0)
+        assert master._wait_used() == 0
+        # self-loop
+        master._write_desc(0, GPA_BASE + REQ_OFF, 51, 1, 0)
+        master._submit(0)
+        assert master._wait_used() == 0
+        # bounds: READ beyond the end takes the sync fallback -> sense
+        result = master.read10(0, NUM_BLOCKS, 1, BLOCK)
+        assert result.status == 2 and result.asc == 0x21
+        # still healthy on the fast path
+        data = os.urandom(BLOCK)
+        assert master.write10(0, 3, data, BLOCK).status == 0
+        assert master.read10(0, 3, 1, BLOCK).data == data
