@@ -815,3 +815,25 @@ class TestPipelinedRobustness:
         data = os.urandom(BLOCK)
         assert master.write10(0, 3, data, BLOCK).status == 0
         assert master.read10(0, 3, 1, BLOCK).data == data
+
+
+class TestVhostIostat:
+    def test_counters_reflect_guest_io(self, vhost_target):
+        client, master = vhost_target
+        before = {s.name: s for s in hipstore.get_bdevs_iostat(client)}
+        data = os.urandom(4 * BLOCK)
+        assert master.write10(0, 40, data, BLOCK).status == 0
+        assert master.read10(0, 40, 4, BLOCK).status == 0
+        # SCSI UNMAP counts as an unmap op
+        descriptors = struct.pack(">QII", 40, 4, 0)
+        param = struct.pack(">HH", 6 + 16, 16) + bytes(4) + descriptors
+        cdb = bytearray(10)
+        cdb[0] = 0x42
+        cdb[7:9] = struct.pack(">H", len(param))
+        assert master.scsi(0, bytes(cdb), data_out=param).status == 0
+        after = {s.name: s for s in hipstore.get_bdevs_iostat(client)}
+        b, a = before["vhb0"], after["vhb0"]
+        assert a.num_write_ops >= b.num_write_ops + 1
+        assert a.num_read_ops >= b.num_read_ops + 1
+        assert a.bytes_written >= b.bytes_written + 4 * BLOCK
+        assert a.num_unmap_ops >= b.num_unmap_ops + 1
