@@ -65,11 +65,9 @@ class RemoteBackend(OIMBackend):
         self.sysfs_block_dir = sysfs_block_dir
         self.dev_dir = dev_dir
         self.device_timeout = device_timeout
-        # Presentation metadata for snapshots created through this
-        # driver: snap id -> (source volume, ctime, size hint). The
-        # oim.v0 API has no stat/list RPC, so sizes come from the
-        # create_volume calls this process made (0 = unknown, which
-        # CSI allows).
+        # Provenance metadata for snapshots created through this
+        # driver: snap id -> (source volume, ctime). Sizes come from
+        # the daemon via the ListMallocBDevs extension.
         self._snap_meta: Dict[str, tuple] = {}
         self._volume_sizes: Dict[str, int] = {}
 
@@ -160,6 +158,17 @@ class RemoteBackend(OIMBackend):
     def supports_snapshots(self) -> bool:
         return True
 
+    def _bdev_size(self, name) -> int:
+        with self._dial_registry() as channel:
+            stub = spec.ControllerStub(channel)
+            reply = stub.ListMallocBDevs(
+                spec.ListMallocBDevsRequest(prefix=name),
+                metadata=self._metadata(), timeout=30)
+        for info in reply.bdevs:
+            if info.name == name:
+                return info.size
+        return 0
+
     def create_snapshot(self, name, source_volume_id):
         import time as _time
 
@@ -182,11 +191,11 @@ class RemoteBackend(OIMBackend):
                     raise LookupError(
                         f"volume {source_volume_id} not found") from None
                 raise
+        size = self._bdev_size(snap_id)
         if existing:
-            return snap_id, existing[2], existing[1]
+            return snap_id, size, existing[1]
         ctime = int(_time.time())
-        size = self._volume_sizes.get(source_volume_id, 0)
-        self._snap_meta[snap_id] = (source_volume_id, ctime, size)
+        self._snap_meta[snap_id] = (source_volume_id, ctime)
         return snap_id, size, ctime
 
     def delete_snapshot(self, snapshot_id) -> None:
@@ -212,7 +221,7 @@ class RemoteBackend(OIMBackend):
             source, ctime = "", 0
             meta = self._snap_meta.get(info.name)
             if meta:
-                source, ctime = meta[0], meta[1]
+                source, ctime = meta
             out.append((info.name, source, info.size, ctime))
         return out
 
@@ -247,8 +256,7 @@ class RemoteBackend(OIMBackend):
                     raise LookupError(
                         f"snapshot {snapshot_id} not found") from None
                 raise
-        meta = self._snap_meta.get(snapshot_id)
-        size = meta[2] if meta else 0
+        size = self._bdev_size(volume_name)
         if size:
             self._volume_sizes[volume_name] = size
         return volume_name, size
