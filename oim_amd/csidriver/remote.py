@@ -69,7 +69,6 @@ class RemoteBackend(OIMBackend):
         # driver: snap id -> (source volume, ctime). Sizes come from
         # the daemon via the ListMallocBDevs extension.
         self._snap_meta: Dict[str, tuple] = {}
-        self._volume_sizes: Dict[str, int] = {}
 
     # --- registry plumbing --------------------------------------------------
 
@@ -108,7 +107,6 @@ class RemoteBackend(OIMBackend):
             stub.ProvisionMallocBDev(
                 spec.ProvisionMallocBDevRequest(bdev_name=name, size=size),
                 metadata=self._metadata(), timeout=30)
-        self._volume_sizes[name] = size
         return name, {}
 
     def delete_volume(self, volume_id: str) -> None:
@@ -148,7 +146,6 @@ class RemoteBackend(OIMBackend):
                     raise LookupError(
                         f"volume {volume_id} not found") from None
                 raise
-        self._volume_sizes[volume_id] = size
         return size
 
     # --- snapshots (CloneMallocBDev extension, docs/spec.md) ----------------
@@ -235,13 +232,7 @@ class RemoteBackend(OIMBackend):
                 if not info.name.startswith(self.SNAP_PREFIX)]
 
     def clone_volume(self, source_volume_id, volume_name):
-        volume_id, size = self.restore_snapshot(source_volume_id,
-                                                volume_name)
-        if not size:
-            size = self._volume_sizes.get(source_volume_id, 0)
-            if size:
-                self._volume_sizes[volume_name] = size
-        return volume_id, size
+        return self.restore_snapshot(source_volume_id, volume_name)
 
     def restore_snapshot(self, snapshot_id, volume_name):
         with self._dial_registry() as channel:
@@ -256,10 +247,7 @@ class RemoteBackend(OIMBackend):
                     raise LookupError(
                         f"snapshot {snapshot_id} not found") from None
                 raise
-        size = self._bdev_size(volume_name)
-        if size:
-            self._volume_sizes[volume_name] = size
-        return volume_name, size
+        return volume_name, self._bdev_size(volume_name)
 
     # --- device lifecycle ---------------------------------------------------
 
