@@ -11,7 +11,6 @@ extended with proxied volume operations).
 """
 
 import argparse
-import sys
 
 import grpc
 

@@ -9,7 +9,7 @@ BDF (from /sys/class/kfd or rocm-smi) as the registry ``<id>/pci`` entry.
 from __future__ import annotations
 
 import re
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 UNSET = 0xFFFF
 

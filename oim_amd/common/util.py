@@ -8,7 +8,7 @@ import os
 import struct
 import subprocess
 import threading
-from typing import Callable, IO, Optional
+from typing import Callable, Optional
 
 from ..log import from_context
 

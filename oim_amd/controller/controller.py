@@ -403,6 +403,7 @@ class Controller(spec.ControllerServicer):
             channel = grpc.secure_channel(target, creds, options=options)
         else:
             channel = grpc.insecure_channel(target)
+        channel = grpc.intercept_channel(channel, LogClientInterceptor())
         try:
             stub = spec.RegistryStub(channel)
             stub.SetValue(

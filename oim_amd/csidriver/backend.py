@@ -5,7 +5,7 @@ via PCI/SCSI discovery)."""
 
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Dict, Tuple
 
 
 class VolumeExistsError(ValueError):

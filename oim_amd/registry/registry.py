@@ -16,7 +16,6 @@ Authorization model (CN = TLS client common name):
 
 from __future__ import annotations
 
-import threading
 from typing import Optional, Sequence
 
 import grpc

@@ -13,7 +13,7 @@ The tiny DSL below keeps the schema declarations readable; see
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, Optional, Sequence, Tuple
 
 from google.protobuf import descriptor_pb2, descriptor_pool, message_factory
 

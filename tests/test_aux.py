@@ -6,7 +6,6 @@ import json
 import subprocess
 import sys
 import threading
-import time
 
 
 from oim_amd import log

@@ -11,7 +11,7 @@ import grpc
 import pytest
 
 from oim_amd import spec
-from oim_amd.common.server import NonBlockingGRPCServer, grpc_target
+from oim_amd.common.server import grpc_target
 from oim_amd.controller import Controller, ControllerServer
 from oim_amd.registry import MemRegistryDB, Registry, RegistryServer
 

@@ -10,7 +10,6 @@ import subprocess
 
 import pytest
 
-from oim_amd import hipstore
 from oim_amd.common.util import get_blk_size64
 from oim_amd.csidriver import LocalBackend, Mounter
 

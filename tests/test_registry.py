@@ -305,7 +305,6 @@ class TestProxyDeadlines:
         """A short client deadline on a proxied call to a stalled
         controller surfaces as DEADLINE_EXCEEDED at the client and
         does not wedge the proxy for later calls."""
-        import threading
         import time as time_mod
 
         from oim_amd import spec

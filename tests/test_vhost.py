@@ -19,7 +19,7 @@ import pytest
 from oim_amd import hipstore
 
 from fixtures import hipstored  # noqa: F401
-from vhost_client import (FEAT_INDIRECT, VhostUserMaster, GET_QUEUE_NUM)
+from vhost_client import VhostUserMaster, GET_QUEUE_NUM
 
 BLOCK = 512
 NUM_BLOCKS = 8192  # 4 MiB
@@ -506,7 +506,6 @@ class TestTrim:
         assert result.status == 2 and result.asc == 0x21
 
     def test_blk_discard(self, hipstored, tmp_path):  # noqa: F811
-        from vhost_client import FEAT_VERSION_1
         with hipstore.Client(hipstored.socket_path) as client:
             hipstore.construct_malloc_bdev(
                 client, num_blocks=2048, block_size=512, name="blktrim")
