@@ -42,13 +42,15 @@ def main(argv=None) -> int:
         ("clone", "clone a malloc bdev (device-side HBM-rate copy)"),
         ("resize", "grow a malloc bdev (offline)"),
         ("list", "list malloc bdevs"),
+        ("stats", "per-bdev I/O counters"),
     ):
         cmd = sub.add_parser(name, help=help_text + " via the registry proxy")
         cmd.add_argument("--controller", required=True,
                          help="controller ID to proxy to")
-        if name == "list":
+        if name in ("list", "stats"):
             cmd.add_argument("volume", nargs="?", default="",
-                             help="name prefix filter")
+                             help="name filter" if name == "stats"
+                             else "name prefix filter")
         else:
             cmd.add_argument("volume")
         if name == "provision":
@@ -124,6 +126,15 @@ def main(argv=None) -> int:
                 for info in reply.bdevs:
                     print(f"{info.name}  {info.size}  bs={info.block_size}  "
                           f"{info.product_name}")
+            elif args.command == "stats":
+                reply = controller.GetIOStats(
+                    spec.GetIOStatsRequest(bdev_name=args.volume),
+                    metadata=metadata, timeout=60)
+                for st in reply.bdevs:
+                    print(f"{st.name}  reads={st.num_read_ops} "
+                          f"writes={st.num_write_ops} "
+                          f"unmaps={st.num_unmap_ops} "
+                          f"rB={st.bytes_read} wB={st.bytes_written}")
             elif args.command == "resize":
                 new_size = parse_size(args.size)
                 controller.ResizeMallocBDev(

@@ -388,6 +388,21 @@ class Controller(spec.ControllerServicer):
                 info.product_name = bdev.product_name
         return reply
 
+    def GetIOStats(self, request, context):
+        """oim-amd extension (docs/spec.md): per-bdev I/O counters."""
+        reply = spec.GetIOStatsReply()
+        with self._client() as client:
+            for stat in hipstore.get_bdevs_iostat(client,
+                                                  request.bdev_name):
+                entry = reply.bdevs.add()
+                entry.name = stat.name
+                entry.num_read_ops = stat.num_read_ops
+                entry.num_write_ops = stat.num_write_ops
+                entry.num_unmap_ops = stat.num_unmap_ops
+                entry.bytes_read = stat.bytes_read
+                entry.bytes_written = stat.bytes_written
+        return reply
+
     # --- self-registration --------------------------------------------------
 
     def register(self) -> None:

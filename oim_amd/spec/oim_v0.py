@@ -78,6 +78,16 @@ MESSAGES = [
              Field("product_name", 4, "string")]),
     Message("ListMallocBDevsReply",
             [Field("bdevs", 1, "BDevInfo", repeated=True)]),
+    Message("GetIOStatsRequest", [Field("bdev_name", 1, "string")]),
+    Message("BDevIOStats",
+            [Field("name", 1, "string"),
+             Field("num_read_ops", 2, "int64"),
+             Field("num_write_ops", 3, "int64"),
+             Field("num_unmap_ops", 4, "int64"),
+             Field("bytes_read", 5, "int64"),
+             Field("bytes_written", 6, "int64")]),
+    Message("GetIOStatsReply",
+            [Field("bdevs", 1, "BDevIOStats", repeated=True)]),
 ]
 
 SERVICES = [
@@ -98,6 +108,7 @@ SERVICES = [
             ("CloneMallocBDev", "CloneMallocBDevRequest", "CloneMallocBDevReply"),
             ("ResizeMallocBDev", "ResizeMallocBDevRequest", "ResizeMallocBDevReply"),
             ("ListMallocBDevs", "ListMallocBDevsRequest", "ListMallocBDevsReply"),
+            ("GetIOStats", "GetIOStatsRequest", "GetIOStatsReply"),
         ],
     ),
 ]
@@ -130,3 +141,6 @@ ResizeMallocBDevReply = _classes["ResizeMallocBDevReply"]
 ListMallocBDevsRequest = _classes["ListMallocBDevsRequest"]
 BDevInfo = _classes["BDevInfo"]
 ListMallocBDevsReply = _classes["ListMallocBDevsReply"]
+GetIOStatsRequest = _classes["GetIOStatsRequest"]
+BDevIOStats = _classes["BDevIOStats"]
+GetIOStatsReply = _classes["GetIOStatsReply"]

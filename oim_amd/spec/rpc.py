@@ -51,6 +51,7 @@ class ControllerStub:
         self.CloneMallocBDev = method("CloneMallocBDev", pb.CloneMallocBDevReply)
         self.ResizeMallocBDev = method("ResizeMallocBDev", pb.ResizeMallocBDevReply)
         self.ListMallocBDevs = method("ListMallocBDevs", pb.ListMallocBDevsReply)
+        self.GetIOStats = method("GetIOStats", pb.GetIOStatsReply)
 
 
 class RegistryServicer:
@@ -87,6 +88,9 @@ class ControllerServicer:
     def ListMallocBDevs(self, request, context):
         context.abort(grpc.StatusCode.UNIMPLEMENTED, "ListMallocBDevs not implemented")
 
+    def GetIOStats(self, request, context):
+        context.abort(grpc.StatusCode.UNIMPLEMENTED, "GetIOStats not implemented")
+
 
 def _unary(fn, request_class):
     return grpc.unary_unary_rpc_method_handler(
@@ -117,6 +121,7 @@ def add_controller_to_server(servicer: ControllerServicer, server: grpc.Server) 
         "CloneMallocBDev": _unary(servicer.CloneMallocBDev, pb.CloneMallocBDevRequest),
         "ResizeMallocBDev": _unary(servicer.ResizeMallocBDev, pb.ResizeMallocBDevRequest),
         "ListMallocBDevs": _unary(servicer.ListMallocBDevs, pb.ListMallocBDevsRequest),
+        "GetIOStats": _unary(servicer.GetIOStats, pb.GetIOStatsRequest),
     }
     server.add_generic_rpc_handlers(
         (grpc.method_handlers_generic_handler(CONTROLLER_SERVICE, handlers),)

@@ -86,6 +86,10 @@ class TestOimctlVolumes:
                                 "--controller", "c9"]) == 0
             out = capsys.readouterr().out
             assert "volx-c" in out
+            assert oimctl.main(["--registry", endpoint, "stats",
+                                "--controller", "c9"]) == 0
+            out = capsys.readouterr().out
+            assert "volx" in out and "reads=" in out
             assert oimctl.main(["--registry", endpoint, "resize",
                                 "--controller", "c9", "volx", "128MiB"]) == 0
             assert oimctl.main(["--registry", endpoint, "provision",
