@@ -101,6 +101,8 @@ PYBIND11_MODULE(_hipstore, m) {
   m.def("create_hbm_bdev", &create_hbm_bdev, py::arg("name"),
         py::arg("block_size"), py::arg("num_blocks"), py::arg("device") = 0,
         py::arg("persistent") = false);
+  m.def("create_file_bdev", &create_file_bdev, py::arg("name"),
+        py::arg("path"), py::arg("block_size") = 512);
   m.def("create_striped_bdev", &create_striped_bdev, py::arg("name"),
         py::arg("children"), py::arg("stripe_size") = 131072);
   m.def("create_replicated_bdev", &create_replicated_bdev, py::arg("name"),
