@@ -62,11 +62,16 @@ def start_stack(rank, local_rank, use_gpu, args, tmp):
     daemon_sock = os.path.join(tmp, "hipstored.sock")
     cmd = [os.path.join(REPO_ROOT, "bin", "hipstored"), "-S", daemon_sock,
            "-d", str(local_rank)]
+    daemon_env = dict(os.environ)
     if args.engine == "persistent":
         cmd.append("-P")
+    elif args.engine == "shared":
+        # one service kernel per device multiplexing all rings
+        cmd.append("-P")
+        daemon_env["HIPSTORE_SHARED"] = "1"
     if not use_gpu:
         cmd.append("-C")
-    daemon = subprocess.Popen(cmd, stderr=subprocess.DEVNULL)
+    daemon = subprocess.Popen(cmd, stderr=subprocess.DEVNULL, env=daemon_env)
     deadline = time.time() + 60
     while not os.path.exists(daemon_sock):
         if daemon.poll() is not None:
@@ -144,9 +149,10 @@ def main() -> int:
     parser.add_argument("--bdev-gb", type=float, default=8.0)
     parser.add_argument("--workload", default="randread")
     parser.add_argument("--engine", default="persistent",
-                        choices=["batched", "persistent"],
-                        help="HBM I/O engine: batched kernel launches or "
-                             "the on-GPU polling service kernel")
+                        choices=["batched", "persistent", "shared"],
+                        help="HBM I/O engine: batched kernel launches, "
+                             "per-queue service kernels, or the shared "
+                             "one-kernel-per-device service")
     args = parser.parse_args()
 
     import torch
