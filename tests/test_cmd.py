@@ -106,3 +106,26 @@ class TestOimctlVolumes:
         assert oimctl.parse_size("64MiB") == 64 << 20
         assert oimctl.parse_size("1G") == 1 << 30
         assert oimctl.parse_size("2k") == 2048
+
+
+class TestOimctlTLS:
+    def test_set_get_over_mtls(self, tmp_path, capsys):
+        import ca as ca_util
+        from oim_amd.registry import MemRegistryDB, Registry, RegistryServer
+
+        trusted = ca_util.make_ca(str(tmp_path / "ca"))
+        registry = Registry(
+            db=MemRegistryDB(),
+            tls=ca_util.tls_config(trusted, "component.registry"))
+        server = RegistryServer(f"unix://{tmp_path}/reg.sock", registry)
+        server.start()
+        try:
+            admin = ca_util.tls_config(trusted, "user.admin")
+            argv_base = ["--registry", server.addr(),
+                         "--ca", admin.ca, "--key", admin.key]
+            assert oimctl.main(argv_base + ["set", "sec-0/address",
+                                            "tcp://x:1"]) == 0
+            assert oimctl.main(argv_base + ["get"]) == 0
+            assert "sec-0/address: tcp://x:1" in capsys.readouterr().out
+        finally:
+            server.stop()
