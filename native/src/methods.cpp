@@ -550,6 +550,34 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         return Json(std::move(o));
       });
 
+  server->register_method("nvmf_get_subsystems", [targets](const Json&) {
+    std::lock_guard<std::mutex> lock(targets->mutex);
+    JsonArray out;
+    for (const auto& [subnqn, entry] : targets->by_nqn) {
+      JsonArray namespaces;
+      int64_t nsid = 1;
+      for (const std::string& bdev : entry.ns_bdevs) {
+        JsonObject ns;
+        ns["nsid"] = Json(nsid++);
+        ns["bdev_name"] = Json(bdev);
+        namespaces.push_back(Json(std::move(ns)));
+      }
+      JsonArray addresses;
+      JsonObject addr;
+      addr["trtype"] = Json(std::string("TCP"));
+      addr["traddr"] = Json(entry.listen_addr);
+      addr["trsvcid"] = Json(std::to_string(entry.target->port()));
+      addresses.push_back(Json(std::move(addr)));
+      JsonObject o;
+      o["nqn"] = Json(subnqn);
+      o["subtype"] = Json(std::string("NVMe"));
+      o["listen_addresses"] = Json(std::move(addresses));
+      o["namespaces"] = Json(std::move(namespaces));
+      out.push_back(Json(std::move(o)));
+    }
+    return Json(std::move(out));
+  });
+
   server->register_method("nvmf_delete_target", [targets](const Json& p) {
     std::lock_guard<std::mutex> lock(targets->mutex);
     auto it = targets->by_nqn.find(p.get_string("subnqn"));

@@ -114,3 +114,24 @@ class TestNvmfRpc:
             hipstore.delete_bdev(client, "remote-a")
             client.invoke("nvmf_delete_target", {"subnqn": SUBNQN + "-rpc"})
             hipstore.delete_bdev(client, "ns-a")
+
+
+class TestSubsystemListing:
+    def test_nvmf_get_subsystems(self, hipstored):  # noqa: F811
+        from oim_amd import hipstore
+        with hipstore.Client(hipstored.socket_path) as client:
+            hipstore.construct_malloc_bdev(client, num_blocks=1024,
+                                           block_size=512, name="nsls")
+            target = client.invoke("nvmf_create_target",
+                                   {"listen_addr": "127.0.0.1", "port": 0,
+                                    "subnqn": "nqn.ls", "bdevs": ["nsls"]})
+            subsystems = client.invoke("nvmf_get_subsystems")
+            match = [s for s in subsystems if s["nqn"] == "nqn.ls"]
+            assert match
+            assert match[0]["namespaces"][0]["bdev_name"] == "nsls"
+            assert match[0]["listen_addresses"][0]["trsvcid"] == \
+                str(target["port"])
+            client.invoke("nvmf_delete_target", {"subnqn": "nqn.ls"})
+            assert not [s for s in client.invoke("nvmf_get_subsystems")
+                        if s["nqn"] == "nqn.ls"]
+            hipstore.delete_bdev(client, "nsls")
