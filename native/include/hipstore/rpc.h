@@ -41,6 +41,8 @@ class RpcServer {
   ~RpcServer();
 
   void register_method(const std::string& name, RpcMethod fn);
+  // Register `alias` to dispatch to already-registered `existing`.
+  void register_alias(const std::string& alias, const std::string& existing);
   bool has_method(const std::string& name) const;
   std::vector<std::string> method_names() const;
 

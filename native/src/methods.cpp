@@ -1053,6 +1053,35 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
     }
     return Json(applied);
   });
+
+  // Modern SPDK method-name aliases (SPDK renamed its RPC surface in
+  // v19.x; the reference client uses the old names, newer stock
+  // tooling the new ones — serve both).
+  for (const auto& [new_name, old_name] :
+       std::initializer_list<std::pair<const char*, const char*>>{
+           {"bdev_get_bdevs", "get_bdevs"},
+           {"bdev_get_iostat", "get_bdevs_iostat"},
+           {"bdev_malloc_create", "construct_malloc_bdev"},
+           {"bdev_malloc_delete", "delete_bdev"},
+           {"bdev_aio_create", "construct_aio_bdev"},
+           {"bdev_aio_delete", "delete_bdev"},
+           {"bdev_rbd_create", "construct_rbd_bdev"},
+           {"bdev_rbd_delete", "delete_bdev"},
+           {"nbd_start_disk", "start_nbd_disk"},
+           {"nbd_get_disks", "get_nbd_disks"},
+           {"nbd_stop_disk", "stop_nbd_disk"},
+           {"vhost_create_scsi_controller",
+            "construct_vhost_scsi_controller"},
+           {"vhost_scsi_controller_add_target", "add_vhost_scsi_lun"},
+           {"vhost_scsi_controller_remove_target",
+            "remove_vhost_scsi_target"},
+           {"vhost_create_blk_controller", "construct_vhost_blk_controller"},
+           {"vhost_delete_controller", "remove_vhost_controller"},
+           {"vhost_get_controllers", "get_vhost_controllers"},
+           {"rpc_get_methods", "get_rpc_methods"},
+       }) {
+    server->register_alias(new_name, old_name);
+  }
 }
 
 }  // namespace hipstore

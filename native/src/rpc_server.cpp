@@ -24,6 +24,14 @@ bool RpcServer::has_method(const std::string& name) const {
   return methods_.count(name) > 0;
 }
 
+void RpcServer::register_alias(const std::string& alias,
+                               const std::string& existing) {
+  std::lock_guard<std::mutex> lock(mutex_);
+  auto it = methods_.find(existing);
+  if (it == methods_.end()) return;
+  methods_[alias] = it->second;
+}
+
 std::vector<std::string> RpcServer::method_names() const {
   std::lock_guard<std::mutex> lock(mutex_);
   std::vector<std::string> names;

@@ -441,3 +441,21 @@ class TestAioBdev:
         # the file is now sparse (fewer blocks than its size implies)
         st = backing.stat()
         assert st.st_blocks * 512 < 1 << 20
+
+
+class TestModernSpdkNames:
+    """SPDK v19+ renamed its RPC surface; both generations dispatch."""
+
+    def test_aliases(self, hipstored):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            client.invoke("bdev_malloc_create",
+                          {"name": "alias0", "num_blocks": 1024,
+                           "block_size": 512})
+            bdevs = client.invoke("bdev_get_bdevs", {"name": "alias0"})
+            assert bdevs[0]["name"] == "alias0"
+            stats = client.invoke("bdev_get_iostat", {})
+            assert any(s["name"] == "alias0" for s in stats["bdevs"])
+            methods = client.invoke("rpc_get_methods")
+            assert "bdev_malloc_create" in methods
+            assert "vhost_create_scsi_controller" in methods
+            client.invoke("bdev_malloc_delete", {"name": "alias0"})
