@@ -198,6 +198,71 @@ class BDevIostat:
     bytes_written: int
 
 
+def construct_striped_malloc_bdev(client: Client, name: str,
+                                  num_blocks: int, block_size: int,
+                                  stripe_size_kb: int = 128,
+                                  count: int = 2,
+                                  devices: Optional[List[int]] = None) -> str:
+    """Striped malloc bdev across GPUs (oim-amd extension)."""
+    params = {"name": name, "num_blocks": num_blocks,
+              "block_size": block_size, "stripe_size_kb": stripe_size_kb}
+    if devices:
+        params["devices"] = devices
+    else:
+        params["count"] = count
+    return client.invoke("construct_striped_malloc_bdev", params)
+
+
+def construct_replicated_malloc_bdev(client: Client, name: str,
+                                     num_blocks: int, block_size: int,
+                                     count: int = 2,
+                                     devices: Optional[List[int]] = None
+                                     ) -> str:
+    """Replicated malloc bdev across GPUs (oim-amd extension)."""
+    params = {"name": name, "num_blocks": num_blocks,
+              "block_size": block_size}
+    if devices:
+        params["devices"] = devices
+    else:
+        params["count"] = count
+    return client.invoke("construct_replicated_malloc_bdev", params)
+
+
+def bdev_clone(client: Client, src: str, name: str) -> str:
+    """Device-side clone (HBM rates; xGMI across GPUs)."""
+    return client.invoke("bdev_clone", {"src": src, "name": name})
+
+
+def resize_malloc_bdev(client: Client, name: str, size: int) -> int:
+    """Offline resize to `size` bytes; returns the new block count."""
+    return client.invoke("resize_malloc_bdev",
+                         {"name": name, "size": size})["num_blocks"]
+
+
+def nvmf_create_target(client: Client, listen_addr: str = "127.0.0.1",
+                       port: int = 0, subnqn: str = "",
+                       bdevs: Optional[List[str]] = None,
+                       digests: bool = True) -> Dict:
+    params = {"listen_addr": listen_addr, "port": port, "digests": digests,
+              "bdevs": bdevs or []}
+    if subnqn:
+        params["subnqn"] = subnqn
+    return client.invoke("nvmf_create_target", params)
+
+
+def nvmf_delete_target(client: Client, subnqn: str) -> None:
+    client.invoke("nvmf_delete_target", {"subnqn": subnqn})
+
+
+def save_config(client: Client) -> Dict:
+    return client.invoke("save_config")
+
+
+def load_config(client: Client, config: Dict) -> int:
+    """Replays a save_config snapshot; returns entries applied."""
+    return client.invoke("load_config", config)
+
+
 def get_bdevs_iostat(client: Client, name: str = "") -> List[BDevIostat]:
     params = {"name": name} if name else {}
     result = client.invoke("get_bdevs_iostat", params)

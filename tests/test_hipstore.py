@@ -459,3 +459,28 @@ class TestModernSpdkNames:
             assert "bdev_malloc_create" in methods
             assert "vhost_create_scsi_controller" in methods
             client.invoke("bdev_malloc_delete", {"name": "alias0"})
+
+
+class TestTypedWrappers:
+    def test_extension_wrappers(self, hipstored, tmp_path):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            hipstore.construct_malloc_bdev(client, num_blocks=1024,
+                                           block_size=512, name="tw0")
+            assert hipstore.bdev_clone(client, "tw0", "tw0c") == "tw0c"
+            assert hipstore.resize_malloc_bdev(client, "tw0",
+                                               2048 * 512) == 2048
+            hipstore.construct_striped_malloc_bdev(
+                client, "tws", num_blocks=512, block_size=512,
+                stripe_size_kb=64, count=2)
+            hipstore.construct_replicated_malloc_bdev(
+                client, "twr", num_blocks=512, block_size=512, count=2)
+            target = hipstore.nvmf_create_target(client, bdevs=["tw0c"],
+                                                 subnqn="nqn.tw")
+            assert target["port"] > 0
+            config = hipstore.save_config(client)
+            assert hipstore.load_config(client, {"subsystems": []}) == 0
+            assert any(sub["subsystem"] == "nvmf"
+                       for sub in config["subsystems"])
+            hipstore.nvmf_delete_target(client, "nqn.tw")
+            for name in ("tws", "twr", "tw0c", "tw0"):
+                hipstore.delete_bdev(client, name)
