@@ -484,3 +484,25 @@ class TestTypedWrappers:
             hipstore.nvmf_delete_target(client, "nqn.tw")
             for name in ("tws", "twr", "tw0c", "tw0"):
                 hipstore.delete_bdev(client, name)
+
+    def test_daemon_refuses_bad_startup_config(self, tmp_path):
+        import subprocess
+
+        import fixtures
+
+        bad = tmp_path / "bad.json"
+        bad.write_text('{"subsystems": [{"subsystem": "bdev", "config": '
+                       '[{"method": "no_such", "params": {}}]}]}')
+        proc = subprocess.run(
+            [fixtures.DEFAULT_BINARY, "-S", str(tmp_path / "x.sock"),
+             "-C", "-c", str(bad)],
+            capture_output=True, text=True, timeout=30)
+        assert proc.returncode == 1
+        assert "config load failed" in proc.stderr
+        garbage = tmp_path / "garbage.json"
+        garbage.write_text("not json at all {")
+        proc = subprocess.run(
+            [fixtures.DEFAULT_BINARY, "-S", str(tmp_path / "y.sock"),
+             "-C", "-c", str(garbage)],
+            capture_output=True, text=True, timeout=30)
+        assert proc.returncode == 1
