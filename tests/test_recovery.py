@@ -242,7 +242,9 @@ class TestKitchenSink:
         import subprocess
         import time as time_mod
 
+        import os
         import fixtures
+        from oim_amd import hipstore
         from vhost_client import VhostUserMaster
 
         backing = tmp_path / "aio.img"
