@@ -203,3 +203,45 @@ class TestCloneAndResizeRaces:
                 spec.CheckMallocBDevRequest(bdev_name="orphan"),
                 metadata=METADATA, timeout=30)
         assert excinfo.value.code() == grpc.StatusCode.NOT_FOUND
+
+
+class TestConfigSnapshotRaces:
+    def test_save_config_during_churn(self, stack, hipstored):  # noqa: F811
+        """save_config stays valid while volumes churn concurrently."""
+        from oim_amd import hipstore
+
+        stop = threading.Event()
+        errors = []
+
+        def churn(i):
+            try:
+                while not stop.is_set():
+                    stack.ProvisionMallocBDev(
+                        spec.ProvisionMallocBDevRequest(
+                            bdev_name=f"churn-{i}", size=1 << 20),
+                        metadata=METADATA, timeout=30)
+                    stack.ProvisionMallocBDev(
+                        spec.ProvisionMallocBDevRequest(
+                            bdev_name=f"churn-{i}", size=0),
+                        metadata=METADATA, timeout=30)
+            except grpc.RpcError as exc:
+                errors.append(exc)
+
+        threads = [threading.Thread(target=churn, args=(i,))
+                   for i in range(3)]
+        for t in threads:
+            t.start()
+        try:
+            with hipstore.Client(hipstored.socket_path) as client:
+                for _ in range(30):
+                    config = client.invoke("save_config")
+                    assert {s["subsystem"] for s in config["subsystems"]} \
+                        == {"bdev", "vhost", "nvmf"}
+                    for sub in config["subsystems"]:
+                        for entry in sub["config"]:
+                            assert "method" in entry
+        finally:
+            stop.set()
+            for t in threads:
+                t.join()
+        assert not errors
