@@ -112,16 +112,47 @@ struct Parser {
               else if (h >= 'A' && h <= 'F') code |= h - 'A' + 10;
               else fail("bad \\u escape");
             }
-            // UTF-8 encode (BMP only; surrogate pairs not needed for RPC).
-            if (code < 0x80) {
-              s += static_cast<char>(code);
-            } else if (code < 0x800) {
-              s += static_cast<char>(0xC0 | (code >> 6));
-              s += static_cast<char>(0x80 | (code & 0x3F));
+            // Surrogate pair: \uD800-\uDBFF must be followed by
+            // \uDC00-\uDFFF; combine to the astral code point.
+            // (Astral characters — emoji in a volume name — arrive
+            // exactly this way from json.dumps; decoding the halves
+            // separately would emit invalid UTF-8 that poisons the
+            // response stream for stricter peers.)
+            uint32_t cp = code;
+            if (code >= 0xD800 && code <= 0xDBFF) {
+              if (end - p < 6 || p[0] != '\\' || p[1] != 'u') {
+                if (end - p < 6) { incomplete = true; return false; }
+                fail("lone high surrogate");
+              }
+              uint32_t low = 0;
+              for (int k = 2; k < 6; ++k) {
+                low <<= 4;
+                char h = p[k];
+                if (h >= '0' && h <= '9') low |= h - '0';
+                else if (h >= 'a' && h <= 'f') low |= h - 'a' + 10;
+                else if (h >= 'A' && h <= 'F') low |= h - 'A' + 10;
+                else fail("bad \\u escape");
+              }
+              if (low < 0xDC00 || low > 0xDFFF) fail("bad surrogate pair");
+              p += 6;
+              cp = 0x10000 + ((code - 0xD800) << 10) + (low - 0xDC00);
+            } else if (code >= 0xDC00 && code <= 0xDFFF) {
+              fail("lone low surrogate");
+            }
+            if (cp < 0x80) {
+              s += static_cast<char>(cp);
+            } else if (cp < 0x800) {
+              s += static_cast<char>(0xC0 | (cp >> 6));
+              s += static_cast<char>(0x80 | (cp & 0x3F));
+            } else if (cp < 0x10000) {
+              s += static_cast<char>(0xE0 | (cp >> 12));
+              s += static_cast<char>(0x80 | ((cp >> 6) & 0x3F));
+              s += static_cast<char>(0x80 | (cp & 0x3F));
             } else {
-              s += static_cast<char>(0xE0 | (code >> 12));
-              s += static_cast<char>(0x80 | ((code >> 6) & 0x3F));
-              s += static_cast<char>(0x80 | (code & 0x3F));
+              s += static_cast<char>(0xF0 | (cp >> 18));
+              s += static_cast<char>(0x80 | ((cp >> 12) & 0x3F));
+              s += static_cast<char>(0x80 | ((cp >> 6) & 0x3F));
+              s += static_cast<char>(0x80 | (cp & 0x3F));
             }
             break;
           }

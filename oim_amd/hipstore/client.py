@@ -83,8 +83,14 @@ class Client:
         while True:
             try:
                 text = self._buffer.decode("utf-8", errors="strict")
-            except UnicodeDecodeError:
-                # recv() split a multi-byte sequence: need more bytes
+            except UnicodeDecodeError as exc:
+                # recv() may split a multi-byte sequence at the TAIL;
+                # invalid bytes anywhere else can never become valid
+                # by reading more — fail instead of waiting forever.
+                if exc.start < len(self._buffer) - 4:
+                    raise RpcError(
+                        ERROR_INTERNAL,
+                        "invalid UTF-8 in daemon response") from None
                 text = ""
             stripped = text.lstrip()
             offset = len(text) - len(stripped)

@@ -231,3 +231,84 @@ class TestNvmfFuzz:
             assert bdev.read(512, 512) == b"\xbb" * 512
         finally:
             target.stop()
+
+
+class TestLoadConfigFuzz:
+    def test_nested_config_fuzz(self, hipstored):  # noqa: F811
+        """Arbitrary nested snapshot structures either apply or fail
+        with a clean error; the daemon survives all of them."""
+        from hypothesis import HealthCheck, given, settings
+        from hypothesis import strategies as st
+
+        scalar = st.one_of(st.none(), st.booleans(), st.integers(
+            min_value=-2**40, max_value=2**40), st.text(max_size=20))
+        entry = st.fixed_dictionaries(
+            {}, optional={
+                "method": st.one_of(scalar, st.sampled_from(
+                    ["construct_malloc_bdev", "no_such", "save_config"])),
+                "params": st.one_of(scalar, st.dictionaries(
+                    st.sampled_from(["name", "num_blocks", "block_size"]),
+                    scalar, max_size=3)),
+            })
+        subsystem = st.fixed_dictionaries(
+            {}, optional={
+                "subsystem": scalar,
+                "config": st.one_of(scalar, st.lists(entry, max_size=3)),
+            })
+        config = st.one_of(
+            scalar,
+            st.fixed_dictionaries({}, optional={
+                "subsystems": st.one_of(scalar,
+                                        st.lists(subsystem, max_size=3))}))
+
+        with hipstore.Client(hipstored.socket_path) as client:
+            @settings(max_examples=100, deadline=None,
+                      suppress_health_check=[
+                          HealthCheck.function_scoped_fixture])
+            @given(config)
+            def fuzz(payload):
+                try:
+                    client.invoke("load_config",
+                                  payload if isinstance(payload, dict)
+                                  else {"subsystems": payload})
+                except hipstore.RpcError:
+                    pass
+
+            fuzz()
+            assert isinstance(client.invoke("get_rpc_methods"), list)
+
+
+class TestAstralCharacters:
+    """json.dumps encodes astral chars as \\uD8xx\\uDCxx surrogate
+    pairs; the daemon must decode pairs (not emit CESU-8) or its
+    response stream poisons strict UTF-8 clients. Regression for a
+    fuzz-found connection wedge."""
+
+    def test_emoji_name_roundtrip(self, hipstored):  # noqa: F811
+        with hipstore.Client(hipstored.socket_path) as client:
+            name = "vol-\U0001F999"
+            client.invoke("construct_malloc_bdev",
+                          {"name": name, "num_blocks": 1024,
+                           "block_size": 512})
+            assert client.invoke("get_bdevs",
+                                 {"name": name})[0]["name"] == name
+            # error-echo path: astral chars in the error message
+            with pytest.raises(hipstore.RpcError):
+                client.invoke("load_config", {"subsystems": [{"config": [
+                    {"method": "x-\U0001F999", "params": {}}]}]})
+            # the connection survives both
+            client.invoke("delete_bdev", {"name": name})
+            assert isinstance(client.invoke("get_rpc_methods"), list)
+
+    def test_lone_surrogate_rejected(self, hipstored):  # noqa: F811
+        import socket as socketmod
+        sock = socketmod.socket(socketmod.AF_UNIX, socketmod.SOCK_STREAM)
+        sock.settimeout(5)
+        sock.connect(hipstored.socket_path)
+        try:
+            sock.sendall(b'{"jsonrpc":"2.0","id":1,"method":"\\ud800x",'
+                         b'"params":{}}')
+            reply = sock.recv(65536)
+            assert b"parse error" in reply or reply == b""
+        finally:
+            sock.close()
