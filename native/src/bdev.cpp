@@ -273,7 +273,9 @@ BdevPtr create_malloc_bdev(const std::string& name, uint64_t block_size,
 
 BdevPtr create_file_bdev(const std::string& name, const std::string& path,
                          uint64_t block_size) {
-  int fd = ::open(path.c_str(), O_RDWR | O_CREAT | O_CLOEXEC, 0600);
+  // The file must already exist (callers size it first); O_CREAT here
+  // would leave stray empty files behind on validation failures.
+  int fd = ::open(path.c_str(), O_RDWR | O_CLOEXEC);
   if (fd < 0) {
     throw std::runtime_error("aio bdev: cannot open " + path);
   }
