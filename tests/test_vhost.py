@@ -836,3 +836,59 @@ class TestVhostIostat:
         assert a.num_read_ops >= b.num_read_ops + 1
         assert a.bytes_written >= b.bytes_written + 4 * BLOCK
         assert a.num_unmap_ops >= b.num_unmap_ops + 1
+
+
+class TestBlkChainFuzz:
+    """Random descriptor chains at a virtio-blk ring: every chain gets
+    a used entry (or is rejected) and the daemon survives."""
+
+    def test_random_chains(self, hipstored, tmp_path):  # noqa: F811
+        import random as randmod
+
+        from vhost_client import (DESC_NEXT, DESC_WRITE, GPA_BASE,
+                                  REQ_OFF)
+        with hipstore.Client(hipstored.socket_path) as client:
+            hipstore.construct_malloc_bdev(
+                client, num_blocks=2048, block_size=512, name="bfz0")
+            client.invoke("construct_vhost_blk_controller",
+                          {"ctrlr": "vbfz", "dev_name": "bfz0"})
+            master = VhostUserMaster(
+                os.path.join(os.path.dirname(hipstored.socket_path),
+                             "vbfz"), queue=0)
+            master.negotiate()
+            rng = randmod.Random(0xB10C)
+            try:
+                for trial in range(200):
+                    nchain = rng.randint(1, 5)
+                    for slot in range(nchain):
+                        gpa = GPA_BASE + REQ_OFF + rng.randrange(0, 0x20000, 4)
+                        length = rng.choice([0, 1, 15, 16, 17, 512, 4096])
+                        flags = (DESC_NEXT if slot < nchain - 1 else 0)
+                        if rng.random() < 0.5:
+                            flags |= DESC_WRITE
+                        if rng.random() < 0.05:
+                            gpa = 0xDEAD00000000  # bad GPA
+                        master._write_desc(slot, gpa, length, flags,
+                                           slot + 1 if slot < nchain - 1
+                                           else 0)
+                    master.mem[REQ_OFF:REQ_OFF + 16] = (
+                        rng.randbytes(16) if rng.random() < 0.5 else
+                        struct.pack("<IIQ", rng.randint(0, 15), 0,
+                                    rng.randint(0, 4096)))
+                    master._submit(0)
+                    master._wait_used()
+                # healthy afterwards
+                header = struct.pack("<IIQ", 0, 0, 0)
+                master.mem[REQ_OFF:REQ_OFF + 16] = header
+                master._write_desc(0, GPA_BASE + REQ_OFF, 16, DESC_NEXT, 1)
+                master._write_desc(1, GPA_BASE + REQ_OFF + 0x1000, 512,
+                                   DESC_WRITE | DESC_NEXT, 2)
+                master._write_desc(2, GPA_BASE + REQ_OFF + 0x2000, 1,
+                                   DESC_WRITE)
+                master._submit(0)
+                master._wait_used()
+                assert master.mem[REQ_OFF + 0x2000] == 0  # clean read
+            finally:
+                master.close()
+            client.invoke("remove_vhost_controller", {"ctrlr": "vbfz"})
+            assert isinstance(client.invoke("get_rpc_methods"), list)
