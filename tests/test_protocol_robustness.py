@@ -312,3 +312,53 @@ class TestAstralCharacters:
             assert b"parse error" in reply or reply == b""
         finally:
             sock.close()
+
+
+class TestNvmfPostHandshakeFuzz:
+    """Valid ICReq, then random framed PDUs into the command parser
+    (the deepest reachable state without a full Fabrics connect)."""
+
+    def test_post_handshake_pdus(self):
+        import struct as structmod
+
+        from hypothesis import HealthCheck, given, settings
+        from hypothesis import strategies as st
+
+        backing = hs.create_malloc_bdev("phfz-ns", 512, 2048)
+        target = hs.start_nvmf_tcp_target("", 0, "nqn.phfz", True)
+        target.add_namespace(backing)
+
+        def icreq():
+            ch = structmod.pack("<BBBBI", 0x00, 0, 128, 0, 128)
+            return ch + structmod.pack("<HBBI", 0, 0, 0, 4) + bytes(112)
+
+        try:
+            @settings(max_examples=80, deadline=None,
+                      suppress_health_check=[
+                          HealthCheck.function_scoped_fixture])
+            @given(st.lists(st.tuples(st.integers(0, 10),
+                                      st.binary(max_size=200)),
+                            min_size=1, max_size=3))
+            def fuzz(pdus):
+                with socket.create_connection(("127.0.0.1", target.port),
+                                              timeout=3) as sock:
+                    sock.sendall(icreq())
+                    try:
+                        sock.recv(128)
+                        for pdu_type, payload in pdus:
+                            plen = 8 + len(payload)
+                            sock.sendall(structmod.pack(
+                                "<BBBBI", pdu_type, 0, 8, 0, plen) + payload)
+                        sock.shutdown(socket.SHUT_WR)
+                        while sock.recv(4096):
+                            pass
+                    except OSError:
+                        pass
+
+            fuzz()
+            bdev = hs.create_nvmf_tcp_bdev("phfz-init", "127.0.0.1",
+                                           target.port, "nqn.phfz")
+            bdev.write(0, b"\x5a" * 512)
+            assert bdev.read(0, 512) == b"\x5a" * 512
+        finally:
+            target.stop()
