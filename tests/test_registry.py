@@ -394,3 +394,42 @@ class TestProxyGarbagePayload:
             ctrl_server.stop()
             reg_server.stop()
             daemon.stop()
+
+
+class TestPathPropertyFuzz:
+    def test_setvalue_path_properties(self, tmp_path):
+        """Any path either stores + round-trips through GetValues or is
+        rejected INVALID_ARGUMENT; the registry survives all of them."""
+        from hypothesis import HealthCheck, given, settings
+        from hypothesis import strategies as st
+
+        from oim_amd.common.server import grpc_target
+
+        registry = Registry(db=MemRegistryDB())
+        server = RegistryServer(f"unix://{tmp_path}/pf.sock", registry)
+        server.start()
+        channel = grpc.insecure_channel(grpc_target(server.addr()))
+        stub = spec.RegistryStub(channel)
+        try:
+            @settings(max_examples=150, deadline=None,
+                      suppress_health_check=[
+                          HealthCheck.function_scoped_fixture])
+            @given(st.text(max_size=40))
+            def fuzz(path):
+                try:
+                    stub.SetValue(spec.SetValueRequest(
+                        value=spec.Value(path=path, value="v")), timeout=10)
+                except grpc.RpcError as exc:
+                    assert exc.code() == grpc.StatusCode.INVALID_ARGUMENT
+                    return
+                reply = stub.GetValues(spec.GetValuesRequest(path=path),
+                                       timeout=10)
+                assert any(v.path == path and v.value == "v"
+                           for v in reply.values), path
+                stub.SetValue(spec.SetValueRequest(
+                    value=spec.Value(path=path, value="")), timeout=10)
+
+            fuzz()
+        finally:
+            channel.close()
+            server.stop()
