@@ -15,6 +15,10 @@ class RegistryPathError(ValueError):
 
 
 def split_registry_path(path: str) -> List[str]:
+    # Lenient like the reference: empty elements (leading/trailing/
+    # doubled slashes) are dropped, so "/a//b/" normalizes to a/b.
+    # Callers get the CANONICAL path back from GetValues — paths are
+    # identifiers up to normalization, not byte strings.
     elements = [e for e in path.split("/") if e != ""]
     for element in elements:
         if element in (".", ".."):

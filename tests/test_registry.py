@@ -414,6 +414,8 @@ class TestPathPropertyFuzz:
             @settings(max_examples=150, deadline=None,
                       suppress_health_check=[
                           HealthCheck.function_scoped_fixture])
+            from oim_amd.common import split_registry_path
+
             @given(st.text(max_size=40))
             def fuzz(path):
                 try:
@@ -422,10 +424,13 @@ class TestPathPropertyFuzz:
                 except grpc.RpcError as exc:
                     assert exc.code() == grpc.StatusCode.INVALID_ARGUMENT
                     return
+                # Accepted paths round-trip in CANONICAL form (empty
+                # elements normalized away, like the reference).
+                canonical = "/".join(split_registry_path(path))
                 reply = stub.GetValues(spec.GetValuesRequest(path=path),
                                        timeout=10)
-                assert any(v.path == path and v.value == "v"
-                           for v in reply.values), path
+                assert any(v.path == canonical and v.value == "v"
+                           for v in reply.values), (path, canonical)
                 stub.SetValue(spec.SetValueRequest(
                     value=spec.Value(path=path, value="")), timeout=10)
 
