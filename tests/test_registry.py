@@ -411,11 +411,11 @@ class TestPathPropertyFuzz:
         channel = grpc.insecure_channel(grpc_target(server.addr()))
         stub = spec.RegistryStub(channel)
         try:
+            from oim_amd.common import split_registry_path
+
             @settings(max_examples=150, deadline=None,
                       suppress_health_check=[
                           HealthCheck.function_scoped_fixture])
-            from oim_amd.common import split_registry_path
-
             @given(st.text(max_size=40))
             def fuzz(path):
                 try:
