@@ -122,7 +122,11 @@ def start_stack(rank, local_rank, use_gpu, args, tmp):
             spec.MapVolumeRequest(volume_id=volume, malloc=spec.MallocParams()),
             metadata=((spec.CONTROLLER_ID_KEY, controller_id),), timeout=60)
 
-    client = hipstore.Client(daemon_sock, timeout=3600)
+    # Bounded RPC timeout: the daemon-side perf step has its own
+    # deadline (~90 s wedge cutoff), so a dead daemon surfaces as a
+    # socket timeout here instead of an unbounded hang (round 1 burned
+    # a 30-minute driver budget on exactly this invoke).
+    client = hipstore.Client(daemon_sock, timeout=180)
 
     def cleanup():
         client.close()
@@ -199,17 +203,23 @@ def main() -> int:
                 if use_gpu:
                     torch.cuda.synchronize()
 
-            for _ in range(args.warmup):
-                run_step(STEP_IOS)
+            for i in range(args.warmup):
+                r = run_step(STEP_IOS)
+                print(f"[bench r{rank}] warmup {i + 1}/{args.warmup}: "
+                      f"{r['iops']:.0f} IOPS", file=sys.stderr, flush=True)
 
             barrier()
             t0 = time.perf_counter()
             p99s = []
             ios_done = 0
-            for _ in range(args.steps):
+            for i in range(args.steps):
                 r = run_step(STEP_IOS)
                 ios_done += r["io_count"]
                 p99s.append(r["lat_p99_us"])
+                # Progress to stderr so a killed run still leaves a tail.
+                print(f"[bench r{rank}] step {i + 1}/{args.steps}: "
+                      f"{r['iops']:.0f} IOPS p99={r['lat_p99_us']:.0f}us",
+                      file=sys.stderr, flush=True)
             barrier()
             elapsed = time.perf_counter() - t0
             client.invoke("perf_session_stop", {"session_id": session})
@@ -236,7 +246,9 @@ def main() -> int:
             "metric": "4KiB_randread_IOPS",
             "value": round(value, 1),
             "unit": "IOPS",
-            "n_gpus": world_size if distributed else args.gpus,
+            # Honest GPU count: the ranks that actually ran (--gpus is
+            # only a hint; torchrun's WORLD_SIZE is ground truth).
+            "n_gpus": world_size,
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": round(elapsed * 1000 / args.steps, 3),

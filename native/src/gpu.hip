@@ -509,6 +509,51 @@ int device_count_cached() {
   return g_device_count;
 }
 
+// Every host-side wait in this file is bounded (round-1 lesson: a
+// resident-but-wedged service kernel turned two 30-minute driver
+// budgets into GPU-busy hangs). Timeouts are env-tunable; on expiry
+// the waiter FAILS the operation and, for teardown, LEAKS rings the
+// GPU may still touch instead of freeing them (freeing pinned memory
+// under a live kernel faults the device).
+double env_seconds(const char* name, double fallback) {
+  const char* env = getenv(name);
+  if (env == nullptr) return fallback;
+  const double v = atof(env);
+  return v > 0 ? v : fallback;
+}
+
+double sync_io_timeout_s() {
+  static const double s = env_seconds("HIPSTORE_SYNC_TIMEOUT_S", 30.0);
+  return s;
+}
+
+double teardown_timeout_s() {
+  static const double s = env_seconds("HIPSTORE_TEARDOWN_TIMEOUT_S", 10.0);
+  return s;
+}
+
+// Bounded stream drain: true if the stream went idle (or errored —
+// nothing left to wait for) within `seconds`.
+bool stream_drain(hipStream_t stream, double seconds) {
+  const auto deadline = std::chrono::steady_clock::now() +
+                        std::chrono::duration<double>(seconds);
+  while (true) {
+    if (hipStreamQuery(stream) != hipErrorNotReady) return true;
+    if (std::chrono::steady_clock::now() > deadline) return false;
+    std::this_thread::sleep_for(std::chrono::microseconds(200));
+  }
+}
+
+void dump_engine_stats(const char* why) {
+  fprintf(stderr,
+          "[hipstore] %s: persistent-engine stats: launches=%llu "
+          "relaunches=%llu stall_queries=%llu\n",
+          why,
+          static_cast<unsigned long long>(g_pers_launches.load()),
+          static_cast<unsigned long long>(g_pers_relaunches.load()),
+          static_cast<unsigned long long>(g_pers_stall_queries.load()));
+}
+
 }  // namespace
 
 uint64_t persistent_stat(int which) {
@@ -610,8 +655,13 @@ class HbmChannel : public HbmChannelBase {
   }
 
   ~HbmChannel() override {
-    // Drain so no kernel touches the ring after it is freed.
-    (void)hipStreamSynchronize(stream_);
+    // Drain so no kernel touches the ring after it is freed. Bounded:
+    // if a launched batch never finishes, leak the ring (a kernel may
+    // still read it) rather than hang or fault.
+    if (!stream_drain(stream_, teardown_timeout_s())) {
+      dump_engine_stats("batched-channel teardown timeout; leaking ring");
+      return;
+    }
     for (auto& batch : inflight_) (void)hipEventDestroy(batch.event);
     for (auto event : event_pool_) (void)hipEventDestroy(event);
     (void)hipHostFree(ring_);
@@ -827,13 +877,24 @@ class HbmPersistentChannel : public HbmChannelBase {
   ~HbmPersistentChannel() override {
     g_per_queue_channels[device_].fetch_sub(1, std::memory_order_relaxed);
     __atomic_store_n(const_cast<uint32_t*>(stop_), 1u, __ATOMIC_RELEASE);
-    (void)hipStreamSynchronize(stream_);
+    // A healthy service kernel sees `stop` within one poll tick; a
+    // wedged one never exits, so the drain is bounded and a timeout
+    // leaks the rings (the resident kernel may still load them).
+    if (!stream_drain(stream_, teardown_timeout_s())) {
+      dump_engine_stats(
+          "persistent-channel teardown timeout; leaking rings");
+      return;
+    }
     (void)hipHostFree(sq_);
     (void)hipHostFree(const_cast<unsigned long long*>(sq_tail_));
     (void)hipHostFree(const_cast<unsigned long long*>(cq_));
     (void)hipHostFree(const_cast<uint32_t*>(stop_));
     (void)hipFree(claim_ctr_);
     (void)hipStreamDestroy(stream_);
+    // Unfired completion callbacks (wedge teardown only): delete, never
+    // call, so their captures are released without touching torn state.
+    std::set<IoState*> leftovers(desc_io_.begin(), desc_io_.end());
+    for (IoState* state : leftovers) delete state;
   }
 
   struct IoState {
@@ -1225,20 +1286,26 @@ class HbmSharedChannel : public HbmChannelBase {
 
   ~HbmSharedChannel() override {
     // Drain so no worker holds a claim below tail, then detach.
-    const auto deadline =
-        std::chrono::steady_clock::now() + std::chrono::seconds(10);
+    const auto deadline = std::chrono::steady_clock::now() +
+                          std::chrono::duration<double>(teardown_timeout_s());
     while (completed_ < tail_ &&
            std::chrono::steady_clock::now() < deadline) {
       poll();
     }
     service_->detach(slot_);
-    (void)hipHostFree(sq_);
-    (void)hipHostFree(const_cast<unsigned long long*>(sq_tail_));
-    (void)hipHostFree(const_cast<unsigned long long*>(cq_));
     std::set<IoState*> leftovers(desc_io_.begin(), desc_io_.end());
     for (IoState* state : leftovers) {
       if (state != nullptr) delete state;  // undrained multi-tile IOs
     }
+    if (completed_ < tail_) {
+      // The drain timed out: a wedged service worker may still hold a
+      // claim into these rings — leak them rather than fault the GPU.
+      dump_engine_stats("shared-channel teardown timeout; leaking rings");
+      return;
+    }
+    (void)hipHostFree(sq_);
+    (void)hipHostFree(const_cast<unsigned long long*>(sq_tail_));
+    (void)hipHostFree(const_cast<unsigned long long*>(cq_));
   }
 
   struct IoState {
@@ -1582,11 +1649,14 @@ BdevPtr create_hbm_bdev(const std::string& name, uint64_t block_size,
   if (device < 0 || device >= gpu_device_count()) {
     throw std::runtime_error("hipstore: bad device index");
   }
-  if (persistent) {
-    // Bring the per-device shared service (and its hardware queues)
-    // up NOW, in this single-threaded moment, rather than lazily from
-    // whichever channel thread first crosses the per-queue cap — see
-    // the stream-warm note in the SharedService constructor.
+  if (persistent && getenv("HIPSTORE_EAGER_SHARED") != nullptr) {
+    // Experimental (wedge bisection only, tools/wedge_experiments.sh):
+    // bring the per-device shared service up NOW, in this
+    // single-threaded moment, rather than lazily from whichever
+    // channel thread first crosses the per-queue cap. Landed as the
+    // default once and is the prime suspect for a whole-suite hang on
+    // a fresh box (round-1 GPUTEST/BENCH rc=124), so it is opt-in
+    // until proven on hardware.
     SharedService::instance(device);
   }
   return std::make_shared<HbmBdev>(name, block_size, num_blocks, device,
@@ -1709,14 +1779,39 @@ int run_sync(Bdev* bdev, IoRequest req) {
     chunk.buffer = req.buffer == nullptr
                        ? nullptr
                        : static_cast<uint8_t*>(req.buffer) + done_bytes;
-    int result = kIoFailed;
-    bool done = false;
-    chunk.on_complete = [&](int status) {
-      result = status;
-      done = true;
+    // Heap state captured by value: on a timeout this frame returns
+    // while the I/O is still outstanding, and a late completion (or
+    // the channel teardown) must not touch dead stack slots.
+    struct SyncState {
+      std::atomic<int> result{kIoFailed};
+      std::atomic<bool> done{false};
     };
+    auto state = std::make_shared<SyncState>();
+    chunk.on_complete = [state](int status) {
+      state->result.store(status, std::memory_order_relaxed);
+      state->done.store(true, std::memory_order_release);
+    };
+    const int op_code = static_cast<int>(chunk.op);
     bdev->submit(channel.get(), std::move(chunk));
-    while (!done) bdev->poll(channel.get());
+    // Bounded wait: a wedged service kernel must fail the I/O in
+    // seconds, not hang the caller at 99% GPU busy (round-1 failure
+    // mode — both driver budgets burned on exactly this loop).
+    const auto deadline = std::chrono::steady_clock::now() +
+                          std::chrono::duration<double>(sync_io_timeout_s());
+    while (!state->done.load(std::memory_order_acquire)) {
+      bdev->poll(channel.get());
+      if (std::chrono::steady_clock::now() > deadline) {
+        fprintf(stderr,
+                "[hipstore] sync I/O timeout after %.0fs "
+                "(op=%d offset=%llu len=%llu)\n",
+                sync_io_timeout_s(), op_code,
+                static_cast<unsigned long long>(req.offset + done_bytes),
+                static_cast<unsigned long long>(total));
+        dump_engine_stats("sync I/O timeout");
+        return kIoFailed;
+      }
+    }
+    const int result = state->result.load(std::memory_order_relaxed);
     if (result != kIoOk || total == 0) return result;
     done_bytes += std::min<uint64_t>(kSyncChunk, total - done_bytes);
   }
@@ -1901,9 +1996,24 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
           ++submitted;
         }
         if (dbg) fprintf(stderr, "[hipstore-dbg] q%d submitted-initial\n", q);
+        // Stall watchdog: if in-flight I/Os stop completing for the
+        // sync-timeout window, abandon the queue (failed) instead of
+        // spinning forever — checked only on empty polls, every 1024th.
+        auto last_progress = clock::now();
+        uint32_t empty_polls = 0;
+        bool stalled = false;
         while (true) {
           if (bdev->poll(channel.get()) == 0) {
             __builtin_ia32_pause();  // spinning submitter hygiene
+            if ((++empty_polls & 0x3FF) == 0 && inflight > 0 &&
+                clock::now() - last_progress >
+                    std::chrono::duration<double>(sync_io_timeout_s())) {
+              stalled = true;
+              break;
+            }
+          } else {
+            last_progress = clock::now();
+            empty_polls = 0;
           }
           if (!stopping &&
               (clock::now() >= deadline ||
@@ -1911,6 +2021,14 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
             stopping = true;
           }
           if (stopping && inflight == 0) break;
+        }
+        if (stalled) {
+          fprintf(stderr,
+                  "[hipstore] bdevperf q%d stalled (%u in flight); "
+                  "abandoning queue\n", q, inflight);
+          dump_engine_stats("bdevperf stall");
+          failed.store(true);
+          return;  // leak buf: outstanding descriptors reference it
         }
         free_pinned(buf);
       } catch (const std::exception&) {
@@ -2040,8 +2158,36 @@ struct PerfSession::Impl {
           ++inflight;
           ++submitted;
         }
+        // Stall watchdog (checked on every 1024th empty poll): a
+        // wedged engine fails the step in seconds instead of pinning
+        // this thread — and step()'s cv.wait — forever.
+        auto last_progress = clock::now();
+        uint32_t empty_polls = 0;
+        bool stalled = false;
         while (inflight > 0) {
-          if (bdev->poll(channel.get()) == 0) __builtin_ia32_pause();
+          if (bdev->poll(channel.get()) == 0) {
+            __builtin_ia32_pause();
+            if ((++empty_polls & 0x3FF) == 0 &&
+                clock::now() - last_progress >
+                    std::chrono::duration<double>(sync_io_timeout_s())) {
+              stalled = true;
+              break;
+            }
+          } else {
+            last_progress = clock::now();
+            empty_polls = 0;
+          }
+        }
+        if (stalled) {
+          fprintf(stderr,
+                  "[hipstore] perf-session q%d stalled (%u in flight); "
+                  "failing the session\n", q, inflight);
+          dump_engine_stats("perf-session stall");
+          std::lock_guard<std::mutex> lock(mutex);
+          st.failed = true;
+          st.done_epoch = ~0ull;  // never blocks step()
+          cv.notify_all();
+          return;  // leak buf: outstanding descriptors reference it
         }
         st.ios = completed;
         {
@@ -2098,13 +2244,22 @@ PerfResult PerfSession::step(uint64_t total_ios) {
   }
   im.cv.notify_all();
   {
+    // Bounded: workers self-abort after the sync-timeout window when
+    // their queue stalls, so this only expires if a worker thread is
+    // itself stuck (e.g. inside a locked HIP runtime call).
+    const auto step_deadline =
+        std::chrono::duration<double>(2 * sync_io_timeout_s() + 30.0);
     std::unique_lock<std::mutex> lock(im.mutex);
-    im.cv.wait(lock, [&] {
+    const bool done = im.cv.wait_for(lock, step_deadline, [&] {
       for (auto& st : im.stats) {
         if (st.done_epoch < this_epoch) return false;
       }
       return true;
     });
+    if (!done) {
+      dump_engine_stats("perf-session step timeout");
+      throw std::runtime_error("perf session: step timed out (wedged queue)");
+    }
   }
   const double elapsed = std::chrono::duration<double>(clock::now() - t0).count();
 
