@@ -113,14 +113,18 @@ def start_stack(rank, local_rank, use_gpu, args, tmp):
         cap.access_mode.mode = csi.ACCESS_MODE_SINGLE_NODE_WRITER
         stub.CreateVolume(request, timeout=120)
     # MapVolume through the proxy: attach the bdev to a SCSI target of
-    # this card (completes the reference's NodeStage control path; the
-    # host-device hotplug itself needs a kernel NBD/virtio path and is
-    # exercised in tests, not in the timed bench).
-    with grpc.insecure_channel(grpc_target(reg_server.addr())) as ch:
-        stub = spec.ControllerStub(ch)
-        stub.MapVolume(
-            spec.MapVolumeRequest(volume_id=volume, malloc=spec.MallocParams()),
-            metadata=((spec.CONTROLLER_ID_KEY, controller_id),), timeout=60)
+    # this card's vhost controller (completes the reference's NodeStage
+    # control path; --frontend vhost then drives that same controller's
+    # virtqueues for the timed region). Skipped for the vhost-blk
+    # frontend, which claims the bdev with its own blk controller.
+    if not (args.frontend == "vhost" and args.personality == "blk"):
+        with grpc.insecure_channel(grpc_target(reg_server.addr())) as ch:
+            stub = spec.ControllerStub(ch)
+            stub.MapVolume(
+                spec.MapVolumeRequest(volume_id=volume,
+                                      malloc=spec.MallocParams()),
+                metadata=((spec.CONTROLLER_ID_KEY, controller_id),),
+                timeout=60)
 
     # Bounded RPC timeout: the daemon-side perf step has its own
     # deadline (~90 s wedge cutoff), so a dead daemon surfaces as a
@@ -139,7 +143,7 @@ def start_stack(rank, local_rank, use_gpu, args, tmp):
         except subprocess.TimeoutExpired:
             daemon.kill()
 
-    return client, volume, cleanup
+    return client, volume, daemon_sock, cleanup
 
 
 def main() -> int:
@@ -157,6 +161,18 @@ def main() -> int:
                         help="HBM I/O engine: batched kernel launches, "
                              "per-queue service kernels, or the shared "
                              "one-kernel-per-device service")
+    parser.add_argument("--frontend", default="daemon",
+                        choices=["daemon", "vhost"],
+                        help="timed data path: in-daemon perf session on "
+                             "the bdev queues (SPDK-bdevperf analog), or "
+                             "through the vhost-user front-end so every "
+                             "timed byte crosses the virtqueue/host "
+                             "boundary")
+    parser.add_argument("--personality", default="scsi",
+                        choices=["scsi", "blk"],
+                        help="vhost frontend personality")
+    parser.add_argument("--vhost-numjobs", type=int, default=4,
+                        help="request rings for --frontend vhost")
     args = parser.parse_args()
 
     import torch
@@ -182,20 +198,67 @@ def main() -> int:
         args.bdev_gb = min(args.bdev_gb, 1.0)
 
     with tempfile.TemporaryDirectory(prefix="oim-bench-") as tmp:
-        client, volume, cleanup = start_stack(rank, local_rank, use_gpu,
-                                              args, tmp)
+        client, volume, daemon_sock, cleanup = start_stack(
+            rank, local_rank, use_gpu, args, tmp)
         try:
-            session = client.invoke("perf_session_start", {
-                "bdev_name": volume,
-                "workload": args.workload,
-                "io_size": args.io_size,
-                "queue_depth": args.queue_depth,
-                "num_queues": args.num_queues,
-            })["session_id"]
+            session = None
+            attachment = None
+            if args.frontend == "vhost":
+                # Timed region crosses the virtqueue: an in-process
+                # vhost-user master (synthetic guest) drives request
+                # rings against the daemon's slave, so every timed
+                # byte traverses descriptor → ring worker → engine →
+                # HBM → used ring (the reference's signature flow,
+                # vhost_scsi.c process_requestq).
+                import threading
 
-            def run_step(n_ios):
-                return client.invoke("perf_session_step", {
-                    "session_id": session, "total_ios": n_ios})
+                from oim_amd.bench.vhost_harness import VhostAttachment
+                if args.personality == "scsi":
+                    # Drive the controller MapVolume attached the bdev
+                    # to — the exact device a guest would see.
+                    attachment = VhostAttachment(
+                        client, daemon_sock, volume, "vhost.0",
+                        "scsi", args.vhost_numjobs, args.queue_depth,
+                        args.io_size, args.workload, create=False)
+                else:
+                    attachment = VhostAttachment(
+                        client, daemon_sock, volume,
+                        f"vhost-bench-{rank}", "blk",
+                        args.vhost_numjobs, args.queue_depth,
+                        args.io_size, args.workload)
+
+                def run_step(n_ios):
+                    jobs = attachment.jobs
+                    per = (n_ios + len(jobs) - 1) // len(jobs)
+                    for j in jobs:
+                        j.lat_us.clear()
+                    threads = [threading.Thread(target=j.run_count,
+                                                args=(per,))
+                               for j in jobs]
+                    t0 = time.perf_counter()
+                    for t in threads:
+                        t.start()
+                    for t in threads:
+                        t.join()
+                    dt = time.perf_counter() - t0
+                    lat = sorted(x for j in jobs for x in j.lat_us)
+                    count = per * len(jobs)
+                    p99 = lat[min(len(lat) - 1, int(len(lat) * 0.99))] \
+                        if lat else 0.0
+                    return {"io_count": count, "iops": count / dt,
+                            "lat_p99_us": p99}
+            else:
+                session = client.invoke("perf_session_start", {
+                    "bdev_name": volume,
+                    "workload": args.workload,
+                    "io_size": args.io_size,
+                    "queue_depth": args.queue_depth,
+                    "num_queues": args.num_queues,
+                })["session_id"]
+
+                def run_step(n_ios):
+                    return client.invoke("perf_session_step", {
+                        "session_id": session, "total_ios": n_ios})
 
             def barrier():
                 if distributed:
@@ -222,7 +285,10 @@ def main() -> int:
                       file=sys.stderr, flush=True)
             barrier()
             elapsed = time.perf_counter() - t0
-            client.invoke("perf_session_stop", {"session_id": session})
+            if session is not None:
+                client.invoke("perf_session_stop", {"session_id": session})
+            if attachment is not None:
+                attachment.close()
         finally:
             cleanup()
 
@@ -270,6 +336,9 @@ def main() -> int:
                 "step_ios_per_gpu": STEP_IOS,
                 "backend": "hbm" if use_gpu else "cpu",
                 "engine": args.engine,
+                "frontend": args.frontend if args.frontend == "daemon"
+                else f"vhost-user-{args.personality}"
+                     f"-x{args.vhost_numjobs}rings",
                 "provisioning": "csi-createvolume+proxy-mapvolume",
                 "p99_us": round(p99, 1),
             },

@@ -122,6 +122,94 @@ class OutstandingRing:
                 self._chain(head // 3)
                 inflight += 1
 
+    def run_count(self, total_ios: int) -> None:
+        """Complete exactly `total_ios` I/Os (step-shaped runs for
+        bench.py's fixed-work steps), keeping `iodepth` outstanding."""
+        initial = min(self.iodepth, total_ios)
+        for tag in range(initial):
+            self._chain(tag)
+        submitted = initial
+        inflight = initial
+        while inflight:
+            self.ring.wait_used()
+            head = self.ring.last_used_head
+            now = time.perf_counter()
+            started = self.submit_ts.pop(head)
+            self.lat_us.append((now - started) * 1e6)
+            self.completed += 1
+            inflight -= 1
+            if submitted < total_ios:
+                self._chain(head // 3)
+                submitted += 1
+                inflight += 1
+
+
+class VhostAttachment:
+    """A vhost-user controller on `socket`'s daemon with `numjobs`
+    request rings and an OutstandingRing job per ring — the reusable
+    core shared by the time-driven CLI below and bench.py's
+    step-driven `--frontend vhost` mode."""
+
+    def __init__(self, client, socket_path: str, bdev_name: str,
+                 ctrlr: str, personality: str, numjobs: int,
+                 iodepth: int, bs: int, rw: str, create: bool = True):
+        """`create=False` attaches to an ALREADY-constructed controller
+        whose LUN 0 / target 0 is the bdev (e.g. the `vhost.0`
+        controller MapVolume set up) instead of building its own."""
+        self.client = client
+        self.ctrlr = ctrlr
+        self.created = create
+        bdev = hipstore.get_bdevs(client, bdev_name)[0]
+        if create:
+            if personality == "blk":
+                client.invoke("construct_vhost_blk_controller",
+                              {"ctrlr": ctrlr, "dev_name": bdev_name})
+            else:
+                client.invoke("construct_vhost_scsi_controller",
+                              {"ctrlr": ctrlr})
+                client.invoke("add_vhost_scsi_lun",
+                              {"ctrlr": ctrlr, "scsi_target_num": 0,
+                               "bdev_name": bdev_name})
+        vhost_path = os.path.join(os.path.dirname(socket_path), ctrlr)
+        max_jobs = 8 if personality == "blk" else 6
+        if not 1 <= numjobs <= max_jobs:
+            raise ValueError(f"numjobs must be 1..{max_jobs} "
+                             f"for {personality}")
+        first_queue = 0 if personality == "blk" else 2
+        qsize = 16
+        while qsize < iodepth * 3 + 1:
+            qsize *= 2
+        # per-ring slab: ring header + 0x100/request slot + data buffers
+        header_guess = ((16 + 2) * qsize + 0x2000 + 8 * qsize + 0xFFF) \
+            & ~0xFFF
+        slab = header_guess + iodepth * 0x100 + iodepth * bs
+        slab = (slab + 0xFFFF) & ~0xFFFF
+        mem = 0x10000 + numjobs * slab
+        self.master = VhostUserMaster(
+            vhost_path, mem_size=max(mem, 8 << 20), qsize=qsize,
+            queue=first_queue)
+        self.master.handshake()
+        self.rings = []
+        self.jobs = []
+        for i in range(numjobs):
+            base = 0x10000 + i * slab
+            ring = self.master.add_ring(first_queue + i, base, qsize)
+            req_base = base + ((ring.header_bytes + 0xFFF) & ~0xFFF)
+            data_base = req_base + iodepth * 0x100
+            self.rings.append(ring)
+            self.jobs.append(OutstandingRing(
+                ring, req_base, data_base, bdev.block_size, bs,
+                bdev.num_blocks, iodepth, rw,
+                personality=personality, seed=0x5EED + i))
+
+    def close(self) -> None:
+        for ring in self.rings:
+            ring.close()
+        self.master.close()
+        if self.created:
+            self.client.invoke("remove_vhost_controller",
+                               {"ctrlr": self.ctrlr})
+
 
 def main(argv=None) -> int:
     parser = argparse.ArgumentParser(
@@ -145,49 +233,15 @@ def main(argv=None) -> int:
     args = parser.parse_args(argv)
 
     with hipstore.Client(args.socket) as client:
-        bdevs = hipstore.get_bdevs(client, args.bdev)
-        bdev = bdevs[0]
-        if args.personality == "blk":
-            client.invoke("construct_vhost_blk_controller",
-                          {"ctrlr": args.ctrlr, "dev_name": args.bdev})
-        else:
-            client.invoke("construct_vhost_scsi_controller",
-                          {"ctrlr": args.ctrlr})
-            client.invoke("add_vhost_scsi_lun",
-                          {"ctrlr": args.ctrlr, "scsi_target_num": 0,
-                           "bdev_name": args.bdev})
-        vhost_path = os.path.join(os.path.dirname(args.socket), args.ctrlr)
-        max_jobs = 8 if args.personality == "blk" else 6
-        if not 1 <= args.numjobs <= max_jobs:
-            raise SystemExit(f"--numjobs must be 1..{max_jobs} "
-                             f"for {args.personality}")
-        first_queue = 0 if args.personality == "blk" else 2
-        qsize = 16
-        while qsize < args.iodepth * 3 + 1:
-            qsize *= 2
-        # per-ring slab: ring header + 0x100/request slot + data buffers
-        header_guess = ((16 + 2) * qsize + 0x2000 + 8 * qsize + 0xFFF) \
-            & ~0xFFF
-        slab = header_guess + args.iodepth * 0x100 + args.iodepth * args.bs
-        slab = (slab + 0xFFFF) & ~0xFFFF
-        mem = 0x10000 + args.numjobs * slab
-        master = VhostUserMaster(
-            vhost_path, mem_size=max(mem, 8 << 20), qsize=qsize,
-            queue=first_queue)
-        master.handshake()
-        rings = []
-        jobs = []
         try:
-            for i in range(args.numjobs):
-                base = 0x10000 + i * slab
-                ring = master.add_ring(first_queue + i, base, qsize)
-                req_base = base + ((ring.header_bytes + 0xFFF) & ~0xFFF)
-                data_base = req_base + args.iodepth * 0x100
-                rings.append(ring)
-                jobs.append(OutstandingRing(
-                    ring, req_base, data_base, bdev.block_size, args.bs,
-                    bdev.num_blocks, args.iodepth, args.rw,
-                    personality=args.personality, seed=0x5EED + i))
+            attachment = VhostAttachment(
+                client, args.socket, args.bdev, args.ctrlr,
+                args.personality, args.numjobs, args.iodepth, args.bs,
+                args.rw)
+        except ValueError as e:
+            raise SystemExit(str(e))
+        jobs = attachment.jobs
+        try:
             import threading
             threads = [threading.Thread(target=j.run, args=(args.runtime,))
                        for j in jobs]
@@ -198,10 +252,7 @@ def main(argv=None) -> int:
                 t.join()
             elapsed = time.perf_counter() - start
         finally:
-            for ring in rings:
-                ring.close()
-            master.close()
-            client.invoke("remove_vhost_controller", {"ctrlr": args.ctrlr})
+            attachment.close()
 
     completed = sum(j.completed for j in jobs)
     iops = completed / elapsed
