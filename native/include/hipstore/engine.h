@@ -10,6 +10,7 @@
 #pragma once
 
 #include <cstdint>
+#include <map>
 #include <string>
 #include <vector>
 
@@ -38,6 +39,15 @@ BdevPtr create_hbm_bdev(const std::string& name, uint64_t block_size,
 // Persistent-engine debug counters: 0=launches, 1=relaunches,
 // 2=stall queries.
 uint64_t persistent_stat(int which);
+
+// Hardware liveness probe for the persistent-service idioms: launches
+// a minimal leader+worker kernel pair and reports which legs of the
+// host<->GPU polling contract work on this box (heartbeats, host-tail
+// visibility, CQ publish, cross-workgroup relay). Keys: launch_err,
+// hb_host_early/final, hb_dev_early/final, relay_final, cq0_ms,
+// cq1_ms (-1 = never seen), stream_drained.
+std::map<std::string, long long> persistent_probe(int device,
+                                                  bool use_atomics);
 
 // HBM capacity of `device`: (total_bytes, free_bytes) via
 // hipMemGetInfo; (0, 0) without a GPU.

@@ -485,6 +485,82 @@ __global__ __launch_bounds__(64) void k_shared_service(SharedCtl ctl) {
   }
 }
 
+// --- liveness probe kernels -------------------------------------------------
+//
+// Minimal reproductions of the persistent-service idioms, used by
+// persistent_probe() to pinpoint WHICH leg of the host<->GPU polling
+// contract fails on a box (round-1 postmortem: the full service kernel
+// sat resident serving nothing on fresh boxes, and the failing leg
+// could not be identified from the outside):
+//   leg 1: does a resident wave run at all? (heartbeat to device mem)
+//   leg 2: GPU -> host visibility (heartbeat to pinned host mem)
+//   leg 3: host -> GPU visibility (sees the host-written tail word)
+//   leg 4: completion publish (system-release store to pinned CQ)
+//   leg 5: cross-workgroup agent-scope visibility (leader->worker)
+
+struct ProbeCtl {
+  volatile unsigned long long* sq_tail;  // pinned (device view)
+  volatile unsigned long long* hb_host;  // pinned (device view)
+  volatile unsigned long long* cq;       // pinned (device view)
+  unsigned long long* hb_dev;            // device
+  unsigned long long* relay;             // device: leader->worker word
+  uint32_t use_atomics;                  // 0 = volatile, 1 = scoped atomics
+  uint32_t max_spins;
+};
+
+__global__ __launch_bounds__(64) void k_probe(ProbeCtl ctl) {
+  const uint32_t lane = threadIdx.x & 63;
+  if (lane != 0) return;
+  if (blockIdx.x == 1) {
+    // Worker-style block: poll the leader's device-memory relay the
+    // way service workers poll known_tail, then publish to cq[1].
+    for (uint32_t i = 0; i < ctl.max_spins; ++i) {
+      unsigned long long v =
+          ctl.use_atomics
+              ? __hip_atomic_load(ctl.relay, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT)
+              : *reinterpret_cast<volatile unsigned long long*>(ctl.relay);
+      if (v != 0) {
+        __hip_atomic_store(&ctl.cq[1], v, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
+        return;
+      }
+      __builtin_amdgcn_s_sleep(8);
+    }
+    return;
+  }
+  // Leader-style block: heartbeat + host-tail poll.
+  for (uint32_t i = 0; i < ctl.max_spins; ++i) {
+    if ((i & 1023) == 0) {
+      const unsigned long long beat = i + 1;
+      __hip_atomic_store(ctl.hb_dev, beat, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      if (ctl.use_atomics) {
+        __hip_atomic_store(ctl.hb_host, beat, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
+      } else {
+        *ctl.hb_host = beat;
+      }
+    }
+    unsigned long long tail =
+        ctl.use_atomics
+            ? __hip_atomic_load(
+                  const_cast<const unsigned long long*>(ctl.sq_tail),
+                  __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM)
+            : *ctl.sq_tail;
+    if (tail != 0) {
+      // Relay to the worker block (leader->worker handoff) and
+      // publish the completion to the host.
+      __hip_atomic_store(ctl.relay, tail, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_store(&ctl.cq[0], tail, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_SYSTEM);
+      return;
+    }
+    __builtin_amdgcn_s_sleep(8);
+  }
+}
+
 int g_device_count = -1;
 
 // ROCm defaults to 4 hardware queues per process; more concurrent
@@ -495,7 +571,13 @@ int g_device_count = -1;
 // initialized HIP; hipstored sets it in main() too). Users can
 // override by exporting their own value.
 struct HwQueueEnvInit {
-  HwQueueEnvInit() { setenv("GPU_MAX_HW_QUEUES", "24", /*overwrite=*/0); }
+  HwQueueEnvInit() {
+    // HIPSTORE_NO_HWQ_INIT opts out (engine_diag.sh probes whether the
+    // raised cap itself breaks service-kernel scheduling on a box).
+    if (getenv("HIPSTORE_NO_HWQ_INIT") == nullptr) {
+      setenv("GPU_MAX_HW_QUEUES", "24", /*overwrite=*/0);
+    }
+  }
 };
 HwQueueEnvInit g_hw_queue_env_init;
 
@@ -998,6 +1080,11 @@ class HbmPersistentChannel : public HbmChannelBase {
     ctl.idle_spins = kIdleSpins;
     hipLaunchKernelGGL(k_persistent_copy, dim3(workers() + 1), dim3(64), 0,
                        stream_, ctl);
+    const hipError_t err = hipGetLastError();
+    if (err != hipSuccess) {
+      fprintf(stderr, "[hipstore] persistent service launch failed: %s\n",
+              hipGetErrorString(err));
+    }
     last_progress_ = std::chrono::steady_clock::now();
   }
 
@@ -2278,6 +2365,88 @@ PerfResult PerfSession::step(uint64_t total_ios) {
       result.io_count * static_cast<double>(im.io_size) / elapsed / 1e6;
   total.fill_result(&result);
   return result;
+}
+
+std::map<std::string, long long> persistent_probe(int device,
+                                                  bool use_atomics) {
+  std::map<std::string, long long> r;
+  r["use_atomics"] = use_atomics;
+  HIP_CHECK(hipSetDevice(device));
+  hipStream_t stream = nullptr;
+  hipStream_t util = nullptr;
+  HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+  HIP_CHECK(hipStreamCreateWithFlags(&util, hipStreamNonBlocking));
+  void* p_tail = nullptr;
+  void* p_hb = nullptr;
+  void* p_cq = nullptr;
+  HIP_CHECK(hipHostMalloc(&p_tail, 8, hipHostMallocMapped));
+  HIP_CHECK(hipHostMalloc(&p_hb, 8, hipHostMallocMapped));
+  HIP_CHECK(hipHostMalloc(&p_cq, 16, hipHostMallocMapped));
+  auto* tail = static_cast<volatile unsigned long long*>(p_tail);
+  auto* hb = static_cast<volatile unsigned long long*>(p_hb);
+  auto* cq = static_cast<volatile unsigned long long*>(p_cq);
+  *tail = 0;
+  *hb = 0;
+  cq[0] = 0;
+  cq[1] = 0;
+  unsigned long long* dev_words = nullptr;  // [0]=hb_dev [1]=relay
+  HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&dev_words), 16));
+  HIP_CHECK(hipMemset(dev_words, 0, 16));
+  HIP_CHECK(hipStreamSynchronize(nullptr));
+  ProbeCtl ctl;
+  ctl.sq_tail = device_view(const_cast<unsigned long long*>(tail));
+  ctl.hb_host = device_view(const_cast<unsigned long long*>(hb));
+  ctl.cq = device_view(const_cast<unsigned long long*>(cq));
+  ctl.hb_dev = dev_words;
+  ctl.relay = dev_words + 1;
+  ctl.use_atomics = use_atomics ? 1 : 0;
+  ctl.max_spins = 4u << 20;  // ~4-8 s of polling
+  hipLaunchKernelGGL(k_probe, dim3(2), dim3(64), 0, stream, ctl);
+  r["launch_err"] = static_cast<long long>(hipGetLastError());
+  using clock = std::chrono::steady_clock;
+  const auto t0 = clock::now();
+  std::this_thread::sleep_for(std::chrono::milliseconds(250));
+  r["hb_host_early"] =
+      static_cast<long long>(__atomic_load_n(hb, __ATOMIC_ACQUIRE));
+  unsigned long long hb_dev_early = 0;
+  (void)hipMemcpyAsync(&hb_dev_early, dev_words, 8, hipMemcpyDeviceToHost,
+                       util);
+  (void)hipStreamSynchronize(util);
+  r["hb_dev_early"] = static_cast<long long>(hb_dev_early);
+  __atomic_store_n(tail, 1ull, __ATOMIC_RELEASE);
+  r["cq0_ms"] = -1;
+  r["cq1_ms"] = -1;
+  while (clock::now() - t0 < std::chrono::seconds(5)) {
+    if (r["cq0_ms"] < 0 && __atomic_load_n(&cq[0], __ATOMIC_ACQUIRE) != 0) {
+      r["cq0_ms"] = std::chrono::duration_cast<std::chrono::milliseconds>(
+                        clock::now() - t0).count();
+    }
+    if (r["cq1_ms"] < 0 && __atomic_load_n(&cq[1], __ATOMIC_ACQUIRE) != 0) {
+      r["cq1_ms"] = std::chrono::duration_cast<std::chrono::milliseconds>(
+                        clock::now() - t0).count();
+    }
+    if (r["cq0_ms"] >= 0 && r["cq1_ms"] >= 0) break;
+    std::this_thread::sleep_for(std::chrono::milliseconds(2));
+  }
+  r["hb_host_final"] =
+      static_cast<long long>(__atomic_load_n(hb, __ATOMIC_ACQUIRE));
+  unsigned long long dev_final[2] = {0, 0};
+  (void)hipMemcpyAsync(dev_final, dev_words, 16, hipMemcpyDeviceToHost,
+                       util);
+  (void)hipStreamSynchronize(util);
+  r["hb_dev_final"] = static_cast<long long>(dev_final[0]);
+  r["relay_final"] = static_cast<long long>(dev_final[1]);
+  const bool drained = stream_drain(stream, 10.0);
+  r["stream_drained"] = drained;
+  if (drained) {
+    (void)hipHostFree(p_tail);
+    (void)hipHostFree(p_hb);
+    (void)hipHostFree(p_cq);
+    (void)hipFree(dev_words);
+    (void)hipStreamDestroy(stream);
+  }  // else: leak — a resident wave may still touch these
+  (void)hipStreamDestroy(util);
+  return r;
 }
 
 }  // namespace hipstore

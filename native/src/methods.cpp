@@ -18,6 +18,7 @@
 #include "hipstore/json.h"
 #include "hipstore/nbd.h"
 #include "hipstore/nvmf.h"
+#include "hipstore/rados.h"
 #include "hipstore/rpc.h"
 #include "hipstore/vhost.h"
 
@@ -213,11 +214,15 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   server->register_method(
       "construct_rbd_bdev",
       [&manager, use_hbm, device, record_creation](const Json& p) {
-        // Ceph RBD path. Without a reachable cluster (this environment
-        // has no network) the image is emulated by a local backing
-        // store of `config.emu_size_mb` (default 1 GiB) — the
-        // ceph-csi parameter plumbing (user_id/pool/image/monitors/
-        // secret) is exercised end-to-end either way.
+        // Ceph RBD path. With `config.mon_host` set, the bdev speaks
+        // the RADOS messenger protocol over TCP to that cluster
+        // endpoint (the in-repo loopback cluster started by
+        // rados_cluster_start, rados_cluster.cpp — this environment
+        // has no external network, so that is the reachable cluster).
+        // Without mon_host the image falls back to a local emulated
+        // store of `config.emu_size_mb` (default 1 GiB); the ceph-csi
+        // parameter plumbing (user_id/pool/image/monitors/secret) is
+        // exercised end-to-end either way.
         const std::string pool = p.get_string("pool_name");
         const std::string image = p.get_string("rbd_name");
         const int64_t block_size = p.get_int("block_size", 512);
@@ -233,10 +238,29 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
           throw RpcError{kInvalidParams, "bdev " + name + " already exists"};
         }
         int64_t emu_mb = 1024;
+        std::string mon_host;
         if (const Json* config = p.get("config")) {
           if (config->is_object()) {
             emu_mb = config->get_int("emu_size_mb", emu_mb);
+            mon_host = config->get_string("mon_host");
           }
+        }
+        if (!mon_host.empty()) {
+          // SPDK's mon_host is comma-separated; this cluster is one
+          // endpoint, so the first entry wins.
+          const size_t comma = mon_host.find(',');
+          if (comma != std::string::npos) mon_host.resize(comma);
+          BdevPtr bdev;
+          try {
+            bdev = create_rbd_bdev(
+                name, mon_host, pool, image, block_size,
+                static_cast<uint64_t>(emu_mb) << 20);
+          } catch (const std::exception& e) {
+            throw RpcError{kInvalidParams, e.what()};
+          }
+          manager.add(bdev);
+          record_creation(name, "construct_rbd_bdev", p);
+          return Json(name);
         }
         const uint64_t num_blocks = emu_mb * 1024 * 1024 / block_size;
         BdevPtr inner;
@@ -606,6 +630,57 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
           throw RpcError{kInternalError, e.what()};
         }
         return Json(name);
+      });
+
+  // --- loopback RADOS cluster (BASELINE config 4) ------------------------
+  // An in-process fake Ceph cluster (msgr-v1 endpoint playing mon+osd,
+  // rados_cluster.cpp) so construct_rbd_bdev's mon_host path has a
+  // reachable peer in this no-network environment. Objects live in an
+  // HBM arena on GPU boxes; data CRCs are GPU-computed there.
+  struct RadosClusters {
+    std::mutex mutex;
+    std::map<uint16_t, std::shared_ptr<RadosCluster>> by_port;
+  };
+  auto rados_clusters = std::make_shared<RadosClusters>();
+
+  server->register_method(
+      "rados_cluster_start",
+      [rados_clusters, use_hbm, device](const Json& p) {
+        const int64_t arena_mb = p.get_int("arena_mb", 512);
+        const int64_t object_mb = p.get_int("object_mb", 4);
+        if (arena_mb <= 0 || object_mb <= 0 || object_mb > arena_mb) {
+          throw RpcError{kInvalidParams, "bad arena/object size"};
+        }
+        std::shared_ptr<RadosCluster> cluster;
+        try {
+          cluster = start_rados_cluster(
+              static_cast<uint16_t>(p.get_int("port", 0)),
+              static_cast<uint64_t>(arena_mb),
+              use_hbm && p.get_bool("use_hbm", true), device,
+              static_cast<uint64_t>(object_mb) << 20);
+        } catch (const std::exception& e) {
+          throw RpcError{kInternalError, e.what()};
+        }
+        std::lock_guard<std::mutex> lock(rados_clusters->mutex);
+        rados_clusters->by_port[cluster->port()] = cluster;
+        JsonObject o;
+        o["port"] = Json(static_cast<int64_t>(cluster->port()));
+        o["mon_host"] =
+            Json("127.0.0.1:" + std::to_string(cluster->port()));
+        return Json(std::move(o));
+      });
+
+  server->register_method(
+      "rados_cluster_stop", [rados_clusters](const Json& p) {
+        std::lock_guard<std::mutex> lock(rados_clusters->mutex);
+        auto it = rados_clusters->by_port.find(
+            static_cast<uint16_t>(p.get_int("port")));
+        if (it == rados_clusters->by_port.end()) {
+          not_found("rados cluster");
+        }
+        it->second->stop();
+        rados_clusters->by_port.erase(it);
+        return Json(JsonObject{});
       });
 
   // --- composite bdevs (BASELINE config 5) -------------------------------

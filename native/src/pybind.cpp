@@ -17,6 +17,7 @@
 #include "hipstore/crc32c.h"
 #include "hipstore/engine.h"
 #include "hipstore/nvmf.h"
+#include "hipstore/rados.h"
 
 namespace py = pybind11;
 using namespace hipstore;
@@ -108,6 +109,20 @@ PYBIND11_MODULE(_hipstore, m) {
   m.def("create_replicated_bdev", &create_replicated_bdev, py::arg("name"),
         py::arg("children"));
 
+  py::class_<RadosCluster, std::shared_ptr<RadosCluster>>(m, "RadosCluster")
+      .def("port", &RadosCluster::port)
+      .def("object_count", &RadosCluster::object_count)
+      .def("stop", &RadosCluster::stop,
+           py::call_guard<py::gil_scoped_release>());
+  m.def("start_rados_cluster", &start_rados_cluster, py::arg("port") = 0,
+        py::arg("arena_mb") = 512, py::arg("use_hbm") = false,
+        py::arg("device") = 0, py::arg("object_bytes") = 4ull << 20);
+  m.def("create_rbd_bdev", &create_rbd_bdev, py::arg("name"),
+        py::arg("mon_host"), py::arg("pool"), py::arg("image"),
+        py::arg("block_size") = 512,
+        py::arg("default_size_bytes") = 64ull << 20,
+        py::arg("object_bytes") = 4ull << 20);
+
   py::class_<NvmfTcpTarget, std::shared_ptr<NvmfTcpTarget>>(m, "NvmfTcpTarget")
       .def_property_readonly("port", &NvmfTcpTarget::port)
       .def("add_namespace", &NvmfTcpTarget::add_namespace, py::arg("bdev"))
@@ -175,6 +190,10 @@ PYBIND11_MODULE(_hipstore, m) {
      py::arg("dst_offset"), py::arg("length"),
      "Device-side range copy between HBM bdevs (LDS-staged kernel "
      "same-device, xGMI peer copy cross-device)");
+
+  m.def("persistent_probe", &persistent_probe, py::arg("device") = 0,
+        py::arg("use_atomics") = true,
+        py::call_guard<py::gil_scoped_release>());
 
   m.def("persistent_stats", [] {
     py::dict d;
