@@ -46,8 +46,9 @@ uint64_t persistent_stat(int which);
 // visibility, CQ publish, cross-workgroup relay). Keys: launch_err,
 // hb_host_early/final, hb_dev_early/final, relay_final, cq0_ms,
 // cq1_ms (-1 = never seen), stream_drained.
-std::map<std::string, long long> persistent_probe(int device,
-                                                  bool use_atomics);
+// flags: 1 atomics, 2 pre-launch pageable H2D memcpyAsync on the
+// kernel's stream, 4 worker agent fetch_adds, 8 service-sized grid.
+std::map<std::string, long long> persistent_probe(int device, int flags);
 
 // HBM capacity of `device`: (total_bytes, free_bytes) via
 // hipMemGetInfo; (0, 0) without a GPU.

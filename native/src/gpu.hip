@@ -305,6 +305,20 @@ __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
 // Debug counters for the persistent engine (read via pybind
 // persistent_stats(); helps separate relaunch thrash from scheduling
 // issues on multi-queue collapses).
+// Writes the persistent channel's three device control words
+// ([claim, known_tail, exit_flag]) from kernel arguments — the
+// launch-path alternative to a pageable-source hipMemcpyAsync on the
+// service stream (engine_diag: candidate root cause of the fresh-box
+// service stall).
+__global__ void k_init_ctl(unsigned long long* dst, unsigned long long a,
+                           unsigned long long b, unsigned long long c) {
+  if (threadIdx.x == 0) {
+    dst[0] = a;
+    dst[1] = b;
+    dst[2] = c;
+  }
+}
+
 std::atomic<uint64_t> g_pers_launches{0};
 std::atomic<uint64_t> g_pers_relaunches{0};
 std::atomic<uint64_t> g_pers_stall_queries{0};
@@ -504,17 +518,26 @@ struct ProbeCtl {
   volatile unsigned long long* cq;       // pinned (device view)
   unsigned long long* hb_dev;            // device
   unsigned long long* relay;             // device: leader->worker word
+  unsigned long long* rmw_counter;       // device: worker fetch_add target
   uint32_t use_atomics;                  // 0 = volatile, 1 = scoped atomics
+  uint32_t do_rmw;                       // worker does agent fetch_adds
   uint32_t max_spins;
 };
 
 __global__ __launch_bounds__(64) void k_probe(ProbeCtl ctl) {
+  __shared__ __attribute__((aligned(16))) uint8_t lds_raw[kTileBytes];
   const uint32_t lane = threadIdx.x & 63;
+  (void)lds_raw;
   if (lane != 0) return;
-  if (blockIdx.x == 1) {
+  if (blockIdx.x != 0) {
     // Worker-style block: poll the leader's device-memory relay the
     // way service workers poll known_tail, then publish to cq[1].
     for (uint32_t i = 0; i < ctl.max_spins; ++i) {
+      if (ctl.do_rmw) {
+        (void)__hip_atomic_fetch_add(ctl.rmw_counter, 1ull,
+                                     __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+      }
       unsigned long long v =
           ctl.use_atomics
               ? __hip_atomic_load(ctl.relay, __ATOMIC_RELAXED,
@@ -1065,9 +1088,23 @@ class HbmPersistentChannel : public HbmChannelBase {
     g_pers_launches.fetch_add(1, std::memory_order_relaxed);
     (void)hipSetDevice(device_);
     // [claim, known_tail, exit_flag] reset to the completed prefix.
-    const unsigned long long init[3] = {completed_, completed_, 0};
-    HIP_CHECK(hipMemcpyAsync(claim_ctr_, init, 24, hipMemcpyHostToDevice,
-                             stream_));
+    // Default: a one-thread init kernel carries the values as kernel
+    // arguments. The previous pageable-source hipMemcpyAsync on this
+    // stream is kept behind HIPSTORE_LAUNCH_INIT=memcpy for A/B runs
+    // (engine_diag found the service stream never advancing past it
+    // on fresh boxes).
+    static const bool use_memcpy = [] {
+      const char* env = getenv("HIPSTORE_LAUNCH_INIT");
+      return env != nullptr && strcmp(env, "memcpy") == 0;
+    }();
+    if (use_memcpy) {
+      const unsigned long long init[3] = {completed_, completed_, 0};
+      HIP_CHECK(hipMemcpyAsync(claim_ctr_, init, 24, hipMemcpyHostToDevice,
+                               stream_));
+    } else {
+      hipLaunchKernelGGL(k_init_ctl, dim3(1), dim3(1), 0, stream_,
+                         claim_ctr_, completed_, completed_, 0ull);
+    }
     PersistentCtl ctl;
     ctl.sq = device_view(sq_);
     ctl.sq_tail = device_view(const_cast<unsigned long long*>(sq_tail_));
@@ -2367,10 +2404,18 @@ PerfResult PerfSession::step(uint64_t total_ios) {
   return result;
 }
 
-std::map<std::string, long long> persistent_probe(int device,
-                                                  bool use_atomics) {
+std::map<std::string, long long> persistent_probe(int device, int flags) {
+  // flags: 1 = scoped atomics (else volatile), 2 = enqueue a
+  // pageable-source hipMemcpyAsync on the KERNEL\'S stream right
+  // before the launch (the service engine\'s launch() pattern),
+  // 4 = workers hammer an agent-scope fetch_add, 8 = 17 workgroups
+  // (the default service grid) instead of 2.
   std::map<std::string, long long> r;
-  r["use_atomics"] = use_atomics;
+  const bool use_atomics = flags & 1;
+  const bool pre_memcpy = flags & 2;
+  const bool do_rmw = flags & 4;
+  const bool big_grid = flags & 8;
+  r["flags"] = flags;
   HIP_CHECK(hipSetDevice(device));
   hipStream_t stream = nullptr;
   hipStream_t util = nullptr;
@@ -2389,9 +2434,9 @@ std::map<std::string, long long> persistent_probe(int device,
   *hb = 0;
   cq[0] = 0;
   cq[1] = 0;
-  unsigned long long* dev_words = nullptr;  // [0]=hb_dev [1]=relay
-  HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&dev_words), 16));
-  HIP_CHECK(hipMemset(dev_words, 0, 16));
+  unsigned long long* dev_words = nullptr;  // [0]=hb_dev [1]=relay [2]=rmw
+  HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&dev_words), 24));
+  HIP_CHECK(hipMemset(dev_words, 0, 24));
   HIP_CHECK(hipStreamSynchronize(nullptr));
   ProbeCtl ctl;
   ctl.sq_tail = device_view(const_cast<unsigned long long*>(tail));
@@ -2399,9 +2444,19 @@ std::map<std::string, long long> persistent_probe(int device,
   ctl.cq = device_view(const_cast<unsigned long long*>(cq));
   ctl.hb_dev = dev_words;
   ctl.relay = dev_words + 1;
+  ctl.rmw_counter = dev_words + 2;
   ctl.use_atomics = use_atomics ? 1 : 0;
+  ctl.do_rmw = do_rmw ? 1 : 0;
   ctl.max_spins = 4u << 20;  // ~4-8 s of polling
-  hipLaunchKernelGGL(k_probe, dim3(2), dim3(64), 0, stream, ctl);
+  if (pre_memcpy) {
+    // The service launch() idiom under test: small H2D from a STACK
+    // array, asynchronously, on the same stream the kernel follows on.
+    const unsigned long long init[3] = {0, 0, 0};
+    HIP_CHECK(hipMemcpyAsync(dev_words, init, 24, hipMemcpyHostToDevice,
+                             stream));
+  }
+  hipLaunchKernelGGL(k_probe, dim3(big_grid ? 17 : 2), dim3(64), 0, stream,
+                     ctl);
   r["launch_err"] = static_cast<long long>(hipGetLastError());
   using clock = std::chrono::steady_clock;
   const auto t0 = clock::now();
@@ -2430,12 +2485,13 @@ std::map<std::string, long long> persistent_probe(int device,
   }
   r["hb_host_final"] =
       static_cast<long long>(__atomic_load_n(hb, __ATOMIC_ACQUIRE));
-  unsigned long long dev_final[2] = {0, 0};
-  (void)hipMemcpyAsync(dev_final, dev_words, 16, hipMemcpyDeviceToHost,
+  unsigned long long dev_final[3] = {0, 0, 0};
+  (void)hipMemcpyAsync(dev_final, dev_words, 24, hipMemcpyDeviceToHost,
                        util);
   (void)hipStreamSynchronize(util);
   r["hb_dev_final"] = static_cast<long long>(dev_final[0]);
   r["relay_final"] = static_cast<long long>(dev_final[1]);
+  r["rmw_final"] = static_cast<long long>(dev_final[2]);
   const bool drained = stream_drain(stream, 10.0);
   r["stream_drained"] = drained;
   if (drained) {

@@ -192,7 +192,7 @@ PYBIND11_MODULE(_hipstore, m) {
      "same-device, xGMI peer copy cross-device)");
 
   m.def("persistent_probe", &persistent_probe, py::arg("device") = 0,
-        py::arg("use_atomics") = true,
+        py::arg("flags") = 1,
         py::call_guard<py::gil_scoped_release>());
 
   m.def("persistent_stats", [] {

@@ -569,16 +569,26 @@ class RbdBdev : public Bdev {
     if (it == channel->inflight.end()) return 0;
     RbdInflight entry = it->second;
     channel->inflight.erase(it);
+    static const bool debug = getenv("HIPSTORE_RADOS_DEBUG") != nullptr;
     int status = kIoOk;
     OsdOpReply reply;
     try {
       if (f.front_crc != crc32c_sw(0, front.data(), front.size()) ||
           (h.data_len && f.data_crc != crc32c_sw(0, data, h.data_len))) {
+        if (debug) {
+          fprintf(stderr, "[rados-client] reply CRC mismatch tid=%llu\n",
+                  static_cast<unsigned long long>(h.tid));
+        }
         status = kIoFailed;
       } else {
         reply = decode_osd_op_reply_front(front);
+        if (debug && reply.result < 0) {
+          fprintf(stderr, "[rados-client] tid=%llu result=%d\n",
+                  static_cast<unsigned long long>(h.tid), reply.result);
+        }
       }
-    } catch (const std::exception&) {
+    } catch (const std::exception& e) {
+      if (debug) fprintf(stderr, "[rados-client] reply decode: %s\n", e.what());
       status = kIoFailed;
     }
     if (status == kIoOk) {

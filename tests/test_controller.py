@@ -10,7 +10,7 @@ import time
 import grpc
 import pytest
 
-from oim_amd import spec
+from oim_amd import hipstore, spec
 from oim_amd.common.server import grpc_target
 from oim_amd.controller import Controller, ControllerServer
 from oim_amd.registry import MemRegistryDB, Registry, RegistryServer
@@ -125,11 +125,17 @@ class TestMapUnmap:
             FakeContext())
         assert ra.scsi_disk.target != rb.scsi_disk.target
 
-    def test_ceph_map_creates_and_unmap_deletes(self, controller):
+    def test_ceph_map_creates_and_unmap_deletes(self, controller, hipstored):  # noqa: F811
+        # Monitors present => the daemon speaks the RADOS wire protocol
+        # to that endpoint (round 2); stand up the loopback cluster the
+        # way a real deployment has a reachable Ceph cluster.
+        with hipstore.Client(hipstored.socket_path) as client:
+            info = client.invoke("rados_cluster_start",
+                                 {"arena_mb": 16, "object_mb": 1})
         req = spec.MapVolumeRequest(
             volume_id="ceph-vol",
             ceph=spec.CephParams(user_id="admin", secret="k",
-                                 monitors="1.2.3.4:6789", pool="rbd",
+                                 monitors=info["mon_host"], pool="rbd",
                                  image="img"),
         )
         reply = controller.MapVolume(req, FakeContext())
