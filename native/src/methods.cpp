@@ -238,23 +238,31 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
           throw RpcError{kInvalidParams, "bdev " + name + " already exists"};
         }
         int64_t emu_mb = 1024;
+        int64_t object_mb = 4;  // RBD default object size (order 22)
         std::string mon_host;
         if (const Json* config = p.get("config")) {
           if (config->is_object()) {
             emu_mb = config->get_int("emu_size_mb", emu_mb);
+            object_mb = config->get_int("object_mb", object_mb);
             mon_host = config->get_string("mon_host");
           }
         }
+        if (object_mb <= 0 || object_mb > 64) {
+          throw RpcError{kInvalidParams, "bad object_mb"};
+        }
         if (!mon_host.empty()) {
           // SPDK's mon_host is comma-separated; this cluster is one
-          // endpoint, so the first entry wins.
+          // endpoint, so the first entry wins. object_mb must not
+          // exceed the cluster's slot granularity
+          // (rados_cluster_start object_mb).
           const size_t comma = mon_host.find(',');
           if (comma != std::string::npos) mon_host.resize(comma);
           BdevPtr bdev;
           try {
             bdev = create_rbd_bdev(
                 name, mon_host, pool, image, block_size,
-                static_cast<uint64_t>(emu_mb) << 20);
+                static_cast<uint64_t>(emu_mb) << 20,
+                static_cast<uint64_t>(object_mb) << 20);
           } catch (const std::exception& e) {
             throw RpcError{kInvalidParams, e.what()};
           }

@@ -49,11 +49,15 @@ def launch_hipstored(tmp_path, cpu: bool = True, device: int = 0) -> HipstoredFi
         cmd.append("-C")
     env = dict(os.environ)
     env.setdefault("GPU_MAX_HW_QUEUES", "24")
-    process = subprocess.Popen(cmd, stderr=subprocess.PIPE, env=env)
+    # stderr goes to a file, not a PIPE: an undrained pipe back-
+    # pressures the daemon after 64 KiB of diagnostics and wedges it.
+    stderr_path = str(tmp_path / "hipstored.stderr")
+    stderr_file = open(stderr_path, "w")  # noqa: SIM115 (daemon lifetime)
+    process = subprocess.Popen(cmd, stderr=stderr_file, env=env)
     deadline = time.time() + 30
     while not os.path.exists(socket_path):
         if process.poll() is not None:
-            err = process.stderr.read().decode()
+            err = open(stderr_path).read()
             pytest.fail(f"hipstored exited early: {err}")
         if time.time() > deadline:
             process.kill()

@@ -56,9 +56,12 @@ class TestBdevs:
         assert excinfo.value.is_not_found()
 
     def test_rbd_emulation(self, client):
+        # No monitors => local emulated store (round 2: with mon_host
+        # set the daemon speaks the RADOS wire protocol instead, and
+        # an unreachable cluster is an error — tests/test_rados.py).
         name = hipstore.construct_rbd_bdev(
             client, pool_name="rbd", rbd_name="img0", block_size=512,
-            user_id="admin", config={"mon_host": "127.0.0.1:6789", "key": "x"},
+            user_id="admin", config={"key": "x"},
         )
         bdevs = hipstore.get_bdevs(client, name=name)
         assert bdevs[0].product_name == "Ceph Rbd Disk"

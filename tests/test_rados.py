@@ -190,7 +190,8 @@ class TestRbdDaemonRpc:
                     client, pool_name="rbd", rbd_name="vol1",
                     block_size=512, name="ceph-vol1", user_id="admin",
                     config={"mon_host": info["mon_host"],
-                            "key": "secret", "emu_size_mb": 8})
+                            "key": "secret", "emu_size_mb": 8,
+                            "object_mb": 1})
                 assert name == "ceph-vol1"
                 bdev = hipstore.get_bdevs(client, "ceph-vol1")[0]
                 assert bdev.product_name == "Ceph Rbd Disk"
