@@ -50,6 +50,12 @@ uint64_t persistent_stat(int which);
 // kernel's stream, 4 worker agent fetch_adds, 8 service-sized grid.
 std::map<std::string, long long> persistent_probe(int device, int flags);
 
+// Launches the real persistent-copy service kernel with a minimal
+// standalone setup and serves one descriptor (0 = pinned->pinned,
+// 1 = pinned->HBM, 2 = HBM->pinned).
+std::map<std::string, long long> persistent_kernel_probe(int device,
+                                                         int variant);
+
 // HBM capacity of `device`: (total_bytes, free_bytes) via
 // hipMemGetInfo; (0, 0) without a GPU.
 std::pair<uint64_t, uint64_t> hbm_info(int device);

@@ -1071,6 +1071,7 @@ class HbmPersistentChannel : public HbmChannelBase {
           g_pers_relaunches.fetch_add(1, std::memory_order_relaxed);
           launch();
         }
+        debug_dump_state();
         last_progress_ = now;  // gate the query itself to 1/ms
       }
     }
@@ -1171,6 +1172,37 @@ class HbmPersistentChannel : public HbmChannelBase {
     }
     return static_cast<uint8_t*>(dev);
   }
+
+  // HIPSTORE_DEBUG: once a second during a stall, read the device
+  // control words back on a utility stream and print them — shows
+  // whether the leader mirrored the tail and how far workers claimed.
+  void debug_dump_state() {
+    static const bool debug = getenv("HIPSTORE_DEBUG") != nullptr;
+    if (!debug) return;
+    const auto now = std::chrono::steady_clock::now();
+    if (now - dbg_last_ < std::chrono::seconds(1)) return;
+    dbg_last_ = now;
+    if (dbg_stream_ == nullptr) {
+      if (hipStreamCreateWithFlags(&dbg_stream_, hipStreamNonBlocking) !=
+          hipSuccess) {
+        return;
+      }
+    }
+    unsigned long long words[3] = {0, 0, 0};
+    (void)hipMemcpyAsync(words, claim_ctr_, 24, hipMemcpyDeviceToHost,
+                         dbg_stream_);
+    (void)hipStreamSynchronize(dbg_stream_);
+    fprintf(stderr,
+            "[hipstore-dbg] perq tail=%llu done=%llu sq_tail=%llu "
+            "dev_claim=%llu dev_known=%llu dev_exit=%llu busy=%d\n",
+            static_cast<unsigned long long>(tail_),
+            static_cast<unsigned long long>(completed_), *sq_tail_,
+            words[0], words[1], words[2],
+            hipStreamQuery(stream_) == hipErrorNotReady);
+  }
+
+  std::chrono::steady_clock::time_point dbg_last_{};
+  hipStream_t dbg_stream_ = nullptr;
 
   uint8_t* base_;
   int device_;
@@ -2504,5 +2536,122 @@ std::map<std::string, long long> persistent_probe(int device, int flags) {
   (void)hipStreamDestroy(util);
   return r;
 }
+
+
+std::map<std::string, long long> persistent_kernel_probe(int device,
+                                                         int variant) {
+  // Launches the REAL k_persistent_copy with a minimal hand-built
+  // setup (same allocation sizes and ctl wiring as
+  // HbmPersistentChannel::launch), serving exactly one descriptor:
+  // variant 0 = pinned->pinned 64 B, 1 = pinned->HBM 4 KiB,
+  // 2 = HBM->pinned 4 KiB. Separates "the kernel is broken" from
+  // "the channel setup is broken".
+  constexpr uint32_t kProbeRing = 32768;
+  std::map<std::string, long long> r;
+  r["variant"] = variant;
+  HIP_CHECK(hipSetDevice(device));
+  hipStream_t stream = nullptr;
+  hipStream_t util = nullptr;
+  HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+  HIP_CHECK(hipStreamCreateWithFlags(&util, hipStreamNonBlocking));
+  BlockDesc* sq = nullptr;
+  HIP_CHECK(hipHostMalloc(reinterpret_cast<void**>(&sq),
+                          kProbeRing * sizeof(BlockDesc),
+                          hipHostMallocMapped));
+  void* p_tail = nullptr;
+  void* p_cq = nullptr;
+  void* p_stop = nullptr;
+  void* p_buf = nullptr;
+  HIP_CHECK(hipHostMalloc(&p_tail, 8, hipHostMallocMapped));
+  HIP_CHECK(hipHostMalloc(&p_cq, kProbeRing * 8, hipHostMallocMapped));
+  HIP_CHECK(hipHostMalloc(&p_stop, 4, hipHostMallocMapped));
+  HIP_CHECK(hipHostMalloc(&p_buf, 8192, hipHostMallocMapped));
+  auto* tail = static_cast<volatile unsigned long long*>(p_tail);
+  auto* cq = static_cast<volatile unsigned long long*>(p_cq);
+  auto* stop = static_cast<volatile uint32_t*>(p_stop);
+  auto* buf = static_cast<uint8_t*>(p_buf);
+  *tail = 0;
+  *stop = 0;
+  memset(const_cast<unsigned long long*>(cq), 0, kProbeRing * 8);
+  memset(buf, 0x5A, 4096);
+  memset(buf + 4096, 0, 4096);
+  uint8_t* hbm = nullptr;
+  HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&hbm), 8192));
+  HIP_CHECK(hipMemset(hbm, 0xA7, 8192));
+  unsigned long long* claim_ctr = nullptr;
+  HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&claim_ctr), 24));
+  HIP_CHECK(hipMemset(claim_ctr, 0, 24));
+  HIP_CHECK(hipStreamSynchronize(nullptr));
+  PersistentCtl ctl;
+  ctl.sq = device_view(sq);
+  ctl.sq_tail = device_view(const_cast<unsigned long long*>(tail));
+  ctl.cq = device_view(const_cast<unsigned long long*>(cq));
+  ctl.stop = device_view(const_cast<uint32_t*>(stop));
+  ctl.claim_counter = claim_ctr;
+  ctl.known_tail = claim_ctr + 1;
+  ctl.exit_flag = reinterpret_cast<uint32_t*>(claim_ctr + 2);
+  ctl.ring_mask = kProbeRing - 1;
+  ctl.idle_spins = 500000;
+  hipLaunchKernelGGL(k_persistent_copy, dim3(17), dim3(64), 0, stream, ctl);
+  r["launch_err"] = static_cast<long long>(hipGetLastError());
+  // One descriptor.
+  uint8_t* buf_dev = device_view(buf);
+  BlockDesc& d = sq[0];
+  switch (variant) {
+    case 0:
+      d.src = buf_dev;
+      d.dst = buf_dev + 4096;
+      d.bytes = 64;
+      break;
+    case 1:
+      d.src = buf_dev;
+      d.dst = hbm;
+      d.bytes = 4096;
+      break;
+    default:
+      d.src = hbm;
+      d.dst = buf_dev + 4096;
+      d.bytes = 4096;
+      break;
+  }
+  d.fill = 0;
+  __atomic_store_n(tail, 1ull, __ATOMIC_RELEASE);
+  using clock = std::chrono::steady_clock;
+  const auto t0 = clock::now();
+  r["cq_ms"] = -1;
+  while (clock::now() - t0 < std::chrono::seconds(4)) {
+    if (__atomic_load_n(&cq[0], __ATOMIC_ACQUIRE) == 1) {
+      r["cq_ms"] = std::chrono::duration_cast<std::chrono::milliseconds>(
+                       clock::now() - t0).count();
+      break;
+    }
+    std::this_thread::sleep_for(std::chrono::milliseconds(2));
+  }
+  unsigned long long words[3] = {0, 0, 0};
+  (void)hipMemcpyAsync(words, claim_ctr, 24, hipMemcpyDeviceToHost, util);
+  (void)hipStreamSynchronize(util);
+  r["dev_claim"] = static_cast<long long>(words[0]);
+  r["dev_known"] = static_cast<long long>(words[1]);
+  r["dev_exit"] = static_cast<long long>(words[2]);
+  r["data_ok"] = variant == 0   ? memcmp(buf, buf + 4096, 64) == 0
+                 : variant == 2 ? buf[4096] == 0xA7 && buf[8191] == 0xA7
+                                : -1;
+  __atomic_store_n(stop, 1u, __ATOMIC_RELEASE);
+  const bool drained = stream_drain(stream, 8.0);
+  r["stream_drained"] = drained;
+  if (drained) {
+    (void)hipHostFree(sq);
+    (void)hipHostFree(p_tail);
+    (void)hipHostFree(p_cq);
+    (void)hipHostFree(p_stop);
+    (void)hipHostFree(p_buf);
+    (void)hipFree(hbm);
+    (void)hipFree(claim_ctr);
+    (void)hipStreamDestroy(stream);
+  }
+  (void)hipStreamDestroy(util);
+  return r;
+}
+
 
 }  // namespace hipstore

@@ -194,6 +194,9 @@ PYBIND11_MODULE(_hipstore, m) {
   m.def("persistent_probe", &persistent_probe, py::arg("device") = 0,
         py::arg("flags") = 1,
         py::call_guard<py::gil_scoped_release>());
+  m.def("persistent_kernel_probe", &persistent_kernel_probe,
+        py::arg("device") = 0, py::arg("variant") = 0,
+        py::call_guard<py::gil_scoped_release>());
 
   m.def("persistent_stats", [] {
     py::dict d;
