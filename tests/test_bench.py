@@ -53,8 +53,24 @@ class TestBenchContract:
         assert proc.returncode == 0, proc.stderr[-2000:]
         result = last_json_line(proc.stdout)
         assert result["n_gpus"] == 2
-        # aggregate over both ranks: 2 ranks x steps x STEP_IOS I/Os
+        # The reported value must be the WHOLE-JOB aggregate:
+        # value [IOPS] x elapsed [s] == 2 ranks x steps x STEP_IOS.
+        expected_ios = 2 * 2 * result["config"]["step_ios_per_gpu"]
+        measured_ios = result["value"] * result["ms_per_step"] \
+            * result["steps"] / 1000
+        assert abs(measured_ios - expected_ios) / expected_ios < 0.01
+
+    def test_vhost_frontend_single_process(self):
+        proc = subprocess.run(
+            [sys.executable, BENCH, "--steps", "1", "--warmup", "0",
+             "--frontend", "vhost", "--queue-depth", "8",
+             "--vhost-numjobs", "2", "--bdev-gb", "0.25"],
+            capture_output=True, text=True, timeout=600, cwd=REPO_ROOT)
+        assert proc.returncode == 0, proc.stderr[-2000:]
+        result = last_json_line(proc.stdout)
+        assert result["config"]["frontend"].startswith("vhost-user-scsi")
         assert result["value"] > 0
+        assert result["config"]["p99_us"] > 0
 
 
 class TestVhostHarness:
