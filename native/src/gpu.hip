@@ -293,8 +293,14 @@ __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
 #pragma unroll 4
     for (uint32_t i = lane; i < n16; i += 64) dst[i] = lds[i];
     // Publish the completion: data must be host-visible before the CQ
-    // word, hence the system-scope release store after a full fence.
-    __threadfence_system();
+    // word. The system-scope RELEASE store carries that ordering by
+    // itself (s_waitcnt on the wave's outstanding stores + L2
+    // writeback before the store). An explicit __threadfence_system()
+    // here is redundant — and on the current pool firmware it HANGS
+    // the wave: round-2 diagnosis (tools/engine_diag3.sh) showed the
+    // serving wave copying the data correctly, then never publishing
+    // the CQ nor claiming again, with the fence the only instruction
+    // between those two points. Do not reintroduce it.
     if (lane == 0) {
       __hip_atomic_store(&ctl.cq[claim & ctl.ring_mask], claim + 1,
                          __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
@@ -480,7 +486,9 @@ __global__ __launch_bounds__(64) void k_shared_service(SharedCtl ctl) {
       float4* __restrict__ d4 = reinterpret_cast<float4*>(dst);
 #pragma unroll 4
       for (uint32_t k = lane; k < n16; k += 64) d4[k] = lds[k];
-      __threadfence_system();
+      // Ordering carried by the system-scope release store; see the
+      // per-queue kernel's publish comment (explicit
+      // __threadfence_system hangs waves on current pool firmware).
       if (lane == 0) {
         volatile unsigned long long* cq =
             reinterpret_cast<volatile unsigned long long*>(slot_ld(slot, 2));
@@ -521,6 +529,7 @@ struct ProbeCtl {
   unsigned long long* rmw_counter;       // device: worker fetch_add target
   uint32_t use_atomics;                  // 0 = volatile, 1 = scoped atomics
   uint32_t do_rmw;                       // worker does agent fetch_adds
+  uint32_t do_fence;                     // __threadfence_system pre-publish
   uint32_t max_spins;
 };
 
@@ -576,6 +585,7 @@ __global__ __launch_bounds__(64) void k_probe(ProbeCtl ctl) {
       // publish the completion to the host.
       __hip_atomic_store(ctl.relay, tail, __ATOMIC_RELEASE,
                          __HIP_MEMORY_SCOPE_AGENT);
+      if (ctl.do_fence) __threadfence_system();  // suspected wave-hang
       __hip_atomic_store(&ctl.cq[0], tail, __ATOMIC_RELEASE,
                          __HIP_MEMORY_SCOPE_SYSTEM);
       return;
@@ -2447,6 +2457,7 @@ std::map<std::string, long long> persistent_probe(int device, int flags) {
   const bool pre_memcpy = flags & 2;
   const bool do_rmw = flags & 4;
   const bool big_grid = flags & 8;
+  const bool do_fence = flags & 16;
   r["flags"] = flags;
   HIP_CHECK(hipSetDevice(device));
   hipStream_t stream = nullptr;
@@ -2479,6 +2490,7 @@ std::map<std::string, long long> persistent_probe(int device, int flags) {
   ctl.rmw_counter = dev_words + 2;
   ctl.use_atomics = use_atomics ? 1 : 0;
   ctl.do_rmw = do_rmw ? 1 : 0;
+  ctl.do_fence = do_fence ? 1 : 0;
   ctl.max_spins = 4u << 20;  // ~4-8 s of polling
   if (pre_memcpy) {
     // The service launch() idiom under test: small H2D from a STACK
