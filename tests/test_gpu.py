@@ -405,3 +405,31 @@ class TestHbmResize:
         assert bdev.read(65535 * 512, 512) == tail
         # grown region arrived zeroed
         assert bdev.read(4096, 512) == b"\0" * 512
+
+
+@pytest.mark.gpu
+class TestEngineContractProbes:
+    """Hardware-contract probes (round-2 diagnosis tooling, kept as
+    regression guards): every leg of the host<->GPU polling contract,
+    and the real service kernel serving a descriptor from a minimal
+    standalone harness. A red here pinpoints WHICH leg broke (e.g. the
+    firmware __threadfence_system wave-hang this caught)."""
+
+    def test_polling_contract_legs(self):
+        r = hs.persistent_probe(0, 1)
+        assert r["launch_err"] == 0
+        assert r["hb_host_early"] > 0, "GPU->host pinned heartbeat"
+        assert r["hb_dev_early"] > 0, "kernel runs at all"
+        assert r["cq0_ms"] >= 0, "host tail -> leader -> CQ publish"
+        assert r["cq1_ms"] >= 0, "leader -> worker agent relay"
+        assert r["stream_drained"] == 1
+
+    @pytest.mark.parametrize("variant", [0, 1, 2])
+    def test_service_kernel_serves(self, variant):
+        r = hs.persistent_kernel_probe(0, variant)
+        assert r["launch_err"] == 0
+        assert r["cq_ms"] >= 0, "service kernel must publish its CQ"
+        if variant != 1:
+            assert r["data_ok"] == 1
+        assert r["dev_known"] == 1
+        assert r["stream_drained"] == 1
