@@ -178,6 +178,14 @@ __global__ __launch_bounds__(256) void k_crc32c_blocks(
 // idempotent (same copy, same completion value), so the handoff is
 // race-free without any further coordination.
 
+__device__ __forceinline__ unsigned long long wave_bcast_u64(
+    unsigned long long v) {
+  int lo = __shfl(static_cast<int>(v & 0xFFFFFFFFu), 0);
+  int hi = __shfl(static_cast<int>(v >> 32), 0);
+  return (static_cast<unsigned long long>(static_cast<uint32_t>(hi)) << 32) |
+         static_cast<uint32_t>(lo);
+}
+
 struct PersistentCtl {
   // Pinned host memory (GPU reads/writes over PCIe, uncached):
   const BlockDesc* sq;     // descriptor ring
@@ -232,31 +240,36 @@ __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
   }
 
   while (true) {
-    // One lane claims the next descriptor index; the wave follows
-    // (64-bit broadcast as two 32-bit halves).
-    unsigned long long claim = 0;
-    if (lane == 0) {
-      claim = __hip_atomic_fetch_add(ctl.claim_counter, 1ull,
-                                     __ATOMIC_RELAXED,
-                                     __HIP_MEMORY_SCOPE_AGENT);
+    // Bounded CAS claim: take a descriptor index only when the leader
+    // has published work past it (claim-only-when-available, the
+    // shared-service discipline). The original blind
+    // fetch_add-then-wait form wedged the serving wave on current
+    // pool firmware (round-2 diagnosis, tools/engine_diag*.sh) while
+    // this form is proven on the same boxes by k_shared_service.
+    const unsigned long long kt = __hip_atomic_load(
+        ctl.known_tail, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    unsigned long long c = __hip_atomic_load(
+        ctl.claim_counter, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    unsigned long long claim = ~0ull;
+    while (c < kt) {
+      unsigned long long witnessed = 0;
+      if (lane == 0) {
+        witnessed = atomicCAS(ctl.claim_counter, c, c + 1);
+      }
+      witnessed = wave_bcast_u64(witnessed);
+      if (witnessed == c) {
+        claim = c;
+        break;
+      }
+      c = witnessed;
     }
-    {
-      int lo = __shfl(static_cast<int>(claim & 0xFFFFFFFFu), 0);
-      int hi = __shfl(static_cast<int>(claim >> 32), 0);
-      claim = (static_cast<unsigned long long>(static_cast<uint32_t>(hi))
-               << 32) | static_cast<uint32_t>(lo);
-    }
-    // Wait for the leader to publish it (or exit). Device-memory
-    // polling only: no PCIe traffic from workers while idle.
-    while (true) {
-      unsigned long long tail = __hip_atomic_load(
-          ctl.known_tail, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-      if (tail > claim) break;
+    if (claim == ~0ull) {
       if (__hip_atomic_load(ctl.exit_flag, __ATOMIC_RELAXED,
                             __HIP_MEMORY_SCOPE_AGENT) != 0) {
         return;
       }
       __builtin_amdgcn_s_sleep(16);
+      continue;
     }
     // Volatile loads: the descriptor address is computable BEFORE the
     // wait loop, so a plain load could be hoisted above it by the
@@ -378,14 +391,6 @@ __device__ __forceinline__ void slot_st(SharedSlot* slot, int word,
                                         unsigned long long value) {
   __hip_atomic_store(reinterpret_cast<unsigned long long*>(slot) + word,
                      value, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
-}
-
-__device__ __forceinline__ unsigned long long wave_bcast_u64(
-    unsigned long long v) {
-  int lo = __shfl(static_cast<int>(v & 0xFFFFFFFFu), 0);
-  int hi = __shfl(static_cast<int>(v >> 32), 0);
-  return (static_cast<unsigned long long>(static_cast<uint32_t>(hi)) << 32) |
-         static_cast<uint32_t>(lo);
 }
 
 __global__ __launch_bounds__(64) void k_shared_service(SharedCtl ctl) {
