@@ -290,21 +290,36 @@ __global__ __launch_bounds__(64) void k_persistent_copy(PersistentCtl ctl) {
     d.dst = reinterpret_cast<uint8_t*>(w1);
     d.bytes = static_cast<uint32_t>(w2 & 0xFFFFFFFFu);
     d.fill = static_cast<uint32_t>(w2 >> 32);
-    const uint32_t n16 = d.bytes >> 4;
+    // Uniform u64 tile path, each lane staging and draining ITS OWN
+    // LDS slots (slot i covered by lane i%64 in both loops — no
+    // cross-lane LDS dependency, so wave-internal lgkmcnt ordering is
+    // the only sync needed). Loads are agent-scope (sc1,
+    // L2-bypassing): a RESIDENT kernel has no dispatch boundaries, so
+    // a plain load can hit a stale line in this XCD's L2 written
+    // around by another XCD's wave (caught as a read-after-write
+    // corruption by TestVhostHbm). Batched kernels don't need this —
+    // each dispatch's implicit acquire invalidates the caches.
+    const uint32_t n8 = d.bytes >> 3;
+    unsigned long long* lds64 = reinterpret_cast<unsigned long long*>(lds);
     if (d.src != nullptr) {
-      const float4* __restrict__ src = reinterpret_cast<const float4*>(d.src);
+      const unsigned long long* __restrict__ src =
+          reinterpret_cast<const unsigned long long*>(d.src);
 #pragma unroll 4
-      for (uint32_t i = lane; i < n16; i += 64) lds[i] = src[i];
+      for (uint32_t i = lane; i < n8; i += 64) {
+        lds64[i] = __hip_atomic_load(src + i, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+      }
     } else {
       const uint32_t b = d.fill & 0xFF;
       const uint32_t word = b | (b << 8) | (b << 16) | (b << 24);
-      const float4 v = {__uint_as_float(word), __uint_as_float(word),
-                        __uint_as_float(word), __uint_as_float(word)};
-      for (uint32_t i = lane; i < n16; i += 64) lds[i] = v;
+      const unsigned long long v =
+          word | (static_cast<unsigned long long>(word) << 32);
+      for (uint32_t i = lane; i < n8; i += 64) lds64[i] = v;
     }
-    float4* __restrict__ dst = reinterpret_cast<float4*>(d.dst);
+    unsigned long long* __restrict__ dst =
+        reinterpret_cast<unsigned long long*>(d.dst);
 #pragma unroll 4
-    for (uint32_t i = lane; i < n16; i += 64) dst[i] = lds[i];
+    for (uint32_t i = lane; i < n8; i += 64) dst[i] = lds64[i];
     // Publish the completion: data must be host-visible before the CQ
     // word. The system-scope RELEASE store carries that ordering by
     // itself (s_waitcnt on the wave's outstanding stores + L2
@@ -476,21 +491,31 @@ __global__ __launch_bounds__(64) void k_shared_service(SharedCtl ctl) {
       const uint8_t* src = reinterpret_cast<const uint8_t*>(w0);
       uint8_t* dst = reinterpret_cast<uint8_t*>(w1);
       const uint32_t bytes = static_cast<uint32_t>(w2 & 0xFFFFFFFFu);
-      const uint32_t n16 = bytes >> 4;
+      // Uniform u64 same-slot tile path with agent-scope loads:
+      // resident-kernel cross-XCD coherence, see the per-queue
+      // kernel's copy loop.
+      const uint32_t n8 = bytes >> 3;
+      unsigned long long* lds64 =
+          reinterpret_cast<unsigned long long*>(lds);
       if (src != nullptr) {
-        const float4* __restrict__ s4 = reinterpret_cast<const float4*>(src);
+        const unsigned long long* __restrict__ s8 =
+            reinterpret_cast<const unsigned long long*>(src);
 #pragma unroll 4
-        for (uint32_t k = lane; k < n16; k += 64) lds[k] = s4[k];
+        for (uint32_t k = lane; k < n8; k += 64) {
+          lds64[k] = __hip_atomic_load(s8 + k, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+        }
       } else {
         const uint32_t b = static_cast<uint32_t>(w2 >> 32) & 0xFF;
         const uint32_t word = b | (b << 8) | (b << 16) | (b << 24);
-        const float4 v = {__uint_as_float(word), __uint_as_float(word),
-                          __uint_as_float(word), __uint_as_float(word)};
-        for (uint32_t k = lane; k < n16; k += 64) lds[k] = v;
+        const unsigned long long v =
+            word | (static_cast<unsigned long long>(word) << 32);
+        for (uint32_t k = lane; k < n8; k += 64) lds64[k] = v;
       }
-      float4* __restrict__ d4 = reinterpret_cast<float4*>(dst);
+      unsigned long long* __restrict__ d8 =
+          reinterpret_cast<unsigned long long*>(dst);
 #pragma unroll 4
-      for (uint32_t k = lane; k < n16; k += 64) d4[k] = lds[k];
+      for (uint32_t k = lane; k < n8; k += 64) d8[k] = lds64[k];
       // Ordering carried by the system-scope release store; see the
       // per-queue kernel's publish comment (explicit
       // __threadfence_system hangs waves on current pool firmware).
