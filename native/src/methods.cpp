@@ -19,6 +19,7 @@
 #include "hipstore/nbd.h"
 #include "hipstore/nvmf.h"
 #include "hipstore/rados.h"
+#include "hipstore/ublk.h"
 #include "hipstore/rpc.h"
 #include "hipstore/vhost.h"
 
@@ -307,6 +308,50 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   server->register_method("stop_nbd_disk", [](const Json& p) {
     nbd_stop(p.get_string("nbd_device"));
     return Json(true);
+  });
+
+  // --- ublk host attach (reference lib/nbd/nbd.c role; the pool's
+  // kernels ship ublk_drv but no nbd module, so this is the path that
+  // yields a real /dev/ublkbN on a GPU box) --------------------------------
+  server->register_method("ublk_start_disk", [&manager](const Json& p) {
+    const std::string bdev_name = p.get_string("bdev_name");
+    if (!manager.find(bdev_name)) not_found("bdev " + bdev_name);
+    if (!ublk_available()) {
+      throw RpcError{kInternalError, "ublk unavailable on this kernel"};
+    }
+    UblkDisk disk;
+    try {
+      disk = ublk_start(bdev_name,
+                        static_cast<int>(p.get_int("queue_depth", 32)));
+    } catch (const std::exception& e) {
+      throw RpcError{kInternalError, e.what()};
+    }
+    JsonObject o;
+    o["dev_id"] = Json(static_cast<int64_t>(disk.dev_id));
+    o["device"] = Json(disk.block_path);
+    o["bdev_name"] = Json(disk.bdev_name);
+    return Json(std::move(o));
+  });
+
+  server->register_method("ublk_stop_disk", [](const Json& p) {
+    try {
+      ublk_stop(static_cast<int>(p.get_int("dev_id")));
+    } catch (const std::exception& e) {
+      throw RpcError{kInvalidParams, e.what()};
+    }
+    return Json(JsonObject{});
+  });
+
+  server->register_method("ublk_get_disks", [](const Json&) {
+    JsonArray out;
+    for (const UblkDisk& disk : ublk_list()) {
+      JsonObject o;
+      o["dev_id"] = Json(static_cast<int64_t>(disk.dev_id));
+      o["device"] = Json(disk.block_path);
+      o["bdev_name"] = Json(disk.bdev_name);
+      out.push_back(Json(std::move(o)));
+    }
+    return Json(std::move(out));
   });
 
   server->register_method("construct_vhost_scsi_controller", [](const Json& p) {

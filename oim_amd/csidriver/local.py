@@ -101,6 +101,9 @@ class LocalBackend(OIMBackend):
                 raise
 
     def _find_exported(self, client, volume_id: str) -> str:
+        for disk in client.invoke("ublk_get_disks") or []:
+            if disk["bdev_name"] == volume_id:
+                return disk["device"]
         for disk in hipstore.get_nbd_disks(client):
             if disk.bdev_name == volume_id:
                 return disk.nbd_device
@@ -112,6 +115,16 @@ class LocalBackend(OIMBackend):
             device = self._find_exported(client, volume_id)
             if device:
                 return device
+            # Preferred path: ublk (the GPU pool's kernels ship
+            # ublk_drv but no nbd module) — the daemon allocates the
+            # device id and mknods /dev/ublkbN itself.
+            try:
+                disk = client.invoke("ublk_start_disk",
+                                     {"bdev_name": volume_id})
+                return disk["device"]
+            except hipstore.RpcError as err:
+                from_context().warn("ublk attach unavailable, trying NBD",
+                                    error=str(err))
             in_use = {d.nbd_device for d in hipstore.get_nbd_disks(client)}
             # Probe /dev/nbd0.. for a free device (local.go:139-176; the
             # reference notes the size==0 probe is racy — daemon-side
@@ -128,10 +141,15 @@ class LocalBackend(OIMBackend):
                                         device=candidate, error=str(err))
                     continue
                 return candidate
-            raise RuntimeError("no free NBD device found")
+            raise RuntimeError("no free NBD/ublk device found")
 
     def delete_device(self, volume_id: str) -> None:
         with self._client() as client:
+            for disk in client.invoke("ublk_get_disks") or []:
+                if disk["bdev_name"] == volume_id:
+                    client.invoke("ublk_stop_disk",
+                                  {"dev_id": disk["dev_id"]})
+                    return
             device = self._find_exported(client, volume_id)
             if device:
                 hipstore.stop_nbd_disk(client, device)
