@@ -316,9 +316,6 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   server->register_method("ublk_start_disk", [&manager](const Json& p) {
     const std::string bdev_name = p.get_string("bdev_name");
     if (!manager.find(bdev_name)) not_found("bdev " + bdev_name);
-    if (!ublk_available()) {
-      throw RpcError{kInternalError, "ublk unavailable on this kernel"};
-    }
     UblkDisk disk;
     try {
       disk = ublk_start(bdev_name,

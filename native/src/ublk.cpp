@@ -357,13 +357,17 @@ void ensure_node(const std::string& path, mode_t type, uint32_t major,
   }
 }
 
-int open_control() {
+// Opens /dev/ublk-control, mknod'ing it from /proc/misc when udev
+// never created it. On failure *reason says which step failed.
+int open_control(std::string* reason = nullptr) {
   const char* path = "/dev/ublk-control";
   int fd = open(path, O_RDWR);
   if (fd >= 0) return fd;
-  // mknod from /proc/misc ("<minor> ublk-control").
   FILE* f = fopen("/proc/misc", "r");
-  if (f == nullptr) return -1;
+  if (f == nullptr) {
+    if (reason) *reason = "/proc/misc unreadable";
+    return -1;
+  }
   char name[64];
   int minor = -1, m;
   while (fscanf(f, "%d %63s", &m, name) == 2) {
@@ -373,9 +377,22 @@ int open_control() {
     }
   }
   fclose(f);
-  if (minor < 0) return -1;
-  (void)mknod(path, S_IFCHR | 0600, makedev(10, minor));
-  return open(path, O_RDWR);
+  if (minor < 0) {
+    if (reason) *reason = "no ublk-control in /proc/misc (driver absent)";
+    return -1;
+  }
+  if (mknod(path, S_IFCHR | 0600, makedev(10, minor)) != 0 &&
+      errno != EEXIST) {
+    if (reason) {
+      *reason = std::string("mknod /dev/ublk-control: ") + strerror(errno);
+    }
+    return -1;
+  }
+  fd = open(path, O_RDWR);
+  if (fd < 0 && reason) {
+    *reason = std::string("open /dev/ublk-control: ") + strerror(errno);
+  }
+  return fd;
 }
 
 // One blocking control command round trip on its own tiny ring.
@@ -410,9 +427,10 @@ class UblkServer {
   ~UblkServer() { stop(); }
 
   UblkDisk start() {
-    ctrl_fd_ = open_control();
+    std::string reason;
+    ctrl_fd_ = open_control(&reason);
     if (ctrl_fd_ < 0) {
-      throw std::runtime_error("ublk: /dev/ublk-control unavailable");
+      throw std::runtime_error("ublk: " + reason);
     }
     ctrl_ring_.init(4);
 
