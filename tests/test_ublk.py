@@ -53,6 +53,10 @@ class TestUblkE2E:
     def test_block_device_roundtrip(self, tmp_path):
         if not _ublk_available():
             pytest.skip("kernel has no ublk support")
+        try:
+            os.close(os.open("/dev/ublk-control", os.O_RDWR))
+        except OSError as e:
+            pytest.skip(f"ublk-control not openable here: {e}")
         daemon = launch_hipstored(tmp_path, cpu=False)
         try:
             with hipstore.Client(daemon.socket_path) as client:
@@ -90,6 +94,10 @@ class TestUblkE2E:
     def test_local_backend_create_device_uses_ublk(self, tmp_path):
         if not _ublk_available():
             pytest.skip("kernel has no ublk support")
+        try:
+            os.close(os.open("/dev/ublk-control", os.O_RDWR))
+        except OSError as e:
+            pytest.skip(f"ublk-control not openable here: {e}")
         from oim_amd.csidriver.local import LocalBackend
 
         daemon = launch_hipstored(tmp_path, cpu=False)
@@ -103,4 +111,72 @@ class TestUblkE2E:
             backend.delete_device("ublk-vol")
             backend.delete_volume("ublk-vol")
         finally:
+            daemon.stop()
+
+    def test_nodestage_yields_mounted_ext4(self, tmp_path):
+        """The reference proves its device path by a guest seeing sda
+        (controller_test.go:306-340). The achievable equivalent here:
+        CSI NodeStageVolume on a GPU box produces a REAL mounted ext4
+        on /dev/ublkbN backed by HBM, with file I/O through the page
+        cache hitting the engine."""
+        if not _ublk_available():
+            pytest.skip("kernel has no ublk support")
+        try:
+            os.close(os.open("/dev/ublk-control", os.O_RDWR))
+        except OSError as e:
+            pytest.skip(f"ublk-control not openable here: {e}")
+        if os.geteuid() != 0:
+            pytest.skip("needs root for mkfs/mount")
+        import grpc
+
+        from oim_amd.common.server import grpc_target
+        from oim_amd.csidriver import LocalBackend, OIMDriver
+        from oim_amd.spec import csi_v1 as csi
+        from oim_amd.spec.rpc_csi import CSIControllerStub, CSINodeStub
+
+        daemon = launch_hipstored(tmp_path, cpu=False)
+        driver = None
+        staging = str(tmp_path / "staging")
+        try:
+            backend = LocalBackend(daemon.socket_path)
+            driver = OIMDriver(driver_name="oim-malloc", node_id="gpu-e2e",
+                               endpoint=f"unix://{tmp_path}/csi.sock",
+                               backend=backend)
+            driver.start()
+            with grpc.insecure_channel(
+                    grpc_target(f"unix://{tmp_path}/csi.sock")) as ch:
+                ctrl = CSIControllerStub(ch)
+                node = CSINodeStub(ch)
+                req = csi.CreateVolumeRequest(name="pvc-ublk")
+                req.capacity_range.required_bytes = 64 << 20
+                cap = req.volume_capabilities.add()
+                cap.mount.fs_type = "ext4"
+                cap.access_mode.mode = csi.ACCESS_MODE_SINGLE_NODE_WRITER
+                ctrl.CreateVolume(req, timeout=60)
+                stage = csi.NodeStageVolumeRequest(
+                    volume_id="pvc-ublk", staging_target_path=staging)
+                stage.volume_capability.mount.fs_type = "ext4"
+                stage.volume_capability.access_mode.mode = \
+                    csi.ACCESS_MODE_SINGLE_NODE_WRITER
+                node.NodeStageVolume(stage, timeout=120)
+                assert os.path.ismount(staging)
+                probe = os.path.join(staging, "hello.bin")
+                payload = os.urandom(1 << 20)
+                with open(probe, "wb") as f:
+                    f.write(payload)
+                    f.flush()
+                    os.fsync(f.fileno())
+                with open(probe, "rb") as f:
+                    assert f.read() == payload
+                node.NodeUnstageVolume(
+                    csi.NodeUnstageVolumeRequest(
+                        volume_id="pvc-ublk",
+                        staging_target_path=staging), timeout=60)
+                assert not os.path.ismount(staging)
+                ctrl.DeleteVolume(
+                    csi.DeleteVolumeRequest(volume_id="pvc-ublk"),
+                    timeout=30)
+        finally:
+            if driver is not None:
+                driver.stop()
             daemon.stop()
