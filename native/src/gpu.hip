@@ -1803,24 +1803,33 @@ class HbmBdev : public Bdev {
   }
 
   int poll(IoChannel* ch) override {
-    if (persistent_) {
-      // No HIP API calls on this path: poll() runs in a tight loop on
-      // every submitter thread, and even hipSetDevice takes the
-      // runtime's global lock — 4+ spinning threads convoyed on it and
-      // starved each other (measured: one of four queues fell to ~10
-      // completions/s). launch()/ensure_running() set the device when
-      // actually needed.
-      if (shared_) return static_cast<HbmSharedChannel*>(ch)->poll();
-      return static_cast<HbmPersistentChannel*>(ch)->poll();
+    // Dispatch on the CHANNEL's kind, exactly like submit(): a
+    // persistent bdev hands out batched/shared channels past the
+    // per-queue cap, and polling those through bdev-level flags cast
+    // them to the wrong class (round-1's "mixed-engine wedge": the
+    // misdispatched poll read garbage — stalled batched fallbacks,
+    // segfaulted shared ones — while the GPU side was healthy all
+    // along). No HIP API calls on the persistent paths: poll() runs
+    // in a tight loop on every submitter thread, and even
+    // hipSetDevice takes the runtime's global lock — 4+ spinning
+    // threads convoyed on it and starved each other.
+    auto* channel = static_cast<HbmChannelBase*>(ch);
+    switch (channel->kind) {
+      case HbmChannelBase::Kind::kShared:
+        return static_cast<HbmSharedChannel*>(channel)->poll();
+      case HbmChannelBase::Kind::kPersistent:
+        return static_cast<HbmPersistentChannel*>(channel)->poll();
+      case HbmChannelBase::Kind::kBatched:
+        break;
     }
-    auto* channel = static_cast<HbmChannel*>(ch);
-    int completed = channel->retire(/*wait=*/false);
-    if (channel->has_pending()) {
+    auto* batched = static_cast<HbmChannel*>(channel);
+    int completed = batched->retire(/*wait=*/false);
+    if (batched->has_pending()) {
       // Launches must come from the bdev's device; only pay the
       // runtime call when there is something to launch.
       (void)hipSetDevice(device_);
-      channel->kick();
-      completed += channel->retire(/*wait=*/false);
+      batched->kick();
+      completed += batched->retire(/*wait=*/false);
     }
     return completed;
   }

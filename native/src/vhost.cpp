@@ -707,9 +707,12 @@ class VhostUserScsiDev {
   }
 
   static bool pipeline_enabled() {
+    // Default ON since round 2's GPU A/B (tools/perf_batch.sh):
+    // pipelined +17% scsi / +4% blk randread through the same master.
+    // HIPSTORE_VHOST_PIPELINE=0 selects the one-at-a-time worker.
     static const bool on = [] {
       const char* env = getenv("HIPSTORE_VHOST_PIPELINE");
-      return env != nullptr && atoi(env) != 0;
+      return env == nullptr || atoi(env) != 0;
     }();
     return on;
   }

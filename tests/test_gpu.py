@@ -338,22 +338,18 @@ class TestInvalidWorkload:
 class TestChannelAutoFallback:
     """Past the per-device HW-queue cap (18 per-queue service kernels),
     new channels transparently fall back to the batched engine; IO on
-    a mixed per-queue + batched channel set stays correct. (The shared
-    service kernel is deliberately NOT mixed with per-queue kernels:
-    both resident on one device stops the shared rings being served —
-    see native/src/gpu.hip get_channel.)
+    a mixed per-queue + batched channel set stays correct.
 
-    Opt-in (HIPSTORE_TEST_FALLBACK=1): mixed-engine runs at >18 queues
-    still wedge on the MI355X pool (see the get_channel NOTE) and a
-    hung test would eat the suite budget; the default-on coverage for
-    the fallback path is the 4+2 case below, which stays within one
-    engine generation per channel set.
+    Round-1's "mixed-engine wedge" was a HOST bug, not a GPU mystery:
+    HbmBdev::poll() dispatched on bdev-level flags, so fallback
+    channels were polled through the wrong class (batched-on-
+    persistent read garbage and stalled; shared fallbacks segfaulted).
+    poll() now dispatches on the channel's kind like submit() always
+    did, and these run by default again. Every wait is bounded, so a
+    regression costs seconds, not the suite budget.
     """
 
     @pytest.mark.timeout(120)
-    @pytest.mark.skipif(os.environ.get("HIPSTORE_TEST_FALLBACK") != "1",
-                        reason="mixed-engine wedge under investigation; "
-                               "opt-in (see class docstring)")
     def test_low_cap_mixed_engines(self, monkeypatch):
         """4 per-queue + 2 batched via HIPSTORE_PERQ_CAP=4 — exercises
         the fallback decision and mixed-kind dispatch with small
@@ -366,14 +362,34 @@ class TestChannelAutoFallback:
         assert proc.returncode == 0, proc.stdout + proc.stderr
         assert "ios=" in proc.stdout
 
-    @pytest.mark.skipif(os.environ.get("HIPSTORE_TEST_FALLBACK") != "1",
-                        reason="mixed engines >18q wedge; opt-in")
+    @pytest.mark.timeout(180)
     def test_22_queues_mixed_engines(self):
         bdev = hs.create_hbm_bdev("fallback-0", 4096, 262144, device=0,
                                   persistent=True)
         result = hs.run_bdevperf(bdev, "randread", 4096, 8, 22, 20.0,
                                  max_ios=200000)
         assert result["io_count"] >= 200000
+        del bdev, result
+        # shared fallback flavour (the segfaulting round-1 config)
+        os.environ["HIPSTORE_FALLBACK"] = "shared"
+        try:
+            import subprocess
+            import sys as sys_mod
+            script = os.path.join(os.path.dirname(__file__),
+                                  "_qsweep_debug.py")
+            env = dict(os.environ, HIPSTORE_PERQ_CAP="4",
+                       HIPSTORE_FALLBACK="shared")
+            proc = subprocess.run([sys_mod.executable, script, "6"],
+                                  env=env, capture_output=True, text=True,
+                                  timeout=100)
+            assert proc.returncode == 0, proc.stdout + proc.stderr
+        finally:
+            os.environ.pop("HIPSTORE_FALLBACK", None)
+        bdev = hs.create_hbm_bdev("fallback-1", 4096, 262144, device=0,
+                                  persistent=True)
+        result = hs.run_bdevperf(bdev, "randread", 4096, 8, 2, 5.0,
+                                 max_ios=20000)
+        assert result["io_count"] >= 20000
         # correctness across the cap boundary
         import random
         rng = random.Random(11)
