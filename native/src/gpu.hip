@@ -964,13 +964,15 @@ inline int per_queue_channel_cap() {
   return cap;
 }
 
-// Which engine serves channels past the cap: "batched" (default; see
-// the mixed-engine wedge note in get_channel) or "shared" (for the
-// round-2 wedge experiments, tools/wedge_experiments.sh).
+// Which engine serves channels past the cap. Default "shared" since
+// round 2: with the poll-dispatch fix in, the mixed-engine matrix
+// (tools/wedge_experiments.sh) measures the shared-service fallback
+// at p99 26-37 us vs 57-258 us for the batched fallback on the same
+// shapes. HIPSTORE_FALLBACK=batched selects the old behavior.
 inline bool fallback_is_shared() {
   static const bool shared = [] {
     const char* env = getenv("HIPSTORE_FALLBACK");
-    return env != nullptr && strcmp(env, "shared") == 0;
+    return env == nullptr || strcmp(env, "batched") != 0;
   }();
   return shared;
 }
@@ -1730,18 +1732,13 @@ class HbmBdev : public Bdev {
     if (persistent_) {
       // Auto-fallback: per-queue service kernels give the best
       // latency but each needs a hardware queue; past the cap, new
-      // channels fall back to the batched engine (transient per-batch
-      // kernel launches coexist with resident service kernels — the
-      // CRC32C and clone paths run that way in production). The slot
-      // is claimed atomically BEFORE constructing — a check-then-create
-      // race let N concurrent creators all pass the check and launch
-      // N > cap service kernels, overrunning GPU_MAX_HW_QUEUES.
-      // NOTE: the shared service kernel is NOT used here — a device
-      // with both per-queue and shared service kernels resident stops
-      // serving the shared rings entirely (reproduced on MI355X even
-      // at 4+2 channels; engines work in isolation at any count).
-      // Until that interaction is understood, the shared service runs
-      // only in all-shared mode (HIPSTORE_SHARED=1).
+      // channels fall back to the shared per-device service kernel
+      // (p99 26-37 us in the mixed matrix vs 57-258 us for batched
+      // fallbacks; round-1's "mixed-engine wedge" was a poll-dispatch
+      // bug, fixed — see poll()). The slot is claimed atomically
+      // BEFORE constructing — a check-then-create race let N
+      // concurrent creators all pass the check and launch N > cap
+      // service kernels, overrunning GPU_MAX_HW_QUEUES.
       if (shared_) {
         return track(std::make_shared<HbmSharedChannel>(device_, base_));
       }
@@ -1752,7 +1749,7 @@ class HbmBdev : public Bdev {
       }
       g_per_queue_channels[device_ & 63].fetch_sub(
           1, std::memory_order_relaxed);
-      if (fallback_is_shared()) {  // wedge experiments only
+      if (fallback_is_shared()) {
         return track(std::make_shared<HbmSharedChannel>(device_, base_));
       }
       return track(
