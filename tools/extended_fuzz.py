@@ -4,7 +4,7 @@ The in-suite 100-example config fuzz found a real bug (JSON surrogate
 pair decoding wedging client streams) — rerun these periodically with
 fresh seeds.
 
-    python tools/extended_fuzz.py [config|vhost|nvmf|all] [examples]
+    python tools/extended_fuzz.py [config|vhost|nvmf|rados|all] [examples]
 
 Each surface runs against its own fresh CPU-mode daemon/target and
 asserts the server stays healthy afterwards.
@@ -165,6 +165,65 @@ def fuzz_nvmf(examples: int) -> None:
         target.stop()
 
 
+def fuzz_rados(examples: int) -> None:
+    """Hostile msgr-v1 sessions against the loopback cluster: random
+    handshake garbage, then (on a well-handshaken session) random
+    tagged frames with corrupt headers/fronts/CRCs. The cluster must
+    drop bad sessions and keep serving good ones."""
+    import oim_amd._hipstore as hs
+
+    cluster = hs.start_rados_cluster(port=0, arena_mb=16, use_hbm=False,
+                                     device=0, object_bytes=1 << 20)
+
+    def handshake(sock):
+        got = b""
+        while len(got) < 9 + 272:
+            chunk = sock.recv(9 + 272 - len(got))
+            if not chunk:
+                raise OSError("early close")
+            got += chunk
+        sock.sendall(b"ceph v027" + bytes(136))
+        sock.sendall(struct.pack("<QIIIIIIB", 0, 8, 1, 1, 24, 0, 0, 0))
+        reply = b""
+        while len(reply) < 26:
+            chunk = sock.recv(26 - len(reply))
+            if not chunk:
+                raise OSError("early close")
+            reply += chunk
+
+    @settings(max_examples=examples, deadline=None,
+              suppress_health_check=SUPPRESS)
+    @given(st.booleans(),
+           st.lists(st.tuples(st.integers(0, 20), st.binary(max_size=300)),
+                    min_size=1, max_size=3))
+    def run(do_handshake, frames):
+        try:
+            with socketmod.create_connection(
+                    ("127.0.0.1", cluster.port()), timeout=3) as sock:
+                if do_handshake:
+                    handshake(sock)
+                for tag, payload in frames:
+                    sock.sendall(bytes([tag]) + payload)
+                sock.shutdown(socketmod.SHUT_WR)
+                while sock.recv(4096):
+                    pass
+        except OSError:
+            pass
+
+    try:
+        run()
+        # the cluster must still serve a well-formed client
+        bdev = hs.create_rbd_bdev("xrfz", f"127.0.0.1:{cluster.port()}",
+                                  "rbd", "xrfz-img", block_size=512,
+                                  default_size_bytes=4 << 20,
+                                  object_bytes=1 << 20)
+        bdev.write(0, b"\xa5" * 4096)
+        assert bdev.read(0, 4096) == b"\xa5" * 4096
+        print(f"rados fuzz: {examples} sessions clean")
+    finally:
+        cluster.stop()
+
+
 def main() -> int:
     surface = sys.argv[1] if len(sys.argv) > 1 else "all"
     examples = int(sys.argv[2]) if len(sys.argv) > 2 else 500
@@ -174,6 +233,8 @@ def main() -> int:
         fuzz_vhost(examples)
     if surface in ("nvmf", "all"):
         fuzz_nvmf(examples)
+    if surface in ("rados", "all"):
+        fuzz_rados(examples)
     return 0
 
 
