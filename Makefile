@@ -20,7 +20,7 @@ B := native/build
 HDRS := $(wildcard native/include/hipstore/*.h)
 
 CORE_OBJS := $(B)/json.o $(B)/bdev.o $(B)/crc32c.o $(B)/rpc_server.o \
-             $(B)/methods.o $(B)/nbd.o $(B)/ublk.o $(B)/composite.o $(B)/vhost.o \
+             $(B)/methods.o $(B)/nbd.o $(B)/ublk.o $(B)/composite.o $(B)/vhost.o $(B)/vhost_master.o \
              $(B)/nvmf_common.o $(B)/nvmf_target.o $(B)/nvmf_initiator.o \
              $(B)/rados_client.o $(B)/rados_cluster.o \
              $(B)/gpu.o

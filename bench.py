@@ -173,6 +173,11 @@ def main() -> int:
                         help="vhost frontend personality")
     parser.add_argument("--vhost-numjobs", type=int, default=4,
                         help="request rings for --frontend vhost")
+    parser.add_argument("--vhost-master", default="native",
+                        choices=["native", "python"],
+                        help="vhost frontend driver: the C++ master "
+                             "(measures the daemon's ceiling) or the "
+                             "Python conformance master")
     args = parser.parse_args()
 
     import torch
@@ -210,43 +215,67 @@ def main() -> int:
                 # byte traverses descriptor → ring worker → engine →
                 # HBM → used ring (the reference's signature flow,
                 # vhost_scsi.c process_requestq).
-                import threading
+                from oim_amd import hipstore as hipstore_mod
 
-                from oim_amd.bench.vhost_harness import VhostAttachment
                 if args.personality == "scsi":
-                    # Drive the controller MapVolume attached the bdev
-                    # to — the exact device a guest would see.
-                    attachment = VhostAttachment(
-                        client, daemon_sock, volume, "vhost.0",
-                        "scsi", args.vhost_numjobs, args.queue_depth,
-                        args.io_size, args.workload, create=False)
+                    ctrlr = "vhost.0"  # MapVolume attached the bdev here
                 else:
-                    attachment = VhostAttachment(
-                        client, daemon_sock, volume,
-                        f"vhost-bench-{rank}", "blk",
-                        args.vhost_numjobs, args.queue_depth,
-                        args.io_size, args.workload)
+                    ctrlr = f"vhost-bench-{rank}"
+                    client.invoke("construct_vhost_blk_controller",
+                                  {"ctrlr": ctrlr, "dev_name": volume})
+                bdev_info = hipstore_mod.get_bdevs(client, volume)[0]
+                vhost_sock = os.path.join(os.path.dirname(daemon_sock),
+                                          ctrlr)
 
-                def run_step(n_ios):
-                    jobs = attachment.jobs
-                    per = (n_ios + len(jobs) - 1) // len(jobs)
-                    for j in jobs:
-                        j.lat_us.clear()
-                    threads = [threading.Thread(target=j.run_count,
-                                                args=(per,))
-                               for j in jobs]
-                    t0 = time.perf_counter()
-                    for t in threads:
-                        t.start()
-                    for t in threads:
-                        t.join()
-                    dt = time.perf_counter() - t0
-                    lat = sorted(x for j in jobs for x in j.lat_us)
-                    count = per * len(jobs)
-                    p99 = lat[min(len(lat) - 1, int(len(lat) * 0.99))] \
-                        if lat else 0.0
-                    return {"io_count": count, "iops": count / dt,
-                            "lat_p99_us": p99}
+                if args.vhost_master == "native":
+                    # C++ master: each step is one native session
+                    # completing n_ios through the virtqueues (the
+                    # Python master tops out ~30k IOPS and would
+                    # measure the interpreter, not the daemon).
+                    from oim_amd import _hipstore as hs_native
+
+                    def run_step(n_ios):
+                        r = hs_native.vhost_master_bench(
+                            vhost_sock, args.personality,
+                            args.vhost_numjobs, args.queue_depth,
+                            args.io_size, args.workload, n_ios,
+                            bdev_info.block_size,
+                            bdev_info.size_bytes)
+                        return {"io_count": r["io_count"],
+                                "iops": r["iops"],
+                                "lat_p99_us": r["lat_p99_us"]}
+
+                    attachment = None
+                else:
+                    import threading
+
+                    from oim_amd.bench.vhost_harness import VhostAttachment
+                    attachment = VhostAttachment(
+                        client, daemon_sock, volume, ctrlr,
+                        args.personality, args.vhost_numjobs,
+                        args.queue_depth, args.io_size, args.workload,
+                        create=False)
+
+                    def run_step(n_ios):
+                        jobs = attachment.jobs
+                        per = (n_ios + len(jobs) - 1) // len(jobs)
+                        for j in jobs:
+                            j.lat_us.clear()
+                        threads = [threading.Thread(target=j.run_count,
+                                                    args=(per,))
+                                   for j in jobs]
+                        t0 = time.perf_counter()
+                        for t in threads:
+                            t.start()
+                        for t in threads:
+                            t.join()
+                        dt = time.perf_counter() - t0
+                        lat = sorted(x for j in jobs for x in j.lat_us)
+                        count = per * len(jobs)
+                        p99 = lat[min(len(lat) - 1, int(len(lat) * 0.99))] \
+                            if lat else 0.0
+                        return {"io_count": count, "iops": count / dt,
+                                "lat_p99_us": p99}
             else:
                 session = client.invoke("perf_session_start", {
                     "bdev_name": volume,
@@ -338,7 +367,7 @@ def main() -> int:
                 "engine": args.engine,
                 "frontend": args.frontend if args.frontend == "daemon"
                 else f"vhost-user-{args.personality}"
-                     f"-x{args.vhost_numjobs}rings",
+                     f"-x{args.vhost_numjobs}rings-{args.vhost_master}",
                 "provisioning": "csi-createvolume+proxy-mapvolume",
                 "p99_us": round(p99, 1),
             },

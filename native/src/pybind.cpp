@@ -18,6 +18,7 @@
 #include "hipstore/engine.h"
 #include "hipstore/nvmf.h"
 #include "hipstore/rados.h"
+#include "hipstore/vhost_master.h"
 
 namespace py = pybind11;
 using namespace hipstore;
@@ -136,6 +137,27 @@ PYBIND11_MODULE(_hipstore, m) {
         py::arg("traddr"), py::arg("trsvcid"), py::arg("subnqn"),
         py::arg("nsid") = 1, py::arg("enable_digests") = true,
         py::call_guard<py::gil_scoped_release>());
+
+  m.def(
+      "vhost_master_bench",
+      [](const std::string& socket_path, const std::string& personality,
+         int num_rings, int iodepth, uint32_t io_size,
+         const std::string& workload, uint64_t total_ios,
+         uint32_t block_size, uint64_t capacity_bytes) {
+        PerfResult r;
+        {
+          py::gil_scoped_release release;
+          r = vhost_master_bench(socket_path, personality, num_rings,
+                                 iodepth, io_size, workload, total_ios,
+                                 block_size, capacity_bytes);
+        }
+        return perf_to_dict(r);
+      },
+      py::arg("socket_path"), py::arg("personality") = "scsi",
+      py::arg("num_rings") = 4, py::arg("iodepth") = 32,
+      py::arg("io_size") = 4096, py::arg("workload") = "randread",
+      py::arg("total_ios") = 100000, py::arg("block_size") = 512,
+      py::arg("capacity_bytes") = 0);
 
   py::class_<PerfSession>(m, "PerfSession")
       .def(py::init<BdevPtr, std::string, uint32_t, uint32_t, int>(),

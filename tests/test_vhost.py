@@ -18,6 +18,7 @@ import pytest
 
 from oim_amd import hipstore
 
+import fixtures
 from fixtures import hipstored  # noqa: F401
 from vhost_client import VhostUserMaster, GET_QUEUE_NUM
 
@@ -892,3 +893,55 @@ class TestBlkChainFuzz:
                 master.close()
             client.invoke("remove_vhost_controller", {"ctrlr": "vbfz"})
             assert isinstance(client.invoke("get_rpc_methods"), list)
+
+
+class TestNativeVhostMaster:
+    """The C++ vhost-user master (native/src/vhost_master.cpp): perf
+    tool counterpart of the Python conformance master above — same
+    handshake and ring layout, measured here for contract only."""
+
+    def test_scsi_and_blk_sessions(self, tmp_path):
+        import oim_amd._hipstore as hs
+
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        try:
+            with hipstore.Client(daemon.socket_path) as client:
+                hipstore.construct_malloc_bdev(
+                    client, num_blocks=65536, block_size=512, name="nm0")
+                client.invoke("construct_vhost_scsi_controller",
+                              {"ctrlr": "nm-scsi"})
+                client.invoke("add_vhost_scsi_lun",
+                              {"ctrlr": "nm-scsi", "scsi_target_num": 0,
+                               "bdev_name": "nm0"})
+                sock = os.path.join(os.path.dirname(daemon.socket_path),
+                                    "nm-scsi")
+                r = hs.vhost_master_bench(sock, "scsi", num_rings=2,
+                                          iodepth=8, io_size=4096,
+                                          workload="randrw",
+                                          total_ios=20000, block_size=512,
+                                          capacity_bytes=65536 * 512)
+                assert r["io_count"] >= 20000
+                assert r["lat_p99_us"] > 0
+                hipstore.construct_malloc_bdev(
+                    client, num_blocks=65536, block_size=512, name="nm1")
+                client.invoke("construct_vhost_blk_controller",
+                              {"ctrlr": "nm-blk", "dev_name": "nm1"})
+                sockb = os.path.join(os.path.dirname(daemon.socket_path),
+                                     "nm-blk")
+                r = hs.vhost_master_bench(sockb, "blk", num_rings=2,
+                                          iodepth=8, io_size=4096,
+                                          workload="randread",
+                                          total_ios=20000, block_size=512,
+                                          capacity_bytes=65536 * 512)
+                assert r["io_count"] >= 20000
+        finally:
+            daemon.stop()
+
+    def test_connect_failure_is_clean(self):
+        import oim_amd._hipstore as hs
+
+        with pytest.raises(RuntimeError):
+            hs.vhost_master_bench("/nonexistent/vhost.sock", "scsi",
+                                  num_rings=1, iodepth=4, io_size=4096,
+                                  workload="randread", total_ios=10,
+                                  block_size=512, capacity_bytes=1 << 20)
