@@ -25,6 +25,12 @@ def main(argv=None) -> int:
                         help="remote mode: controller to talk to")
     parser.add_argument("--emulate", default="",
                         help="volume-parameter emulation (ceph-csi)")
+    parser.add_argument("--csiversion", default="1.0",
+                        choices=["1.0", "0.3"],
+                        help="CSI personality to serve (reference "
+                             "main.go --csiversion; 0.3 is the legacy "
+                             "twin for pre-1.0 kubelets / ceph-csi "
+                             "v0.3 emulation)")
     parser.add_argument("--ca", default="")
     parser.add_argument("--key", default="")
     log.add_flags(parser)
@@ -55,7 +61,8 @@ def main(argv=None) -> int:
         )
     driver = OIMDriver(
         driver_name=args.drivername, node_id=args.nodeid,
-        endpoint=args.endpoint, backend=backend)
+        endpoint=args.endpoint, backend=backend,
+        csi_version=args.csiversion)
     driver.run()
     return 0
 

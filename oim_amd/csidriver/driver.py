@@ -530,14 +530,27 @@ class NodeServer(CSINodeServicer):
 
 class OIMDriver:
     """Assembles the three services on one endpoint
-    (reference Start, oim-driver.go:275-292)."""
+    (reference Start, oim-driver.go:275-292). `csi_version` selects
+    the personality: "1.0" (default) registers the csi.v1 services,
+    "0.3" the legacy twins (reference oimDriver03, driver0.go) — same
+    servicers underneath, adapted in oim_amd.csidriver.driver03."""
 
     def __init__(self, driver_name: str, node_id: str, endpoint: str,
-                 backend: OIMBackend, mounter: Optional[Mounter] = None):
+                 backend: OIMBackend, mounter: Optional[Mounter] = None,
+                 csi_version: str = "1.0"):
+        if csi_version not in ("1.0", "0.3"):
+            raise ValueError(f"unsupported CSI version {csi_version!r}")
+        self.csi_version = csi_version
         self.identity = IdentityServer(
             driver_name, offline_expansion=backend.supports_expansion())
         self.controller = ControllerServer(backend)
         self.node = NodeServer(node_id, backend, mounter)
+        if csi_version == "0.3":
+            from .driver03 import (ControllerServer0, IdentityServer0,
+                                   NodeServer0)
+            self.identity0 = IdentityServer0(driver_name)
+            self.controller0 = ControllerServer0(self.controller)
+            self.node0 = NodeServer0(node_id, self.node)
         # CSI requests carry `secrets` maps (NodeStage, CreateVolume):
         # payload logging must redact them (the reference used
         # protosanitizer StripSecrets the same way).
@@ -545,14 +558,23 @@ class OIMDriver:
             endpoint=endpoint,
             interceptors=[LogServerInterceptor(strip_secrets_formatter)])
 
-    def start(self) -> None:
-        def register(server):
-            add_csi_identity_to_server(self.identity, server)
-            add_csi_controller_to_server(self.controller, server)
-            add_csi_node_to_server(self.node, server)
+    def _register(self, server) -> None:
+        if self.csi_version == "0.3":
+            from ..spec.rpc_csi0 import (add_csi0_controller_to_server,
+                                         add_csi0_identity_to_server,
+                                         add_csi0_node_to_server)
+            add_csi0_identity_to_server(self.identity0, server)
+            add_csi0_controller_to_server(self.controller0, server)
+            add_csi0_node_to_server(self.node0, server)
+            return
+        add_csi_identity_to_server(self.identity, server)
+        add_csi_controller_to_server(self.controller, server)
+        add_csi_node_to_server(self.node, server)
 
-        self.server.start(register)
-        from_context().info("CSI driver started", endpoint=self.server.addr())
+    def start(self) -> None:
+        self.server.start(self._register)
+        from_context().info("CSI driver started", endpoint=self.server.addr(),
+                            csi_version=self.csi_version)
 
     def addr(self) -> str:
         return self.server.addr()
@@ -561,12 +583,7 @@ class OIMDriver:
         self.server.stop()
 
     def run(self) -> None:
-        def register(server):
-            add_csi_identity_to_server(self.identity, server)
-            add_csi_controller_to_server(self.controller, server)
-            add_csi_node_to_server(self.node, server)
-
-        self.server.run(register)
+        self.server.run(self._register)
 
 
 def make_params_mapper(emulate: str):
