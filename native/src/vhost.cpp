@@ -704,6 +704,12 @@ class VhostUserScsiDev {
     }
     unmap_regions();
     negotiated_features_ = 0;
+    if (getenv("HIPSTORE_DEBUG") != nullptr) {
+      fprintf(stderr, "[vhost %s] session served async=%llu sync=%llu\n",
+              name_.c_str(),
+              static_cast<unsigned long long>(async_count_.load()),
+              static_cast<unsigned long long>(sync_count_.load()));
+    }
   }
 
   static bool pipeline_enabled() {
@@ -791,7 +797,11 @@ class VhostUserScsiDev {
           if (ring.last_avail == avail_idx) break;
           uint16_t head = ring.avail->ring[ring.last_avail % ring.num];
           ring.last_avail++;
-          if (submit_async(index, head, channels, inflight_box)) continue;
+          if (submit_async(index, head, channels, inflight_box)) {
+            async_count_.fetch_add(1, std::memory_order_relaxed);
+            continue;
+          }
+          sync_count_.fetch_add(1, std::memory_order_relaxed);
           // Fallback: synchronous command (probe, trim, errors...)
           uint32_t written = 0;
           try {
@@ -1532,6 +1542,12 @@ class VhostUserScsiDev {
   int stop_pipe_[2] = {-1, -1};
   std::thread accept_thread_;
   std::atomic<bool> stopping_{false};
+  // Pipelined-worker dispatch mix (HIPSTORE_DEBUG prints on session
+  // reset): fast-path async submissions vs per-command sync fallbacks
+  // — the decisive counter when a front-end measures slower than the
+  // engine can serve.
+  std::atomic<uint64_t> async_count_{0};
+  std::atomic<uint64_t> sync_count_{0};
 
   uint64_t negotiated_features_ = 0;
   uint64_t protocol_features_ = 0;
