@@ -228,19 +228,21 @@ def main() -> int:
                                           ctrlr)
 
                 if args.vhost_master == "native":
-                    # C++ master: each step is one native session
-                    # completing n_ios through the virtqueues (the
-                    # Python master tops out ~30k IOPS and would
-                    # measure the interpreter, not the daemon).
+                    # C++ master, ONE standing session: the handshake
+                    # and the slave's hipHostRegister of guest memory
+                    # happen before the timed region; each step drives
+                    # n_ios through the standing rings. (The Python
+                    # master tops out ~30k IOPS and would measure the
+                    # interpreter, not the daemon.)
                     from oim_amd import _hipstore as hs_native
+                    vhost_session = hs_native.VhostMasterSession(
+                        vhost_sock, args.personality,
+                        args.vhost_numjobs, args.queue_depth,
+                        args.io_size, bdev_info.block_size,
+                        bdev_info.size_bytes)
 
                     def run_step(n_ios):
-                        r = hs_native.vhost_master_bench(
-                            vhost_sock, args.personality,
-                            args.vhost_numjobs, args.queue_depth,
-                            args.io_size, args.workload, n_ios,
-                            bdev_info.block_size,
-                            bdev_info.size_bytes)
+                        r = vhost_session.run(n_ios, args.workload)
                         return {"io_count": r["io_count"],
                                 "iops": r["iops"],
                                 "lat_p99_us": r["lat_p99_us"]}

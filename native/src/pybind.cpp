@@ -138,6 +138,24 @@ PYBIND11_MODULE(_hipstore, m) {
         py::arg("nsid") = 1, py::arg("enable_digests") = true,
         py::call_guard<py::gil_scoped_release>());
 
+  py::class_<VhostMasterSession>(m, "VhostMasterSession")
+      .def(py::init<const std::string&, const std::string&, int, int,
+                    uint32_t, uint32_t, uint64_t>(),
+           py::arg("socket_path"), py::arg("personality") = "scsi",
+           py::arg("num_rings") = 4, py::arg("iodepth") = 32,
+           py::arg("io_size") = 4096, py::arg("block_size") = 512,
+           py::arg("capacity_bytes") = 0,
+           py::call_guard<py::gil_scoped_release>())
+      .def("run", [](VhostMasterSession& s, uint64_t total_ios,
+                     const std::string& workload) {
+        PerfResult r;
+        {
+          py::gil_scoped_release release;
+          r = s.run(total_ios, workload);
+        }
+        return perf_to_dict(r);
+      }, py::arg("total_ios"), py::arg("workload") = "randread");
+
   m.def(
       "vhost_master_bench",
       [](const std::string& socket_path, const std::string& personality,

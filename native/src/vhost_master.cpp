@@ -142,25 +142,20 @@ struct Hist {
   }
 };
 
-struct RingCtx {
-  uint8_t* mem;          // region base (uaddr 0 of the region)
-  uint64_t base;         // ring slab offset within the region
-  uint32_t qsize;
-  int kick = -1;
-  int call = -1;
-  uint64_t desc_off, avail_off, used_off, req_off, data_off;
-  uint16_t avail_idx = 0;
-  uint16_t used_idx = 0;
-};
-
 }  // namespace
 
-PerfResult vhost_master_bench(const std::string& socket_path,
-                              const std::string& personality,
-                              int num_rings, int iodepth, uint32_t io_size,
-                              const std::string& workload,
-                              uint64_t total_ios, uint32_t block_size,
-                              uint64_t capacity_bytes) {
+VhostMasterSession::VhostMasterSession(const std::string& socket_path,
+                                       const std::string& personality,
+                                       int num_rings, int iodepth,
+                                       uint32_t io_size,
+                                       uint32_t block_size,
+                                       uint64_t capacity_bytes)
+    : personality_(personality),
+      num_rings_(num_rings),
+      iodepth_(iodepth),
+      io_size_(io_size),
+      block_size_(block_size),
+      capacity_bytes_(capacity_bytes) {
   const bool blk = personality == "blk";
   const int first_queue = blk ? 0 : 2;
   const int max_rings = blk ? 8 : 6;
@@ -207,7 +202,7 @@ PerfResult vhost_master_bench(const std::string& socket_path,
     throw std::runtime_error("vhost master: mmap failed");
   }
 
-  std::vector<RingCtx> rings(num_rings);
+  std::vector<VhostMasterRing> rings(num_rings);
   PerfResult result;
   try {
     // Handshake (GET/SET features, owner, memory table).
@@ -236,7 +231,7 @@ PerfResult vhost_master_bench(const std::string& socket_path,
 
     // Ring setup.
     for (int i = 0; i < num_rings; ++i) {
-      RingCtx& ring = rings[i];
+      VhostMasterRing& ring = rings[i];
       ring.mem = mem;
       ring.qsize = qsize;
       ring.base = 0x10000 + slab * i;
@@ -267,6 +262,48 @@ PerfResult vhost_master_bench(const std::string& socket_path,
       send_msg(sock, kSetVringEnable, &u32x2, 8);
     }
 
+    // Rings configured; session stays up across run() calls.
+    (void)0;
+    sock_ = sock;
+    memfd_ = memfd;
+    mem_ = mem;
+    mem_size_ = mem_size;
+    rings_ = std::move(rings);
+  } catch (...) {
+    for (auto& ring : rings) {
+      if (ring.kick >= 0) close(ring.kick);
+      if (ring.call >= 0) close(ring.call);
+    }
+    munmap(mem, mem_size);
+    close(memfd);
+    close(sock);
+    throw;
+  }
+}
+
+VhostMasterSession::~VhostMasterSession() {
+  for (auto& ring : rings_) {
+    if (ring.kick >= 0) close(ring.kick);
+    if (ring.call >= 0) close(ring.call);
+  }
+  if (mem_ != nullptr) munmap(mem_, mem_size_);
+  if (memfd_ >= 0) close(memfd_);
+  if (sock_ >= 0) close(sock_);
+}
+
+PerfResult VhostMasterSession::run(uint64_t total_ios,
+                                   const std::string& workload) {
+  const bool blk = personality_ == "blk";
+  const int num_rings = num_rings_;
+  const int iodepth = iodepth_;
+  const uint32_t io_size = io_size_;
+  const uint32_t block_size = block_size_;
+  const uint64_t capacity_bytes = capacity_bytes_;
+  uint8_t* mem = mem_;
+  std::vector<VhostMasterRing>& rings = rings_;
+  ++run_seq_;
+  PerfResult result;
+  {
     // Drive.
     const bool do_write_only = workload == "randwrite";
     const bool do_mix = workload == "randrw";
@@ -281,9 +318,9 @@ PerfResult vhost_master_bench(const std::string& socket_path,
     for (int r = 0; r < num_rings; ++r) {
       threads.emplace_back([&, r] {
         using clock = std::chrono::steady_clock;
-        RingCtx& ring = rings[r];
+        VhostMasterRing& ring = rings[r];
         Hist& hist = hists[r];
-        std::mt19937_64 rng(0x5EEDull + r);
+        std::mt19937_64 rng(0x5EEDull + r + run_seq_ * 1315423911ull);
         std::vector<clock::time_point> t_submit(iodepth);
         auto* descs = reinterpret_cast<VringDesc*>(mem + ring.desc_off);
         auto* avail_idx_p =
@@ -428,24 +465,19 @@ PerfResult vhost_master_bench(const std::string& socket_path,
       result.lat_p999_us = total.pct(0.999);
       result.lat_max_us = total.max;
     }
-  } catch (...) {
-    for (auto& ring : rings) {
-      if (ring.kick >= 0) close(ring.kick);
-      if (ring.call >= 0) close(ring.call);
-    }
-    munmap(mem, mem_size);
-    close(memfd);
-    close(sock);
-    throw;
   }
-  for (auto& ring : rings) {
-    close(ring.kick);
-    close(ring.call);
-  }
-  munmap(mem, mem_size);
-  close(memfd);
-  close(sock);
   return result;
+}
+
+PerfResult vhost_master_bench(const std::string& socket_path,
+                              const std::string& personality,
+                              int num_rings, int iodepth, uint32_t io_size,
+                              const std::string& workload,
+                              uint64_t total_ios, uint32_t block_size,
+                              uint64_t capacity_bytes) {
+  VhostMasterSession session(socket_path, personality, num_rings, iodepth,
+                             io_size, block_size, capacity_bytes);
+  return session.run(total_ios, workload);
 }
 
 }  // namespace hipstore
