@@ -1028,7 +1028,8 @@ class VhostUserScsiDev {
       __atomic_store_n(&ring.used->idx, ++ring.used_idx, __ATOMIC_RELEASE);
       did_work = true;
     }
-    if (did_work && ring.call >= 0) {
+    if (did_work && ring.call >= 0 &&
+        !(__atomic_load_n(&ring.avail->flags, __ATOMIC_ACQUIRE) & 1)) {
       uint64_t one = 1;
       (void)!write(ring.call, &one, 8);
     }
