@@ -13,6 +13,8 @@
 #include <sys/mman.h>
 #include <sys/socket.h>
 #include <sys/un.h>
+#include <pthread.h>
+#include <sched.h>
 #include <unistd.h>
 
 #include <hip/hip_runtime.h>
@@ -723,7 +725,27 @@ class VhostUserScsiDev {
     return on;
   }
 
+  // Optional SPDK-reactor-style pinning for the spinning ring
+  // workers (HIPSTORE_VHOST_AFFINITY_BASE=<core>): unpinned hot
+  // pollers migrate under CFS and crowd each other once several rings
+  // spin concurrently (the vhost scale probe's global plateau).
+  static void maybe_pin_ring_worker(unsigned index) {
+    static std::atomic<int> next_slot{0};
+    const char* env = getenv("HIPSTORE_VHOST_AFFINITY_BASE");
+    if (env == nullptr) return;
+    const long ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+    if (ncpu <= 0) return;
+    (void)index;
+    const int core =
+        (atoi(env) + next_slot.fetch_add(1)) % static_cast<int>(ncpu);
+    cpu_set_t set;
+    CPU_ZERO(&set);
+    CPU_SET(core, &set);
+    (void)pthread_setaffinity_np(pthread_self(), sizeof(set), &set);
+  }
+
   void ring_worker(unsigned index) {
+    maybe_pin_ring_worker(index);
     ChannelCache channels;  // this worker's engine channels
     if (pipeline_enabled()) {
       ring_worker_pipelined(index, &channels);
