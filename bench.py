@@ -36,9 +36,19 @@ def _set_affinity_base(num_queues: int) -> None:
     ncpu = os.cpu_count() or 1
     world = int(os.environ.get("WORLD_SIZE", "1"))
     # Only pin when there are enough cores for every rank's queues.
-    if ncpu >= world * (num_queues + 2):
+    # Disjoint ranges per rank: perf-session queues first, then the
+    # vhost slave's ring workers, then the native master's pollers
+    # (spinning threads migrate under CFS — measured 3.4x on the vhost
+    # path, tools/vhost_scale_probe.py).
+    span = num_queues + 2
+    if ncpu >= world * span:
         os.environ.setdefault("HIPSTORE_AFFINITY_BASE",
-                              str(local_rank * (num_queues + 2)))
+                              str(local_rank * span))
+    if ncpu >= world * (span + 16):
+        base = world * span + local_rank * 16
+        os.environ.setdefault("HIPSTORE_VHOST_AFFINITY_BASE", str(base))
+        os.environ.setdefault("HIPSTORE_MASTER_AFFINITY_BASE",
+                              str(base + 8))
 
 REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO_ROOT)

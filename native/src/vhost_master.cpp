@@ -13,6 +13,8 @@
 // outstanding — the same shape as the Python harness, minus the GIL.
 
 #include <poll.h>
+#include <pthread.h>
+#include <sched.h>
 #include <sys/eventfd.h>
 #include <sys/mman.h>
 #include <sys/socket.h>
@@ -323,6 +325,19 @@ PerfResult VhostMasterSession::run(uint64_t total_ios,
     for (int r = 0; r < num_rings; ++r) {
       threads.emplace_back([&, r] {
         using clock = std::chrono::steady_clock;
+        // Busy-polling master threads migrate under CFS like the
+        // slave's ring workers; same optional pinning cure
+        // (HIPSTORE_MASTER_AFFINITY_BASE=<core>).
+        if (const char* env = getenv("HIPSTORE_MASTER_AFFINITY_BASE")) {
+          const long ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+          if (ncpu > 0) {
+            cpu_set_t set;
+            CPU_ZERO(&set);
+            CPU_SET((atoi(env) + r) % static_cast<int>(ncpu), &set);
+            (void)pthread_setaffinity_np(pthread_self(), sizeof(set),
+                                         &set);
+          }
+        }
         VhostMasterRing& ring = rings[r];
         Hist& hist = hists[r];
         std::mt19937_64 rng(0x5EEDull + r + run_seq_ * 1315423911ull);
