@@ -19,6 +19,7 @@
 #include <cerrno>
 #include <cstdio>
 #include <map>
+#include <set>
 #include <mutex>
 #include <stdexcept>
 #include <vector>
@@ -344,6 +345,13 @@ class RbdChannel : public IoChannel {
 
   ~RbdChannel() override {
     if (fd >= 0) close(fd);
+    // Teardown with outstanding I/O: release the per-IO states
+    // without firing their callbacks (a dead connection's callers are
+    // gone; mirrors the engine channels' leftover handling).
+    std::set<RbdIoState*> leftovers;
+    for (auto& [tid, entry] : inflight) leftovers.insert(entry.state);
+    for (RbdIoState* state : leftovers) delete state;
+    for (auto& [state, tids] : flushes) delete state;
   }
 };
 
@@ -381,7 +389,10 @@ class RbdBdev : public Bdev {
       }
       return;
     }
-    if (!check_bounds(req)) {
+    if (!check_bounds(req) ||
+        (req.op == IoOp::kFill && req.fill != 0)) {
+      // RADOS zero op cannot express a nonzero fill pattern: reject
+      // BEFORE sending (an invalid request must not mutate data).
       channel->immediate.emplace_back(std::move(req.on_complete),
                                       kIoInvalid);
       return;
@@ -422,11 +433,8 @@ class RbdBdev : public Bdev {
           op.payload_len = static_cast<uint32_t>(len);
           payload = static_cast<const uint8_t*>(req.buffer) + done;
           break;
-        default:  // kFill: zeroes only over the wire (RADOS zero op)
+        default:  // kFill (fill==0, validated above): RADOS zero op
           op.op = kOsdOpZero;
-          if (req.fill != 0) {
-            state->status = kIoInvalid;
-          }
           break;
       }
       const uint64_t tid = channel->next_tid++;
