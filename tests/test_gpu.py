@@ -370,21 +370,18 @@ class TestChannelAutoFallback:
                                  max_ios=200000)
         assert result["io_count"] >= 200000
         del bdev, result
-        # shared fallback flavour (the segfaulting round-1 config)
-        os.environ["HIPSTORE_FALLBACK"] = "shared"
-        try:
-            import subprocess
-            import sys as sys_mod
-            script = os.path.join(os.path.dirname(__file__),
-                                  "_qsweep_debug.py")
-            env = dict(os.environ, HIPSTORE_PERQ_CAP="4",
-                       HIPSTORE_FALLBACK="shared")
-            proc = subprocess.run([sys_mod.executable, script, "6"],
-                                  env=env, capture_output=True, text=True,
-                                  timeout=100)
-            assert proc.returncode == 0, proc.stdout + proc.stderr
-        finally:
-            os.environ.pop("HIPSTORE_FALLBACK", None)
+        # shared-fallback flavour at a tiny cap (the round-1
+        # segfaulting config), isolated in its own process
+        import subprocess
+        import sys as sys_mod
+        script = os.path.join(os.path.dirname(__file__),
+                              "_qsweep_debug.py")
+        env = dict(os.environ, HIPSTORE_PERQ_CAP="4",
+                   HIPSTORE_FALLBACK="shared")
+        proc = subprocess.run([sys_mod.executable, script, "6"],
+                              env=env, capture_output=True, text=True,
+                              timeout=100)
+        assert proc.returncode == 0, proc.stdout + proc.stderr
         bdev = hs.create_hbm_bdev("fallback-1", 4096, 262144, device=0,
                                   persistent=True)
         result = hs.run_bdevperf(bdev, "randread", 4096, 8, 2, 5.0,
