@@ -11,6 +11,8 @@
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/socket.h>
+#include <pthread.h>
+#include <sched.h>
 #include <unistd.h>
 
 #include <atomic>
@@ -132,6 +134,23 @@ class TargetImpl : public NvmfTcpTarget,
       auto done = std::make_shared<std::atomic<bool>>(false);
       connections_.emplace_back(
           std::thread([this, fd, done] {
+            // Optional pinning for this hot-polling connection thread
+            // (HIPSTORE_NVMF_AFFINITY_BASE=<core>): unpinned spinners
+            // migrate under CFS and crowd each other — the same
+            // plateau measured and fixed on the vhost ring workers.
+            if (const char* env = getenv("HIPSTORE_NVMF_AFFINITY_BASE")) {
+              static std::atomic<int> next_slot{0};
+              const long ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+              if (ncpu > 0) {
+                cpu_set_t set;
+                CPU_ZERO(&set);
+                CPU_SET((atoi(env) + next_slot.fetch_add(1)) %
+                            static_cast<int>(ncpu),
+                        &set);
+                (void)pthread_setaffinity_np(pthread_self(), sizeof(set),
+                                             &set);
+              }
+            }
             serve(fd);
             close(fd);
             done->store(true);
