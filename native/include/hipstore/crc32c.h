@@ -19,6 +19,9 @@ namespace hipstore {
 // fresh digest); internally applies the standard ~ pre/post conditioning
 // so crc32c_sw(0, ...) matches the usual CRC-32C value.
 uint32_t crc32c_sw(uint32_t init, const void* data, size_t len);
+// Byte-table-only variant (the bit-exactness reference; crc32c_sw
+// auto-selects the SSE4.2 hardware instruction when available).
+uint32_t crc32c_table(uint32_t init, const void* data, size_t len);
 
 // CRC of a concatenation: crc32c(A||B) from crc32c(A), crc32c(B) and
 // len(B) (zlib crc32_combine construction on the Castagnoli poly).

@@ -31,7 +31,46 @@ const Table& table() {
 
 }  // namespace
 
+// Hardware path: x86 SSE4.2 carries a Castagnoli CRC instruction
+// (the same accelerator SPDK's spdk_crc32c_update uses, reference
+// lib/util/crc32c.c:56-80). ~0.1 cycle/byte vs ~4 for the byte table
+// — at NVMe/TCP data-digest rates that is ~10 us/IO saved across the
+// two ends. Table path remains the bit-exactness reference (KATs).
+__attribute__((target("sse4.2")))
+static uint32_t crc32c_hw(uint32_t crc, const uint8_t* p, size_t len) {
+  while (len >= 8) {
+    crc = static_cast<uint32_t>(
+        __builtin_ia32_crc32di(crc, *reinterpret_cast<const uint64_t*>(p)));
+    p += 8;
+    len -= 8;
+  }
+  while (len > 0) {
+    crc = __builtin_ia32_crc32qi(crc, *p);
+    ++p;
+    --len;
+  }
+  return crc;
+}
+
+static bool have_sse42() {
+  static const bool ok = __builtin_cpu_supports("sse4.2");
+  return ok;
+}
+
 uint32_t crc32c_sw(uint32_t init, const void* data, size_t len) {
+  const uint8_t* p = static_cast<const uint8_t*>(data);
+  uint32_t crc = ~init;
+  if (have_sse42()) return ~crc32c_hw(crc, p, len);
+  const Table& tab = table();
+  for (size_t i = 0; i < len; ++i) {
+    crc = (crc >> 8) ^ tab.t[(crc ^ p[i]) & 0xFF];
+  }
+  return ~crc;
+}
+
+// Table-only variant for tests: the bit-exactness reference the
+// hardware and GPU paths are checked against.
+uint32_t crc32c_table(uint32_t init, const void* data, size_t len) {
   const Table& tab = table();
   const uint8_t* p = static_cast<const uint8_t*>(data);
   uint32_t crc = ~init;

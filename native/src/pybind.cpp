@@ -249,6 +249,11 @@ PYBIND11_MODULE(_hipstore, m) {
   m.def("crc32c_combine", &crc32c_combine, py::arg("crc1"), py::arg("crc2"),
         py::arg("len2"));
 
+  m.def("crc32c_table", [](py::buffer data, uint32_t init) {
+    py::buffer_info info = data.request();
+    return crc32c_table(init, info.ptr,
+                        static_cast<size_t>(info.size) * info.itemsize);
+  }, py::arg("data"), py::arg("init") = 0);
   m.def("crc32c", [](py::buffer data, uint32_t init) {
     py::buffer_info info = data.request();
     return crc32c_sw(init, info.ptr, static_cast<size_t>(info.size) * info.itemsize);
