@@ -687,9 +687,13 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   // rados_cluster.cpp) so construct_rbd_bdev's mon_host path has a
   // reachable peer in this no-network environment. Objects live in an
   // HBM arena on GPU boxes; data CRCs are GPU-computed there.
+  struct RadosClusterEntry {
+    std::shared_ptr<RadosCluster> cluster;
+    Json params;  // replayable rados_cluster_start params
+  };
   struct RadosClusters {
     std::mutex mutex;
-    std::map<uint16_t, std::shared_ptr<RadosCluster>> by_port;
+    std::map<uint16_t, RadosClusterEntry> by_port;
   };
   auto rados_clusters = std::make_shared<RadosClusters>();
 
@@ -712,7 +716,13 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
           throw RpcError{kInternalError, e.what()};
         }
         std::lock_guard<std::mutex> lock(rados_clusters->mutex);
-        rados_clusters->by_port[cluster->port()] = cluster;
+        JsonObject replay;
+        replay["port"] = Json(static_cast<int64_t>(cluster->port()));
+        replay["arena_mb"] = Json(arena_mb);
+        replay["object_mb"] = Json(object_mb);
+        replay["use_hbm"] = Json(p.get_bool("use_hbm", true));
+        rados_clusters->by_port[cluster->port()] =
+            RadosClusterEntry{cluster, Json(std::move(replay))};
         JsonObject o;
         o["port"] = Json(static_cast<int64_t>(cluster->port()));
         o["mon_host"] =
@@ -728,7 +738,7 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
         if (it == rados_clusters->by_port.end()) {
           not_found("rados cluster");
         }
-        it->second->stop();
+        it->second.cluster->stop();
         rados_clusters->by_port.erase(it);
         return Json(JsonObject{});
       });
@@ -1044,8 +1054,20 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
   // this is the SPDK-compatible control-state checkpoint. Composite
   // and RBD bdevs are not yet emitted (construction params are not
   // retained for them).
-  server->register_method("save_config", [&manager, targets,
-                                          creations](const Json&) {
+  server->register_method("save_config", [&manager, targets, creations,
+                                          rados_clusters](const Json&) {
+    // Loopback RADOS clusters replay FIRST (with their bound ports)
+    // so rbd bdevs whose mon_host points at them reconnect.
+    JsonArray rados_cfg;
+    {
+      std::lock_guard<std::mutex> lock(rados_clusters->mutex);
+      for (const auto& [port, entry] : rados_clusters->by_port) {
+        JsonObject e;
+        e["method"] = Json(std::string("rados_cluster_start"));
+        e["params"] = entry.params;
+        rados_cfg.push_back(Json(std::move(e)));
+      }
+    }
     JsonArray bdev_cfg;
     std::vector<std::pair<uint64_t, Json>> recorded;
     {
@@ -1140,10 +1162,32 @@ void register_storage_methods(RpcServer* server, bool use_hbm, int device,
       o["config"] = Json(std::move(cfg));
       return Json(std::move(o));
     };
+    JsonArray nbd_cfg;
+    for (const auto& [bdev_name, device] : nbd_list()) {
+      JsonObject params;
+      params["bdev_name"] = Json(bdev_name);
+      params["nbd_device"] = Json(device);
+      JsonObject e;
+      e["method"] = Json(std::string("start_nbd_disk"));
+      e["params"] = Json(std::move(params));
+      nbd_cfg.push_back(Json(std::move(e)));
+    }
+    JsonArray ublk_cfg;
+    for (const UblkDisk& disk : ublk_list()) {
+      JsonObject params;
+      params["bdev_name"] = Json(disk.bdev_name);
+      JsonObject e;
+      e["method"] = Json(std::string("ublk_start_disk"));
+      e["params"] = Json(std::move(params));
+      ublk_cfg.push_back(Json(std::move(e)));
+    }
     JsonArray subsystems;
+    subsystems.push_back(subsystem("rados", std::move(rados_cfg)));
     subsystems.push_back(subsystem("bdev", std::move(bdev_cfg)));
     subsystems.push_back(subsystem("vhost", std::move(vhost_cfg)));
     subsystems.push_back(subsystem("nvmf", std::move(nvmf_cfg)));
+    subsystems.push_back(subsystem("nbd", std::move(nbd_cfg)));
+    subsystems.push_back(subsystem("ublk", std::move(ublk_cfg)));
     JsonObject out;
     out["subsystems"] = Json(std::move(subsystems));
     return Json(std::move(out));

@@ -221,7 +221,7 @@ class TestConfigSnapshot:
             config = client.invoke("save_config")
         daemon.stop()
         assert {s["subsystem"] for s in config["subsystems"]} == \
-            {"bdev", "vhost", "nvmf"}
+            {"rados", "bdev", "vhost", "nvmf", "nbd", "ublk"}
 
         config_path = tmp_path / "cfg.json"
         config_path.write_text(jsonmod.dumps(config))

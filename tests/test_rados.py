@@ -238,3 +238,46 @@ class TestRadosGpu:
             assert r["io_count"] >= 2000
         finally:
             c.stop()
+
+
+class TestRadosConfigReplay:
+    def test_cluster_and_rbd_survive_snapshot_replay(self, tmp_path):
+        """save_config captures the loopback cluster (with its bound
+        port) BEFORE the bdev subsystem, so an rbd bdev whose mon_host
+        points at it reconnects on `load_config` in a fresh daemon."""
+        import pathlib
+
+        import fixtures as fx
+
+        dir_a = pathlib.Path(tmp_path) / "a"
+        dir_a.mkdir()
+        daemon = fx.launch_hipstored(dir_a)
+        try:
+            with hipstore.Client(daemon.socket_path) as client:
+                info = client.invoke("rados_cluster_start",
+                                     {"arena_mb": 16, "object_mb": 1})
+                hipstore.construct_rbd_bdev(
+                    client, pool_name="rbd", rbd_name="imgr",
+                    block_size=512, name="rbd-replay",
+                    config={"mon_host": info["mon_host"],
+                            "emu_size_mb": 4, "object_mb": 1})
+                config = client.invoke("save_config")
+        finally:
+            daemon.stop()
+        # The cluster died with the daemon; replaying in a new daemon
+        # restarts it on the SAME port and reconnects the bdev.
+        dir_b = pathlib.Path(tmp_path) / "b"
+        dir_b.mkdir()
+        daemon2 = fx.launch_hipstored(dir_b)
+        try:
+            with hipstore.Client(daemon2.socket_path) as client:
+                client.invoke("load_config", config)
+                bdev = hipstore.get_bdevs(client, "rbd-replay")[0]
+                assert bdev.product_name == "Ceph Rbd Disk"
+                assert bdev.num_blocks * bdev.block_size == 4 << 20
+                r = hipstore.perf_run(client, "rbd-replay", io_size=4096,
+                                      queue_depth=2, num_queues=1,
+                                      seconds=0.2, workload="randwrite")
+                assert r["io_count"] > 0
+        finally:
+            daemon2.stop()
