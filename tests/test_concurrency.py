@@ -236,7 +236,7 @@ class TestConfigSnapshotRaces:
                 for _ in range(30):
                     config = client.invoke("save_config")
                     assert {s["subsystem"] for s in config["subsystems"]} \
-                        == {"bdev", "vhost", "nvmf"}
+                        == {"rados", "bdev", "vhost", "nvmf", "nbd", "ublk"}
                     for sub in config["subsystems"]:
                         for entry in sub["config"]:
                             assert "method" in entry
