@@ -52,6 +52,50 @@ class TestMainModules:
             oim_csi_driver.main([
                 "--nodeid", "n1", "--hipstored-socket", "/x",
                 "--oim-registry-address", "tcp://y:1"])  # both modes
+        with pytest.raises(SystemExit):
+            oim_csi_driver.main([
+                "--nodeid", "n1", "--hipstored-socket", "/x",
+                "--csiversion", "0.9"])  # unknown personality
+
+    def test_csi_driver_serves_03_personality(self, tmp_path):
+        """--csiversion 0.3 (reference main.go flag) serves the legacy
+        twins: a csi.v0 GetPluginInfo round-trips over the wire."""
+        import threading
+        import time as time_mod
+
+        import fixtures
+        import grpc
+        from oim_amd.cmd import oim_csi_driver
+        from oim_amd.common.server import grpc_target
+        from oim_amd.spec import csi_v0 as csi0
+        from oim_amd.spec.rpc_csi0 import CSI0IdentityStub, CSI0NodeStub
+
+        daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
+        endpoint = f"unix://{tmp_path}/csi03.sock"
+        thread = threading.Thread(
+            target=oim_csi_driver.main,
+            args=([["--nodeid", "n03", "--drivername", "oim-03",
+                    "--hipstored-socket", daemon.socket_path,
+                    "--endpoint", endpoint,
+                    "--csiversion", "0.3"]][0],),
+            daemon=True)
+        thread.start()
+        try:
+            deadline = time_mod.time() + 10
+            sock_path = endpoint[len("unix://"):]
+            import os as os_mod
+            while not os_mod.path.exists(sock_path):
+                assert time_mod.time() < deadline, "driver never listened"
+                time_mod.sleep(0.05)
+            with grpc.insecure_channel(grpc_target(endpoint)) as ch:
+                info = CSI0IdentityStub(ch).GetPluginInfo(
+                    csi0.GetPluginInfoRequest(), timeout=5)
+                assert info.name == "oim-03"
+                node_id = CSI0NodeStub(ch).NodeGetId(
+                    csi0.NodeGetIdRequest(), timeout=5).node_id
+                assert node_id == "n03"
+        finally:
+            daemon.stop()
 
 
 class TestOimctlVolumes:
