@@ -60,31 +60,29 @@ class TestMainModules:
     def test_csi_driver_serves_03_personality(self, tmp_path):
         """--csiversion 0.3 (reference main.go flag) serves the legacy
         twins: a csi.v0 GetPluginInfo round-trips over the wire."""
-        import threading
+        import os as os_mod
+        import subprocess
         import time as time_mod
 
         import fixtures
         import grpc
-        from oim_amd.cmd import oim_csi_driver
         from oim_amd.common.server import grpc_target
         from oim_amd.spec import csi_v0 as csi0
         from oim_amd.spec.rpc_csi0 import CSI0IdentityStub, CSI0NodeStub
 
         daemon = fixtures.launch_hipstored(tmp_path, cpu=True)
         endpoint = f"unix://{tmp_path}/csi03.sock"
-        thread = threading.Thread(
-            target=oim_csi_driver.main,
-            args=([["--nodeid", "n03", "--drivername", "oim-03",
-                    "--hipstored-socket", daemon.socket_path,
-                    "--endpoint", endpoint,
-                    "--csiversion", "0.3"]][0],),
-            daemon=True)
-        thread.start()
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "oim_amd.cmd.oim_csi_driver",
+             "--nodeid", "n03", "--drivername", "oim-03",
+             "--hipstored-socket", daemon.socket_path,
+             "--endpoint", endpoint, "--csiversion", "0.3"],
+            stderr=subprocess.DEVNULL)
         try:
-            deadline = time_mod.time() + 10
+            deadline = time_mod.time() + 15
             sock_path = endpoint[len("unix://"):]
-            import os as os_mod
             while not os_mod.path.exists(sock_path):
+                assert proc.poll() is None, "driver exited early"
                 assert time_mod.time() < deadline, "driver never listened"
                 time_mod.sleep(0.05)
             with grpc.insecure_channel(grpc_target(endpoint)) as ch:
@@ -95,6 +93,8 @@ class TestMainModules:
                     csi0.NodeGetIdRequest(), timeout=5).node_id
                 assert node_id == "n03"
         finally:
+            proc.terminate()
+            proc.wait(timeout=10)
             daemon.stop()
 
 
