@@ -35,10 +35,18 @@ namespace {
 
 using namespace nvmf;
 
-// CRC32C of an aligned HBM range via the GPU kernel, else software.
+// CRC32C of a range the target is about to transmit. The host copy
+// always exists here (TCP needs host bytes), and SSE4.2 runs at
+// ~15 GB/s, so the GPU per-4KiB kernel only wins once the extent is
+// large enough to amortize a launch+sync round trip (~100 us when
+// resident service kernels hold the HW queues) against re-touching
+// the bytes: measured crossover is around 1 MiB. Below that the
+// hardware-CRC host path is strictly faster (128 KiB digest path:
+// 0.57 -> ~5 GB/s).
 uint32_t range_crc32c(Bdev* bdev, uint64_t offset, const void* host_copy,
                       uint32_t len) {
-  if (bdev->device_base() != nullptr && len >= 16384 && len % 4096 == 0 &&
+  if (bdev->device_base() != nullptr &&
+      (host_copy == nullptr || len >= (1u << 20)) && len % 4096 == 0 &&
       offset % 4096 == 0) {
     const uint32_t blocks = len / 4096;
     std::vector<uint32_t> crcs(blocks);

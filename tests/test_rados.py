@@ -231,8 +231,14 @@ class TestRadosGpu:
             blob = bytes(random.Random(3).getrandbits(8)
                          for _ in range(64 * 1024))
             off = (1 << 20) - 8192
-            b.write(off, blob)       # GPU verifies the landed CRC
-            assert b.read(off, len(blob)) == blob  # GPU computes reply CRC
+            b.write(off, blob)       # small op: SSE4.2 verify on readback
+            assert b.read(off, len(blob)) == blob
+            # Whole-object aligned op (>= 1 MiB): the GPU per-4KiB CRC
+            # kernel verifies the landed extent / hashes the reply.
+            big = bytes(random.Random(4).getrandbits(8)
+                        for _ in range(1 << 20))
+            b.write(4 << 20, big)
+            assert b.read(4 << 20, len(big)) == big
             r = hs.run_bdevperf(b, "randrw", 4096, 8, 2, 1.0,
                                 max_ios=2000)
             assert r["io_count"] >= 2000
