@@ -93,8 +93,12 @@ class RadosClusterImpl : public RadosCluster,
     }
     hbm_ = use_hbm && gpu_available();
     if (hbm_) {
+      // Persistent engine: each OSD op costs one arena round trip
+      // under the cluster mutex, so per-op latency IS the ceiling
+      // (batched launches measured ~37 us/op -> 27k IOPS).
       arena_ = create_hbm_bdev("rados-arena", 4096,
-                               arena_bytes / 4096, device);
+                               arena_bytes / 4096, device,
+                               /*persistent=*/true);
     } else {
       arena_ = create_malloc_bdev("rados-arena", 4096, arena_bytes / 4096);
     }
@@ -312,10 +316,18 @@ class RadosClusterImpl : public RadosCluster,
                            reply_data->data() + reply_off, len) != kIoOk) {
           return kEio;
         }
-        // Outbound data CRC from HBM via the GPU kernel (whole reply
-        // data is this op's payload: the client sends one op/message).
+        // Outbound data CRC (whole reply data is this op's payload:
+        // the client sends one op/message). Large aligned extents
+        // hash in place in HBM on the GPU; otherwise SSE4.2 over the
+        // reply bytes just read — never a second arena readback.
         if (reply_off == 0) {
-          *reply_data_crc = arena_crc(channel, bounce, arena_off, len);
+          if (hbm_ && arena_off % 4096 == 0 && len % 4096 == 0 &&
+              len >= (1ull << 20)) {
+            *reply_data_crc = arena_crc(channel, bounce, arena_off, len);
+          } else {
+            *reply_data_crc =
+                crc32c_sw(0, reply_data->data() + reply_off, len);
+          }
         } else {
           *reply_data_crc = ~0u;  // multi-op reply: software CRC
         }
@@ -354,10 +366,16 @@ class RadosClusterImpl : public RadosCluster,
                             op->length) != kIoOk) {
           return kEio;
         }
-        // GPU-side integrity check: CRC the landed HBM extent and
-        // compare against the messenger footer (single-op messages
-        // only — the footer covers the whole data section).
-        if (frame.data.size() == op->payload_len) {
+        // Data-CRC check against the messenger footer. Whole-object
+        // aligned writes verify the LANDED HBM extent with the GPU
+        // kernel (end-to-end: wire + DMA); small ops verify the
+        // received buffer with SSE4.2, which is what Ceph's
+        // messenger itself does — a per-small-op GPU launch under
+        // the cluster mutex measured ~200 us and collapsed
+        // randwrite to 5k IOPS.
+        if (hbm_ && frame.data.size() == op->payload_len &&
+            arena_off % 4096 == 0 && op->length % 4096 == 0 &&
+            op->length >= (1ull << 20)) {
           const uint32_t crc =
               arena_crc(channel, bounce, arena_off, op->length);
           if (crc != frame.footer_data_crc) {
