@@ -174,3 +174,31 @@ class TestCrc32cHardwarePath:
         import oim_amd._hipstore as hs
 
         assert hs.crc32c(b"123456789", 0) == 0xE3069283
+
+    def test_combine_matches_whole_buffer_crc(self):
+        """GF(2) combine ladder: crc(a||b) from crc(a), crc(b), len(b).
+        This is what folds the GPU per-4KiB block CRCs into one
+        NVMe/TCP DDGST / RADOS data CRC, so it must agree with the
+        straight-line CRC for arbitrary split points."""
+        import os as _os
+        import random as _random
+
+        import oim_amd._hipstore as hs
+
+        rng = _random.Random(42)
+        blob = _os.urandom(50000)
+        whole = hs.crc32c(blob, 0)
+        for _ in range(20):
+            cut = rng.randrange(0, len(blob) + 1)
+            a, b = blob[:cut], blob[cut:]
+            got = hs.crc32c_combine(hs.crc32c(a, 0), hs.crc32c(b, 0), len(b))
+            assert got == whole, cut
+        # zero-length pieces on either side are identities
+        assert hs.crc32c_combine(whole, hs.crc32c(b"", 0), 0) == whole
+        assert hs.crc32c_combine(hs.crc32c(b"", 0), whole, len(blob)) == whole
+        # many-piece fold at 4 KiB granularity (the GPU kernel's shape)
+        crc = hs.crc32c(blob[:4096], 0)
+        for off in range(4096, len(blob), 4096):
+            piece = blob[off:off + 4096]
+            crc = hs.crc32c_combine(crc, hs.crc32c(piece, 0), len(piece))
+        assert crc == whole
