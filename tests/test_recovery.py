@@ -306,3 +306,45 @@ class TestKitchenSink:
         finally:
             proc.terminate()
             proc.wait(timeout=10)
+
+
+class TestDaemonHardKill:
+    """SIGKILL the data daemon with a client mid-conversation: the
+    client must fail with an error promptly (bounded-everything
+    discipline — never a hang), and the socket path must be reusable
+    by a fresh daemon that then replays the saved config (the SPDK
+    crash-recovery model: state lives in the config snapshot, not the
+    process)."""
+
+    def test_kill_mid_session_fails_fast_and_socket_reusable(self, tmp_path):
+        import os
+        import signal
+        import time
+
+        import fixtures as fx
+        from oim_amd import hipstore
+
+        daemon = fx.launch_hipstored(tmp_path)
+        try:
+            with hipstore.Client(daemon.socket_path) as client:
+                hipstore.construct_malloc_bdev(
+                    client, num_blocks=4096, block_size=512, name="hk")
+                config = client.invoke("save_config")
+                daemon.process.send_signal(signal.SIGKILL)
+                daemon.process.wait(timeout=10)
+                t0 = time.monotonic()
+                with pytest.raises((hipstore.RpcError, OSError)):
+                    for _ in range(3):  # first call may see buffered EOF
+                        client.invoke("get_bdevs")
+                assert time.monotonic() - t0 < 5.0  # failed, not hung
+        finally:
+            daemon.stop()
+        # Same socket path, fresh daemon, replayed topology.
+        os.unlink(daemon.socket_path)
+        daemon2 = fx.launch_hipstored(tmp_path)
+        try:
+            with hipstore.Client(daemon2.socket_path) as client:
+                client.invoke("load_config", config)
+                assert hipstore.get_bdevs(client, "hk")[0].num_blocks == 4096
+        finally:
+            daemon2.stop()
