@@ -11,6 +11,7 @@
 #include <fcntl.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <poll.h>
 #include <sys/socket.h>
 #include <unistd.h>
 
@@ -33,6 +34,17 @@ using namespace nvmf;
 constexpr const char* kHostNqn = "nqn.2014-08.org.nvmexpress:uuid:"
                                  "8f6a52f2-0000-4000-8000-oimamdinitiat";
 
+// Setup-path wait bound: a target that accepts and then goes silent
+// during ICReq/Fabrics/Identify (or a blackholed address) must fail
+// create_nvmf_tcp_bdev, not hang the daemon. The I/O path switches
+// the fd to non-blocking polled mode afterwards, where these
+// timeouts are inert.
+int setup_timeout_s() {
+  const char* env = getenv("HIPSTORE_NVMF_SETUP_TIMEOUT");
+  const int v = env != nullptr ? atoi(env) : 0;
+  return v > 0 ? v : 10;
+}
+
 int tcp_connect(const std::string& addr, uint16_t port) {
   int fd = socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
   if (fd < 0) throw std::runtime_error("nvmf: socket failed");
@@ -41,10 +53,25 @@ int tcp_connect(const std::string& addr, uint16_t port) {
   sa.sin_port = htons(port);
   sa.sin_addr.s_addr = addr.empty() ? htonl(INADDR_LOOPBACK)
                                     : inet_addr(addr.c_str());
-  if (connect(fd, reinterpret_cast<sockaddr*>(&sa), sizeof(sa)) < 0) {
+  const int flags = fcntl(fd, F_GETFL, 0);
+  fcntl(fd, F_SETFL, flags | O_NONBLOCK);
+  int rc = connect(fd, reinterpret_cast<sockaddr*>(&sa), sizeof(sa));
+  if (rc < 0 && errno == EINPROGRESS) {
+    pollfd pfd{fd, POLLOUT, 0};
+    rc = ::poll(&pfd, 1, setup_timeout_s() * 1000);
+    int soerr = 0;
+    socklen_t slen = sizeof(soerr);
+    if (rc == 1) getsockopt(fd, SOL_SOCKET, SO_ERROR, &soerr, &slen);
+    rc = (rc == 1 && soerr == 0) ? 0 : -1;
+  }
+  if (rc < 0) {
     close(fd);
     throw std::runtime_error("nvmf: connect to " + addr + " failed");
   }
+  fcntl(fd, F_SETFL, flags);
+  timeval tv{setup_timeout_s(), 0};
+  setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+  setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof(tv));
   int one = 1;
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
   return fd;

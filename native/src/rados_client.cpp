@@ -12,6 +12,7 @@
 #include <fcntl.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <poll.h>
 #include <sys/socket.h>
 #include <unistd.h>
 
@@ -223,6 +224,15 @@ bool msgr_recv(int fd, MsgrFrame* frame, bool verify_data_crc) {
 
 namespace {
 
+// Setup-path wait bound (seconds): connect + every blocking
+// handshake/header read. A monitor that accepts and then goes silent
+// (or a blackholed address) must fail the RPC, not hang the daemon.
+int setup_timeout_s() {
+  const char* env = getenv("HIPSTORE_RADOS_SETUP_TIMEOUT");
+  const int v = env != nullptr ? atoi(env) : 0;
+  return v > 0 ? v : 10;
+}
+
 int tcp_connect(const std::string& host, uint16_t port) {
   int fd = socket(AF_INET, SOCK_STREAM, 0);
   if (fd < 0) throw std::runtime_error("rados: socket failed");
@@ -231,10 +241,27 @@ int tcp_connect(const std::string& host, uint16_t port) {
   sa.sin_port = htons(port);
   sa.sin_addr.s_addr =
       host.empty() ? htonl(INADDR_LOOPBACK) : inet_addr(host.c_str());
-  if (connect(fd, reinterpret_cast<sockaddr*>(&sa), sizeof(sa)) < 0) {
+  // Bounded connect: non-blocking + poll, then back to blocking with
+  // send/recv timeouts so no later read_exact/write_exact can hang.
+  const int flags = fcntl(fd, F_GETFL, 0);
+  fcntl(fd, F_SETFL, flags | O_NONBLOCK);
+  int rc = connect(fd, reinterpret_cast<sockaddr*>(&sa), sizeof(sa));
+  if (rc < 0 && errno == EINPROGRESS) {
+    pollfd pfd{fd, POLLOUT, 0};
+    rc = ::poll(&pfd, 1, setup_timeout_s() * 1000);
+    int soerr = 0;
+    socklen_t slen = sizeof(soerr);
+    if (rc == 1) getsockopt(fd, SOL_SOCKET, SO_ERROR, &soerr, &slen);
+    rc = (rc == 1 && soerr == 0) ? 0 : -1;
+  }
+  if (rc < 0) {
     close(fd);
     throw std::runtime_error("rados: connect to " + host + " failed");
   }
+  fcntl(fd, F_SETFL, flags);
+  timeval tv{setup_timeout_s(), 0};
+  setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+  setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof(tv));
   int one = 1;
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
   return fd;

@@ -362,3 +362,48 @@ class TestNvmfPostHandshakeFuzz:
             assert bdev.read(0, 512) == b"\x5a" * 512
         finally:
             target.stop()
+
+
+class TestSetupWaitBounds:
+    """A server that accepts and then goes silent must fail bdev
+    construction within the setup timeout — never hang the daemon
+    (bounded-everything discipline; timeout via
+    HIPSTORE_{RADOS,NVMF}_SETUP_TIMEOUT, default 10 s)."""
+
+    @pytest.fixture
+    def silent_server(self):
+        import socket as socket_mod
+
+        s = socket_mod.socket()
+        s.bind(("127.0.0.1", 0))
+        s.listen(4)
+        port = s.getsockname()[1]
+        yield port
+        s.close()
+
+    def test_rados_silent_monitor_times_out(self, silent_server,
+                                            monkeypatch):
+        import time
+
+        import oim_amd._hipstore as hs
+
+        monkeypatch.setenv("HIPSTORE_RADOS_SETUP_TIMEOUT", "1")
+        t0 = time.monotonic()
+        with pytest.raises(RuntimeError):
+            hs.create_rbd_bdev("rbd-silent", f"127.0.0.1:{silent_server}",
+                               "rbd", "x", block_size=512,
+                               default_size_bytes=1 << 20)
+        assert time.monotonic() - t0 < 8.0
+
+    def test_nvmf_silent_target_times_out(self, silent_server,
+                                          monkeypatch):
+        import time
+
+        import oim_amd._hipstore as hs
+
+        monkeypatch.setenv("HIPSTORE_NVMF_SETUP_TIMEOUT", "1")
+        t0 = time.monotonic()
+        with pytest.raises(RuntimeError):
+            hs.create_nvmf_tcp_bdev("nvmf-silent", "127.0.0.1",
+                                    silent_server, "nqn.silent")
+        assert time.monotonic() - t0 < 8.0
