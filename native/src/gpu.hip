@@ -2176,7 +2176,13 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
           req.buffer = buf + static_cast<size_t>(slot) * io_size;
           submit_ts[slot] = clock::now();
           req.on_complete = [&, slot](int status) {
-            if (status != kIoOk) failed.store(true);
+            if (status != kIoOk) {
+              // Abort the run on the first error (SPDK bdevperf
+              // semantics): drain and stop rather than hammering a
+              // dead/failing bdev for the rest of the duration.
+              failed.store(true);
+              stopping = true;
+            }
             const auto now = clock::now();
             st.lat.record(static_cast<uint32_t>(std::min<int64_t>(
                 std::chrono::duration_cast<std::chrono::microseconds>(
@@ -2218,7 +2224,8 @@ PerfResult run_bdevperf(Bdev* bdev, const std::string& workload,
             empty_polls = 0;
           }
           if (!stopping &&
-              (clock::now() >= deadline ||
+              (failed.load(std::memory_order_relaxed) ||
+               clock::now() >= deadline ||
                (per_queue_cap && submitted >= per_queue_cap))) {
             stopping = true;
           }

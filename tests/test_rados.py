@@ -287,3 +287,28 @@ class TestRadosConfigReplay:
                 assert r["io_count"] > 0
         finally:
             daemon2.stop()
+
+
+class TestClusterDeathMidWorkload:
+    def test_run_fails_fast_when_cluster_stops(self, monkeypatch):
+        """Stopping the cluster under a running workload must surface
+        an I/O failure within the bounded-wait window — dead
+        connections fail inflight ops (fail_all), they never hang the
+        perf loop."""
+        import threading
+        import time
+
+        monkeypatch.setenv("HIPSTORE_SYNC_TIMEOUT_S", "5")
+        c = hs.start_rados_cluster(port=0, arena_mb=32, use_hbm=False,
+                                   device=0, object_bytes=1 << 20)
+        b = make_image(c, "img-die", size_mb=8, name="rbd-die")
+        b.write(0, b"\x5a" * 4096)
+        killer = threading.Timer(0.3, c.stop)
+        killer.start()
+        t0 = time.monotonic()
+        try:
+            with pytest.raises(RuntimeError):
+                hs.run_bdevperf(b, "randread", 4096, 8, 2, 30.0)
+        finally:
+            killer.join()
+        assert time.monotonic() - t0 < 25.0  # failed fast, not 30 s
