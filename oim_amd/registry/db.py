@@ -78,8 +78,23 @@ class FileRegistryDB(RegistryDB):
         self._mutex = threading.Lock()
         self._data: Dict[str, str] = {}
         if os.path.exists(path):
-            with open(path) as f:
-                self._data = json.load(f)
+            # Refuse to start on a corrupt DB rather than silently
+            # discarding registry state (writes are atomic-rename, so
+            # corruption means something external broke the file —
+            # surface it with the path, don't guess).
+            try:
+                with open(path) as f:
+                    data = json.load(f)
+            except (json.JSONDecodeError, UnicodeDecodeError) as exc:
+                raise RuntimeError(
+                    f"registry db {path} is corrupt: {exc}") from exc
+            if not isinstance(data, dict) or not all(
+                    isinstance(k, str) and isinstance(v, str)
+                    for k, v in data.items()):
+                raise RuntimeError(
+                    f"registry db {path} is corrupt: not a "
+                    "string-to-string map")
+            self._data = data
 
     def _flush_locked(self) -> None:
         tmp = self._path + ".tmp"

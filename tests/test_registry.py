@@ -57,6 +57,30 @@ class TestFileDB:
         db3 = FileRegistryDB(path)
         assert db3.lookup(["c1", "address"]) is None
 
+    def test_corrupt_file_refused_with_path_in_error(self, tmp_path):
+        """A corrupt DB must fail startup loudly (state is durable;
+        silently discarding it would un-register every controller),
+        and the error must say which file."""
+        path = str(tmp_path / "reg.json")
+        with open(path, "w") as f:
+            f.write("{truncated")
+        with pytest.raises(RuntimeError, match="reg.json"):
+            FileRegistryDB(path)
+        with open(path, "w") as f:
+            f.write('["a", "list", "not", "a", "map"]')
+        with pytest.raises(RuntimeError, match="string-to-string"):
+            FileRegistryDB(path)
+        # a stale .tmp from a torn flush is harmless: ignored, and the
+        # next write replaces it
+        with open(path, "w") as f:
+            f.write('{"c1/address": "tcp://x:9"}')
+        with open(path + ".tmp", "w") as f:
+            f.write("{garbage")
+        db = FileRegistryDB(path)
+        assert db.lookup(["c1", "address"]) == "tcp://x:9"
+        db.store(["c2", "address"], "tcp://y:1")
+        assert FileRegistryDB(path).lookup(["c2", "address"]) == "tcp://y:1"
+
 
 class MockController(spec.ControllerServicer):
     """Records requests (reference registry_test.go MockController)."""
