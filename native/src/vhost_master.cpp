@@ -205,7 +205,6 @@ VhostMasterSession::VhostMasterSession(const std::string& socket_path,
   }
 
   std::vector<VhostMasterRing> rings(num_rings);
-  PerfResult result;
   try {
     // Handshake (GET/SET features, owner, memory table).
     send_msg(sock, kGetFeatures, nullptr, 0);
