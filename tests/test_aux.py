@@ -85,3 +85,70 @@ class TestFioHarness:
         out = capsys.readouterr().out
         assert "IOPS=" in out
         assert "[Finished:Performance]" in out
+
+
+class TestPayloadFormatters:
+    """gRPC payload logging (reference pkg/oim-common/tracing.go +
+    protosanitizer StripSecrets): secrets must never reach logs, and
+    the interceptors must pass calls through untouched."""
+
+    def _msg(self):
+        from oim_amd.spec import csi_v1 as csi
+
+        return csi.NodeStageVolumeRequest(
+            volume_id="vol-1",
+            staging_target_path="/mnt/stage",
+            secrets={"admin": "hunter2"})
+
+    def test_strip_secrets_redacts(self):
+        from oim_amd.common.tracing import strip_secrets_formatter
+
+        out = strip_secrets_formatter(self._msg())
+        assert "hunter2" not in out
+        assert "secrets=<redacted>" in out
+        assert "vol-1" in out  # non-secret fields still logged
+
+    def test_complete_formatter_includes_everything(self):
+        from oim_amd.common.tracing import complete_formatter
+
+        out = complete_formatter(self._msg())
+        assert "hunter2" in out and "vol-1" in out
+
+    def test_null_formatter_and_non_proto(self):
+        from oim_amd.common.tracing import (null_formatter,
+                                            strip_secrets_formatter)
+
+        assert null_formatter(self._msg()) == "<omitted>"
+        assert strip_secrets_formatter("plain") == "'plain'"
+
+    def test_server_interceptor_passthrough(self, tmp_path):
+        """A servicer behind LogServerInterceptor(strip_secrets) must
+        behave identically; the log line carries the redacted form."""
+        import grpc
+
+        from oim_amd.common.server import NonBlockingGRPCServer
+        from oim_amd.common.tracing import (LogServerInterceptor,
+                                            strip_secrets_formatter)
+        from oim_amd.spec import csi_v1 as csi
+        from oim_amd.spec.rpc_csi import (CSIIdentityServicer,
+                                          CSIIdentityStub,
+                                          add_csi_identity_to_server)
+
+        class Identity(CSIIdentityServicer):
+            def GetPluginInfo(self, request, context):
+                return csi.GetPluginInfoResponse(name="t", vendor_version="1")
+
+        endpoint = f"unix://{tmp_path}/trace.sock"
+        server = NonBlockingGRPCServer(
+            endpoint,
+            interceptors=[LogServerInterceptor(strip_secrets_formatter)])
+        server.start(lambda srv: add_csi_identity_to_server(Identity(),
+                                                             srv))
+        try:
+            with grpc.insecure_channel(
+                    f"unix:{tmp_path}/trace.sock") as channel:
+                stub = CSIIdentityStub(channel)
+                resp = stub.GetPluginInfo(csi.GetPluginInfoRequest())
+            assert resp.name == "t"
+        finally:
+            server.stop()
