@@ -350,7 +350,10 @@ def main() -> int:
     value = total_ios / elapsed
     if rank == 0:
         out = {
-            "metric": "4KiB_randread_IOPS",
+            # Defaults name BASELINE.json's metric (4KiB_randread_IOPS);
+            # overriding --workload/--io-size relabels rather than
+            # misreporting a different measurement under that name.
+            "metric": f"{args.io_size // 1024}KiB_{args.workload}_IOPS",
             "value": round(value, 1),
             "unit": "IOPS",
             # Honest GPU count: the ranks that actually ran (--gpus is
