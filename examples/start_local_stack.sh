@@ -92,7 +92,8 @@ echo "--- vhost-user (VM-attach) data-path benchmark (on a clone:"
 echo "    demo-vol itself is claimed by its SCSI mapping) ---"
 oimctl_cmd clone --controller demo-gpu0 demo-vol demo-vbench
 python -m oim_amd.bench.vhost_harness --socket "$WORK/hipstored.sock" \
-    --bdev demo-vbench --iodepth 16 --runtime 2 --ctrlr demo-vhbench
+    --bdev demo-vbench --iodepth 16 --runtime 2 --ctrlr demo-vhbench \
+    --master native
 oimctl_cmd provision --controller demo-gpu0 demo-vbench 0
 
 echo "demo complete; logs in $WORK"

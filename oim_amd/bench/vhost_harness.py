@@ -211,6 +211,76 @@ class VhostAttachment:
                                {"ctrlr": self.ctrlr})
 
 
+def _native_main(args) -> int:
+    """Drive the controller with the C++ vhost-user master
+    (native/src/vhost_master.cpp): one standing session, repeated
+    fixed-count runs until --runtime elapses. This is the same driver
+    bench.py --frontend vhost uses."""
+    from oim_amd import _hipstore as hs
+
+    with hipstore.Client(args.socket) as client:
+        bdev = hipstore.get_bdevs(client, args.bdev)[0]
+        if args.personality == "blk":
+            client.invoke("construct_vhost_blk_controller",
+                          {"ctrlr": args.ctrlr, "dev_name": args.bdev})
+        else:
+            client.invoke("construct_vhost_scsi_controller",
+                          {"ctrlr": args.ctrlr})
+            client.invoke("add_vhost_scsi_lun",
+                          {"ctrlr": args.ctrlr, "scsi_target_num": 0,
+                           "bdev_name": args.bdev})
+        vhost_path = os.path.join(os.path.dirname(args.socket),
+                                  args.ctrlr)
+        try:
+            session = hs.VhostMasterSession(
+                vhost_path, args.personality, args.numjobs,
+                args.iodepth, args.bs, bdev.block_size,
+                bdev.num_blocks * bdev.block_size)
+            session.run(max(args.numjobs * args.iodepth * 16, 4096),
+                        args.rw)  # warm: rings built, memory registered
+            chunk = max(args.numjobs * args.iodepth * 256, 20000)
+            total = 0
+            start = time.perf_counter()
+            while time.perf_counter() - start < args.runtime:
+                r = session.run(chunk, args.rw)
+                total += r["io_count"]
+            elapsed = time.perf_counter() - start
+            del session
+        finally:
+            client.invoke("remove_vhost_controller",
+                          {"ctrlr": args.ctrlr})
+    iops = total / elapsed
+    print(f"{args.bdev} via vhost ({args.personality}, native master): "
+          f"rw={args.rw}, bs={args.bs}, iodepth={args.iodepth}, "
+          f"numjobs={args.numjobs}")
+    print(f"  IOPS={iops:,.0f}, BW={iops * args.bs / 1e6:,.1f} MB/s "
+          f"({total} ios in {elapsed:.2f}s)")
+    print(f"  lat (usec, last chunk): avg={r['lat_avg_us']:.1f}, "
+          f"p50={r['lat_p50_us']:.0f}, p99={r['lat_p99_us']:.0f}")
+    if args.perfdash:
+        payload = {
+            "version": "v1",
+            "dataItems": [{
+                "data": {
+                    "iops": iops,
+                    "throughput_mbps": iops * args.bs / 1e6,
+                    "lat_p50_us": r["lat_p50_us"],
+                    "lat_p99_us": r["lat_p99_us"],
+                },
+                "unit": "mixed",
+                "labels": {"bdev": args.bdev, "rw": args.rw,
+                           "bs": str(args.bs),
+                           "iodepth": str(args.iodepth),
+                           "numjobs": str(args.numjobs),
+                           "master": "native",
+                           "path": f"vhost-user-{args.personality}"},
+            }],
+            "labels": {"suite": "hipstored-vhost"},
+        }
+        print("[Finished:Performance] " + json.dumps(payload))
+    return 0
+
+
 def main(argv=None) -> int:
     parser = argparse.ArgumentParser(
         description="vhost-user-scsi data-path benchmark")
@@ -229,8 +299,17 @@ def main(argv=None) -> int:
     parser.add_argument("--ctrlr", default="vhost-bench")
     parser.add_argument("--personality", default="scsi",
                         choices=["scsi", "blk"])
+    parser.add_argument("--master", default="python",
+                        choices=["python", "native"],
+                        help="ring driver: the Python conformance "
+                             "master (interpreter-bound, ~50-200k "
+                             "IOPS) or the C++ native master, which "
+                             "measures the daemon's data path")
     parser.add_argument("--perfdash", action="store_true")
     args = parser.parse_args(argv)
+
+    if args.master == "native":
+        return _native_main(args)
 
     with hipstore.Client(args.socket) as client:
         try:

@@ -945,3 +945,33 @@ class TestNativeVhostMaster:
                                   num_rings=1, iodepth=4, io_size=4096,
                                   workload="randread", total_ios=10,
                                   block_size=512, capacity_bytes=1 << 20)
+
+
+class TestVhostHarnessNativeCli:
+    def test_native_master_cli(self, hipstored):  # noqa: F811
+        """`vhost_harness --master native` drives the C++ master end
+        to end and emits the perfdash shape."""
+        import json as json_mod
+        import subprocess
+        import sys
+
+        with hipstore.Client(hipstored.socket_path) as client:
+            try:
+                hipstore.get_bdevs(client, "vhn")
+            except hipstore.RpcError:
+                client.invoke("construct_malloc_bdev",
+                              {"num_blocks": 1 << 14, "block_size": 512,
+                               "name": "vhn"})
+        r = subprocess.run(
+            [sys.executable, "-m", "oim_amd.bench.vhost_harness",
+             "--socket", hipstored.socket_path, "--bdev", "vhn",
+             "--iodepth", "4", "--runtime", "0.5", "--ctrlr", "vhn-c",
+             "--master", "native", "--perfdash"],
+            capture_output=True, text=True, timeout=120,
+            cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+        assert r.returncode == 0, r.stderr[-800:]
+        line = [ln for ln in r.stdout.splitlines()
+                if ln.startswith("[Finished:Performance]")][0]
+        payload = json_mod.loads(line.split(" ", 1)[1])
+        assert payload["dataItems"][0]["data"]["iops"] > 0
+        assert payload["dataItems"][0]["labels"]["master"] == "native"
