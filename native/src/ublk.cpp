@@ -24,6 +24,7 @@
 #include <unistd.h>
 
 #include <atomic>
+#include <chrono>
 #include <cerrno>
 #include <cstdio>
 #include <cstring>
@@ -581,6 +582,7 @@ class UblkServer {
       std::vector<std::pair<int, int>> done;  // (tag, result-bytes-or-neg)
       std::mutex done_mutex;  // completions fire on this thread (poll)
       bool abort_seen = false;
+      std::chrono::steady_clock::time_point stop_observed_{};
       while (true) {
         // 1) kernel completions -> bdev submissions
         IoUringCqe cqe{};
@@ -647,9 +649,18 @@ class UblkServer {
           done.clear();
         }
         if (ring.submit(0) < 0 && errno != EBUSY) break;
-        if (stopping_.load(std::memory_order_acquire) && abort_seen &&
-            inflight_bdev == 0) {
-          break;
+        if (stopping_.load(std::memory_order_acquire)) {
+          // Normal teardown: STOP_DEV retires every tag with an abort
+          // CQE. Bound the wait anyway — a kernel that never delivers
+          // them (or a wedged bdev) must not hang stop()'s join.
+          using clock = std::chrono::steady_clock;
+          if (stop_observed_ == clock::time_point{}) {
+            stop_observed_ = clock::now();
+          }
+          if ((abort_seen && inflight_bdev == 0) ||
+              clock::now() - stop_observed_ > std::chrono::seconds(5)) {
+            break;
+          }
         }
         if (!any) {
           std::this_thread::sleep_for(std::chrono::microseconds(50));
